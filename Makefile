@@ -1,0 +1,52 @@
+# nezha — MI355X-native build.
+# Everything is compiled with hipcc as HIP C++ for gfx950; the built
+# artifacts live in build/ (gitignored; ships to the GPU box via the gpurun
+# snapshot since they are not gpurun-ignored).
+
+HIPCC    ?= hipcc
+ARCH     ?= gfx950
+BUILD    ?= build
+
+CXXFLAGS := -O3 -std=c++17 -fPIC -I include -I . --offload-arch=$(ARCH) \
+            -Wall -Wno-unused-function -x hip
+LDFLAGS  := -L/opt/rocm/lib -lrccl -lz
+
+HOST_SRCS := src/core/tensor.cc src/core/ops.cc src/cpu/cpu_ops.cc \
+             src/util/util.cc src/ckpt/npz.cc src/nn/modules.cc src/nn/optim.cc \
+             src/models/models.cc src/train/trainer.cc src/train/ddp.cc \
+             src/comm/rccl_comm.cc src/comm/fake_comm.cc src/comm/tcp_comm.cc \
+             src/capi.cc
+HIP_SRCS  := src/hip/elementwise.hip src/hip/gemm.hip
+
+OBJS := $(HOST_SRCS:%.cc=$(BUILD)/%.o) $(HIP_SRCS:%.hip=$(BUILD)/%.o)
+
+LIB := $(BUILD)/libnezha.so
+
+CMDS := $(BUILD)/nezha-train $(BUILD)/nezha-bench $(BUILD)/nezha-info
+TESTS_CPP := $(BUILD)/test_core $(BUILD)/test_ddp_fake
+
+.PHONY: all lib cmds tests clean
+all: lib cmds tests
+lib: $(LIB)
+cmds: $(CMDS)
+tests: $(TESTS_CPP)
+
+$(BUILD)/%.o: %.cc
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(BUILD)/%.o: %.hip
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(LIB): $(OBJS)
+	$(HIPCC) -shared -fPIC $(OBJS) $(LDFLAGS) -o $@
+
+$(BUILD)/nezha-%: cmd/%.cc $(LIB)
+	$(HIPCC) $(CXXFLAGS) $< -o $@ -L$(BUILD) -lnezha $(LDFLAGS) -Wl,-rpath,'$$ORIGIN'
+
+$(BUILD)/test_%: tests/cpp/test_%.cc $(LIB)
+	$(HIPCC) $(CXXFLAGS) $< -o $@ -L$(BUILD) -lnezha $(LDFLAGS) -Wl,-rpath,'$$ORIGIN'
+
+clean:
+	rm -rf $(BUILD)

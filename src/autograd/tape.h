@@ -1,0 +1,77 @@
+// Record-replay reverse-mode autodiff.
+// Forward op calls push backward closures onto the Tape; Tape::backward()
+// replays them in reverse. Parameter gradients land in Param::grad (a view
+// into a flat optimizer/DDP bucket once an optimizer is attached) and fire
+// a grad-ready hook that feeds the DDP bucketer.
+#pragma once
+
+#include <functional>
+#include <memory>
+#include <vector>
+
+#include "src/core/ops.h"
+
+namespace nz {
+
+struct Param {
+  std::string name;
+  Tensor data;    // bf16 on HIP, f32 on CPU
+  Tensor grad;    // same dtype as data; may be a bucket view
+  bool grad_init = false;  // false => next grad write uses beta=0
+  std::function<void(Param&)> ready_hook;  // set by Ddp
+
+  int64_t numel() const { return data.numel(); }
+  // Returns beta for the next in-place grad write and flips the flag.
+  float grad_beta() {
+    float b = grad_init ? 1.0f : 0.0f;
+    grad_init = true;
+    return b;
+  }
+  void mark_ready() {
+    if (ready_hook) ready_hook(*this);
+  }
+};
+
+struct Var {
+  Tensor value;
+  Tensor grad;            // lazily allocated, same dtype as value
+  bool grad_init = false;
+  bool needs_grad = true; // false for the input leaf
+
+  Tensor& ensure_grad() {
+    if (!grad.defined())
+      grad = Tensor(value.ctx(), value.dtype(), value.shape());
+    return grad;
+  }
+  float grad_beta() {
+    ensure_grad();
+    float b = grad_init ? 1.0f : 0.0f;
+    grad_init = true;
+    return b;
+  }
+};
+using VarPtr = std::shared_ptr<Var>;
+
+inline VarPtr make_var(Tensor t, bool needs_grad = true) {
+  auto v = std::make_shared<Var>();
+  v->value = std::move(t);
+  v->needs_grad = needs_grad;
+  return v;
+}
+
+class Tape {
+ public:
+  void push(std::function<void()> fn) { nodes_.push_back(std::move(fn)); }
+  // Replays nodes newest-first, then clears the tape.
+  void backward() {
+    for (auto it = nodes_.rbegin(); it != nodes_.rend(); ++it) (*it)();
+    nodes_.clear();
+  }
+  void clear() { nodes_.clear(); }
+  size_t size() const { return nodes_.size(); }
+
+ private:
+  std::vector<std::function<void()>> nodes_;
+};
+
+}  // namespace nz

@@ -1,0 +1,107 @@
+// Dispatch layer: tensor.device() -> exactly one backend (C7: no registry,
+// no multi-backend dispatch). Profiling scopes hook in here.
+#include "src/core/ops.h"
+
+#include "src/cpu/cpu_ops.h"
+#include "src/hip/hip_ops.h"
+
+namespace nz {
+namespace ops {
+
+static Device dev(const Tensor& t) { return t.device(); }
+
+#define NZ_DISPATCH(fn, first, ...)                  \
+  do {                                               \
+    if (dev(first) == Device::CPU)                   \
+      cpu::fn(__VA_ARGS__);                          \
+    else                                             \
+      hip::fn(__VA_ARGS__);                          \
+  } while (0)
+
+void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
+          float alpha, float beta, const GemmEpilogue* epi) {
+  NZ_DISPATCH(gemm, A, A, B, C, transA, transB, alpha, beta, epi);
+}
+void gemm_batched(const Tensor& A, const Tensor& B, Tensor& C, bool transA,
+                  bool transB, float alpha, int64_t batch, int64_t M, int64_t N,
+                  int64_t K, int64_t sA, int64_t sB, int64_t sC) {
+  NZ_DISPATCH(gemm_batched, A, A, B, C, transA, transB, alpha, batch, M, N, K, sA, sB, sC);
+}
+void bias_gelu_fwd(const Tensor& x, const Tensor* bias, Tensor& y, Tensor* save_pre) {
+  NZ_DISPATCH(bias_gelu_fwd, x, x, bias, y, save_pre);
+}
+void bias_gelu_bwd(const Tensor& dy, const Tensor& z, Tensor& dz, Tensor* dbias) {
+  NZ_DISPATCH(bias_gelu_bwd, dy, dy, z, dz, dbias);
+}
+void layernorm_fwd(const Tensor& x, const Tensor& gamma, const Tensor& beta,
+                   Tensor& y, Tensor& mean, Tensor& rstd, float eps) {
+  NZ_DISPATCH(layernorm_fwd, x, x, gamma, beta, y, mean, rstd, eps);
+}
+void layernorm_bwd(const Tensor& dy, const Tensor& x, const Tensor& gamma,
+                   const Tensor& mean, const Tensor& rstd, Tensor& dx,
+                   Tensor& dgamma, Tensor& dbeta, float beta_dx) {
+  NZ_DISPATCH(layernorm_bwd, dy, dy, x, gamma, mean, rstd, dx, dgamma, dbeta, beta_dx);
+}
+void softmax_fwd(const Tensor& x, Tensor& y, float scale) {
+  NZ_DISPATCH(softmax_fwd, x, x, y, scale);
+}
+void softmax_bwd(const Tensor& dy, const Tensor& y, Tensor& dx, float scale) {
+  NZ_DISPATCH(softmax_bwd, dy, dy, y, dx, scale);
+}
+void adam_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor& m,
+               Tensor& v, float lr, float b1, float b2, float eps, int step,
+               float grad_scale, float wd) {
+  NZ_DISPATCH(adam_step, param, param, grad, master, m, v, lr, b1, b2, eps, step, grad_scale, wd);
+}
+void sgd_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor* momentum,
+              float lr, float mu, float grad_scale) {
+  NZ_DISPATCH(sgd_step, param, param, grad, master, momentum, lr, mu, grad_scale);
+}
+void fill(Tensor& t, float value) { NZ_DISPATCH(fill, t, t, value); }
+void rand_uniform(Tensor& t, uint64_t seed, float lo, float hi) {
+  NZ_DISPATCH(rand_uniform, t, t, seed, lo, hi);
+}
+void rand_normal(Tensor& t, uint64_t seed, float mean, float std) {
+  NZ_DISPATCH(rand_normal, t, t, seed, mean, std);
+}
+void cast_copy(Tensor& dst, const Tensor& src) { NZ_DISPATCH(cast_copy, dst, dst, src); }
+void add_inplace(Tensor& dst, const Tensor& src, float alpha) {
+  NZ_DISPATCH(add_inplace, dst, dst, src, alpha);
+}
+void scale_inplace(Tensor& dst, float alpha) { NZ_DISPATCH(scale_inplace, dst, dst, alpha); }
+void transpose2d(const Tensor& src, Tensor& dst) { NZ_DISPATCH(transpose2d, src, src, dst); }
+void mse_loss(const Tensor& y, const Tensor& target, Tensor& loss_out, Tensor* dy,
+              float grad_scale) {
+  NZ_DISPATCH(mse_loss, y, y, target, loss_out, dy, grad_scale);
+}
+void ce_loss(const Tensor& logits, const Tensor& labels, Tensor& loss_out,
+             Tensor* dlogits, float grad_scale) {
+  NZ_DISPATCH(ce_loss, logits, logits, labels, loss_out, dlogits, grad_scale);
+}
+void reduce_sum(const Tensor& src, Tensor& out) { NZ_DISPATCH(reduce_sum, src, src, out); }
+void l2norm_sq(const Tensor& src, Tensor& out) { NZ_DISPATCH(l2norm_sq, src, src, out); }
+
+}  // namespace ops
+}  // namespace nz
+
+namespace nz { namespace ops {
+void colsum(const Tensor& src, Tensor& dst, float beta) {
+  NZ_DISPATCH(colsum, src, src, dst, beta);
+}
+void attn_split_qkv(const Tensor& qkv, Tensor& q, Tensor& k, Tensor& v,
+                    int64_t B, int64_t S, int64_t heads, int64_t dh) {
+  NZ_DISPATCH(attn_split_qkv, qkv, qkv, q, k, v, B, S, heads, dh);
+}
+void attn_split_qkv_bwd(const Tensor& dq, const Tensor& dk, const Tensor& dv,
+                        Tensor& dqkv, int64_t B, int64_t S, int64_t heads, int64_t dh) {
+  NZ_DISPATCH(attn_split_qkv_bwd, dq, dq, dk, dv, dqkv, B, S, heads, dh);
+}
+void attn_merge(const Tensor& ctx, Tensor& out, int64_t B, int64_t S,
+                int64_t heads, int64_t dh) {
+  NZ_DISPATCH(attn_merge, ctx, ctx, out, B, S, heads, dh);
+}
+void attn_merge_bwd(const Tensor& dout, Tensor& dctx, int64_t B, int64_t S,
+                    int64_t heads, int64_t dh) {
+  NZ_DISPATCH(attn_merge_bwd, dout, dout, dctx, B, S, heads, dh);
+}
+}}

@@ -1,0 +1,798 @@
+// CDNA4 kernel set, part 1: K2 (bias+GELU), K3 (LayerNorm), K4 (softmax),
+// K6/K7 (Adam/SGD), K8 (elementwise family), K9 (losses), K10 (reductions),
+// attention layout ops. Design notes per /opt/skills/guides/cdna_hip_programming.md:
+// 64-wide waves (no 32-lane warp idioms), bf16 vectorized as ushort4/short8
+// where hot (Guideline 13), two-stage deterministic reductions (no atomics
+// on the loss scalar), block-per-row norm/softmax with wave shuffles.
+#include "src/hip/hip_common.h"
+#include "src/hip/hip_ops.h"
+
+namespace nz {
+namespace hip {
+
+// ------------------------------------------------------------------ fill ---
+template <typename T>
+__global__ void fill_k(T* p, int64_t n, float v) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) st(p, i, v);
+}
+
+void fill(Tensor& t, float value) {
+  hipStream_t s = stream_of(t);
+  int64_t n = t.numel();
+  dim3 g(grid_1d(n, 256)), b(256);
+  if (t.dtype() == DType::F32)
+    fill_k<float><<<g, b, 0, s>>>((float*)t.data(), n, value);
+  else
+    fill_k<uint16_t><<<g, b, 0, s>>>((uint16_t*)t.data(), n, value);
+  NZ_KERNEL_CHECK();
+}
+
+// ------------------------------------------------------------------- rng ---
+template <typename T, bool NORMAL>
+__global__ void rand_k(T* p, int64_t n, uint64_t seed, float a, float b) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float v = NORMAL ? (a + b * rng_normal(seed, i))
+                     : (a + (b - a) * rng_uniform01(seed, i));
+    st(p, i, v);
+  }
+}
+
+void rand_uniform(Tensor& t, uint64_t seed, float lo, float hi) {
+  hipStream_t s = stream_of(t);
+  int64_t n = t.numel();
+  dim3 g(grid_1d(n, 256)), b(256);
+  if (t.dtype() == DType::F32)
+    rand_k<float, false><<<g, b, 0, s>>>((float*)t.data(), n, seed, lo, hi);
+  else
+    rand_k<uint16_t, false><<<g, b, 0, s>>>((uint16_t*)t.data(), n, seed, lo, hi);
+  NZ_KERNEL_CHECK();
+}
+void rand_normal(Tensor& t, uint64_t seed, float mean, float std) {
+  hipStream_t s = stream_of(t);
+  int64_t n = t.numel();
+  dim3 g(grid_1d(n, 256)), b(256);
+  if (t.dtype() == DType::F32)
+    rand_k<float, true><<<g, b, 0, s>>>((float*)t.data(), n, seed, mean, std);
+  else
+    rand_k<uint16_t, true><<<g, b, 0, s>>>((uint16_t*)t.data(), n, seed, mean, std);
+  NZ_KERNEL_CHECK();
+}
+
+// ------------------------------------------------------------- cast/copy ---
+__global__ void cast_f2b_k(uint16_t* d, const float* src, int64_t n) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) d[i] = f2b(src[i]);
+}
+__global__ void cast_b2f_k(float* d, const uint16_t* src, int64_t n) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) d[i] = b2f(src[i]);
+}
+
+void cast_copy(Tensor& dst, const Tensor& src) {
+  NZ_CHECK(dst.numel() == src.numel(), "cast_copy numel mismatch");
+  hipStream_t s = stream_of(dst);
+  int64_t n = src.numel();
+  if (dst.dtype() == src.dtype()) {
+    NZ_HIP_CHECK_K(hipMemcpyAsync(dst.data(), src.data(), src.bytes(),
+                                  hipMemcpyDeviceToDevice, s));
+    return;
+  }
+  dim3 g(grid_1d(n, 256)), b(256);
+  if (src.dtype() == DType::F32)
+    cast_f2b_k<<<g, b, 0, s>>>((uint16_t*)dst.data(), (const float*)src.data(), n);
+  else
+    cast_b2f_k<<<g, b, 0, s>>>((float*)dst.data(), (const uint16_t*)src.data(), n);
+  NZ_KERNEL_CHECK();
+}
+
+// ------------------------------------------------------------- add/scale ---
+// bf16 += bf16 vectorized: 4 bf16 per thread (ushort4 = 8 B/lane).
+__global__ void add_b16_k(uint16_t* d, const uint16_t* src, int64_t n, float a) {
+  int64_t i4 = (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (; i4 + 3 < n; i4 += stride) {
+    ushort4 dv = *(const ushort4*)(d + i4);
+    ushort4 sv = *(const ushort4*)(src + i4);
+    dv.x = f2b(b2f(dv.x) + a * b2f(sv.x));
+    dv.y = f2b(b2f(dv.y) + a * b2f(sv.y));
+    dv.z = f2b(b2f(dv.z) + a * b2f(sv.z));
+    dv.w = f2b(b2f(dv.w) + a * b2f(sv.w));
+    *(ushort4*)(d + i4) = dv;
+  }
+  // tail (n % 4 elements), done once by the single global thread 0
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    for (int64_t i = n & ~int64_t(3); i < n; ++i)
+      d[i] = f2b(b2f(d[i]) + a * b2f(src[i]));
+}
+__global__ void add_f32_k(float* d, const float* src, int64_t n, float a) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) d[i] += a * src[i];
+}
+__global__ void add_mixed_k(uint16_t* d, const float* src, int64_t n, float a) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) d[i] = f2b(b2f(d[i]) + a * src[i]);
+}
+
+void add_inplace(Tensor& dst, const Tensor& src, float alpha) {
+  NZ_CHECK(dst.numel() == src.numel(), "add numel mismatch");
+  hipStream_t s = stream_of(dst);
+  int64_t n = dst.numel();
+  dim3 b(256);
+  if (dst.dtype() == DType::BF16 && src.dtype() == DType::BF16) {
+    dim3 g(grid_1d((n + 3) / 4, 256));
+    add_b16_k<<<g, b, 0, s>>>((uint16_t*)dst.data(), (const uint16_t*)src.data(), n, alpha);
+  } else if (dst.dtype() == DType::F32 && src.dtype() == DType::F32) {
+    add_f32_k<<<dim3(grid_1d(n, 256)), b, 0, s>>>((float*)dst.data(),
+                                                  (const float*)src.data(), n, alpha);
+  } else if (dst.dtype() == DType::BF16 && src.dtype() == DType::F32) {
+    add_mixed_k<<<dim3(grid_1d(n, 256)), b, 0, s>>>((uint16_t*)dst.data(),
+                                                    (const float*)src.data(), n, alpha);
+  } else {
+    NZ_CHECK(false, "unsupported add dtype combo");
+  }
+  NZ_KERNEL_CHECK();
+}
+
+template <typename T>
+__global__ void scale_k(T* d, int64_t n, float a) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) st(d, i, ld(d, i) * a);
+}
+void scale_inplace(Tensor& dst, float alpha) {
+  hipStream_t s = stream_of(dst);
+  int64_t n = dst.numel();
+  dim3 g(grid_1d(n, 256)), b(256);
+  if (dst.dtype() == DType::F32)
+    scale_k<float><<<g, b, 0, s>>>((float*)dst.data(), n, alpha);
+  else
+    scale_k<uint16_t><<<g, b, 0, s>>>((uint16_t*)dst.data(), n, alpha);
+  NZ_KERNEL_CHECK();
+}
+
+// ------------------------------------------------------------- transpose ---
+// LDS-tiled 64x64 transpose with +1-element padded rows (bank-conflict fix).
+template <typename T>
+__global__ void transpose_k(const T* __restrict__ src, T* __restrict__ dst,
+                            int64_t R, int64_t C) {
+  __shared__ T tile[64][65];
+  int64_t c0 = blockIdx.x * 64, r0 = blockIdx.y * 64;
+  int tx = threadIdx.x & 63, ty = threadIdx.x >> 6;  // 256 thr: 64 x 4
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    int64_t r = r0 + ty + i * 4, c = c0 + tx;
+    if (r < R && c < C) tile[ty + i * 4][tx] = src[r * C + c];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    int64_t r = c0 + ty + i * 4, c = r0 + tx;  // transposed coords
+    if (r < C && c < R) dst[r * R + c] = tile[tx][ty + i * 4];
+  }
+}
+void transpose2d(const Tensor& src, Tensor& dst) {
+  int64_t R = src.rows(), C = src.cols();
+  NZ_CHECK(dst.rows() == C && dst.cols() == R, "transpose shape mismatch");
+  hipStream_t s = stream_of(src);
+  dim3 g((C + 63) / 64, (R + 63) / 64), b(256);
+  if (src.dtype() == DType::F32)
+    transpose_k<float><<<g, b, 0, s>>>((const float*)src.data(), (float*)dst.data(), R, C);
+  else
+    transpose_k<uint16_t><<<g, b, 0, s>>>((const uint16_t*)src.data(),
+                                          (uint16_t*)dst.data(), R, C);
+  NZ_KERNEL_CHECK();
+}
+
+// ---------------------------------------------------------------- colsum ---
+// dst[j] = beta*dst[j] + sum_m src[m,j]. One thread per column; consecutive
+// threads read consecutive columns -> fully coalesced row sweeps.
+template <typename T, typename TD>
+__global__ void colsum_k(const T* __restrict__ src, TD* __restrict__ dst,
+                         int64_t M, int64_t N, float beta) {
+  int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  if (j >= N) return;
+  float acc = 0.f;
+  for (int64_t i = 0; i < M; ++i) acc += ld(src, i * N + j);
+  float prev = beta != 0.f ? beta * ld(dst, j) : 0.f;
+  st(dst, j, prev + acc);
+}
+void colsum(const Tensor& src, Tensor& dst, float beta) {
+  int64_t M = src.rows(), N = src.cols();
+  NZ_CHECK(dst.numel() == N, "colsum dst size");
+  hipStream_t s = stream_of(src);
+  dim3 g((N + 255) / 256), b(256);
+  if (src.dtype() == DType::BF16)
+    colsum_k<uint16_t, uint16_t><<<g, b, 0, s>>>((const uint16_t*)src.data(),
+                                                 (uint16_t*)dst.data(), M, N, beta);
+  else
+    colsum_k<float, float><<<g, b, 0, s>>>((const float*)src.data(),
+                                           (float*)dst.data(), M, N, beta);
+  NZ_KERNEL_CHECK();
+}
+
+// --------------------------------------------------------- K2: bias+GELU ---
+template <typename T>
+__global__ void bias_gelu_fwd_k(const T* __restrict__ x, const T* __restrict__ bias,
+                                T* __restrict__ y, T* __restrict__ z, int64_t n,
+                                int64_t N) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float v = ld(x, i) + (bias ? ld(bias, i % N) : 0.f);
+    if (z) st(z, i, v);
+    st(y, i, gelu_fwd_scalar(v));
+  }
+}
+void bias_gelu_fwd(const Tensor& x, const Tensor* bias, Tensor& y, Tensor* save_pre) {
+  hipStream_t s = stream_of(x);
+  int64_t n = x.numel(), N = x.cols();
+  dim3 g(grid_1d(n, 256)), b(256);
+  if (x.dtype() == DType::BF16)
+    bias_gelu_fwd_k<uint16_t><<<g, b, 0, s>>>(
+        (const uint16_t*)x.data(), bias ? (const uint16_t*)bias->data() : nullptr,
+        (uint16_t*)y.data(), save_pre ? (uint16_t*)save_pre->data() : nullptr, n, N);
+  else
+    bias_gelu_fwd_k<float><<<g, b, 0, s>>>(
+        (const float*)x.data(), bias ? (const float*)bias->data() : nullptr,
+        (float*)y.data(), save_pre ? (float*)save_pre->data() : nullptr, n, N);
+  NZ_KERNEL_CHECK();
+}
+
+// dz = dy * gelu'(z), vectorized 4-wide for bf16.
+__global__ void gelu_bwd_b16_k(const uint16_t* __restrict__ dy,
+                               const uint16_t* __restrict__ z,
+                               uint16_t* __restrict__ dz, int64_t n4) {
+  int64_t i4 = (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (; i4 < n4; i4 += stride) {
+    ushort4 dyv = *(const ushort4*)(dy + i4);
+    ushort4 zv = *(const ushort4*)(z + i4);
+    ushort4 o;
+    o.x = f2b(b2f(dyv.x) * gelu_bwd_scalar(b2f(zv.x)));
+    o.y = f2b(b2f(dyv.y) * gelu_bwd_scalar(b2f(zv.y)));
+    o.z = f2b(b2f(dyv.z) * gelu_bwd_scalar(b2f(zv.z)));
+    o.w = f2b(b2f(dyv.w) * gelu_bwd_scalar(b2f(zv.w)));
+    *(ushort4*)(dz + i4) = o;
+  }
+}
+template <typename T>
+__global__ void gelu_bwd_k(const T* __restrict__ dy, const T* __restrict__ z,
+                           T* __restrict__ dz, int64_t n) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) st(dz, i, ld(dy, i) * gelu_bwd_scalar(ld(z, i)));
+}
+void bias_gelu_bwd(const Tensor& dy, const Tensor& z, Tensor& dz, Tensor* dbias) {
+  hipStream_t s = stream_of(dy);
+  int64_t n = dy.numel();
+  dim3 b(256);
+  if (dy.dtype() == DType::BF16 && (n & 3) == 0) {
+    gelu_bwd_b16_k<<<dim3(grid_1d(n / 4, 256)), b, 0, s>>>(
+        (const uint16_t*)dy.data(), (const uint16_t*)z.data(),
+        (uint16_t*)dz.data(), n);
+  } else if (dy.dtype() == DType::BF16) {
+    gelu_bwd_k<uint16_t><<<dim3(grid_1d(n, 256)), b, 0, s>>>(
+        (const uint16_t*)dy.data(), (const uint16_t*)z.data(), (uint16_t*)dz.data(), n);
+  } else {
+    gelu_bwd_k<float><<<dim3(grid_1d(n, 256)), b, 0, s>>>(
+        (const float*)dy.data(), (const float*)z.data(), (float*)dz.data(), n);
+  }
+  NZ_KERNEL_CHECK();
+  if (dbias) colsum(dz, *dbias, 0.f);
+}
+
+// --------------------------------------------------------- K3: LayerNorm ---
+// One 256-thread block per row; f32 accumulation; mean/rstd cached for bwd.
+template <typename T>
+__global__ void layernorm_fwd_k(const T* __restrict__ x, const T* __restrict__ g,
+                                const T* __restrict__ be, T* __restrict__ y,
+                                float* __restrict__ mean, float* __restrict__ rstd,
+                                int64_t R, int64_t N, float eps) {
+  __shared__ float lds4[4];
+  for (int64_t row = blockIdx.x; row < R; row += gridDim.x) {
+    const T* xr = x + row * N;
+    float s = 0.f;
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x) s += ld(xr, j);
+    float mu = block_sum<256>(s, lds4) / N;
+    __syncthreads();
+    float v = 0.f;
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x) {
+      float d = ld(xr, j) - mu;
+      v += d * d;
+    }
+    float rs = rsqrtf(block_sum<256>(v, lds4) / N + eps);
+    if (threadIdx.x == 0) { mean[row] = mu; rstd[row] = rs; }
+    T* yr = y + row * N;
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x)
+      st(yr, j, (ld(xr, j) - mu) * rs * ld(g, j) + ld(be, j));
+    __syncthreads();
+  }
+}
+void layernorm_fwd(const Tensor& x, const Tensor& gamma, const Tensor& beta,
+                   Tensor& y, Tensor& mean, Tensor& rstd, float eps) {
+  hipStream_t s = stream_of(x);
+  int64_t R = x.rows(), N = x.cols();
+  dim3 g(grid_1d(R, 1, 2048)), b(256);
+  if (x.dtype() == DType::BF16)
+    layernorm_fwd_k<uint16_t><<<g, b, 0, s>>>(
+        (const uint16_t*)x.data(), (const uint16_t*)gamma.data(),
+        (const uint16_t*)beta.data(), (uint16_t*)y.data(), (float*)mean.data(),
+        (float*)rstd.data(), R, N, eps);
+  else
+    layernorm_fwd_k<float><<<g, b, 0, s>>>(
+        (const float*)x.data(), (const float*)gamma.data(),
+        (const float*)beta.data(), (float*)y.data(), (float*)mean.data(),
+        (float*)rstd.data(), R, N, eps);
+  NZ_KERNEL_CHECK();
+}
+
+template <typename T>
+__global__ void layernorm_dx_k(const T* __restrict__ dy, const T* __restrict__ x,
+                               const T* __restrict__ g, const float* __restrict__ mean,
+                               const float* __restrict__ rstd, T* __restrict__ dx,
+                               int64_t R, int64_t N, float beta_dx) {
+  __shared__ float lds4[4];
+  for (int64_t row = blockIdx.x; row < R; row += gridDim.x) {
+    const T* xr = x + row * N;
+    const T* dyr = dy + row * N;
+    float mu = mean[row], rs = rstd[row];
+    float s1 = 0.f, s2 = 0.f;
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x) {
+      float xh = (ld(xr, j) - mu) * rs;
+      float dyg = ld(dyr, j) * ld(g, j);
+      s1 += dyg;
+      s2 += dyg * xh;
+    }
+    float sum_dy = block_sum<256>(s1, lds4);
+    __syncthreads();
+    float sum_dyxh = block_sum<256>(s2, lds4);
+    T* dxr = dx + row * N;
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x) {
+      float xh = (ld(xr, j) - mu) * rs;
+      float dyg = ld(dyr, j) * ld(g, j);
+      float v = rs * (dyg - sum_dy / N - xh * sum_dyxh / N);
+      float prev = beta_dx != 0.f ? beta_dx * ld(dxr, j) : 0.f;
+      st(dxr, j, prev + v);
+    }
+    __syncthreads();
+  }
+}
+// dgamma[j] = sum_i dy[i,j]*xhat[i,j]; dbeta[j] = sum_i dy[i,j]
+template <typename T>
+__global__ void layernorm_dgb_k(const T* __restrict__ dy, const T* __restrict__ x,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ rstd,
+                                T* __restrict__ dgamma, T* __restrict__ dbeta,
+                                int64_t R, int64_t N) {
+  int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  if (j >= N) return;
+  float sg = 0.f, sb = 0.f;
+  for (int64_t i = 0; i < R; ++i) {
+    float dyv = ld(dy, i * N + j);
+    float xh = (ld(x, i * N + j) - mean[i]) * rstd[i];
+    sg += dyv * xh;
+    sb += dyv;
+  }
+  st(dgamma, j, sg);
+  st(dbeta, j, sb);
+}
+void layernorm_bwd(const Tensor& dy, const Tensor& x, const Tensor& gamma,
+                   const Tensor& mean, const Tensor& rstd, Tensor& dx,
+                   Tensor& dgamma, Tensor& dbeta, float beta_dx) {
+  hipStream_t s = stream_of(dy);
+  int64_t R = x.rows(), N = x.cols();
+  dim3 b(256);
+  if (x.dtype() == DType::BF16) {
+    layernorm_dx_k<uint16_t><<<dim3(grid_1d(R, 1, 2048)), b, 0, s>>>(
+        (const uint16_t*)dy.data(), (const uint16_t*)x.data(),
+        (const uint16_t*)gamma.data(), (const float*)mean.data(),
+        (const float*)rstd.data(), (uint16_t*)dx.data(), R, N, beta_dx);
+    layernorm_dgb_k<uint16_t><<<dim3((N + 255) / 256), b, 0, s>>>(
+        (const uint16_t*)dy.data(), (const uint16_t*)x.data(),
+        (const float*)mean.data(), (const float*)rstd.data(),
+        (uint16_t*)dgamma.data(), (uint16_t*)dbeta.data(), R, N);
+  } else {
+    layernorm_dx_k<float><<<dim3(grid_1d(R, 1, 2048)), b, 0, s>>>(
+        (const float*)dy.data(), (const float*)x.data(), (const float*)gamma.data(),
+        (const float*)mean.data(), (const float*)rstd.data(), (float*)dx.data(),
+        R, N, beta_dx);
+    layernorm_dgb_k<float><<<dim3((N + 255) / 256), b, 0, s>>>(
+        (const float*)dy.data(), (const float*)x.data(), (const float*)mean.data(),
+        (const float*)rstd.data(), (float*)dgamma.data(), (float*)dbeta.data(), R, N);
+  }
+  NZ_KERNEL_CHECK();
+}
+
+// ----------------------------------------------------------- K4: softmax ---
+// Block per row, online-style: pass1 max, pass2 exp+sum (f32), pass3 scale.
+template <typename T>
+__global__ void softmax_fwd_k(const T* __restrict__ x, T* __restrict__ y,
+                              int64_t R, int64_t N, float scale) {
+  __shared__ float lds4[4];
+  for (int64_t row = blockIdx.x; row < R; row += gridDim.x) {
+    const T* xr = x + row * N;
+    float m = -INFINITY;
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x)
+      m = fmaxf(m, scale * ld(xr, j));
+    m = block_max<256>(m, lds4);
+    __syncthreads();
+    float sum = 0.f;
+    T* yr = y + row * N;
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x) {
+      float e = __expf(scale * ld(xr, j) - m);
+      st(yr, j, e);
+      sum += e;
+    }
+    float inv = 1.0f / block_sum<256>(sum, lds4);
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x)
+      st(yr, j, ld(yr, j) * inv);
+    __syncthreads();
+  }
+}
+void softmax_fwd(const Tensor& x, Tensor& y, float scale) {
+  hipStream_t s = stream_of(x);
+  int64_t R = x.rows(), N = x.cols();
+  dim3 g(grid_1d(R, 1, 4096)), b(256);
+  if (x.dtype() == DType::BF16)
+    softmax_fwd_k<uint16_t><<<g, b, 0, s>>>((const uint16_t*)x.data(),
+                                            (uint16_t*)y.data(), R, N, scale);
+  else
+    softmax_fwd_k<float><<<g, b, 0, s>>>((const float*)x.data(), (float*)y.data(),
+                                         R, N, scale);
+  NZ_KERNEL_CHECK();
+}
+
+template <typename T>
+__global__ void softmax_bwd_k(const T* __restrict__ dy, const T* __restrict__ y,
+                              T* __restrict__ dx, int64_t R, int64_t N, float scale) {
+  __shared__ float lds4[4];
+  for (int64_t row = blockIdx.x; row < R; row += gridDim.x) {
+    const T* dyr = dy + row * N;
+    const T* yr = y + row * N;
+    float dot = 0.f;
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x)
+      dot += ld(dyr, j) * ld(yr, j);
+    dot = block_sum<256>(dot, lds4);
+    T* dxr = dx + row * N;
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x)
+      st(dxr, j, scale * ld(yr, j) * (ld(dyr, j) - dot));
+    __syncthreads();
+  }
+}
+void softmax_bwd(const Tensor& dy, const Tensor& y, Tensor& dx, float scale) {
+  hipStream_t s = stream_of(dy);
+  int64_t R = y.rows(), N = y.cols();
+  dim3 g(grid_1d(R, 1, 4096)), b(256);
+  if (y.dtype() == DType::BF16)
+    softmax_bwd_k<uint16_t><<<g, b, 0, s>>>((const uint16_t*)dy.data(),
+                                            (const uint16_t*)y.data(),
+                                            (uint16_t*)dx.data(), R, N, scale);
+  else
+    softmax_bwd_k<float><<<g, b, 0, s>>>((const float*)dy.data(),
+                                         (const float*)y.data(), (float*)dx.data(),
+                                         R, N, scale);
+  NZ_KERNEL_CHECK();
+}
+
+// --------------------------------------------------------- K6/K7: optim ---
+// HBM-bound: vectorized 4 elements/thread (float4 state, ushort4 params).
+template <typename TP, typename TG>
+__global__ void adam_k(TP* __restrict__ param, const TG* __restrict__ grad,
+                       float* __restrict__ master, float* __restrict__ m,
+                       float* __restrict__ v, int64_t n, float lr, float b1,
+                       float b2, float eps, float bc1, float bc2, float gs,
+                       float wd) {
+  int64_t i4 = (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (; i4 < n; i4 += stride) {
+    int lim = (int)min((int64_t)4, n - i4);
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      if (u >= lim) break;
+      int64_t i = i4 + u;
+      float g = ld(grad, i) * gs;
+      float ma = master[i];
+      if (wd != 0.f) g += wd * ma;
+      float mi = b1 * m[i] + (1 - b1) * g;
+      float vi = b2 * v[i] + (1 - b2) * g * g;
+      m[i] = mi;
+      v[i] = vi;
+      float upd = lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+      ma -= upd;
+      master[i] = ma;
+      st(param, i, ma);
+    }
+  }
+}
+void adam_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor& m,
+               Tensor& v, float lr, float b1, float b2, float eps, int step,
+               float gs, float wd) {
+  hipStream_t s = stream_of(param);
+  int64_t n = param.numel();
+  float bc1 = 1.f - powf(b1, (float)step), bc2 = 1.f - powf(b2, (float)step);
+  dim3 g(grid_1d((n + 3) / 4, 256)), b(256);
+  if (param.dtype() == DType::BF16)
+    adam_k<uint16_t, uint16_t><<<g, b, 0, s>>>(
+        (uint16_t*)param.data(), (const uint16_t*)grad.data(),
+        (float*)master.data(), (float*)m.data(), (float*)v.data(), n, lr, b1, b2,
+        eps, bc1, bc2, gs, wd);
+  else
+    adam_k<float, float><<<g, b, 0, s>>>(
+        (float*)param.data(), (const float*)grad.data(), (float*)master.data(),
+        (float*)m.data(), (float*)v.data(), n, lr, b1, b2, eps, bc1, bc2, gs, wd);
+  NZ_KERNEL_CHECK();
+}
+
+template <typename TP>
+__global__ void sgd_k(TP* __restrict__ param, const TP* __restrict__ grad,
+                      float* __restrict__ master, float* __restrict__ mom,
+                      int64_t n, float lr, float mu, float gs) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float g = ld(grad, i) * gs;
+    if (mom) {
+      float mi = mu * mom[i] + g;
+      mom[i] = mi;
+      g = mi;
+    }
+    float ma = master[i] - lr * g;
+    master[i] = ma;
+    st(param, i, ma);
+  }
+}
+void sgd_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor* momentum,
+              float lr, float mu, float gs) {
+  hipStream_t s = stream_of(param);
+  int64_t n = param.numel();
+  dim3 g(grid_1d(n, 256)), b(256);
+  if (param.dtype() == DType::BF16)
+    sgd_k<uint16_t><<<g, b, 0, s>>>((uint16_t*)param.data(),
+                                    (const uint16_t*)grad.data(),
+                                    (float*)master.data(),
+                                    momentum ? (float*)momentum->data() : nullptr,
+                                    n, lr, mu, gs);
+  else
+    sgd_k<float><<<g, b, 0, s>>>((float*)param.data(), (const float*)grad.data(),
+                                 (float*)master.data(),
+                                 momentum ? (float*)momentum->data() : nullptr, n,
+                                 lr, mu, gs);
+  NZ_KERNEL_CHECK();
+}
+
+// ------------------------------------------------------------ K9: losses ---
+// Two-stage deterministic reduction: per-block partials, then single-block
+// combine (no float atomics -> bit-stable run to run).
+template <typename T>
+__global__ void mse_partial_k(const T* __restrict__ y, const T* __restrict__ t,
+                              T* __restrict__ dy, float* __restrict__ partials,
+                              int64_t n, float gs) {
+  __shared__ float lds4[4];
+  float acc = 0.f;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float inv_n = 1.0f / float(n);
+  for (; i < n; i += stride) {
+    float d = ld(y, i) - ld(t, i);
+    acc += d * d;
+    if (dy) st(dy, i, 2.0f * d * inv_n * gs);
+  }
+  float bs = block_sum<256>(acc, lds4);
+  if (threadIdx.x == 0) partials[blockIdx.x] = bs;
+}
+__global__ void combine_partials_k(const float* __restrict__ partials, int nb,
+                                   float* __restrict__ out, float scale) {
+  __shared__ float lds4[4];
+  float acc = 0.f;
+  for (int i = threadIdx.x; i < nb; i += blockDim.x) acc += partials[i];
+  float s = block_sum<256>(acc, lds4);
+  if (threadIdx.x == 0) out[0] = s * scale;
+}
+
+void mse_loss(const Tensor& y, const Tensor& target, Tensor& loss_out, Tensor* dy,
+              float gs) {
+  hipStream_t s = stream_of(y);
+  int64_t n = y.numel();
+  int nb = (int)grid_1d((n + 3) / 4, 256, 1024);
+  Tensor partials(y.ctx(), DType::F32, Shape{nb});
+  dim3 g(nb), b(256);
+  if (y.dtype() == DType::BF16)
+    mse_partial_k<uint16_t><<<g, b, 0, s>>>(
+        (const uint16_t*)y.data(), (const uint16_t*)target.data(),
+        dy ? (uint16_t*)dy->data() : nullptr, (float*)partials.data(), n, gs);
+  else
+    mse_partial_k<float><<<g, b, 0, s>>>(
+        (const float*)y.data(), (const float*)target.data(),
+        dy ? (float*)dy->data() : nullptr, (float*)partials.data(), n, gs);
+  combine_partials_k<<<dim3(1), b, 0, s>>>((const float*)partials.data(), nb,
+                                           (float*)loss_out.data(), 1.0f / n);
+  NZ_KERNEL_CHECK();
+}
+
+// CE over rows (f32 logits; labels hold integral class ids in f32).
+__global__ void ce_loss_k(const float* __restrict__ logits,
+                          const float* __restrict__ labels,
+                          float* __restrict__ dlogits,
+                          float* __restrict__ partials, int64_t R, int64_t N,
+                          float gs) {
+  __shared__ float lds4[4];
+  float row_losses = 0.f;
+  for (int64_t row = blockIdx.x; row < R; row += gridDim.x) {
+    const float* lr = logits + row * N;
+    float m = -INFINITY;
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x) m = fmaxf(m, lr[j]);
+    m = block_max<256>(m, lds4);
+    __syncthreads();
+    float z = 0.f;
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x) z += __expf(lr[j] - m);
+    z = block_sum<256>(z, lds4);
+    int64_t y = (int64_t)labels[row];
+    if (threadIdx.x == 0)
+      row_losses += -(lr[y] - m - __logf(z));
+    if (dlogits) {
+      float* dlr = dlogits + row * N;
+      for (int64_t j = threadIdx.x; j < N; j += blockDim.x) {
+        float p = __expf(lr[j] - m) / z;
+        dlr[j] = (p - (j == y ? 1.f : 0.f)) / float(R) * gs;
+      }
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) partials[blockIdx.x] = row_losses;
+}
+void ce_loss(const Tensor& logits, const Tensor& labels, Tensor& loss_out,
+             Tensor* dlogits, float gs) {
+  NZ_CHECK(logits.dtype() == DType::F32, "hip ce_loss expects f32 logits");
+  hipStream_t s = stream_of(logits);
+  int64_t R = logits.rows(), N = logits.cols();
+  int nb = (int)grid_1d(R, 1, 1024);
+  Tensor partials(logits.ctx(), DType::F32, Shape{nb});
+  ce_loss_k<<<dim3(nb), dim3(256), 0, s>>>(
+      (const float*)logits.data(), (const float*)labels.data(),
+      dlogits ? (float*)dlogits->data() : nullptr, (float*)partials.data(), R, N, gs);
+  combine_partials_k<<<dim3(1), dim3(256), 0, s>>>(
+      (const float*)partials.data(), nb, (float*)loss_out.data(), 1.0f / R);
+  NZ_KERNEL_CHECK();
+}
+
+// ------------------------------------------------------- K10: reductions ---
+template <typename T, bool SQ>
+__global__ void reduce_partial_k(const T* __restrict__ src,
+                                 float* __restrict__ partials, int64_t n) {
+  __shared__ float lds4[4];
+  float acc = 0.f;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float v = ld(src, i);
+    acc += SQ ? v * v : v;
+  }
+  float bs = block_sum<256>(acc, lds4);
+  if (threadIdx.x == 0) partials[blockIdx.x] = bs;
+}
+template <bool SQ>
+static void reduce_impl(const Tensor& src, Tensor& out) {
+  hipStream_t s = stream_of(src);
+  int64_t n = src.numel();
+  int nb = (int)grid_1d((n + 3) / 4, 256, 1024);
+  Tensor partials(src.ctx(), DType::F32, Shape{nb});
+  if (src.dtype() == DType::BF16)
+    reduce_partial_k<uint16_t, SQ><<<dim3(nb), dim3(256), 0, s>>>(
+        (const uint16_t*)src.data(), (float*)partials.data(), n);
+  else
+    reduce_partial_k<float, SQ><<<dim3(nb), dim3(256), 0, s>>>(
+        (const float*)src.data(), (float*)partials.data(), n);
+  combine_partials_k<<<dim3(1), dim3(256), 0, s>>>((const float*)partials.data(),
+                                                   nb, (float*)out.data(), 1.0f);
+  NZ_KERNEL_CHECK();
+}
+void reduce_sum(const Tensor& src, Tensor& out) { reduce_impl<false>(src, out); }
+void l2norm_sq(const Tensor& src, Tensor& out) { reduce_impl<true>(src, out); }
+
+// -------------------------------------------- attention layout (K8 fam) ---
+// One thread per dh-element group of 2 (ushort2/float2 moves); dh >= 2.
+template <typename T, bool FORWARD>
+__global__ void attn_split_k(const T* __restrict__ a, T* __restrict__ q,
+                             T* __restrict__ k, T* __restrict__ v, int64_t total,
+                             int64_t S, int64_t heads, int64_t dh) {
+  int64_t H = heads * dh;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    int64_t d = i % dh, h = (i / dh) % heads, which = (i / (dh * heads)) % 3;
+    int64_t s_ = (i / (3 * H)) % S, b = i / (3 * H * S);
+    int64_t bhsd = ((b * heads + h) * S + s_) * dh + d;
+    T* dst3 = which == 0 ? q : (which == 1 ? k : v);
+    if (FORWARD)
+      dst3[bhsd] = a[i];
+    else
+      const_cast<T*>(a)[i] = dst3[bhsd];
+  }
+}
+void attn_split_qkv(const Tensor& qkv, Tensor& q, Tensor& k, Tensor& v,
+                    int64_t B, int64_t S, int64_t heads, int64_t dh) {
+  hipStream_t s = stream_of(qkv);
+  int64_t total = qkv.numel();
+  dim3 g(grid_1d(total, 256, 4096)), b(256);
+  if (qkv.dtype() == DType::BF16)
+    attn_split_k<uint16_t, true><<<g, b, 0, s>>>(
+        (const uint16_t*)qkv.data(), (uint16_t*)q.data(), (uint16_t*)k.data(),
+        (uint16_t*)v.data(), total, S, heads, dh);
+  else
+    attn_split_k<float, true><<<g, b, 0, s>>>(
+        (const float*)qkv.data(), (float*)q.data(), (float*)k.data(),
+        (float*)v.data(), total, S, heads, dh);
+  NZ_KERNEL_CHECK();
+}
+void attn_split_qkv_bwd(const Tensor& dq, const Tensor& dk, const Tensor& dv,
+                        Tensor& dqkv, int64_t B, int64_t S, int64_t heads,
+                        int64_t dh) {
+  hipStream_t s = stream_of(dqkv);
+  int64_t total = dqkv.numel();
+  dim3 g(grid_1d(total, 256, 4096)), b(256);
+  if (dqkv.dtype() == DType::BF16)
+    attn_split_k<uint16_t, false><<<g, b, 0, s>>>(
+        (const uint16_t*)dqkv.data(), (uint16_t*)dq.data(), (uint16_t*)dk.data(),
+        (uint16_t*)dv.data(), total, S, heads, dh);
+  else
+    attn_split_k<float, false><<<g, b, 0, s>>>(
+        (const float*)dqkv.data(), (float*)dq.data(), (float*)dk.data(),
+        (float*)dv.data(), total, S, heads, dh);
+  NZ_KERNEL_CHECK();
+}
+
+template <typename T, bool FORWARD>
+__global__ void attn_merge_k(const T* __restrict__ ctx, T* __restrict__ out,
+                             int64_t total, int64_t S, int64_t heads, int64_t dh) {
+  int64_t H = heads * dh;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {  // i indexes [B,S,H]
+    int64_t d = i % dh, h = (i / dh) % heads, s_ = (i / H) % S, b = i / (H * S);
+    int64_t bhsd = ((b * heads + h) * S + s_) * dh + d;
+    if (FORWARD)
+      out[i] = ctx[bhsd];
+    else
+      const_cast<T*>(ctx)[bhsd] = out[i];
+  }
+}
+void attn_merge(const Tensor& ctx, Tensor& out, int64_t B, int64_t S,
+                int64_t heads, int64_t dh) {
+  hipStream_t s = stream_of(ctx);
+  int64_t total = out.numel();
+  dim3 g(grid_1d(total, 256, 4096)), b(256);
+  if (ctx.dtype() == DType::BF16)
+    attn_merge_k<uint16_t, true><<<g, b, 0, s>>>((const uint16_t*)ctx.data(),
+                                                 (uint16_t*)out.data(), total, S,
+                                                 heads, dh);
+  else
+    attn_merge_k<float, true><<<g, b, 0, s>>>((const float*)ctx.data(),
+                                              (float*)out.data(), total, S, heads, dh);
+  NZ_KERNEL_CHECK();
+}
+void attn_merge_bwd(const Tensor& dout, Tensor& dctx, int64_t B, int64_t S,
+                    int64_t heads, int64_t dh) {
+  hipStream_t s = stream_of(dout);
+  int64_t total = dout.numel();
+  dim3 g(grid_1d(total, 256, 4096)), b(256);
+  if (dout.dtype() == DType::BF16)
+    attn_merge_k<uint16_t, false><<<g, b, 0, s>>>((const uint16_t*)dctx.data(),
+                                                  (uint16_t*)dout.data(), total, S,
+                                                  heads, dh);
+  else
+    attn_merge_k<float, false><<<g, b, 0, s>>>((const float*)dctx.data(),
+                                               (float*)dout.data(), total, S,
+                                               heads, dh);
+  NZ_KERNEL_CHECK();
+}
+
+}  // namespace hip
+}  // namespace nz

@@ -1,0 +1,258 @@
+// K1: hand-written CDNA4 bf16 GEMM (MFMA v_mfma_f32_16x16x32_bf16, f32
+// accumulate) with fused bias(+GELU) epilogue and strided-batch mode.
+//
+// Tier A structure ("correct-first", guide §5 anatomy): 128x128 macro-tile,
+// BK=64, 256 threads = 4 waves in a 2x2 grid, each wave owns a 64x64
+// sub-tile = 4x4 fragments of 16x16. A staged in LDS as [BM][BK], B staged
+// TRANSPOSED as [BN][BK] so both fragment reads are contiguous 16-byte
+// ds_read_b128 along k. Rows padded +8 bf16 (16 B) to break bank conflicts
+// while keeping 16-B alignment (guide §6 Guideline 4/17). Double-buffered
+// LDS with global loads issued before compute (T14 issue-early/write-late).
+// Tier B (glds + XCD swizzle, guide §5 ladder step 3) comes after this is
+// proven correct on hardware.
+#include "src/hip/hip_common.h"
+#include "src/hip/hip_ops.h"
+
+namespace nz {
+namespace hip {
+
+using short8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+namespace {
+
+constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int LDS_STRIDE = BK + 8;  // 72 elems = 144 B, 16-B aligned rows
+constexpr int THREADS = 256;
+
+struct GemmArgs {
+  const uint16_t* A;
+  const uint16_t* B;
+  uint16_t* C;
+  const uint16_t* bias;   // null or [N]
+  uint16_t* save_pre;     // null or [M,N]
+  int64_t M, N, K;
+  int64_t lda, ldb;       // row strides of the STORED matrices
+  int64_t sA, sB, sC;     // batch strides (elements)
+  float alpha, beta;
+  int fuse_gelu;
+};
+
+// Stage op(A) tile [BM][BK] (rows m, cols k) into lds with row stride
+// LDS_STRIDE. TRANS=false: stored [.,lda] row-major [M,K] -> vector loads
+// along k. TRANS=true: stored [K,M] -> vector loads along m, scatter to lds.
+template <bool TRANS>
+__device__ void stage_a(const uint16_t* __restrict__ A, uint16_t* __restrict__ lds,
+                        int64_t m0, int64_t k0, int64_t M, int64_t K, int64_t lda) {
+  constexpr int ELEMS = BM * BK / THREADS;  // 32 elems per thread, as 4 x 8
+#pragma unroll
+  for (int rep = 0; rep < ELEMS / 8; ++rep) {
+    int e = (rep * THREADS + (int)threadIdx.x) * 8;
+    if (!TRANS) {
+      int m = e / BK, k = e % BK;
+      int64_t gm = m0 + m, gk = k0 + k;
+      short8 v = {};
+      if (gm < M) v = *(const short8*)(A + gm * lda + gk);
+      *(short8*)(lds + m * LDS_STRIDE + k) = v;
+    } else {
+      int k = e / BM, m = e % BM;
+      int64_t gk = k0 + k, gm = m0 + m;
+      short8 v = {};
+      if (gm + 7 < M) {
+        v = *(const short8*)(A + gk * lda + gm);
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          v[i] = (gm + i < M) ? (short)A[gk * lda + gm + i] : (short)0;
+      }
+#pragma unroll
+      for (int i = 0; i < 8; ++i) lds[(m + i) * LDS_STRIDE + k] = (uint16_t)v[i];
+    }
+  }
+}
+
+// Stage op(B) tile into lds TRANSPOSED: lds[n][k] = op(B)[k][n].
+// TRANS=false: stored [K,N] -> vector loads along n, scatter.
+// TRANS=true:  stored [N,K] -> vector loads along k, direct.
+template <bool TRANS>
+__device__ void stage_b(const uint16_t* __restrict__ B, uint16_t* __restrict__ lds,
+                        int64_t n0, int64_t k0, int64_t N, int64_t K, int64_t ldb) {
+  constexpr int ELEMS = BN * BK / THREADS;
+#pragma unroll
+  for (int rep = 0; rep < ELEMS / 8; ++rep) {
+    int e = (rep * THREADS + (int)threadIdx.x) * 8;
+    if (!TRANS) {
+      int k = e / BN, n = e % BN;
+      int64_t gk = k0 + k, gn = n0 + n;
+      short8 v = {};
+      if (gn + 7 < N) {
+        v = *(const short8*)(B + gk * ldb + gn);
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          v[i] = (gn + i < N) ? (short)B[gk * ldb + gn + i] : (short)0;
+      }
+#pragma unroll
+      for (int i = 0; i < 8; ++i) lds[(n + i) * LDS_STRIDE + k] = (uint16_t)v[i];
+    } else {
+      int n = e / BK, k = e % BK;
+      int64_t gn = n0 + n, gk = k0 + k;
+      short8 v = {};
+      if (gn < N) v = *(const short8*)(B + gn * ldb + gk);
+      *(short8*)(lds + n * LDS_STRIDE + k) = v;
+    }
+  }
+}
+
+template <bool TA, bool TB>
+__launch_bounds__(THREADS) __global__ void gemm_bf16_k(GemmArgs args) {
+  __shared__ uint16_t As[2][BM * LDS_STRIDE];
+  __shared__ uint16_t Bs[2][BN * LDS_STRIDE];
+
+  const int64_t batch = blockIdx.z;
+  const uint16_t* A = args.A + batch * args.sA;
+  const uint16_t* B = args.B + batch * args.sB;
+  uint16_t* C = args.C + batch * args.sC;
+  uint16_t* save_pre = args.save_pre ? args.save_pre + batch * args.sC : nullptr;
+
+  const int64_t m0 = (int64_t)blockIdx.y * BM;
+  const int64_t n0 = (int64_t)blockIdx.x * BN;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;      // 4 waves
+  const int wr = wid >> 1, wc = wid & 1; // 2x2 wave grid, 64x64 each
+
+  f32x4 acc[4][4] = {};
+
+  const int64_t ntiles = args.K / BK;
+  stage_a<TA>(A, As[0], m0, 0, args.M, args.K, args.lda);
+  stage_b<TB>(B, Bs[0], n0, 0, args.N, args.K, args.ldb);
+  __syncthreads();
+
+  int cur = 0;
+  for (int64_t t = 0; t < ntiles; ++t) {
+    // Issue next tile's staging AFTER compute reads current LDS? No:
+    // stage into the other buffer now (loads overlap MFMA via scheduler),
+    // barrier afterwards protects buffer reuse two iterations out.
+    if (t + 1 < ntiles) {
+      stage_a<TA>(A, As[cur ^ 1], m0, (t + 1) * BK, args.M, args.K, args.lda);
+      stage_b<TB>(B, Bs[cur ^ 1], n0, (t + 1) * BK, args.N, args.K, args.ldb);
+    }
+    const uint16_t* Ab = As[cur];
+    const uint16_t* Bb = Bs[cur];
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 32) {
+      short8 a_frag[4], b_frag[4];
+      const int kbase = kk + (lane >> 4) * 8;
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        int row = wr * 64 + fm * 16 + (lane & 15);
+        a_frag[fm] = *(const short8*)(Ab + row * LDS_STRIDE + kbase);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = wc * 64 + fn * 16 + (lane & 15);
+        b_frag[fn] = *(const short8*)(Bb + col * LDS_STRIDE + kbase);
+      }
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[fm], b_frag[fn], acc[fm][fn], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // Epilogue: alpha/beta, +bias, save_pre, gelu, bf16 store.
+  // C/D fragment map (guide §3): col = lane&15, row = (lane>>4)*4 + r.
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int64_t grow = m0 + wr * 64 + fm * 16 + (lane >> 4) * 4 + r;
+        int64_t gcol = n0 + wc * 64 + fn * 16 + (lane & 15);
+        if (grow >= args.M || gcol >= args.N) continue;
+        int64_t idx = grow * args.N + gcol;
+        float v = args.alpha * acc[fm][fn][r];
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+        if (args.bias) v += b2f(args.bias[gcol]);
+        if (save_pre) save_pre[idx] = f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        C[idx] = f2b(v);
+      }
+    }
+  }
+}
+
+void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
+                 hipStream_t s) {
+  NZ_CHECK(args.K % BK == 0, "hip gemm requires K % 64 == 0, got K=" +
+                                 std::to_string(args.K));
+  dim3 grid((args.N + BN - 1) / BN, (args.M + BM - 1) / BM, batch);
+  dim3 block(THREADS);
+  if (!transA && !transB)
+    gemm_bf16_k<false, false><<<grid, block, 0, s>>>(args);
+  else if (!transA && transB)
+    gemm_bf16_k<false, true><<<grid, block, 0, s>>>(args);
+  else if (transA && !transB)
+    gemm_bf16_k<true, false><<<grid, block, 0, s>>>(args);
+  else
+    gemm_bf16_k<true, true><<<grid, block, 0, s>>>(args);
+  NZ_KERNEL_CHECK();
+}
+
+}  // namespace
+
+void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
+          float alpha, float beta, const ops::GemmEpilogue* epi) {
+  NZ_CHECK(A.dtype() == DType::BF16 && B.dtype() == DType::BF16 &&
+               C.dtype() == DType::BF16,
+           "hip gemm: bf16 tensors required");
+  GemmArgs args{};
+  args.M = C.rows();
+  args.N = C.cols();
+  args.K = transA ? A.rows() : A.cols();
+  NZ_CHECK((transA ? A.cols() : A.rows()) == args.M, "gemm M mismatch");
+  NZ_CHECK((transB ? B.rows() : B.cols()) == args.N, "gemm N mismatch");
+  NZ_CHECK((transB ? B.cols() : B.rows()) == args.K, "gemm K mismatch");
+  args.A = (const uint16_t*)A.data();
+  args.B = (const uint16_t*)B.data();
+  args.C = (uint16_t*)C.data();
+  args.lda = A.cols();
+  args.ldb = B.cols();
+  args.alpha = alpha;
+  args.beta = beta;
+  if (epi) {
+    args.bias = epi->bias ? (const uint16_t*)epi->bias->data() : nullptr;
+    args.save_pre = epi->save_pre ? (uint16_t*)epi->save_pre->data() : nullptr;
+    args.fuse_gelu = epi->gelu ? 1 : 0;
+  }
+  launch_gemm(args, transA, transB, 1, stream_of(C));
+}
+
+void gemm_batched(const Tensor& A, const Tensor& B, Tensor& C, bool transA,
+                  bool transB, float alpha, int64_t batch, int64_t M, int64_t N,
+                  int64_t K, int64_t sA, int64_t sB, int64_t sC) {
+  NZ_CHECK(A.dtype() == DType::BF16, "hip gemm_batched: bf16 required");
+  GemmArgs args{};
+  args.A = (const uint16_t*)A.data();
+  args.B = (const uint16_t*)B.data();
+  args.C = (uint16_t*)C.data();
+  args.M = M;
+  args.N = N;
+  args.K = K;
+  args.lda = transA ? M : K;
+  args.ldb = transB ? K : N;
+  args.sA = sA;
+  args.sB = sB;
+  args.sC = sC;
+  args.alpha = alpha;
+  args.beta = 0.f;
+  launch_gemm(args, transA, transB, batch, stream_of(C));
+}
+
+}  // namespace hip
+}  // namespace nz

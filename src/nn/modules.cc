@@ -1,0 +1,225 @@
+#include "src/nn/modules.h"
+
+#include <cmath>
+
+namespace nz {
+
+static DType param_dtype(Context* ctx) {
+  return ctx->device() == Device::HIP ? DType::BF16 : DType::F32;
+}
+
+// Accumulate gradient tensor g into var's grad (first write = copy).
+static void accum_into(Var& v, const Tensor& g) {
+  if (!v.grad_init) {
+    v.ensure_grad();
+    ops::cast_copy(v.grad, g);
+    v.grad_init = true;
+  } else {
+    ops::add_inplace(v.grad, g, 1.0f);
+  }
+}
+
+// ---------------------------------------------------------------- Linear ---
+Linear::Linear(Context* ctx, std::string name, int64_t in, int64_t out,
+               bool bias, bool gelu, uint64_t seed)
+    : has_bias(bias), fuse_gelu(gelu), in_dim(in), out_dim(out), ctx_(ctx) {
+  DType pd = param_dtype(ctx);
+  w.name = name + ".w";
+  w.data = Tensor(ctx, pd, Shape{in, out});
+  float std = std::sqrt(2.0f / float(in + out));  // Xavier
+  ops::rand_normal(w.data, seed, 0.f, std);
+  w.grad = Tensor(ctx, pd, Shape{in, out});
+  if (has_bias) {
+    b.name = name + ".b";
+    b.data = Tensor(ctx, pd, Shape{out});
+    ops::fill(b.data, 0.f);
+    b.grad = Tensor(ctx, pd, Shape{out});
+  }
+}
+
+void Linear::collect_params(std::vector<Param*>& out) {
+  out.push_back(&w);
+  if (has_bias) out.push_back(&b);
+}
+
+VarPtr Linear::forward(Tape& tape, VarPtr x) {
+  int64_t M = x->value.rows();
+  DType pd = param_dtype(ctx_);
+  auto y = make_var(Tensor(ctx_, pd, Shape{M, out_dim}));
+  Tensor z;  // pre-activation, saved for GELU backward
+  ops::GemmEpilogue epi;
+  epi.bias = has_bias ? &b.data : nullptr;
+  epi.gelu = fuse_gelu;
+  if (fuse_gelu) {
+    z = Tensor(ctx_, pd, Shape{M, out_dim});
+    epi.save_pre = &z;
+  }
+  ops::gemm(x->value, w.data, y->value, false, false, 1.0f, 0.0f, &epi);
+
+  tape.push([this, x, y, z]() {
+    Context* ctx = ctx_;
+    DType pd = param_dtype(ctx);
+    Tensor& dy = y->grad;
+    NZ_CHECK(dy.defined(), "linear backward without upstream grad");
+    Tensor dz;
+    if (fuse_gelu) {
+      dz = Tensor(ctx, pd, dy.shape());
+      NZ_CHECK(!has_bias || !b.grad_init, "param " + b.name + " used twice");
+      ops::bias_gelu_bwd(dy, z, dz, has_bias ? &b.grad : nullptr);
+      if (has_bias) { b.grad_init = true; b.mark_ready(); }
+    } else {
+      dz = dy;
+      if (has_bias) {
+        ops::colsum(dy, b.grad, b.grad_beta());
+        b.mark_ready();
+      }
+    }
+    // wgrad: dW[in,out] = x^T . dz
+    ops::gemm(x->value, dz, w.grad, true, false, 1.0f, w.grad_beta(), nullptr);
+    w.mark_ready();
+    // dgrad: dx[M,in] = dz . W^T
+    if (x->needs_grad) {
+      x->ensure_grad();
+      ops::gemm(dz, w.data, x->grad, false, true, 1.0f, x->grad_beta(), nullptr);
+    }
+  });
+  return y;
+}
+
+// ------------------------------------------------------------- LayerNorm ---
+LayerNorm::LayerNorm(Context* ctx, std::string name, int64_t d)
+    : dim(d), ctx_(ctx) {
+  DType pd = param_dtype(ctx);
+  gamma.name = name + ".gamma";
+  gamma.data = Tensor(ctx, pd, Shape{d});
+  ops::fill(gamma.data, 1.f);
+  gamma.grad = Tensor(ctx, pd, Shape{d});
+  beta.name = name + ".beta";
+  beta.data = Tensor(ctx, pd, Shape{d});
+  ops::fill(beta.data, 0.f);
+  beta.grad = Tensor(ctx, pd, Shape{d});
+}
+
+void LayerNorm::collect_params(std::vector<Param*>& out) {
+  out.push_back(&gamma);
+  out.push_back(&beta);
+}
+
+VarPtr LayerNorm::forward(Tape& tape, VarPtr x) {
+  int64_t R = x->value.rows();
+  auto y = make_var(Tensor(ctx_, x->value.dtype(), x->value.shape()));
+  Tensor mean(ctx_, DType::F32, Shape{R});
+  Tensor rstd(ctx_, DType::F32, Shape{R});
+  ops::layernorm_fwd(x->value, gamma.data, beta.data, y->value, mean, rstd, 1e-5f);
+
+  tape.push([this, x, y, mean, rstd]() {
+    Tensor& dy = y->grad;
+    NZ_CHECK(dy.defined(), "layernorm backward without upstream grad");
+    NZ_CHECK(!gamma.grad_init && !beta.grad_init, "LN params used twice");
+    if (x->needs_grad) {
+      x->ensure_grad();
+      ops::layernorm_bwd(dy, x->value, gamma.data, mean, rstd, x->grad,
+                         gamma.grad, beta.grad, x->grad_beta());
+    } else {
+      Tensor scratch_dx(ctx_, x->value.dtype(), x->value.shape());
+      ops::layernorm_bwd(dy, x->value, gamma.data, mean, rstd, scratch_dx,
+                         gamma.grad, beta.grad, 0.f);
+    }
+    gamma.grad_init = beta.grad_init = true;
+    gamma.mark_ready();
+    beta.mark_ready();
+  });
+  return y;
+}
+
+// --------------------------------------------------- MultiHeadAttention ---
+MultiHeadAttention::MultiHeadAttention(Context* ctx, std::string name,
+                                       int64_t hidden_, int64_t heads_,
+                                       int64_t seq_, int64_t batch_, uint64_t seed)
+    : qkv_proj(ctx, name + ".qkv", hidden_, 3 * hidden_, true, false, seed),
+      out_proj(ctx, name + ".out", hidden_, hidden_, true, false, seed + 1),
+      hidden(hidden_), heads(heads_), dh(hidden_ / heads_), seq(seq_),
+      batch(batch_), ctx_(ctx) {
+  NZ_CHECK(hidden % heads == 0, "hidden not divisible by heads");
+}
+
+void MultiHeadAttention::collect_params(std::vector<Param*>& out) {
+  qkv_proj.collect_params(out);
+  out_proj.collect_params(out);
+}
+
+VarPtr MultiHeadAttention::forward(Tape& tape, VarPtr x) {
+  // x: [B*S, H]
+  DType pd = x->value.dtype();
+  int64_t B = batch, S = seq, Hh = heads, D = dh;
+  float scale = 1.0f / std::sqrt(float(D));
+  VarPtr qkv = qkv_proj.forward(tape, x);  // [B*S, 3H]
+
+  Tensor q(ctx_, pd, Shape{B, Hh, S, D}), k(ctx_, pd, Shape{B, Hh, S, D}),
+      v(ctx_, pd, Shape{B, Hh, S, D});
+  ops::attn_split_qkv(qkv->value, q, k, v, B, S, Hh, D);
+  Tensor scores(ctx_, pd, Shape{B * Hh * S, S});
+  ops::gemm_batched(q, k, scores, false, true, 1.0f, B * Hh, S, S, D,
+                    S * D, S * D, S * S);
+  Tensor probs(ctx_, pd, Shape{B * Hh * S, S});
+  ops::softmax_fwd(scores, probs, scale);
+  Tensor ctxt(ctx_, pd, Shape{B, Hh, S, D});
+  ops::gemm_batched(probs, v, ctxt, false, false, 1.0f, B * Hh, S, D, S,
+                    S * S, S * D, S * D);
+  auto merged = make_var(Tensor(ctx_, pd, Shape{B * S, hidden}));
+  ops::attn_merge(ctxt, merged->value, B, S, Hh, D);
+
+  tape.push([this, qkv, merged, q, k, v, probs, pd, B, S, Hh, D, scale]() {
+    Tensor& dmerged = merged->grad;
+    NZ_CHECK(dmerged.defined(), "attention backward without upstream grad");
+    Tensor dctx(ctx_, pd, Shape{B, Hh, S, D});
+    ops::attn_merge_bwd(dmerged, dctx, B, S, Hh, D);
+    Tensor dprobs(ctx_, pd, Shape{B * Hh * S, S});
+    ops::gemm_batched(dctx, v, dprobs, false, true, 1.0f, B * Hh, S, S, D,
+                      S * D, S * D, S * S);
+    Tensor dv(ctx_, pd, Shape{B, Hh, S, D});
+    ops::gemm_batched(probs, dctx, dv, true, false, 1.0f, B * Hh, S, D, S,
+                      S * S, S * D, S * D);
+    Tensor dscores(ctx_, pd, Shape{B * Hh * S, S});
+    ops::softmax_bwd(dprobs, probs, dscores, scale);
+    Tensor dq(ctx_, pd, Shape{B, Hh, S, D});
+    ops::gemm_batched(dscores, k, dq, false, false, 1.0f, B * Hh, S, D, S,
+                      S * S, S * D, S * D);
+    Tensor dk(ctx_, pd, Shape{B, Hh, S, D});
+    ops::gemm_batched(dscores, q, dk, true, false, 1.0f, B * Hh, S, D, S,
+                      S * S, S * D, S * D);
+    qkv->ensure_grad();
+    NZ_CHECK(!qkv->grad_init, "qkv grad written twice");
+    ops::attn_split_qkv_bwd(dq, dk, dv, qkv->grad, B, S, Hh, D);
+    qkv->grad_init = true;
+  });
+  return out_proj.forward(tape, merged);
+}
+
+// ---------------------------------------------------------- residual add ---
+VarPtr residual_add(Tape& tape, VarPtr a, VarPtr b) {
+  Context* ctx = a->value.ctx();
+  auto out = make_var(Tensor(ctx, a->value.dtype(), a->value.shape()));
+  ops::cast_copy(out->value, a->value);
+  ops::add_inplace(out->value, b->value, 1.0f);
+  tape.push([a, b, out]() {
+    NZ_CHECK(out->grad.defined(), "residual backward without upstream grad");
+    accum_into(*a, out->grad);
+    accum_into(*b, out->grad);
+  });
+  return out;
+}
+
+// ------------------------------------------------------------------ loss ---
+Tensor mse_loss(Tape& tape, VarPtr y, const Tensor& target) {
+  Context* ctx = y->value.ctx();
+  Tensor loss(ctx, DType::F32, Shape{1});
+  NZ_CHECK(!y->grad_init, "loss applied to var with existing grad");
+  y->ensure_grad();
+  y->grad_init = true;
+  ops::mse_loss(y->value, target, loss, &y->grad, 1.0f);
+  (void)tape;
+  return loss;
+}
+
+}  // namespace nz

@@ -1,0 +1,38 @@
+// DDP engine (call stack §3.3): the ONE parallelism
+// strategy (C5). Gradient buckets (shared with the optimizer, C6 sizing)
+// are all-reduced on the comm stream as soon as their last grad is
+// produced, overlapped with the rest of backward (C14). On CPU contexts
+// (FakeComm/TcpComm tests) the all-reduce runs inline and synchronously.
+#pragma once
+
+#include <unordered_map>
+
+#include "src/comm/comm.h"
+#include "src/nn/optim.h"
+
+namespace nz {
+
+class Ddp {
+ public:
+  Ddp(Context* ctx, Optimizer* opt, Comm* comm);
+
+  // Broadcast rank-0 parameter values (flat data + f32 master) to all ranks.
+  void sync_params();
+  void pre_step();      // reset per-bucket pending counters
+  void finish_grads();  // compute stream waits for outstanding all-reduces
+  float grad_scale() const { return 1.0f / float(comm_->nranks()); }
+  Comm* comm() const { return comm_; }
+  int64_t comm_launches() const { return comm_launches_; }
+
+ private:
+  void on_ready(Param& p);
+  void launch_allreduce(Bucket& b);
+
+  Context* ctx_;
+  Optimizer* opt_;
+  Comm* comm_;
+  std::unordered_map<Param*, int> bucket_of_;
+  int64_t comm_launches_ = 0;
+};
+
+}  // namespace nz

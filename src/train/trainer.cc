@@ -1,0 +1,165 @@
+#include "src/train/trainer.h"
+
+#include "src/ckpt/npz.h"
+#include "src/util/util.h"
+
+namespace nz {
+
+Trainer::Trainer(Context* ctx, std::unique_ptr<Model> model,
+                 Optimizer::Kind optim, float lr, int64_t bucket_bytes,
+                 uint64_t data_seed)
+    : ctx_(ctx), model_(std::move(model)) {
+  opt_ = std::make_unique<Optimizer>(ctx, model_->params(), optim, lr, bucket_bytes);
+  DType dt = ctx->device() == Device::HIP ? DType::BF16 : DType::F32;
+  input_ = Tensor(ctx, dt, model_->input_shape());
+  target_ = Tensor(ctx, dt, model_->output_shape());
+  ops::rand_normal(input_, data_seed, 0.f, 1.f);
+  ops::rand_normal(target_, data_seed + 1, 0.f, 1.f);
+}
+
+void Trainer::reseed_data(uint64_t seed) {
+  ops::rand_normal(input_, seed, 0.f, 1.f);
+  ops::rand_normal(target_, seed + 1, 0.f, 1.f);
+}
+
+void Trainer::attach_ddp(Comm* comm) {
+  ddp_ = std::make_unique<Ddp>(ctx_, opt_.get(), comm);
+  ddp_->sync_params();
+}
+
+void Trainer::step() {
+  if (ddp_) ddp_->pre_step();
+  tape_.clear();
+  VarPtr x = make_var(input_, /*needs_grad=*/false);
+  VarPtr y = model_->forward(tape_, x);
+  loss_ = mse_loss(tape_, y, target_);
+  tape_.backward();
+  if (ddp_) ddp_->finish_grads();
+  opt_->step(ddp_ ? ddp_->grad_scale() : 1.0f);
+}
+
+float Trainer::last_loss() {
+  NZ_CHECK(loss_.defined(), "no step taken yet");
+  float v = 0.f;
+  loss_.download(&v, 4);
+  return v;
+}
+
+// ------------------------------------------------------------ checkpoint ---
+static NpyArray slice_to_npy(const std::vector<char>& flat_host, DType dt,
+                             const Shape& shape, int64_t elem_offset) {
+  NpyArray a;
+  a.dtype = dt;
+  a.shape = shape.dims;
+  size_t es = dtype_size(dt);
+  size_t off = size_t(elem_offset) * es;
+  size_t n = size_t(shape.numel()) * es;
+  a.data.assign(flat_host.begin() + off, flat_host.begin() + off + n);
+  return a;
+}
+
+void Trainer::save_checkpoint(const std::string& path) {
+  if (ddp_ && ddp_->comm()->rank() != 0) {
+    ddp_->comm()->barrier();
+    return;
+  }
+  std::map<std::string, NpyArray> arrays;
+  for (auto& b : opt_->buckets()) {
+    std::vector<char> data_host(b.data_flat.bytes()), master_host(b.master.bytes());
+    b.data_flat.download(data_host.data(), data_host.size());
+    b.master.download(master_host.data(), master_host.size());
+    std::vector<char> m_host, v_host;
+    if (b.m.defined()) {
+      m_host.resize(b.m.bytes());
+      v_host.resize(b.v.bytes());
+      b.m.download(m_host.data(), m_host.size());
+      b.v.download(v_host.data(), v_host.size());
+    }
+    for (size_t i = 0; i < b.params.size(); ++i) {
+      Param* p = b.params[i];
+      int64_t off = b.offsets[i];
+      arrays["param/" + p->name] =
+          slice_to_npy(data_host, p->data.dtype(), p->data.shape(), off);
+      arrays["master/" + p->name] =
+          slice_to_npy(master_host, DType::F32, p->data.shape(), off);
+      if (!m_host.empty()) {
+        arrays["adam_m/" + p->name] =
+            slice_to_npy(m_host, DType::F32, p->data.shape(), off);
+        arrays["adam_v/" + p->name] =
+            slice_to_npy(v_host, DType::F32, p->data.shape(), off);
+      }
+    }
+  }
+  NpyArray stepa;
+  stepa.dtype = DType::F32;
+  stepa.shape = {1};
+  float st = (float)opt_->step_count();
+  stepa.data.resize(4);
+  std::memcpy(stepa.data.data(), &st, 4);
+  arrays["meta/step"] = stepa;
+  npz_save(path, arrays);
+  if (ddp_) ddp_->comm()->barrier();
+}
+
+void Trainer::load_checkpoint(const std::string& path) {
+  bool is_root = !ddp_ || ddp_->comm()->rank() == 0;
+  if (is_root) {
+    auto arrays = npz_load(path);
+    auto fetch = [&](const std::string& key) -> NpyArray& {
+      auto it = arrays.find(key);
+      NZ_CHECK(it != arrays.end(), "checkpoint missing member " + key);
+      return it->second;
+    };
+    for (auto& b : opt_->buckets()) {
+      std::vector<char> data_host(b.data_flat.bytes()), master_host(b.master.bytes());
+      std::vector<char> m_host(b.m.defined() ? b.m.bytes() : 0),
+          v_host(b.v.defined() ? b.v.bytes() : 0);
+      for (size_t i = 0; i < b.params.size(); ++i) {
+        Param* p = b.params[i];
+        size_t es = dtype_size(p->data.dtype());
+        size_t off = size_t(b.offsets[i]) * es;
+        auto& pa = fetch("param/" + p->name);
+        NZ_CHECK(pa.numel() == p->numel(), "shape mismatch for " + p->name);
+        std::memcpy(data_host.data() + off, pa.data.data(), pa.data.size());
+        auto& ma = fetch("master/" + p->name);
+        std::memcpy(master_host.data() + b.offsets[i] * 4, ma.data.data(),
+                    ma.data.size());
+        if (!m_host.empty()) {
+          std::memcpy(m_host.data() + b.offsets[i] * 4,
+                      fetch("adam_m/" + p->name).data.data(), p->numel() * 4);
+          std::memcpy(v_host.data() + b.offsets[i] * 4,
+                      fetch("adam_v/" + p->name).data.data(), p->numel() * 4);
+        }
+      }
+      b.data_flat.upload(data_host.data(), data_host.size());
+      b.master.upload(master_host.data(), master_host.size());
+      if (!m_host.empty()) {
+        b.m.upload(m_host.data(), m_host.size());
+        b.v.upload(v_host.data(), v_host.size());
+      }
+    }
+    auto it = arrays.find("meta/step");
+    if (it != arrays.end()) {
+      float st;
+      std::memcpy(&st, it->second.data.data(), 4);
+      opt_->set_step_count((int)st);
+    }
+  }
+  if (ddp_) {
+    // rank 0 loaded; broadcast params + master; broadcast step via scalar.
+    ddp_->sync_params();
+    for (auto& b : opt_->buckets()) {
+      if (b.m.defined()) {
+        Stream* s = ctx_->device() == Device::HIP ? ctx_->compute_stream() : nullptr;
+        ddp_->comm()->broadcast(b.m.data(), b.numel, DType::F32, 0, s);
+        ddp_->comm()->broadcast(b.v.data(), b.numel, DType::F32, 0, s);
+      }
+    }
+    float st = ddp_->comm()->all_reduce_scalar(
+        is_root ? (float)opt_->step_count() : 0.f, ReduceOp::MAX);
+    opt_->set_step_count((int)st);
+    if (ctx_->device() == Device::HIP) ctx_->compute_stream()->sync();
+  }
+}
+
+}  // namespace nz

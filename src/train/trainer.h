@@ -1,0 +1,49 @@
+// Trainer (call stack §3.2): synthetic-data training
+// loop. Input/target batches are device-resident and reused — zero
+// host<->device transfers in steady state except optional loss readback.
+#pragma once
+
+#include <memory>
+
+#include "src/models/models.h"
+#include "src/nn/optim.h"
+#include "src/train/ddp.h"
+
+namespace nz {
+
+class Trainer {
+ public:
+  Trainer(Context* ctx, std::unique_ptr<Model> model, Optimizer::Kind optim,
+          float lr, int64_t bucket_bytes, uint64_t data_seed);
+
+  // Attach DDP: installs hooks and broadcasts initial params.
+  void attach_ddp(Comm* comm);
+
+  // Regenerate the synthetic input/target batch (per-rank data shards).
+  void reseed_data(uint64_t seed);
+
+  // One training step: fwd -> loss -> bwd (-> overlapped all-reduce) -> opt.
+  // Asynchronous on HIP (no sync, no download).
+  void step();
+  float last_loss();   // downloads the loss scalar (syncs on HIP)
+  int step_count() const { return opt_->step_count(); }
+
+  Model* model() { return model_.get(); }
+  Optimizer* optimizer() { return opt_.get(); }
+  Ddp* ddp() { return ddp_.get(); }
+  Context* ctx() { return ctx_; }
+
+  // Checkpoint: rank 0 writes; resume loads + broadcasts.
+  void save_checkpoint(const std::string& path);
+  void load_checkpoint(const std::string& path);
+
+ private:
+  Context* ctx_;
+  std::unique_ptr<Model> model_;
+  std::unique_ptr<Optimizer> opt_;
+  std::unique_ptr<Ddp> ddp_;
+  Tape tape_;
+  Tensor input_, target_, loss_;
+};
+
+}  // namespace nz
