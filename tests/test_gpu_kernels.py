@@ -1,0 +1,260 @@
+"""T3: every HIP kernel vs the T2-validated CPU/numpy oracle.
+bf16 tolerance policy: engine computes on bf16 inputs with f32 accumulation;
+the reference computes in f32 on the same bf16-rounded inputs; outputs are
+compared with rel-err <= 3e-2 elementwise vs the output scale. Determinism:
+two identical runs must be bit-equal.
+"""
+import numpy as np
+import pytest
+
+import nezha_amd as nz
+
+pytestmark = pytest.mark.gpu
+
+RNG = np.random.default_rng(42)
+
+
+def randb(*shape):
+    """random f32 pre-rounded to bf16 so engine and oracle see same values"""
+    x = RNG.normal(size=shape).astype(np.float32)
+    return nz.bf16_to_f32(nz.f32_to_bf16(x))
+
+
+def check(actual, ref, rtol=0.02):
+    scale = max(np.abs(ref).max(), 1e-3)
+    err = np.abs(actual - ref).max() / scale
+    assert err < rtol, f"max rel err {err:.4f} (scale {scale:.3f})"
+
+
+@pytest.mark.parametrize("transA,transB", [(False, False), (False, True),
+                                           (True, False), (True, True)])
+def test_gemm_trans(hip_ctx, transA, transB):
+    M, N, K = 100, 96, 128
+    A = randb(*(K, M) if transA else (M, K))
+    B = randb(*(N, K) if transB else (K, N))
+    ref = (A.T if transA else A) @ (B.T if transB else B)
+    a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
+    c = hip_ctx.tensor((M, N), nz.BF16)
+    nz.gemm(a, b, c, transA, transB)
+    check(c.numpy(), ref)
+
+
+def test_gemm_large_shape(hip_ctx):
+    M, N, K = 512, 384, 512
+    A, B = randb(M, K), randb(K, N)
+    a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
+    c = hip_ctx.tensor((M, N), nz.BF16)
+    nz.gemm(a, b, c)
+    check(c.numpy(), A @ B)
+
+
+def test_gemm_identity_asymmetric(hip_ctx):
+    """A=I with asymmetric B catches transposed C-writes (guide §3)."""
+    K = 128
+    A = np.eye(K, dtype=np.float32)
+    B = randb(K, K)
+    a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
+    c = hip_ctx.tensor((K, K), nz.BF16)
+    nz.gemm(a, b, c)
+    check(c.numpy(), B, rtol=0.01)
+
+
+def test_gemm_epilogue(hip_ctx):
+    import torch
+    M, N, K = 64, 128, 64
+    A, B, bias = randb(M, K), randb(K, N), randb(N)
+    a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
+    bi = hip_ctx.from_numpy_bf16(bias)
+    c = hip_ctx.tensor((M, N), nz.BF16)
+    pre = hip_ctx.tensor((M, N), nz.BF16)
+    nz.gemm(a, b, c, bias=bi, gelu=True, save_pre=pre)
+    z_ref = A @ B + bias
+    y_ref = torch.nn.functional.gelu(torch.from_numpy(z_ref),
+                                     approximate="tanh").numpy()
+    check(pre.numpy(), z_ref)
+    check(c.numpy(), y_ref)
+
+
+def test_gemm_beta(hip_ctx):
+    M, N, K = 64, 64, 64
+    A, B, C0 = randb(M, K), randb(K, N), randb(M, N)
+    a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
+    c = hip_ctx.from_numpy_bf16(C0)
+    nz.gemm(a, b, c, alpha=0.5, beta=1.0)
+    check(c.numpy(), 0.5 * (A @ B) + C0)
+
+
+def test_gemm_batched(hip_ctx):
+    G, M, N, K = 6, 64, 64, 64
+    A, B = randb(G, M, K), randb(G, K, N)
+    a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
+    c = hip_ctx.tensor((G, M, N), nz.BF16)
+    nz.gemm_batched(a, b, c, False, False, 1.0, G, M, N, K, M * K, K * N, M * N)
+    check(c.numpy(), np.einsum("gmk,gkn->gmn", A, B))
+
+
+def test_gemm_determinism(hip_ctx):
+    M, N, K = 256, 256, 256
+    A, B = randb(M, K), randb(K, N)
+    a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
+    c1 = hip_ctx.tensor((M, N), nz.BF16)
+    c2 = hip_ctx.tensor((M, N), nz.BF16)
+    nz.gemm(a, b, c1)
+    nz.gemm(a, b, c2)
+    np.testing.assert_array_equal(c1.raw(), c2.raw())
+
+
+def test_rng_matches_cpu(hip_ctx):
+    """GPU and CPU RNG are the same counter-based function: bitwise equal."""
+    cpu = nz.Ctx(nz.CPU)
+    a = cpu.tensor((4096,), nz.F32).rand_normal(99).numpy()
+    g = hip_ctx.tensor((4096,), nz.F32).rand_normal(99).numpy()
+    np.testing.assert_array_equal(a, g)
+
+
+def test_bias_gelu_fwd_bwd(hip_ctx):
+    import torch
+    M, N = 128, 256
+    X, bias, dY = randb(M, N), randb(N), randb(M, N)
+    xt = torch.from_numpy(X).requires_grad_(True)
+    bt = torch.from_numpy(bias).requires_grad_(True)
+    yt = torch.nn.functional.gelu(xt + bt, approximate="tanh")
+    yt.backward(torch.from_numpy(dY))
+
+    x, b = hip_ctx.from_numpy_bf16(X), hip_ctx.from_numpy_bf16(bias)
+    y = hip_ctx.tensor((M, N), nz.BF16)
+    z = hip_ctx.tensor((M, N), nz.BF16)
+    nz.bias_gelu_fwd(x, b, y, z)
+    check(y.numpy(), yt.detach().numpy())
+
+    dy = hip_ctx.from_numpy_bf16(dY)
+    dz, db = hip_ctx.tensor((M, N), nz.BF16), hip_ctx.tensor((N,), nz.BF16)
+    nz.bias_gelu_bwd(dy, z, dz, db)
+    check(dz.numpy(), xt.grad.numpy())
+    check(db.numpy(), bt.grad.numpy())
+
+
+def test_layernorm_fwd_bwd(hip_ctx):
+    import torch
+    R, N = 128, 768
+    X, G, Be, dY = randb(R, N), randb(N), randb(N), randb(R, N)
+    xt = torch.from_numpy(X).requires_grad_(True)
+    gt = torch.from_numpy(G).requires_grad_(True)
+    bt = torch.from_numpy(Be).requires_grad_(True)
+    yt = torch.nn.functional.layer_norm(xt, (N,), gt, bt, eps=1e-5)
+    yt.backward(torch.from_numpy(dY))
+
+    x, g, b = (hip_ctx.from_numpy_bf16(X), hip_ctx.from_numpy_bf16(G),
+               hip_ctx.from_numpy_bf16(Be))
+    y = hip_ctx.tensor((R, N), nz.BF16)
+    mean, rstd = hip_ctx.tensor((R,)), hip_ctx.tensor((R,))
+    nz.layernorm_fwd(x, g, b, y, mean, rstd)
+    check(y.numpy(), yt.detach().numpy())
+
+    dy = hip_ctx.from_numpy_bf16(dY)
+    dx = hip_ctx.tensor((R, N), nz.BF16)
+    dg, db = hip_ctx.tensor((N,), nz.BF16), hip_ctx.tensor((N,), nz.BF16)
+    nz.layernorm_bwd(dy, x, g, mean, rstd, dx, dg, db)
+    check(dx.numpy(), xt.grad.numpy(), rtol=0.03)
+    check(dg.numpy(), gt.grad.numpy(), rtol=0.03)
+    check(db.numpy(), bt.grad.numpy(), rtol=0.03)
+
+
+def test_softmax_fwd_bwd(hip_ctx):
+    import torch
+    R, N, scale = 384, 512, 0.125
+    X, dY = randb(R, N), randb(R, N)
+    xt = torch.from_numpy(X).requires_grad_(True)
+    yt = torch.softmax(scale * xt, dim=-1)
+    yt.backward(torch.from_numpy(dY))
+    x = hip_ctx.from_numpy_bf16(X)
+    y = hip_ctx.tensor((R, N), nz.BF16)
+    nz.softmax_fwd(x, y, scale)
+    check(y.numpy(), yt.detach().numpy())
+    dy = hip_ctx.from_numpy_bf16(dY)
+    dx = hip_ctx.tensor((R, N), nz.BF16)
+    nz.softmax_bwd(dy, y, dx, scale)
+    check(dx.numpy(), xt.grad.numpy())
+
+
+def test_adam_step(hip_ctx):
+    n = 100003  # odd size: exercises the vector tail
+    P = randb(n)
+    Gr = randb(n)
+    p = hip_ctx.from_numpy_bf16(P)
+    g = hip_ctx.from_numpy_bf16(Gr)
+    master = hip_ctx.from_numpy(P.copy())
+    m = hip_ctx.tensor((n,)).fill(0)
+    v = hip_ctx.tensor((n,)).fill(0)
+    nz.adam_step(p, g, master, m, v, lr=1e-3, step=1)
+    # reference in numpy (f32)
+    mr = 0.1 * Gr
+    vr = 0.001 * Gr * Gr
+    ref = P - 1e-3 * (mr / 0.1) / (np.sqrt(vr / 0.001) + 1e-8)
+    check(master.numpy(), ref, rtol=0.001)
+    check(p.numpy(), ref, rtol=0.01)
+
+
+def test_sgd_step(hip_ctx):
+    n = 5000
+    P, Gr = randb(n), randb(n)
+    p = hip_ctx.from_numpy_bf16(P)
+    g = hip_ctx.from_numpy_bf16(Gr)
+    master = hip_ctx.from_numpy(P.copy())
+    nz.sgd_step(p, g, master, None, lr=0.1)
+    check(master.numpy(), P - 0.1 * Gr, rtol=0.001)
+
+
+def test_mse_loss(hip_ctx):
+    Y, T = randb(256, 128), randb(256, 128)
+    y, t = hip_ctx.from_numpy_bf16(Y), hip_ctx.from_numpy_bf16(T)
+    loss = hip_ctx.tensor((1,))
+    dy = hip_ctx.tensor((256, 128), nz.BF16)
+    nz.mse_loss(y, t, loss, dy)
+    assert abs(loss.numpy()[0] - np.mean((Y - T) ** 2)) < 1e-3
+    check(dy.numpy(), 2 * (Y - T) / Y.size, rtol=0.03)
+
+
+def test_elementwise_family(hip_ctx):
+    X, S = randb(333, 64), randb(333, 64)
+    x = hip_ctx.from_numpy_bf16(X)
+    s = hip_ctx.from_numpy_bf16(S)
+    nz.add_inplace(x, s, 0.5)
+    check(x.numpy(), X + 0.5 * S, rtol=0.01)
+    nz.scale_inplace(x, 2.0)
+    check(x.numpy(), 2 * (X + 0.5 * S), rtol=0.01)
+    # colsum
+    cs = hip_ctx.tensor((64,), nz.BF16)
+    nz.colsum(hip_ctx.from_numpy_bf16(X), cs)
+    check(cs.numpy(), X.sum(0), rtol=0.02)
+    # transpose
+    tr = hip_ctx.tensor((64, 333), nz.BF16)
+    nz.transpose2d(hip_ctx.from_numpy_bf16(X), tr)
+    np.testing.assert_array_equal(tr.numpy(), X.T)
+    # cast bf16 -> f32 -> bf16
+    f = hip_ctx.tensor((333, 64), nz.F32)
+    nz.cast_copy(f, hip_ctx.from_numpy_bf16(X))
+    np.testing.assert_array_equal(f.numpy(), X)
+    # reductions
+    out = hip_ctx.tensor((1,))
+    nz.reduce_sum(hip_ctx.from_numpy_bf16(X), out)
+    assert abs(out.numpy()[0] - X.sum()) < abs(X.sum()) * 0.01 + 0.5
+    nz.l2norm_sq(hip_ctx.from_numpy_bf16(X), out)
+    assert abs(out.numpy()[0] - (X * X).sum()) < (X * X).sum() * 0.01
+
+
+def test_attn_split_merge(hip_ctx):
+    B, S, H, D = 2, 16, 4, 32
+    QKV = randb(B * S, 3 * H * D)
+    qkv = hip_ctx.from_numpy_bf16(QKV)
+    q = hip_ctx.tensor((B, H, S, D), nz.BF16)
+    k = hip_ctx.tensor((B, H, S, D), nz.BF16)
+    v = hip_ctx.tensor((B, H, S, D), nz.BF16)
+    nz.attn_split_qkv(qkv, q, k, v, B, S, H, D)
+    ref = QKV.reshape(B, S, 3, H, D).transpose(2, 0, 3, 1, 4)
+    np.testing.assert_array_equal(q.numpy(), ref[0])
+    np.testing.assert_array_equal(v.numpy(), ref[2])
+    out = hip_ctx.tensor((B * S, H * D), nz.BF16)
+    nz.attn_merge(q, out, B, S, H, D)
+    np.testing.assert_array_equal(
+        out.numpy(), ref[0].transpose(0, 2, 1, 3).reshape(B * S, H * D))
