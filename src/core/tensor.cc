@@ -176,6 +176,9 @@ void Tensor::upload(const void* host_src, size_t n) {
     std::memcpy(data(), host_src, n);
   } else {
     ctx_->make_current();
+    // The compute stream is non-blocking w.r.t. the null stream: sync it
+    // first so queued kernels (e.g. init) cannot overwrite the upload.
+    ctx_->compute_stream()->sync();
     NZ_HIP_CHECK(hipMemcpy(data(), host_src, n, hipMemcpyHostToDevice));
   }
 }

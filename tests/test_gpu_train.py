@@ -11,7 +11,7 @@ pytestmark = pytest.mark.gpu
 
 
 def test_mlp_training_loss_decreases(hip_ctx):
-    tr = nz.Trainer.mlp(hip_ctx, layers=4, dim=1024, batch=2048, lr=1e-4,
+    tr = nz.Trainer.mlp(hip_ctx, layers=4, dim=1024, batch=2048, lr=1e-3,
                         seed=17)
     tr.step()
     l0 = tr.loss()
