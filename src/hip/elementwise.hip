@@ -159,7 +159,40 @@ void scale_inplace(Tensor& dst, float alpha) {
 }
 
 // ------------------------------------------------------------- transpose ---
-// LDS-tiled 64x64 transpose with +1-element padded rows (bank-conflict fix).
+// Fast path (shapes % 64): 64x64 LDS tile, 16-B vector global loads AND
+// stores (Guideline 13 — scalar bf16 is 2-2.5x slower); rows padded one
+// vector to break bank conflicts on the gather pass. Also runs batched
+// (grid.z) to serve the attention/PV lowering.
+template <typename T, int VEC>
+__global__ void transpose_vec_k(const T* __restrict__ src, T* __restrict__ dst,
+                                int64_t R, int64_t C, int64_t sSrc, int64_t sDst) {
+  using vec_t = __attribute__((ext_vector_type(VEC))) T;
+  static_assert(sizeof(T) * VEC == 16, "vec width must be 16 B");
+  __shared__ T tile[64][64 + VEC];
+  const T* sp = src + blockIdx.z * sSrc;
+  T* dp = dst + blockIdx.z * sDst;
+  int64_t c0 = (int64_t)blockIdx.x * 64, r0 = (int64_t)blockIdx.y * 64;
+  constexpr int CHUNKS_PER_ROW = 64 / VEC;
+#pragma unroll
+  for (int rep = 0; rep < 64 * CHUNKS_PER_ROW / 256; ++rep) {
+    int cidx = rep * 256 + threadIdx.x;
+    int row = cidx / CHUNKS_PER_ROW, cg = cidx % CHUNKS_PER_ROW;
+    *(vec_t*)&tile[row][cg * VEC] =
+        *(const vec_t*)(sp + (r0 + row) * C + c0 + cg * VEC);
+  }
+  __syncthreads();
+#pragma unroll
+  for (int rep = 0; rep < 64 * CHUNKS_PER_ROW / 256; ++rep) {
+    int cidx = rep * 256 + threadIdx.x;
+    int orow = cidx / CHUNKS_PER_ROW, og = cidx % CHUNKS_PER_ROW;
+    T vals[VEC];
+#pragma unroll
+    for (int i = 0; i < VEC; ++i) vals[i] = tile[og * VEC + i][orow];
+    *(vec_t*)(dp + (c0 + orow) * R + r0 + og * VEC) = *(vec_t*)vals;
+  }
+}
+
+// generic fallback for ragged shapes (scalar, guarded)
 template <typename T>
 __global__ void transpose_k(const T* __restrict__ src, T* __restrict__ dst,
                             int64_t R, int64_t C) {
@@ -178,9 +211,29 @@ __global__ void transpose_k(const T* __restrict__ src, T* __restrict__ dst,
     if (r < C && c < R) dst[r * R + c] = tile[tx][ty + i * 4];
   }
 }
+
+void transpose2d_batched(const Tensor& src, Tensor& dst, int64_t batch,
+                         int64_t R, int64_t C, int64_t sSrc, int64_t sDst) {
+  hipStream_t s = stream_of(src);
+  NZ_CHECK(R % 64 == 0 && C % 64 == 0, "batched transpose needs %64 shapes");
+  dim3 g(C / 64, R / 64, batch), b(256);
+  if (src.dtype() == DType::F32)
+    transpose_vec_k<float, 4><<<g, b, 0, s>>>((const float*)src.data(),
+                                              (float*)dst.data(), R, C, sSrc, sDst);
+  else
+    transpose_vec_k<uint16_t, 8><<<g, b, 0, s>>>((const uint16_t*)src.data(),
+                                                 (uint16_t*)dst.data(), R, C,
+                                                 sSrc, sDst);
+  NZ_KERNEL_CHECK();
+}
+
 void transpose2d(const Tensor& src, Tensor& dst) {
   int64_t R = src.rows(), C = src.cols();
   NZ_CHECK(dst.rows() == C && dst.cols() == R, "transpose shape mismatch");
+  if (R % 64 == 0 && C % 64 == 0) {
+    transpose2d_batched(src, dst, 1, R, C, 0, 0);
+    return;
+  }
   hipStream_t s = stream_of(src);
   dim3 g((C + 63) / 64, (R + 63) / 64), b(256);
   if (src.dtype() == DType::F32)

@@ -187,16 +187,152 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_k(GemmArgs args) {
   }
 }
 
+// ---------------------------------------------------------------------------
+// Tier B: fast NT kernel (C[M,N] = A[M,K] . B[N,K]^T), both operands
+// k-contiguous, staged by direct global->LDS DMA (global_load_lds_dwordx4,
+// guide §5 ladder step 3: width=16 took 517->874 TF). 128x128 tile, BK=64,
+// double-buffered 64 KB LDS (2 blocks/CU), issue-next-tile-before-compute
+// (T3 minimum 2-phase), s_setprio(1) around the MFMA cluster (T5), and a
+// bijective XCD-aware block swizzle (T1) so neighbor tiles share an XCD L2.
+// Out-of-range rows/cols are CLAMPED at the staging loads (glds cannot mask
+// lanes); the epilogue masks stores, so garbage lanes never escape.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ void glds16(const uint16_t* gsrc, uint16_t* lds_dst) {
+  // direct global->LDS DMA, 16 B per lane (global_load_lds_dwordx4).
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) uint32_t*)gsrc,
+      (__attribute__((address_space(3))) uint32_t*)lds_dst, 16, 0, 0);
+}
+
+__launch_bounds__(THREADS) __global__ void gemm_bf16_nt_k(GemmArgs args,
+                                                          int nwg_n) {
+  constexpr int NT_BK = 64;
+  // ONE shared object (guide §5 ".s-level traps" (a): a second __shared__
+  // array would force a vmcnt(0) before every ds_read, de-pipelining glds).
+  // Layout: [buf][A tile 128x64 | B tile 128x64].
+  __shared__ __attribute__((aligned(16))) uint16_t smem[2 * (BM + BN) * NT_BK];
+  constexpr int BUF_STRIDE = (BM + BN) * NT_BK;
+  constexpr int B_OFF = BM * NT_BK;
+#define NT_AS(buf) (smem + (buf) * BUF_STRIDE)
+#define NT_BS(buf) (smem + (buf) * BUF_STRIDE + B_OFF)
+
+  const int64_t batch = blockIdx.y;
+  const uint16_t* __restrict__ A = args.A + batch * args.sA;
+  const uint16_t* __restrict__ B = args.B + batch * args.sB;
+  uint16_t* C = args.C + batch * args.sC;
+  uint16_t* save_pre = args.save_pre ? args.save_pre + batch * args.sC : nullptr;
+
+  // XCD-aware bijective block swizzle (guide T1): contiguous grid chunk per
+  // XCD so neighboring tiles (sharing A rows / B panels) hit the same L2.
+  int nwg = gridDim.x;
+  int orig = blockIdx.x;
+  int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+  int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+  const int64_t m0 = (int64_t)(wgid / nwg_n) * BM;
+  const int64_t n0 = (int64_t)(wgid % nwg_n) * BN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+
+  // Per-lane global element offsets for the 4 glds chunks of each operand:
+  // chunk c covers bytes [wid*4096 + c*1024 + lane*16]; elem e = byte/2,
+  // row = e/64, k = e%64 (lane-linear LDS image [128][64], rows of 128 B).
+  int e0 = (wid * 2048) + lane * 8;  // elements; +c*512 per chunk
+
+  f32x4 acc[4][4] = {};
+  const int64_t ntiles = args.K / NT_BK;
+
+  auto stage = [&](int buf, int64_t t) {
+    int64_t k0 = t * NT_BK;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int e = e0 + c * 512;
+      int row = e >> 6, k = e & 63;
+      int64_t ar = m0 + row;
+      if (ar > Mc) ar = Mc;  // clamp: masked at the epilogue
+      glds16(A + ar * lda + k0 + k, NT_AS(buf) + e);
+      int64_t br = n0 + row;
+      if (br > Nc) br = Nc;
+      glds16(B + br * ldb + k0 + k, NT_BS(buf) + e);
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();  // carries vmcnt(0): tile 0 landed
+
+  int cur = 0;
+  for (int64_t t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) stage(cur ^ 1, t + 1);  // issue BEFORE compute (T3)
+#pragma unroll
+    for (int kk = 0; kk < NT_BK; kk += 32) {
+      short8 a_frag[4], b_frag[4];
+      const int kbase = kk + (lane >> 4) * 8;
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        int row = wr * 64 + fm * 16 + (lane & 15);
+        a_frag[fm] = *(const short8*)(NT_AS(cur) + row * NT_BK + kbase);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = wc * 64 + fn * 16 + (lane & 15);
+        b_frag[fn] = *(const short8*)(NT_BS(cur) + col * NT_BK + kbase);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[fm], b_frag[fn], acc[fm][fn], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();  // drains the in-flight glds (vmcnt(0)) + barrier
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+#pragma unroll
+      for (int rI = 0; rI < 4; ++rI) {
+        int64_t grow = m0 + wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+        int64_t gcol = n0 + wc * 64 + fn * 16 + (lane & 15);
+        if (grow >= args.M || gcol >= args.N) continue;
+        int64_t idx = grow * args.N + gcol;
+        float v = args.alpha * acc[fm][fn][rI];
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+        if (args.bias) v += b2f(args.bias[gcol]);
+        if (save_pre) save_pre[idx] = f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        C[idx] = f2b(v);
+      }
+    }
+  }
+}
+
 void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
                  hipStream_t s) {
   NZ_CHECK(args.K % BK == 0, "hip gemm requires K % 64 == 0, got K=" +
                                  std::to_string(args.K));
+  if (!transA && transB) {
+    // fast NT path: glds-staged, XCD-swizzled (see gemm_bf16_nt_k)
+    int nwg_m = (int)((args.M + BM - 1) / BM);
+    int nwg_n = (int)((args.N + BN - 1) / BN);
+    dim3 grid(nwg_m * nwg_n, batch);
+    gemm_bf16_nt_k<<<grid, dim3(THREADS), 0, s>>>(args, nwg_n);
+    NZ_KERNEL_CHECK();
+    return;
+  }
   dim3 grid((args.N + BN - 1) / BN, (args.M + BM - 1) / BM, batch);
   dim3 block(THREADS);
   if (!transA && !transB)
     gemm_bf16_k<false, false><<<grid, block, 0, s>>>(args);
-  else if (!transA && transB)
-    gemm_bf16_k<false, true><<<grid, block, 0, s>>>(args);
   else if (transA && !transB)
     gemm_bf16_k<true, false><<<grid, block, 0, s>>>(args);
   else
@@ -205,6 +341,16 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
 }
 
 }  // namespace
+
+// The fast kernel is the NT form (both operands k-contiguous -> lane-linear
+// glds staging). NN and TN are LOWERED to NT by transposing the k-major
+// operand(s) through the vectorized tile-transpose — the transpose costs
+// ~2 passes over the operand (~tens of us) vs the GEMM's hundreds, and NT
+// then runs ~4x the scatter-staged generic kernel. TT stays on the generic
+// tiled kernel (test-only form).
+static bool can_fast_transpose(int64_t R, int64_t C) {
+  return R % 64 == 0 && C % 64 == 0;
+}
 
 void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
           float alpha, float beta, const ops::GemmEpilogue* epi) {
@@ -230,6 +376,31 @@ void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
     args.save_pre = epi->save_pre ? (uint16_t*)epi->save_pre->data() : nullptr;
     args.fuse_gelu = epi->gelu ? 1 : 0;
   }
+  Tensor at_scratch, bt_scratch;  // keep lowered operands alive to launch
+  if (args.K % 64 == 0) {
+    if (!transA && !transB && can_fast_transpose(args.K, args.N)) {
+      // NN: C = A.B = NT(A, B^T)
+      bt_scratch = Tensor(C.ctx(), DType::BF16, Shape{args.N, args.K});
+      transpose2d(B, bt_scratch);
+      args.B = (const uint16_t*)bt_scratch.data();
+      args.ldb = args.K;
+      transB = true;
+      transA = false;
+    } else if (transA && !transB && can_fast_transpose(args.K, args.M) &&
+               can_fast_transpose(args.K, args.N)) {
+      // TN: C = A^T.B = NT(A^T, B^T)
+      at_scratch = Tensor(C.ctx(), DType::BF16, Shape{args.M, args.K});
+      transpose2d(A, at_scratch);
+      bt_scratch = Tensor(C.ctx(), DType::BF16, Shape{args.N, args.K});
+      transpose2d(B, bt_scratch);
+      args.A = (const uint16_t*)at_scratch.data();
+      args.B = (const uint16_t*)bt_scratch.data();
+      args.lda = args.K;
+      args.ldb = args.K;
+      transA = false;
+      transB = true;
+    }
+  }
   launch_gemm(args, transA, transB, 1, stream_of(C));
 }
 
@@ -251,6 +422,33 @@ void gemm_batched(const Tensor& A, const Tensor& B, Tensor& C, bool transA,
   args.sC = sC;
   args.alpha = alpha;
   args.beta = 0.f;
+  // Lower batched NN / TN to the fast NT kernel via batched transposes
+  // (attention P.V, dV, dK paths). Scratch = [batch, rows, cols] per op.
+  Tensor at_scratch, bt_scratch;
+  if (K % 64 == 0) {
+    if (!transA && !transB && can_fast_transpose(K, N)) {
+      bt_scratch = Tensor(C.ctx(), DType::BF16, Shape{batch, N, K});
+      transpose2d_batched(B, bt_scratch, batch, K, N, sB, N * K);
+      args.B = (const uint16_t*)bt_scratch.data();
+      args.ldb = K;
+      args.sB = N * K;
+      transB = true;
+    } else if (transA && !transB && can_fast_transpose(K, M) &&
+               can_fast_transpose(K, N)) {
+      at_scratch = Tensor(C.ctx(), DType::BF16, Shape{batch, M, K});
+      transpose2d_batched(A, at_scratch, batch, K, M, sA, M * K);
+      bt_scratch = Tensor(C.ctx(), DType::BF16, Shape{batch, N, K});
+      transpose2d_batched(B, bt_scratch, batch, K, N, sB, N * K);
+      args.A = (const uint16_t*)at_scratch.data();
+      args.B = (const uint16_t*)bt_scratch.data();
+      args.lda = K;
+      args.ldb = K;
+      args.sA = M * K;
+      args.sB = N * K;
+      transA = false;
+      transB = true;
+    }
+  }
   launch_gemm(args, transA, transB, batch, stream_of(C));
 }
 
