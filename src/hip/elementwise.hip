@@ -4,6 +4,8 @@
 // 64-wide waves (no 32-lane warp idioms), bf16 vectorized as ushort4/short8
 // where hot (Guideline 13), two-stage deterministic reductions (no atomics
 // on the loss scalar), block-per-row norm/softmax with wave shuffles.
+#include <algorithm>
+
 #include "src/hip/hip_common.h"
 #include "src/hip/hip_ops.h"
 
@@ -245,15 +247,31 @@ void transpose2d(const Tensor& src, Tensor& dst) {
 }
 
 // ---------------------------------------------------------------- colsum ---
-// dst[j] = beta*dst[j] + sum_m src[m,j]. One thread per column; consecutive
-// threads read consecutive columns -> fully coalesced row sweeps.
-template <typename T, typename TD>
-__global__ void colsum_k(const T* __restrict__ src, TD* __restrict__ dst,
-                         int64_t M, int64_t N, float beta) {
+// dst[j] = beta*dst[j] + sum_m src[m,j]. Two-stage: stage 1 splits the rows
+// into CHUNK groups so the grid fills all 256 CUs (a single thread-per-
+// column pass left 15/16 of the chip idle and ran 160x slower), each block
+// accumulating its 256 columns over its row range (coalesced 512-B row
+// sweeps); stage 2 folds the f32 partials.
+template <typename T>
+__global__ void colsum_stage1_k(const T* __restrict__ src,
+                                float* __restrict__ partials, int64_t M,
+                                int64_t N, int64_t rows_per_chunk) {
+  int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  if (j >= N) return;
+  int64_t r0 = blockIdx.y * rows_per_chunk;
+  int64_t r1 = min(M, r0 + rows_per_chunk);
+  float acc = 0.f;
+  for (int64_t i = r0; i < r1; ++i) acc += ld(src, i * N + j);
+  partials[blockIdx.y * N + j] = acc;
+}
+template <typename TD>
+__global__ void colsum_stage2_k(const float* __restrict__ partials,
+                                TD* __restrict__ dst, int64_t N, int chunks,
+                                float beta) {
   int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   if (j >= N) return;
   float acc = 0.f;
-  for (int64_t i = 0; i < M; ++i) acc += ld(src, i * N + j);
+  for (int c = 0; c < chunks; ++c) acc += partials[c * N + j];
   float prev = beta != 0.f ? beta * ld(dst, j) : 0.f;
   st(dst, j, prev + acc);
 }
@@ -261,13 +279,29 @@ void colsum(const Tensor& src, Tensor& dst, float beta) {
   int64_t M = src.rows(), N = src.cols();
   NZ_CHECK(dst.numel() == N, "colsum dst size");
   hipStream_t s = stream_of(src);
-  dim3 g((N + 255) / 256), b(256);
+  // enough chunks that colblocks*chunks fills the chip (256 CUs)
+  int64_t colblocks = (N + 255) / 256;
+  int chunks = (int)std::min<int64_t>((M + 63) / 64,
+                                      std::max<int64_t>(1, 2048 / colblocks));
+  int64_t rows_per_chunk = (M + chunks - 1) / chunks;
+  Tensor partials(src.ctx(), DType::F32, Shape{(int64_t)chunks, N});
+  dim3 g1(colblocks, chunks), b(256);
   if (src.dtype() == DType::BF16)
-    colsum_k<uint16_t, uint16_t><<<g, b, 0, s>>>((const uint16_t*)src.data(),
-                                                 (uint16_t*)dst.data(), M, N, beta);
+    colsum_stage1_k<uint16_t><<<g1, b, 0, s>>>((const uint16_t*)src.data(),
+                                               (float*)partials.data(), M, N,
+                                               rows_per_chunk);
   else
-    colsum_k<float, float><<<g, b, 0, s>>>((const float*)src.data(),
-                                           (float*)dst.data(), M, N, beta);
+    colsum_stage1_k<float><<<g1, b, 0, s>>>((const float*)src.data(),
+                                            (float*)partials.data(), M, N,
+                                            rows_per_chunk);
+  dim3 g2(colblocks);
+  if (dst.dtype() == DType::BF16)
+    colsum_stage2_k<uint16_t><<<g2, b, 0, s>>>((const float*)partials.data(),
+                                               (uint16_t*)dst.data(), N, chunks,
+                                               beta);
+  else
+    colsum_stage2_k<float><<<g2, b, 0, s>>>((const float*)partials.data(),
+                                            (float*)dst.data(), N, chunks, beta);
   NZ_KERNEL_CHECK();
 }
 
@@ -418,21 +452,38 @@ __global__ void layernorm_dx_k(const T* __restrict__ dy, const T* __restrict__ x
     __syncthreads();
   }
 }
-// dgamma[j] = sum_i dy[i,j]*xhat[i,j]; dbeta[j] = sum_i dy[i,j]
+// dgamma[j] = sum_i dy[i,j]*xhat[i,j]; dbeta[j] = sum_i dy[i,j].
+// Two-stage like colsum (row chunks fill the chip), partials [chunk][2][N].
 template <typename T>
-__global__ void layernorm_dgb_k(const T* __restrict__ dy, const T* __restrict__ x,
-                                const float* __restrict__ mean,
-                                const float* __restrict__ rstd,
-                                T* __restrict__ dgamma, T* __restrict__ dbeta,
-                                int64_t R, int64_t N) {
+__global__ void layernorm_dgb1_k(const T* __restrict__ dy, const T* __restrict__ x,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ rstd,
+                                 float* __restrict__ partials, int64_t R,
+                                 int64_t N, int64_t rows_per_chunk) {
   int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   if (j >= N) return;
+  int64_t r0 = blockIdx.y * rows_per_chunk;
+  int64_t r1 = min(R, r0 + rows_per_chunk);
   float sg = 0.f, sb = 0.f;
-  for (int64_t i = 0; i < R; ++i) {
+  for (int64_t i = r0; i < r1; ++i) {
     float dyv = ld(dy, i * N + j);
     float xh = (ld(x, i * N + j) - mean[i]) * rstd[i];
     sg += dyv * xh;
     sb += dyv;
+  }
+  partials[(blockIdx.y * 2 + 0) * N + j] = sg;
+  partials[(blockIdx.y * 2 + 1) * N + j] = sb;
+}
+template <typename T>
+__global__ void layernorm_dgb2_k(const float* __restrict__ partials,
+                                 T* __restrict__ dgamma, T* __restrict__ dbeta,
+                                 int64_t N, int chunks) {
+  int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  if (j >= N) return;
+  float sg = 0.f, sb = 0.f;
+  for (int c = 0; c < chunks; ++c) {
+    sg += partials[(c * 2 + 0) * N + j];
+    sb += partials[(c * 2 + 1) * N + j];
   }
   st(dgamma, j, sg);
   st(dbeta, j, sb);
@@ -443,23 +494,35 @@ void layernorm_bwd(const Tensor& dy, const Tensor& x, const Tensor& gamma,
   hipStream_t s = stream_of(dy);
   int64_t R = x.rows(), N = x.cols();
   dim3 b(256);
+  int64_t colblocks = (N + 255) / 256;
+  int chunks = (int)std::min<int64_t>((R + 63) / 64,
+                                      std::max<int64_t>(1, 2048 / colblocks));
+  int64_t rows_per_chunk = (R + chunks - 1) / chunks;
+  Tensor partials(x.ctx(), DType::F32, Shape{(int64_t)chunks * 2, N});
+  dim3 g1(colblocks, chunks), g2(colblocks);
   if (x.dtype() == DType::BF16) {
     layernorm_dx_k<uint16_t><<<dim3(grid_1d(R, 1, 2048)), b, 0, s>>>(
         (const uint16_t*)dy.data(), (const uint16_t*)x.data(),
         (const uint16_t*)gamma.data(), (const float*)mean.data(),
         (const float*)rstd.data(), (uint16_t*)dx.data(), R, N, beta_dx);
-    layernorm_dgb_k<uint16_t><<<dim3((N + 255) / 256), b, 0, s>>>(
+    layernorm_dgb1_k<uint16_t><<<g1, b, 0, s>>>(
         (const uint16_t*)dy.data(), (const uint16_t*)x.data(),
         (const float*)mean.data(), (const float*)rstd.data(),
-        (uint16_t*)dgamma.data(), (uint16_t*)dbeta.data(), R, N);
+        (float*)partials.data(), R, N, rows_per_chunk);
+    layernorm_dgb2_k<uint16_t><<<g2, b, 0, s>>>(
+        (const float*)partials.data(), (uint16_t*)dgamma.data(),
+        (uint16_t*)dbeta.data(), N, chunks);
   } else {
     layernorm_dx_k<float><<<dim3(grid_1d(R, 1, 2048)), b, 0, s>>>(
         (const float*)dy.data(), (const float*)x.data(), (const float*)gamma.data(),
         (const float*)mean.data(), (const float*)rstd.data(), (float*)dx.data(),
         R, N, beta_dx);
-    layernorm_dgb_k<float><<<dim3((N + 255) / 256), b, 0, s>>>(
+    layernorm_dgb1_k<float><<<g1, b, 0, s>>>(
         (const float*)dy.data(), (const float*)x.data(), (const float*)mean.data(),
-        (const float*)rstd.data(), (float*)dgamma.data(), (float*)dbeta.data(), R, N);
+        (const float*)rstd.data(), (float*)partials.data(), R, N, rows_per_chunk);
+    layernorm_dgb2_k<float><<<g2, b, 0, s>>>(
+        (const float*)partials.data(), (float*)dgamma.data(),
+        (float*)dbeta.data(), N, chunks);
   }
   NZ_KERNEL_CHECK();
 }
@@ -536,33 +599,63 @@ void softmax_bwd(const Tensor& dy, const Tensor& y, Tensor& dx, float scale) {
 }
 
 // --------------------------------------------------------- K6/K7: optim ---
-// HBM-bound: vectorized 4 elements/thread (float4 state, ushort4 params).
+// HBM-bound (C3 "one kernel launch per bucket"): fully vectorized — 16-B
+// float4 loads/stores on the f32 state, 8-B ushort4 on bf16 param/grad.
+__global__ void adam_vec_k(ushort4* __restrict__ param,
+                           const ushort4* __restrict__ grad,
+                           float4* __restrict__ master, float4* __restrict__ m,
+                           float4* __restrict__ v, int64_t n4, float lr,
+                           float b1, float b2, float eps, float bc1, float bc2,
+                           float gs, float wd) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) {
+    ushort4 gv = grad[i];
+    float4 ma = master[i], mv = m[i], vv = v[i];
+    float gf[4] = {b2f(gv.x) * gs, b2f(gv.y) * gs, b2f(gv.z) * gs,
+                   b2f(gv.w) * gs};
+    float* map = &ma.x;
+    float* mvp = &mv.x;
+    float* vvp = &vv.x;
+    ushort4 pv;
+    uint16_t* pvp = &pv.x;
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      float g = gf[u];
+      if (wd != 0.f) g += wd * map[u];
+      float mi = b1 * mvp[u] + (1 - b1) * g;
+      float vi = b2 * vvp[u] + (1 - b2) * g * g;
+      mvp[u] = mi;
+      vvp[u] = vi;
+      map[u] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+      pvp[u] = f2b(map[u]);
+    }
+    master[i] = ma;
+    m[i] = mv;
+    v[i] = vv;
+    param[i] = pv;
+  }
+}
+
 template <typename TP, typename TG>
 __global__ void adam_k(TP* __restrict__ param, const TG* __restrict__ grad,
                        float* __restrict__ master, float* __restrict__ m,
                        float* __restrict__ v, int64_t n, float lr, float b1,
                        float b2, float eps, float bc1, float bc2, float gs,
                        float wd) {
-  int64_t i4 = (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) * 4;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
-  for (; i4 < n; i4 += stride) {
-    int lim = (int)min((int64_t)4, n - i4);
-#pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      if (u >= lim) break;
-      int64_t i = i4 + u;
-      float g = ld(grad, i) * gs;
-      float ma = master[i];
-      if (wd != 0.f) g += wd * ma;
-      float mi = b1 * m[i] + (1 - b1) * g;
-      float vi = b2 * v[i] + (1 - b2) * g * g;
-      m[i] = mi;
-      v[i] = vi;
-      float upd = lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
-      ma -= upd;
-      master[i] = ma;
-      st(param, i, ma);
-    }
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float g = ld(grad, i) * gs;
+    float ma = master[i];
+    if (wd != 0.f) g += wd * ma;
+    float mi = b1 * m[i] + (1 - b1) * g;
+    float vi = b2 * v[i] + (1 - b2) * g * g;
+    m[i] = mi;
+    v[i] = vi;
+    ma -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    master[i] = ma;
+    st(param, i, ma);
   }
 }
 void adam_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor& m,
@@ -572,7 +665,12 @@ void adam_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor& m,
   int64_t n = param.numel();
   float bc1 = 1.f - powf(b1, (float)step), bc2 = 1.f - powf(b2, (float)step);
   dim3 g(grid_1d((n + 3) / 4, 256)), b(256);
-  if (param.dtype() == DType::BF16)
+  if (param.dtype() == DType::BF16 && (n & 3) == 0)
+    adam_vec_k<<<g, b, 0, s>>>(
+        (ushort4*)param.data(), (const ushort4*)grad.data(),
+        (float4*)master.data(), (float4*)m.data(), (float4*)v.data(), n / 4, lr,
+        b1, b2, eps, bc1, bc2, gs, wd);
+  else if (param.dtype() == DType::BF16)
     adam_k<uint16_t, uint16_t><<<g, b, 0, s>>>(
         (uint16_t*)param.data(), (const uint16_t*)grad.data(),
         (float*)master.data(), (float*)m.data(), (float*)v.data(), n, lr, b1, b2,
