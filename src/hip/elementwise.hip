@@ -739,6 +739,34 @@ __global__ void mse_partial_k(const T* __restrict__ y, const T* __restrict__ t,
   float bs = block_sum<256>(acc, lds4);
   if (threadIdx.x == 0) partials[blockIdx.x] = bs;
 }
+
+// bf16 fast path: ushort4 vector loads/stores (HBM-bound op, Guideline 13).
+__global__ void mse_partial_b16_k(const ushort4* __restrict__ y,
+                                  const ushort4* __restrict__ t,
+                                  ushort4* __restrict__ dy,
+                                  float* __restrict__ partials, int64_t n4,
+                                  float gs_over_n) {
+  __shared__ float lds4[4];
+  float acc = 0.f;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) {
+    ushort4 yv = y[i], tv = t[i];
+    float d0 = b2f(yv.x) - b2f(tv.x), d1 = b2f(yv.y) - b2f(tv.y);
+    float d2 = b2f(yv.z) - b2f(tv.z), d3 = b2f(yv.w) - b2f(tv.w);
+    acc += d0 * d0 + d1 * d1 + d2 * d2 + d3 * d3;
+    if (dy) {
+      ushort4 o;
+      o.x = f2b(2.0f * d0 * gs_over_n);
+      o.y = f2b(2.0f * d1 * gs_over_n);
+      o.z = f2b(2.0f * d2 * gs_over_n);
+      o.w = f2b(2.0f * d3 * gs_over_n);
+      dy[i] = o;
+    }
+  }
+  float bs = block_sum<256>(acc, lds4);
+  if (threadIdx.x == 0) partials[blockIdx.x] = bs;
+}
 __global__ void combine_partials_k(const float* __restrict__ partials, int nb,
                                    float* __restrict__ out, float scale) {
   __shared__ float lds4[4];
@@ -755,7 +783,12 @@ void mse_loss(const Tensor& y, const Tensor& target, Tensor& loss_out, Tensor* d
   int nb = (int)grid_1d((n + 3) / 4, 256, 1024);
   Tensor partials(y.ctx(), DType::F32, Shape{nb});
   dim3 g(nb), b(256);
-  if (y.dtype() == DType::BF16)
+  if (y.dtype() == DType::BF16 && (n & 3) == 0)
+    mse_partial_b16_k<<<g, b, 0, s>>>(
+        (const ushort4*)y.data(), (const ushort4*)target.data(),
+        dy ? (ushort4*)dy->data() : nullptr, (float*)partials.data(), n / 4,
+        gs / float(n));
+  else if (y.dtype() == DType::BF16)
     mse_partial_k<uint16_t><<<g, b, 0, s>>>(
         (const uint16_t*)y.data(), (const uint16_t*)target.data(),
         dy ? (uint16_t*)dy->data() : nullptr, (float*)partials.data(), n, gs);

@@ -247,12 +247,19 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_nt_k(GemmArgs args,
   f32x4 acc[4][4] = {};
   const int64_t ntiles = args.K / NT_BK;
 
+  // LDS XOR swizzle (guide §6 G4 / §5.4 rule 21): without it every 16-lane
+  // ds_read_b128 group of the fragment reads lands on 2 bank groups (8-way
+  // conflict, rows are 128 B). The involution kbyte ^= (row&7)<<4 spreads a
+  // group over 8 slots; with glds the LDS image is lane-linear, so the
+  // permutation is applied to the per-lane global SOURCE address here and
+  // to the ds_read byte address below (same involution both sides).
   auto stage = [&](int buf, int64_t t) {
     int64_t k0 = t * NT_BK;
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       int e = e0 + c * 512;
-      int row = e >> 6, k = e & 63;
+      int row = e >> 6;
+      int k = (e & 63) ^ ((row & 7) << 3);  // element swizzle (16B = 8 elems)
       int64_t ar = m0 + row;
       if (ar > Mc) ar = Mc;  // clamp: masked at the epilogue
       glds16(A + ar * lda + k0 + k, NT_AS(buf) + e);
@@ -275,12 +282,14 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_nt_k(GemmArgs args,
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm) {
         int row = wr * 64 + fm * 16 + (lane & 15);
-        a_frag[fm] = *(const short8*)(NT_AS(cur) + row * NT_BK + kbase);
+        int ks = kbase ^ ((row & 7) << 3);  // read-side of the involution
+        a_frag[fm] = *(const short8*)(NT_AS(cur) + row * NT_BK + ks);
       }
 #pragma unroll
       for (int fn = 0; fn < 4; ++fn) {
         int col = wc * 64 + fn * 16 + (lane & 15);
-        b_frag[fn] = *(const short8*)(NT_BS(cur) + col * NT_BK + kbase);
+        int ks = kbase ^ ((col & 7) << 3);
+        b_frag[fn] = *(const short8*)(NT_BS(cur) + col * NT_BK + ks);
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
