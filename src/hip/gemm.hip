@@ -325,10 +325,157 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_nt_k(GemmArgs args,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Tier B+: deep-pipelined NT kernel. 256x128 tile, 512 threads (8 waves as
+// 4Mx2N, 64x64 per wave), BK=64, THREE LDS buffers (144 KB, 1 block/CU)
+// with counted s_waitcnt vmcnt(N) and raw s_barriers so two tiles of glds
+// stay in flight across every barrier (guide §5 "Pipelining across
+// barriers": a __syncthreads here would carry vmcnt(0) and drain the DMA
+// queue — the ~20% structural stall of the 2-buffer version). Same XOR
+// swizzle and XCD-aware block remap as gemm_bf16_nt_k.
+// Per-thread glds per tile: 4 for A (32 KB) + 2 for B (16 KB) = 6, so
+// "one tile still in flight" == vmcnt(6).
+// ---------------------------------------------------------------------------
+__launch_bounds__(512) __global__ void gemm_bf16_nt_deep_k(GemmArgs args,
+                                                           int nwg_n) {
+  constexpr int DBM = 256, DBN = 128, DBK = 64;
+  constexpr int ABYTES = DBM * DBK;              // elements per A tile (16K)
+  constexpr int BBYTES = DBN * DBK;              // elements per B tile (8K)
+  constexpr int BUFE = ABYTES + BBYTES;          // 24K elements = 48 KB
+  __shared__ __attribute__((aligned(16))) uint16_t smem[3 * BUFE];
+
+  const uint16_t* __restrict__ A = args.A;
+  const uint16_t* __restrict__ B = args.B;
+  uint16_t* C = args.C;
+  uint16_t* save_pre = args.save_pre;
+
+  int nwg = gridDim.x;
+  int orig = blockIdx.x;
+  int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+  int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+  const int64_t m0 = (int64_t)(wgid / nwg_n) * DBM;
+  const int64_t n0 = (int64_t)(wgid % nwg_n) * DBN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;        // 8 waves
+  const int wr = wid >> 1, wc = wid & 1;   // 4x2 wave grid, 64x64 each
+
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  const int64_t ntiles = args.K / DBK;
+
+  // per-lane element offsets: A chunks c=0..3 at e = tid*8 + c*4096,
+  // B chunks c=0..1 at e = tid*8 + c*4096 (8192 elems per 512-thread pass).
+  const int ea = (int)threadIdx.x * 8;
+
+  auto stage = [&](int buf, int64_t t) {
+    int64_t k0 = t * DBK;
+    uint16_t* as = smem + buf * BUFE;
+    uint16_t* bs = as + ABYTES;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int e = ea + c * 4096;
+      int row = e >> 6;
+      int k = (e & 63) ^ ((row & 7) << 3);
+      int64_t ar = m0 + row;
+      if (ar > Mc) ar = Mc;
+      glds16(A + ar * lda + k0 + k, as + e);
+    }
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int e = ea + c * 4096;
+      int row = e >> 6;
+      int k = (e & 63) ^ ((row & 7) << 3);
+      int64_t br = n0 + row;
+      if (br > Nc) br = Nc;
+      glds16(B + br * ldb + k0 + k, bs + e);
+    }
+  };
+
+  f32x4 acc[4][4] = {};
+
+  stage(0, 0);
+  if (ntiles > 1) stage(1, 1);
+
+  for (int64_t t = 0; t < ntiles; ++t) {
+    // open: tile t landed (one younger tile may stay in flight), all waves
+    if (t + 1 < ntiles)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+    // issue tile t+2 into the buffer released at the END of iteration t-1
+    if (t + 2 < ntiles) stage((int)((t + 2) % 3), t + 2);
+
+    const uint16_t* as = smem + (t % 3) * BUFE;
+    const uint16_t* bs = as + ABYTES;
+#pragma unroll
+    for (int kk = 0; kk < DBK; kk += 32) {
+      short8 a_frag[4], b_frag[4];
+      const int kbase = kk + (lane >> 4) * 8;
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        int row = wr * 64 + fm * 16 + (lane & 15);
+        int ks = kbase ^ ((row & 7) << 3);
+        a_frag[fm] = *(const short8*)(as + row * DBK + ks);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = wc * 64 + fn * 16 + (lane & 15);
+        int ks = kbase ^ ((col & 7) << 3);
+        b_frag[fn] = *(const short8*)(bs + col * DBK + ks);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[fm], b_frag[fn], acc[fm][fn], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    // close: every wave done reading buf[t%3]; the DMA writes issued next
+    // iteration may then land in it. lgkmcnt(0) so no ds_read is pending.
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+  }
+
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn) {
+#pragma unroll
+      for (int rI = 0; rI < 4; ++rI) {
+        int64_t grow = m0 + wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+        int64_t gcol = n0 + wc * 64 + fn * 16 + (lane & 15);
+        if (grow >= args.M || gcol >= args.N) continue;
+        int64_t idx = grow * args.N + gcol;
+        float v = args.alpha * acc[fm][fn][rI];
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+        if (args.bias) v += b2f(args.bias[gcol]);
+        if (save_pre) save_pre[idx] = f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        C[idx] = f2b(v);
+      }
+    }
+  }
+}
+
 void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
                  hipStream_t s) {
   NZ_CHECK(args.K % BK == 0, "hip gemm requires K % 64 == 0, got K=" +
                                  std::to_string(args.K));
+  if (!transA && transB && batch == 1 && args.K / 64 >= 4) {
+    // deep-pipelined NT (single-batch large-K shapes)
+    int nwg_m = (int)((args.M + 255) / 256);
+    int nwg_n = (int)((args.N + 127) / 128);
+    dim3 grid(nwg_m * nwg_n);
+    gemm_bf16_nt_deep_k<<<grid, dim3(512), 0, s>>>(args, nwg_n);
+    NZ_KERNEL_CHECK();
+    return;
+  }
   if (!transA && transB) {
     // fast NT path: glds-staged, XCD-swizzled (see gemm_bf16_nt_k)
     int nwg_m = (int)((args.M + BM - 1) / BM);
