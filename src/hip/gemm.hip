@@ -463,12 +463,237 @@ __launch_bounds__(512) __global__ void gemm_bf16_nt_deep_k(GemmArgs args,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Tier B++: 8-phase 256x256 NT kernel (the guide §5 256^2 template class,
+// re-derived for the NT layout with a provably-safe wait schedule).
+//
+// Geometry: BM=BN=256, BK=64, 512 threads = 8 waves (2 wrow x 4 wcol).
+// LDS = 2 tile-buffers x 4 half-tiles (A rows [0,128)/[128,256), B cols
+// same) x 16 KB = 128 KB, ONE __shared__ array. An iteration covers two
+// K-tiles (t even in buf0, t+1 in buf1) as 8 phases; phase p computes one
+// 128x128 C-quadrant (HA,HB) of one tile over its full K=64 (16 MFMA/wave)
+// and issues ONE half-tile DMA stage (2 glds/thread) into the slot whose
+// last reader finished at the previous phase barrier:
+//
+//   p0 t  Q(0,0) stage A0(t+1)   p4 t+1 Q(0,0) stage A0(t+2)
+//   p1 t  Q(1,0) stage B1(t+1)   p5 t+1 Q(1,0) stage B1(t+2)
+//   p2 t  Q(1,1) stage B0(t+2)   p6 t+1 Q(1,1) stage B0(t+3)
+//   p3 t  Q(0,1) stage A1(t+2)   p7 t+1 Q(0,1) stage A1(t+3)
+//
+// Counted waits (vmcnt in 2-load stage units, issue-order landing):
+// p0/p4 wait vmcnt(6) (3 stage-units in flight -> the tile's A0 landed),
+// p2/p6 wait vmcnt(8) (4 units in flight -> its B1 landed); the last two
+// iterations drain with vmcnt(0). Each wait precedes the phase barrier so
+// it covers every wave's DMA, not just the waiting wave's.
+// ---------------------------------------------------------------------------
+namespace ph8 {
+constexpr int PBM = 256, PBN = 256, PBK = 64;
+constexpr int HALF_E = 128 * PBK;            // elements per half-tile (8192)
+constexpr int BUF_E = 4 * HALF_E;            // elements per tile-buffer
+// slot offsets inside smem: buf*BUF_E + (isB*2 + half)*HALF_E
+__device__ __forceinline__ uint16_t* slot(uint16_t* smem, int buf, int isB,
+                                          int half) {
+  return smem + buf * BUF_E + (isB * 2 + half) * HALF_E;
+}
+}  // namespace ph8
+
+__launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase_k(GemmArgs args,
+                                                                int nwg_n) {
+  using namespace ph8;
+  __shared__ __attribute__((aligned(16))) uint16_t smem[2 * BUF_E];
+
+  const uint16_t* __restrict__ A = args.A;
+  const uint16_t* __restrict__ B = args.B;
+  uint16_t* C = args.C;
+  uint16_t* save_pre = args.save_pre;
+
+  int nwg = gridDim.x;
+  int orig = blockIdx.x;
+  int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+  int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+  const int64_t m0 = (int64_t)(wgid / nwg_n) * PBM;
+  const int64_t n0 = (int64_t)(wgid % nwg_n) * PBN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;  // 2x4 wave grid
+
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  const int64_t ntiles = args.K / PBK;  // even, >= 6 (launch guard)
+  const int ea = (int)threadIdx.x * 8;  // first element of this thread's chunk
+
+  // stage one half-tile (128 rows x 64 k) of operand A (isB=0) or B (isB=1),
+  // half h, of K-tile t, into buf = t & 1. 2 glds per thread.
+  auto stage = [&](int isB, int h, int64_t t) {
+    int64_t k0 = t * PBK;
+    uint16_t* dst = slot(smem, (int)(t & 1), isB, h);
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int e = ea + c * 4096;
+      int row = e >> 6;
+      int k = (e & 63) ^ ((row & 7) << 3);
+      if (!isB) {
+        int64_t ar = m0 + h * 128 + row;
+        if (ar > Mc) ar = Mc;
+        glds16(A + ar * lda + k0 + k, dst + e);
+      } else {
+        int64_t br = n0 + h * 128 + row;
+        if (br > Nc) br = Nc;
+        glds16(B + br * ldb + k0 + k, dst + e);
+      }
+    }
+  };
+
+  // acc[HA][HB][fm 0..3][fn 0..1]
+  f32x4 acc[2][2][4][2] = {};
+
+  // one compute phase: quadrant (HA,HB) of tile t (buffer t&1)
+  auto compute = [&](int HA, int HB, int64_t t, auto& accq) {
+    const uint16_t* as = slot(smem, (int)(t & 1), 0, HA);
+    const uint16_t* bs = slot(smem, (int)(t & 1), 1, HB);
+#pragma unroll
+    for (int kk = 0; kk < PBK; kk += 32) {
+      short8 a_frag[4], b_frag[2];
+      const int kbase = kk + (lane >> 4) * 8;
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        int row = wr * 64 + fm * 16 + (lane & 15);
+        int ks = kbase ^ ((row & 7) << 3);
+        a_frag[fm] = *(const short8*)(as + row * PBK + ks);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = wc * 32 + fn * 16 + (lane & 15);
+        int ks = kbase ^ ((col & 7) << 3);
+        b_frag[fn] = *(const short8*)(bs + col * PBK + ks);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn)
+          accq[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[fm], b_frag[fn], accq[fm][fn], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  };
+
+  // prologue: 6 stage-units in the order the wait schedule assumes
+  stage(1, 0, 0);  // B0(0)
+  stage(0, 1, 0);  // A1(0)
+  stage(0, 0, 0);  // A0(0)
+  stage(1, 1, 0);  // B1(0)
+  stage(1, 0, 1);  // B0(1)
+  stage(0, 1, 1);  // A1(1)
+
+#define PH8_WAIT(n)                                        \
+  do {                                                     \
+    if (tail)                                              \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");     \
+    else                                                   \
+      asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory"); \
+    __builtin_amdgcn_s_barrier();                          \
+    asm volatile("" ::: "memory");                         \
+  } while (0)
+#define PH8_BAR()                                          \
+  do {                                                     \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");     \
+    __builtin_amdgcn_s_barrier();                          \
+    asm volatile("" ::: "memory");                         \
+  } while (0)
+
+  for (int64_t t = 0; t < ntiles; t += 2) {
+    const bool tail = (t + 4 >= ntiles);
+    const bool s2 = (t + 2 < ntiles), s3 = (t + 3 < ntiles);
+    // phase 0: Q(0,0) of t; stage A0(t+1)
+    PH8_WAIT(6);
+    stage(0, 0, t + 1);  // A0 of the odd tile: staged same-iteration (p0)
+    compute(0, 0, t, acc[0][0]);
+    PH8_BAR();
+    // phase 1: Q(1,0) of t; stage B1(t+1)
+    stage(1, 1, t + 1);
+    compute(1, 0, t, acc[1][0]);
+    PH8_BAR();
+    // phase 2: Q(1,1) of t; stage B0(t+2)
+    PH8_WAIT(8);
+    if (s2) stage(1, 0, t + 2);
+    compute(1, 1, t, acc[1][1]);
+    PH8_BAR();
+    // phase 3: Q(0,1) of t; stage A1(t+2)
+    if (s2) stage(0, 1, t + 2);
+    compute(0, 1, t, acc[0][1]);
+    PH8_BAR();
+    // phase 4: Q(0,0) of t+1; stage A0(t+2)
+    PH8_WAIT(6);
+    if (s2) stage(0, 0, t + 2);
+    compute(0, 0, t + 1, acc[0][0]);
+    PH8_BAR();
+    // phase 5: Q(1,0) of t+1; stage B1(t+2)
+    if (s2) stage(1, 1, t + 2);
+    compute(1, 0, t + 1, acc[1][0]);
+    PH8_BAR();
+    // phase 6: Q(1,1) of t+1; stage B0(t+3)
+    PH8_WAIT(8);
+    if (s3) stage(1, 0, t + 3);
+    compute(1, 1, t + 1, acc[1][1]);
+    PH8_BAR();
+    // phase 7: Q(0,1) of t+1; stage A1(t+3)
+    if (s3) stage(0, 1, t + 3);
+    compute(0, 1, t + 1, acc[0][1]);
+    PH8_BAR();
+  }
+#undef PH8_WAIT
+#undef PH8_BAR
+
+  // epilogue over the four quadrants
+#pragma unroll
+  for (int ha = 0; ha < 2; ++ha) {
+#pragma unroll
+    for (int hb = 0; hb < 2; ++hb) {
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+          for (int rI = 0; rI < 4; ++rI) {
+            int64_t grow =
+                m0 + ha * 128 + wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+            int64_t gcol = n0 + hb * 128 + wc * 32 + fn * 16 + (lane & 15);
+            if (grow >= args.M || gcol >= args.N) continue;
+            int64_t idx = grow * args.N + gcol;
+            float v = args.alpha * acc[ha][hb][fm][fn][rI];
+            if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+            if (args.bias) v += b2f(args.bias[gcol]);
+            if (save_pre) save_pre[idx] = f2b(v);
+            if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+            C[idx] = f2b(v);
+          }
+        }
+      }
+    }
+  }
+}
+
 void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
                  hipStream_t s) {
   NZ_CHECK(args.K % BK == 0, "hip gemm requires K % 64 == 0, got K=" +
                                  std::to_string(args.K));
-  if (!transA && transB && batch == 1 && args.K / 64 >= 4) {
-    // deep-pipelined NT (single-batch large-K shapes)
+  static const int kNtVariant = [] {
+    const char* v = getenv("NEZHA_GEMM_VARIANT");
+    return v ? atoi(v) : 2;  // 0=2-buf 128x128, 1=3-buf deep, 2=8-phase 256^2
+  }();
+  if (!transA && transB && batch == 1 && kNtVariant == 2 &&
+      args.K % 128 == 0 && args.K / 64 >= 6) {
+    int nwg_m = (int)((args.M + 255) / 256);
+    int nwg_n = (int)((args.N + 255) / 256);
+    dim3 grid(nwg_m * nwg_n);
+    gemm_bf16_nt_8phase_k<<<grid, dim3(512), 0, s>>>(args, nwg_n);
+    NZ_KERNEL_CHECK();
+    return;
+  }
+  if (!transA && transB && batch == 1 && kNtVariant == 1 && args.K / 64 >= 4) {
+    // deep-pipelined NT (kept for A/B comparisons; 2-buf won on occupancy)
     int nwg_m = (int)((args.M + 255) / 256);
     int nwg_n = (int)((args.N + 127) / 128);
     dim3 grid(nwg_m * nwg_n);
