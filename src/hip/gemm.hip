@@ -646,30 +646,58 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase_k(GemmArgs args,
 #undef PH8_WAIT
 #undef PH8_BAR
 
-  // epilogue over the four quadrants
+  // Epilogue: stage the 256x256 bf16 C tile through LDS (exactly reuses the
+  // 128 KB smem, DMA already drained), then stream it out with 16-B vector
+  // stores — the scatter of 128 scalar 2-B stores per lane is store-ISSUE
+  // bound (guide T21); this halves+ the tail at equal bytes.
 #pragma unroll
-  for (int ha = 0; ha < 2; ++ha) {
+  for (int ha = 0; ha < 2; ++ha)
 #pragma unroll
-    for (int hb = 0; hb < 2; ++hb) {
+    for (int hb = 0; hb < 2; ++hb)
 #pragma unroll
-      for (int fm = 0; fm < 4; ++fm) {
+      for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
-        for (int fn = 0; fn < 2; ++fn) {
+        for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
           for (int rI = 0; rI < 4; ++rI) {
-            int64_t grow =
-                m0 + ha * 128 + wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
-            int64_t gcol = n0 + hb * 128 + wc * 32 + fn * 16 + (lane & 15);
-            if (grow >= args.M || gcol >= args.N) continue;
-            int64_t idx = grow * args.N + gcol;
-            float v = args.alpha * acc[ha][hb][fm][fn][rI];
-            if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
-            if (args.bias) v += b2f(args.bias[gcol]);
-            if (save_pre) save_pre[idx] = f2b(v);
-            if (args.fuse_gelu) v = gelu_fwd_scalar(v);
-            C[idx] = f2b(v);
+            int lrow = ha * 128 + wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+            int lcol = hb * 128 + wc * 32 + fn * 16 + (lane & 15);
+            smem[lrow * 256 + lcol] = f2b(args.alpha * acc[ha][hb][fm][fn][rI]);
           }
-        }
+  __syncthreads();
+  const bool interior = (m0 + PBM <= args.M) && (n0 + PBN <= args.N);
+#pragma unroll 4
+  for (int c = 0; c < 16; ++c) {  // 256*256 elems / (512 threads * 8) = 16
+    int e = (c * 512 + (int)threadIdx.x) * 8;
+    int row = e >> 8, col = e & 255;
+    int64_t grow = m0 + row;
+    int64_t gcol = n0 + col;
+    short8 vv = *(const short8*)(smem + e);
+    if (interior) {
+      int64_t idx = grow * args.N + gcol;
+      short8 zout, yout;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        zout[i] = (short)f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        yout[i] = (short)f2b(v);
+      }
+      if (save_pre) *(short8*)(save_pre + idx) = zout;
+      *(short8*)(C + idx) = yout;
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        if (grow >= args.M || gcol + i >= args.N) continue;
+        int64_t idx = grow * args.N + gcol + i;
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        if (save_pre) save_pre[idx] = f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        C[idx] = f2b(v);
       }
     }
   }
@@ -687,10 +715,14 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
       args.K % 128 == 0 && args.K / 64 >= 6) {
     int nwg_m = (int)((args.M + 255) / 256);
     int nwg_n = (int)((args.N + 255) / 256);
-    dim3 grid(nwg_m * nwg_n);
-    gemm_bf16_nt_8phase_k<<<grid, dim3(512), 0, s>>>(args, nwg_n);
-    NZ_KERNEL_CHECK();
-    return;
+    // 256^2 tiles only when the grid fills the chip (1 block/CU, 256 CUs);
+    // small-N/M shapes run faster on the 128^2 2-buffer kernel below.
+    if (nwg_m * (int64_t)nwg_n >= 224) {
+      dim3 grid(nwg_m * nwg_n);
+      gemm_bf16_nt_8phase_k<<<grid, dim3(512), 0, s>>>(args, nwg_n);
+      NZ_KERNEL_CHECK();
+      return;
+    }
   }
   if (!transA && transB && batch == 1 && kNtVariant == 1 && args.K / 64 >= 4) {
     // deep-pipelined NT (kept for A/B comparisons; 2-buf won on occupancy)
