@@ -12,7 +12,7 @@ CXXFLAGS := -O3 -std=c++17 -fPIC -I include -I . --offload-arch=$(ARCH) \
 LDFLAGS  := -L/opt/rocm/lib -lrccl -lz
 
 HOST_SRCS := src/core/tensor.cc src/core/ops.cc src/cpu/cpu_ops.cc \
-             src/util/util.cc src/ckpt/npz.cc src/nn/modules.cc src/nn/optim.cc \
+             src/util/util.cc src/util/prof.cc src/ckpt/npz.cc src/nn/modules.cc src/nn/optim.cc \
              src/models/models.cc src/train/trainer.cc src/train/ddp.cc \
              src/comm/rccl_comm.cc src/comm/fake_comm.cc src/comm/tcp_comm.cc \
              src/capi.cc

@@ -33,6 +33,8 @@ nz_status nz_ctx_sync(nz_ctx*);
 int nz_hip_device_count(void);
 /* allocator high-water mark in bytes */
 int64_t nz_ctx_mem_high_water(nz_ctx*);
+/* write the NEZHA_TRACE chrome-trace file, if tracing is enabled */
+nz_status nz_prof_flush(nz_ctx*);
 
 /* ------------------------------------------------------------- tensors -- */
 typedef struct nz_tensor nz_tensor;

@@ -7,6 +7,7 @@
 #include "src/comm/comm.h"
 #include "src/models/models.h"
 #include "src/train/trainer.h"
+#include "src/util/prof.h"
 #include "src/util/util.h"
 
 using namespace nz;
@@ -58,6 +59,11 @@ extern "C" nz_status nz_ctx_sync(nz_ctx* c) {
   NZ_API_END
 }
 extern "C" int nz_hip_device_count(void) { return hip_device_count(); }
+extern "C" nz_status nz_prof_flush(nz_ctx* c) {
+  NZ_API_BEGIN
+  nz::prof::flush(&c->impl);
+  NZ_API_END
+}
 extern "C" int64_t nz_ctx_mem_high_water(nz_ctx* c) {
   return (int64_t)c->impl.arena()->high_water();
 }

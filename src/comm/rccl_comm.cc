@@ -3,6 +3,11 @@
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
 
+#include <unistd.h>
+
+#include <atomic>
+#include <thread>
+
 #include "src/comm/comm.h"
 #include "src/util/util.h"
 
@@ -40,8 +45,27 @@ class RcclComm : public Comm {
     NZ_NCCL_CHECK(ncclCommInitRank(&comm_, nranks, id, rank));
     NZ_LOG_INFO("RCCL comm initialized: rank %d/%d on device %d", rank, nranks,
                 ctx->device_index());
+    // Failure-detection watchdog: a dead peer must not
+    // hang the node — poll ncclCommGetAsyncError and abort+exit on error.
+    if (nranks > 1) {
+      watchdog_ = std::thread([this] {
+        while (!stop_.load(std::memory_order_relaxed)) {
+          ncclResult_t async = ncclSuccess;
+          if (ncclCommGetAsyncError(comm_, &async) == ncclSuccess &&
+              async != ncclSuccess && async != ncclInProgress) {
+            NZ_LOG_ERROR("RCCL async error on rank %d: %s — aborting comm",
+                         rank_, ncclGetErrorString(async));
+            (void)ncclCommAbort(comm_);
+            _exit(3);
+          }
+          std::this_thread::sleep_for(std::chrono::milliseconds(200));
+        }
+      });
+    }
   }
   ~RcclComm() override {
+    stop_.store(true);
+    if (watchdog_.joinable()) watchdog_.join();
     if (comm_) (void)ncclCommDestroy(comm_);
   }
   int rank() const override { return rank_; }
@@ -81,6 +105,8 @@ class RcclComm : public Comm {
   int nranks_, rank_;
   ncclComm_t comm_ = nullptr;
   Tensor scratch_;
+  std::thread watchdog_;
+  std::atomic<bool> stop_{false};
 };
 
 }  // namespace

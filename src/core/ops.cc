@@ -4,6 +4,7 @@
 
 #include "src/cpu/cpu_ops.h"
 #include "src/hip/hip_ops.h"
+#include "src/util/prof.h"
 
 namespace nz {
 namespace ops {
@@ -12,6 +13,7 @@ static Device dev(const Tensor& t) { return t.device(); }
 
 #define NZ_DISPATCH(fn, first, ...)                  \
   do {                                               \
+    prof::OpScope nz_scope_(#fn, (first).ctx());     \
     if (dev(first) == Device::CPU)                   \
       cpu::fn(__VA_ARGS__);                          \
     else                                             \

@@ -3,6 +3,8 @@
 #include "src/ckpt/npz.h"
 #include "src/util/util.h"
 
+#include <unistd.h>
+
 namespace nz {
 
 Trainer::Trainer(Context* ctx, std::unique_ptr<Model> model,
@@ -28,6 +30,21 @@ void Trainer::attach_ddp(Comm* comm) {
 }
 
 void Trainer::step() {
+  // Fault injection: NEZHA_FAULT_STEP=k (optionally
+  // NEZHA_FAULT_RANK=r) kills this process at step k, exercising the
+  // watchdog + restart-from-checkpoint path in tests.
+  static const long fault_step = [] {
+    const char* v = getenv("NEZHA_FAULT_STEP");
+    return v ? atol(v) : -1L;
+  }();
+  if (fault_step >= 0 && opt_->step_count() + 1 == fault_step) {
+    const char* fr = getenv("NEZHA_FAULT_RANK");
+    int frank = fr ? atoi(fr) : 0;
+    if (!ddp_ || ddp_->comm()->rank() == frank) {
+      NZ_LOG_ERROR("NEZHA_FAULT_STEP: injected fault at step %ld", fault_step);
+      _exit(42);
+    }
+  }
   if (ddp_) ddp_->pre_step();
   tape_.clear();
   VarPtr x = make_var(input_, /*needs_grad=*/false);
