@@ -17,6 +17,8 @@ Trainer::Trainer(Context* ctx, std::unique_ptr<Model> model,
   target_ = Tensor(ctx, dt, model_->output_shape());
   ops::rand_normal(input_, data_seed, 0.f, 1.f);
   ops::rand_normal(target_, data_seed + 1, 0.f, 1.f);
+  if (const char* mp = getenv("NEZHA_METRICS"))
+    metrics_ = std::make_unique<MetricsWriter>(mp);
 }
 
 void Trainer::reseed_data(uint64_t seed) {
@@ -47,12 +49,29 @@ void Trainer::step() {
   }
   if (ddp_) ddp_->pre_step();
   tape_.clear();
+  const bool mx = metrics_ && ctx_->device() == Device::HIP;
+  if (mx) ev_start_.record(*ctx_->compute_stream());
   VarPtr x = make_var(input_, /*needs_grad=*/false);
   VarPtr y = model_->forward(tape_, x);
   loss_ = mse_loss(tape_, y, target_);
+  if (mx) ev_fwd_.record(*ctx_->compute_stream());
   tape_.backward();
   if (ddp_) ddp_->finish_grads();
+  if (mx) ev_bwd_.record(*ctx_->compute_stream());
   opt_->step(ddp_ ? ddp_->grad_scale() : 1.0f);
+  if (mx) {
+    ev_opt_.record(*ctx_->compute_stream());
+    ev_opt_.sync();
+    metrics_->write({{"step", (double)opt_->step_count()},
+                     {"loss", last_loss()},
+                     {"fwd_ms", ev_fwd_.elapsed_ms_since(ev_start_)},
+                     {"bwd_ms", ev_bwd_.elapsed_ms_since(ev_fwd_)},
+                     {"opt_ms", ev_opt_.elapsed_ms_since(ev_bwd_)},
+                     {"mem_hw_gb", ctx_->arena()->high_water() / 1e9}});
+  } else if (metrics_) {
+    metrics_->write({{"step", (double)opt_->step_count()},
+                     {"loss", last_loss()}});
+  }
 }
 
 float Trainer::last_loss() {

@@ -8,6 +8,7 @@
 #include "src/models/models.h"
 #include "src/nn/optim.h"
 #include "src/train/ddp.h"
+#include "src/util/util.h"
 
 namespace nz {
 
@@ -44,6 +45,11 @@ class Trainer {
   std::unique_ptr<Ddp> ddp_;
   Tape tape_;
   Tensor input_, target_, loss_;
+  // per-step metrics: enabled by NEZHA_METRICS=<path>;
+  // fwd/bwd/opt ms splits via hipEvents (syncs each step — observability
+  // mode, off on the bench path).
+  std::unique_ptr<MetricsWriter> metrics_;
+  Event ev_start_, ev_fwd_, ev_bwd_, ev_opt_;
 };
 
 }  // namespace nz
