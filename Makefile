@@ -8,7 +8,7 @@ ARCH     ?= gfx950
 BUILD    ?= build
 
 CXXFLAGS := -O3 -std=c++17 -fPIC -I include -I . --offload-arch=$(ARCH) \
-            -Wall -Wno-unused-function -x hip
+            -Wall -Wno-unused-function -x hip -MMD -MP
 LDFLAGS  := -L/opt/rocm/lib -lrccl -lz
 
 HOST_SRCS := src/core/tensor.cc src/core/ops.cc src/cpu/cpu_ops.cc \
@@ -47,6 +47,8 @@ $(BUILD)/nezha-%: cmd/%.cc $(LIB)
 
 $(BUILD)/test_%: tests/cpp/test_%.cc $(LIB)
 	$(HIPCC) $(CXXFLAGS) $< -o $@ -L$(BUILD) -lnezha $(LDFLAGS) -Wl,-rpath,'$$ORIGIN'
+
+-include $(OBJS:.o=.d)
 
 clean:
 	rm -rf $(BUILD)
