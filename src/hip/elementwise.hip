@@ -279,10 +279,12 @@ void colsum(const Tensor& src, Tensor& dst, float beta) {
   int64_t M = src.rows(), N = src.cols();
   NZ_CHECK(dst.numel() == N, "colsum dst size");
   hipStream_t s = stream_of(src);
-  // enough chunks that colblocks*chunks fills the chip (256 CUs)
+  // enough chunks that colblocks*chunks fills the chip, but capped so the
+  // stage-2 fold stays short (BERT: 512 chunks made stage2 the hot kernel)
   int64_t colblocks = (N + 255) / 256;
-  int chunks = (int)std::min<int64_t>((M + 63) / 64,
-                                      std::max<int64_t>(1, 2048 / colblocks));
+  int chunks = (int)std::min<int64_t>(
+      (M + 63) / 64,
+      std::min<int64_t>(64, std::max<int64_t>(1, 512 / colblocks)));
   int64_t rows_per_chunk = (M + chunks - 1) / chunks;
   Tensor partials(src.ctx(), DType::F32, Shape{(int64_t)chunks, N});
   dim3 g1(colblocks, chunks), b(256);
@@ -495,8 +497,9 @@ void layernorm_bwd(const Tensor& dy, const Tensor& x, const Tensor& gamma,
   int64_t R = x.rows(), N = x.cols();
   dim3 b(256);
   int64_t colblocks = (N + 255) / 256;
-  int chunks = (int)std::min<int64_t>((R + 63) / 64,
-                                      std::max<int64_t>(1, 2048 / colblocks));
+  int chunks = (int)std::min<int64_t>(
+      (R + 63) / 64,
+      std::min<int64_t>(64, std::max<int64_t>(1, 512 / colblocks)));
   int64_t rows_per_chunk = (R + chunks - 1) / chunks;
   Tensor partials(x.ctx(), DType::F32, Shape{(int64_t)chunks * 2, N});
   dim3 g1(colblocks, chunks), g2(colblocks);
@@ -553,10 +556,84 @@ __global__ void softmax_fwd_k(const T* __restrict__ x, T* __restrict__ y,
     __syncthreads();
   }
 }
+// One wave per row for narrow rows (attention scores S<=256): the row
+// lives in registers (<=4 elems/lane), one pass, wave shuffles only.
+template <typename T, int EPL>  // elements per lane (N = EPL*64)
+__global__ void softmax_fwd_wave_k(const T* __restrict__ x, T* __restrict__ y,
+                                   int64_t R, int64_t N, float scale) {
+  int64_t row = blockIdx.x * (int64_t)(blockDim.x / 64) + (threadIdx.x >> 6);
+  int lane = threadIdx.x & 63;
+  if (row >= R) return;
+  const T* xr = x + row * N;
+  float vals[EPL];
+  float m = -INFINITY;
+#pragma unroll
+  for (int i = 0; i < EPL; ++i) {
+    int64_t j = lane * EPL + i;  // contiguous per lane -> 2*EPL B/lane loads
+    vals[i] = j < N ? scale * ld(xr, j) : -INFINITY;
+    m = fmaxf(m, vals[i]);
+  }
+  m = wave_max(m);
+  float sum = 0.f;
+#pragma unroll
+  for (int i = 0; i < EPL; ++i) {
+    vals[i] = __expf(vals[i] - m);
+    sum += vals[i];
+  }
+  sum = wave_sum(sum);
+  float inv = 1.0f / sum;
+  T* yr = y + row * N;
+#pragma unroll
+  for (int i = 0; i < EPL; ++i) {
+    int64_t j = lane * EPL + i;
+    if (j < N) st(yr, j, vals[i] * inv);
+  }
+}
+
+template <typename T, int EPL>
+__global__ void softmax_bwd_wave_k(const T* __restrict__ dy,
+                                   const T* __restrict__ y, T* __restrict__ dx,
+                                   int64_t R, int64_t N, float scale) {
+  int64_t row = blockIdx.x * (int64_t)(blockDim.x / 64) + (threadIdx.x >> 6);
+  int lane = threadIdx.x & 63;
+  if (row >= R) return;
+  const T* dyr = dy + row * N;
+  const T* yr = y + row * N;
+  float dv[EPL], yv[EPL], dot = 0.f;
+#pragma unroll
+  for (int i = 0; i < EPL; ++i) {
+    int64_t j = lane * EPL + i;
+    dv[i] = j < N ? ld(dyr, j) : 0.f;
+    yv[i] = j < N ? ld(yr, j) : 0.f;
+    dot += dv[i] * yv[i];
+  }
+  dot = wave_sum(dot);
+  T* dxr = dx + row * N;
+#pragma unroll
+  for (int i = 0; i < EPL; ++i) {
+    int64_t j = lane * EPL + i;
+    if (j < N) st(dxr, j, scale * yv[i] * (dv[i] - dot));
+  }
+}
+
 void softmax_fwd(const Tensor& x, Tensor& y, float scale) {
   hipStream_t s = stream_of(x);
   int64_t R = x.rows(), N = x.cols();
   dim3 g(grid_1d(R, 1, 4096)), b(256);
+  if (N <= 128 && x.dtype() == DType::BF16) {
+    dim3 gw(grid_1d((R + 3) / 4, 1, (int64_t)1 << 30));
+    softmax_fwd_wave_k<uint16_t, 2><<<gw, b, 0, s>>>(
+        (const uint16_t*)x.data(), (uint16_t*)y.data(), R, N, scale);
+    NZ_KERNEL_CHECK();
+    return;
+  }
+  if (N <= 256 && x.dtype() == DType::BF16) {
+    dim3 gw(grid_1d((R + 3) / 4, 1, (int64_t)1 << 30));
+    softmax_fwd_wave_k<uint16_t, 4><<<gw, b, 0, s>>>(
+        (const uint16_t*)x.data(), (uint16_t*)y.data(), R, N, scale);
+    NZ_KERNEL_CHECK();
+    return;
+  }
   if (x.dtype() == DType::BF16)
     softmax_fwd_k<uint16_t><<<g, b, 0, s>>>((const uint16_t*)x.data(),
                                             (uint16_t*)y.data(), R, N, scale);
@@ -587,6 +664,22 @@ void softmax_bwd(const Tensor& dy, const Tensor& y, Tensor& dx, float scale) {
   hipStream_t s = stream_of(dy);
   int64_t R = y.rows(), N = y.cols();
   dim3 g(grid_1d(R, 1, 4096)), b(256);
+  if (N <= 128 && y.dtype() == DType::BF16) {
+    dim3 gw(grid_1d((R + 3) / 4, 1, (int64_t)1 << 30));
+    softmax_bwd_wave_k<uint16_t, 2><<<gw, b, 0, s>>>(
+        (const uint16_t*)dy.data(), (const uint16_t*)y.data(),
+        (uint16_t*)dx.data(), R, N, scale);
+    NZ_KERNEL_CHECK();
+    return;
+  }
+  if (N <= 256 && y.dtype() == DType::BF16) {
+    dim3 gw(grid_1d((R + 3) / 4, 1, (int64_t)1 << 30));
+    softmax_bwd_wave_k<uint16_t, 4><<<gw, b, 0, s>>>(
+        (const uint16_t*)dy.data(), (const uint16_t*)y.data(),
+        (uint16_t*)dx.data(), R, N, scale);
+    NZ_KERNEL_CHECK();
+    return;
+  }
   if (y.dtype() == DType::BF16)
     softmax_bwd_k<uint16_t><<<g, b, 0, s>>>((const uint16_t*)dy.data(),
                                             (const uint16_t*)y.data(),
@@ -901,12 +994,41 @@ __global__ void attn_split_k(const T* __restrict__ a, T* __restrict__ q,
       const_cast<T*>(a)[i] = dst3[bhsd];
   }
 }
+
+// bf16 fast path: one thread moves 8 contiguous elements along dh
+// (dh % 8 == 0) with one index computation per 16-B chunk.
+template <bool FORWARD>
+__global__ void attn_split_vec_k(uint16_t* __restrict__ a,
+                                 uint16_t* __restrict__ q,
+                                 uint16_t* __restrict__ k,
+                                 uint16_t* __restrict__ v, int64_t chunks,
+                                 int64_t S, int64_t heads, int64_t dh) {
+  using v8 = __attribute__((ext_vector_type(8))) short;
+  int64_t dh8 = dh / 8;
+  int64_t H8 = heads * dh8;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < chunks; i += stride) {
+    int64_t d = i % dh8, h = (i / dh8) % heads, which = (i / H8) % 3;
+    int64_t s_ = (i / (3 * H8)) % S, b = i / (3 * H8 * S);
+    int64_t bhsd = (((b * heads + h) * S + s_) * dh8 + d) * 8;
+    uint16_t* dst3 = which == 0 ? q : (which == 1 ? k : v);
+    if (FORWARD)
+      *(v8*)(dst3 + bhsd) = *(const v8*)(a + i * 8);
+    else
+      *(v8*)(a + i * 8) = *(const v8*)(dst3 + bhsd);
+  }
+}
 void attn_split_qkv(const Tensor& qkv, Tensor& q, Tensor& k, Tensor& v,
                     int64_t B, int64_t S, int64_t heads, int64_t dh) {
   hipStream_t s = stream_of(qkv);
   int64_t total = qkv.numel();
   dim3 g(grid_1d(total, 256, 4096)), b(256);
-  if (qkv.dtype() == DType::BF16)
+  if (qkv.dtype() == DType::BF16 && dh % 8 == 0)
+    attn_split_vec_k<true><<<dim3(grid_1d(total / 8, 256, 4096)), b, 0, s>>>(
+        (uint16_t*)qkv.data(), (uint16_t*)q.data(), (uint16_t*)k.data(),
+        (uint16_t*)v.data(), total / 8, S, heads, dh);
+  else if (qkv.dtype() == DType::BF16)
     attn_split_k<uint16_t, true><<<g, b, 0, s>>>(
         (const uint16_t*)qkv.data(), (uint16_t*)q.data(), (uint16_t*)k.data(),
         (uint16_t*)v.data(), total, S, heads, dh);
@@ -922,7 +1044,11 @@ void attn_split_qkv_bwd(const Tensor& dq, const Tensor& dk, const Tensor& dv,
   hipStream_t s = stream_of(dqkv);
   int64_t total = dqkv.numel();
   dim3 g(grid_1d(total, 256, 4096)), b(256);
-  if (dqkv.dtype() == DType::BF16)
+  if (dqkv.dtype() == DType::BF16 && dh % 8 == 0)
+    attn_split_vec_k<false><<<dim3(grid_1d(total / 8, 256, 4096)), b, 0, s>>>(
+        (uint16_t*)dqkv.data(), (uint16_t*)dq.data(), (uint16_t*)dk.data(),
+        (uint16_t*)dv.data(), total / 8, S, heads, dh);
+  else if (dqkv.dtype() == DType::BF16)
     attn_split_k<uint16_t, false><<<g, b, 0, s>>>(
         (const uint16_t*)dqkv.data(), (uint16_t*)dq.data(), (uint16_t*)dk.data(),
         (uint16_t*)dv.data(), total, S, heads, dh);
@@ -948,12 +1074,35 @@ __global__ void attn_merge_k(const T* __restrict__ ctx, T* __restrict__ out,
       const_cast<T*>(ctx)[bhsd] = out[i];
   }
 }
+
+template <bool FORWARD>
+__global__ void attn_merge_vec_k(uint16_t* __restrict__ ctx,
+                                 uint16_t* __restrict__ out, int64_t chunks,
+                                 int64_t S, int64_t heads, int64_t dh) {
+  using v8 = __attribute__((ext_vector_type(8))) short;
+  int64_t dh8 = dh / 8;
+  int64_t H8 = heads * dh8;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < chunks; i += stride) {  // i indexes chunks of [B,S,H]
+    int64_t d = i % dh8, h = (i / dh8) % heads, s_ = (i / H8) % S,
+            b = i / (H8 * S);
+    int64_t bhsd = (((b * heads + h) * S + s_) * dh8 + d) * 8;
+    if (FORWARD)
+      *(v8*)(out + i * 8) = *(const v8*)(ctx + bhsd);
+    else
+      *(v8*)(ctx + bhsd) = *(const v8*)(out + i * 8);
+  }
+}
 void attn_merge(const Tensor& ctx, Tensor& out, int64_t B, int64_t S,
                 int64_t heads, int64_t dh) {
   hipStream_t s = stream_of(ctx);
   int64_t total = out.numel();
   dim3 g(grid_1d(total, 256, 4096)), b(256);
-  if (ctx.dtype() == DType::BF16)
+  if (ctx.dtype() == DType::BF16 && dh % 8 == 0)
+    attn_merge_vec_k<true><<<dim3(grid_1d(total / 8, 256, 4096)), b, 0, s>>>(
+        (uint16_t*)ctx.data(), (uint16_t*)out.data(), total / 8, S, heads, dh);
+  else if (ctx.dtype() == DType::BF16)
     attn_merge_k<uint16_t, true><<<g, b, 0, s>>>((const uint16_t*)ctx.data(),
                                                  (uint16_t*)out.data(), total, S,
                                                  heads, dh);
@@ -967,7 +1116,10 @@ void attn_merge_bwd(const Tensor& dout, Tensor& dctx, int64_t B, int64_t S,
   hipStream_t s = stream_of(dout);
   int64_t total = dout.numel();
   dim3 g(grid_1d(total, 256, 4096)), b(256);
-  if (dout.dtype() == DType::BF16)
+  if (dout.dtype() == DType::BF16 && dh % 8 == 0)
+    attn_merge_vec_k<false><<<dim3(grid_1d(total / 8, 256, 4096)), b, 0, s>>>(
+        (uint16_t*)dctx.data(), (uint16_t*)dout.data(), total / 8, S, heads, dh);
+  else if (dout.dtype() == DType::BF16)
     attn_merge_k<uint16_t, false><<<g, b, 0, s>>>((const uint16_t*)dctx.data(),
                                                   (uint16_t*)dout.data(), total, S,
                                                   heads, dh);
