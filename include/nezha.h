@@ -74,6 +74,9 @@ nz_status nz_op_adam_step(nz_tensor* param, nz_tensor* grad, nz_tensor* master,
                           float weight_decay);
 nz_status nz_op_sgd_step(nz_tensor* param, nz_tensor* grad, nz_tensor* master,
                          nz_tensor* momentum, float lr, float mu, float grad_scale);
+nz_status nz_op_dropout_fwd(nz_tensor* x, nz_tensor* y, nz_tensor* mask,
+                            float p, uint64_t seed);
+nz_status nz_op_dropout_bwd(nz_tensor* dy, nz_tensor* mask, nz_tensor* dx, float p);
 nz_status nz_op_mse_loss(nz_tensor* y, nz_tensor* target, nz_tensor* loss,
                          nz_tensor* dy, float grad_scale);
 nz_status nz_op_ce_loss(nz_tensor* logits, nz_tensor* labels, nz_tensor* loss,

@@ -73,6 +73,8 @@ _lib.nz_op_softmax_bwd.argtypes = [ctypes.c_void_p] * 3 + [ctypes.c_float]
 _lib.nz_op_adam_step.argtypes = [ctypes.c_void_p] * 5 + [ctypes.c_float] * 4 + [
     ctypes.c_int, ctypes.c_float, ctypes.c_float]
 _lib.nz_op_sgd_step.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_float] * 3
+_lib.nz_op_dropout_fwd.argtypes = [ctypes.c_void_p] * 3 + [ctypes.c_float, ctypes.c_uint64]
+_lib.nz_op_dropout_bwd.argtypes = [ctypes.c_void_p] * 3 + [ctypes.c_float]
 _lib.nz_op_mse_loss.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_float]
 _lib.nz_op_ce_loss.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_float]
 _lib.nz_op_add_inplace.argtypes = [ctypes.c_void_p] * 2 + [ctypes.c_float]
@@ -302,6 +304,14 @@ def adam_step(param, grad, master, m, v, lr, beta1=0.9, beta2=0.999, eps=1e-8,
 def sgd_step(param, grad, master, momentum, lr, mu=0.0, grad_scale=1.0):
     _ck(_lib.nz_op_sgd_step(_h(param), _h(grad), _h(master), _h(momentum), lr,
                             mu, grad_scale))
+
+
+def dropout_fwd(x, y, mask, p, seed):
+    _ck(_lib.nz_op_dropout_fwd(_h(x), _h(y), _h(mask), p, seed))
+
+
+def dropout_bwd(dy, mask, dx, p):
+    _ck(_lib.nz_op_dropout_bwd(_h(dy), _h(mask), _h(dx), p))
 
 
 def mse_loss(y, target, loss, dy=None, grad_scale=1.0):

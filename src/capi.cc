@@ -195,6 +195,18 @@ extern "C" nz_status nz_op_sgd_step(nz_tensor* param, nz_tensor* grad,
                 lr, mu, gs);
   NZ_API_END
 }
+extern "C" nz_status nz_op_dropout_fwd(nz_tensor* x, nz_tensor* y,
+                                       nz_tensor* mask, float p, uint64_t seed) {
+  NZ_API_BEGIN
+  ops::dropout_fwd(x->t, y->t, mask->t, p, seed);
+  NZ_API_END
+}
+extern "C" nz_status nz_op_dropout_bwd(nz_tensor* dy, nz_tensor* mask,
+                                       nz_tensor* dx, float p) {
+  NZ_API_BEGIN
+  ops::dropout_bwd(dy->t, mask->t, dx->t, p);
+  NZ_API_END
+}
 extern "C" nz_status nz_op_mse_loss(nz_tensor* y, nz_tensor* target,
                                     nz_tensor* loss, nz_tensor* dy, float gs) {
   NZ_API_BEGIN

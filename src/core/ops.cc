@@ -72,6 +72,12 @@ void add_inplace(Tensor& dst, const Tensor& src, float alpha) {
 }
 void scale_inplace(Tensor& dst, float alpha) { NZ_DISPATCH(scale_inplace, dst, dst, alpha); }
 void transpose2d(const Tensor& src, Tensor& dst) { NZ_DISPATCH(transpose2d, src, src, dst); }
+void dropout_fwd(const Tensor& x, Tensor& y, Tensor& mask, float p, uint64_t seed) {
+  NZ_DISPATCH(dropout_fwd, x, x, y, mask, p, seed);
+}
+void dropout_bwd(const Tensor& dy, const Tensor& mask, Tensor& dx, float p) {
+  NZ_DISPATCH(dropout_bwd, dy, dy, mask, dx, p);
+}
 void mse_loss(const Tensor& y, const Tensor& target, Tensor& loss_out, Tensor* dy,
               float grad_scale) {
   NZ_DISPATCH(mse_loss, y, y, target, loss_out, dy, grad_scale);

@@ -62,6 +62,13 @@ void add_inplace(Tensor& dst, const Tensor& src, float alpha);  // dst += alpha*
 void scale_inplace(Tensor& dst, float alpha);
 void transpose2d(const Tensor& src, Tensor& dst);     // dst[j,i] = src[i,j]
 
+// Dropout (K8 family; off by default in the contract models). Forward
+// writes y = x * mask / (1-p) and the bf16/f32 mask itself (counter-based
+// philox-style RNG keyed by seed — deterministic, replayable); backward is
+// dx = dy * mask / (1-p) via the same call with the saved mask.
+void dropout_fwd(const Tensor& x, Tensor& y, Tensor& mask, float p, uint64_t seed);
+void dropout_bwd(const Tensor& dy, const Tensor& mask, Tensor& dx, float p);
+
 // ---- K9: losses. loss_out: f32 scalar tensor on device. dy optional.
 void mse_loss(const Tensor& y, const Tensor& target, Tensor& loss_out,
               Tensor* dy, float grad_scale);

@@ -392,6 +392,37 @@ void transpose2d(const Tensor& src, Tensor& dst) {
       std::memcpy(d + (j * R + i) * es, s + (i * C + j) * es, es);
 }
 
+void dropout_fwd(const Tensor& x, Tensor& y, Tensor& mask, float p, uint64_t seed) {
+  int64_t n = x.numel();
+  float keep = 1.0f - p, inv = keep > 0 ? 1.0f / keep : 0.f;
+  auto get = [&](const Tensor& t, int64_t i) {
+    return t.dtype() == DType::F32 ? ((const float*)t.data())[i]
+                                   : bf16_to_f32(((const uint16_t*)t.data())[i]);
+  };
+  auto put = [&](Tensor& t, int64_t i, float v) {
+    if (t.dtype() == DType::F32) ((float*)t.data())[i] = v;
+    else ((uint16_t*)t.data())[i] = f32_to_bf16(v);
+  };
+  for (int64_t i = 0; i < n; ++i) {
+    float m = rng_uniform01(seed, (uint64_t)i) >= p ? 1.0f : 0.0f;
+    put(mask, i, m);
+    put(y, i, get(x, i) * m * inv);
+  }
+}
+void dropout_bwd(const Tensor& dy, const Tensor& mask, Tensor& dx, float p) {
+  int64_t n = dy.numel();
+  float keep = 1.0f - p, inv = keep > 0 ? 1.0f / keep : 0.f;
+  auto get = [&](const Tensor& t, int64_t i) {
+    return t.dtype() == DType::F32 ? ((const float*)t.data())[i]
+                                   : bf16_to_f32(((const uint16_t*)t.data())[i]);
+  };
+  for (int64_t i = 0; i < n; ++i) {
+    float v = get(dy, i) * get(mask, i) * inv;
+    if (dx.dtype() == DType::F32) ((float*)dx.data())[i] = v;
+    else ((uint16_t*)dx.data())[i] = f32_to_bf16(v);
+  }
+}
+
 // ------------------------------------------------------------- k9 losses ---
 void mse_loss(const Tensor& y, const Tensor& target, Tensor& loss_out, Tensor* dy,
               float grad_scale) {

@@ -35,6 +35,8 @@ void cast_copy(Tensor& dst, const Tensor& src);
 void add_inplace(Tensor& dst, const Tensor& src, float alpha);
 void scale_inplace(Tensor& dst, float alpha);
 void transpose2d(const Tensor& src, Tensor& dst);
+void dropout_fwd(const Tensor& x, Tensor& y, Tensor& mask, float p, uint64_t seed);
+void dropout_bwd(const Tensor& dy, const Tensor& mask, Tensor& dx, float p);
 void mse_loss(const Tensor& y, const Tensor& target, Tensor& loss_out, Tensor* dy,
               float grad_scale);
 void ce_loss(const Tensor& logits, const Tensor& labels, Tensor& loss_out,

@@ -812,6 +812,56 @@ void sgd_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor* momentu
   NZ_KERNEL_CHECK();
 }
 
+// ----------------------------------------------------------- K8: dropout ---
+template <typename T>
+__global__ void dropout_fwd_k(const T* __restrict__ x, T* __restrict__ y,
+                              T* __restrict__ mask, int64_t n, float p,
+                              float inv, uint64_t seed) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float m = rng_uniform01(seed, (uint64_t)i) >= p ? 1.0f : 0.0f;
+    st(mask, i, m);
+    st(y, i, ld(x, i) * m * inv);
+  }
+}
+template <typename T>
+__global__ void dropout_bwd_k(const T* __restrict__ dy, const T* __restrict__ mask,
+                              T* __restrict__ dx, int64_t n, float inv) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) st(dx, i, ld(dy, i) * ld(mask, i) * inv);
+}
+void dropout_fwd(const Tensor& x, Tensor& y, Tensor& mask, float p, uint64_t seed) {
+  hipStream_t s = stream_of(x);
+  int64_t n = x.numel();
+  float inv = p < 1.f ? 1.0f / (1.0f - p) : 0.f;
+  dim3 g(grid_1d(n, 256)), b(256);
+  if (x.dtype() == DType::BF16)
+    dropout_fwd_k<uint16_t><<<g, b, 0, s>>>((const uint16_t*)x.data(),
+                                            (uint16_t*)y.data(),
+                                            (uint16_t*)mask.data(), n, p, inv, seed);
+  else
+    dropout_fwd_k<float><<<g, b, 0, s>>>((const float*)x.data(), (float*)y.data(),
+                                         (float*)mask.data(), n, p, inv, seed);
+  NZ_KERNEL_CHECK();
+}
+void dropout_bwd(const Tensor& dy, const Tensor& mask, Tensor& dx, float p) {
+  hipStream_t s = stream_of(dy);
+  int64_t n = dy.numel();
+  float inv = p < 1.f ? 1.0f / (1.0f - p) : 0.f;
+  dim3 g(grid_1d(n, 256)), b(256);
+  if (dy.dtype() == DType::BF16)
+    dropout_bwd_k<uint16_t><<<g, b, 0, s>>>((const uint16_t*)dy.data(),
+                                            (const uint16_t*)mask.data(),
+                                            (uint16_t*)dx.data(), n, inv);
+  else
+    dropout_bwd_k<float><<<g, b, 0, s>>>((const float*)dy.data(),
+                                         (const float*)mask.data(),
+                                         (float*)dx.data(), n, inv);
+  NZ_KERNEL_CHECK();
+}
+
 // ------------------------------------------------------------ K9: losses ---
 // Two-stage deterministic reduction: per-block partials, then single-block
 // combine (no float atomics -> bit-stable run to run).

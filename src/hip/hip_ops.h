@@ -37,6 +37,8 @@ void transpose2d(const Tensor& src, Tensor& dst);
 // batched fast transpose (R,C % 64): dst[b][c][r] = src[b][r][c]
 void transpose2d_batched(const Tensor& src, Tensor& dst, int64_t batch,
                          int64_t R, int64_t C, int64_t sSrc, int64_t sDst);
+void dropout_fwd(const Tensor& x, Tensor& y, Tensor& mask, float p, uint64_t seed);
+void dropout_bwd(const Tensor& dy, const Tensor& mask, Tensor& dx, float p);
 void mse_loss(const Tensor& y, const Tensor& target, Tensor& loss_out, Tensor* dy,
               float grad_scale);
 void ce_loss(const Tensor& logits, const Tensor& labels, Tensor& loss_out,

@@ -201,3 +201,21 @@ def test_attn_split_merge(cpu_ctx):
     nz.attn_merge(q, out, B, S, H, D)
     merged_ref = ref[0].transpose(0, 2, 1, 3).reshape(B * S, H * D)
     np.testing.assert_array_equal(out.numpy(), merged_ref)
+
+
+def test_dropout(cpu_ctx):
+    n = 10000
+    X = randf(n)
+    x = cpu_ctx.from_numpy(X)
+    y, mask = cpu_ctx.tensor((n,)), cpu_ctx.tensor((n,))
+    nz.dropout_fwd(x, y, mask, 0.3, seed=11)
+    m = mask.numpy()
+    assert abs(m.mean() - 0.7) < 0.03
+    np.testing.assert_allclose(y.numpy(), X * m / 0.7, rtol=1e-6, atol=1e-7)
+    dy = cpu_ctx.from_numpy(X)
+    dx = cpu_ctx.tensor((n,))
+    nz.dropout_bwd(dy, mask, dx, 0.3)
+    np.testing.assert_allclose(dx.numpy(), X * m / 0.7, rtol=1e-6, atol=1e-7)
+    # p=0 is identity with all-ones mask
+    nz.dropout_fwd(x, y, mask, 0.0, seed=11)
+    np.testing.assert_allclose(y.numpy(), X, rtol=1e-6)

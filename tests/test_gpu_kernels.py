@@ -263,3 +263,20 @@ def test_attn_split_merge(hip_ctx):
     nz.attn_merge(q, out, B, S, H, D)
     np.testing.assert_array_equal(
         out.numpy(), ref[0].transpose(0, 2, 1, 3).reshape(B * S, H * D))
+
+
+def test_dropout_gpu(hip_ctx):
+    n = 65536
+    X = randb(n)
+    x = hip_ctx.from_numpy_bf16(X)
+    y, mask = hip_ctx.tensor((n,), nz.BF16), hip_ctx.tensor((n,), nz.BF16)
+    nz.dropout_fwd(x, y, mask, 0.25, seed=3)
+    m = mask.numpy()
+    assert abs(m.mean() - 0.75) < 0.02
+    check(y.numpy(), X * m / 0.75, rtol=0.01)
+    # GPU mask equals CPU mask bitwise (same counter-based RNG)
+    cpu = nz.Ctx(nz.CPU)
+    xc = cpu.from_numpy(np.zeros(n, dtype=np.float32))
+    yc, mc = cpu.tensor((n,)), cpu.tensor((n,))
+    nz.dropout_fwd(xc, yc, mc, 0.25, seed=3)
+    np.testing.assert_array_equal(m, mc.numpy())
