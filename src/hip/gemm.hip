@@ -703,6 +703,223 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase_k(GemmArgs args,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Variant 3: 8-phase kernel with fragment reads PIPELINED one phase ahead
+// (the guide's "per-phase interleave is the lever"): body p issues the
+// ds_reads of phase p+1 right before the MFMAs of phase p, so LDS reads,
+// in-flight half-tile DMA and the matrix pipe overlap. Body structure:
+//   [vmcnt_p?; s_barrier; stage S(p); lgkmcnt(0); R(p+1); MFMA(p)]
+// The stage moves AFTER the barrier (all waves' R(p-1) reads completed
+// before the slot is overwritten); the wait schedule shifts one phase
+// earlier than variant 2 (derivation in the body comments):
+//   before p1: vmcnt(6)  before p3: vmcnt(4)
+//   before p5: vmcnt(6)  before p7: vmcnt(4)
+// Dual fragment register sets (p&1) keep reads and compute independent.
+// ---------------------------------------------------------------------------
+__launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase2_k(GemmArgs args,
+                                                                 int nwg_n) {
+  using namespace ph8;
+  __shared__ __attribute__((aligned(16))) uint16_t smem[2 * BUF_E];
+
+  const uint16_t* __restrict__ A = args.A;
+  const uint16_t* __restrict__ B = args.B;
+  uint16_t* C = args.C;
+  uint16_t* save_pre = args.save_pre;
+
+  int nwg = gridDim.x;
+  int orig = blockIdx.x;
+  int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+  int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+  const int64_t m0 = (int64_t)(wgid / nwg_n) * PBM;
+  const int64_t n0 = (int64_t)(wgid % nwg_n) * PBN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;
+
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  const int64_t ntiles = args.K / PBK;
+  const int ea = (int)threadIdx.x * 8;
+
+  auto stage = [&](int isB, int h, int64_t t) {
+    int64_t k0 = t * PBK;
+    uint16_t* dst = slot(smem, (int)(t & 1), isB, h);
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int e = ea + c * 4096;
+      int row = e >> 6;
+      int k = (e & 63) ^ ((row & 7) << 3);
+      if (!isB) {
+        int64_t ar = m0 + h * 128 + row;
+        if (ar > Mc) ar = Mc;
+        glds16(A + ar * lda + k0 + k, dst + e);
+      } else {
+        int64_t br = n0 + h * 128 + row;
+        if (br > Nc) br = Nc;
+        glds16(B + br * ldb + k0 + k, dst + e);
+      }
+    }
+  };
+
+  f32x4 acc[2][2][4][2] = {};
+  // two half-phase (one kk step) fragment register sets, rotated A/B
+  short8 fa[2][4], fb[2][2];
+
+  // issue the ds_reads for kk-half KH of quadrant (HA,HB) of tile t into
+  // register set S; the compiler tracks the dependency and inserts the
+  // counted lgkm waits before the consuming MFMAs (guide: hipcc's
+  // ds_read->MFMA scheduling is near-optimal — issue early, don't wait).
+  auto rd = [&](int S, int KH, int HA, int HB, int64_t t) {
+    const uint16_t* as = slot(smem, (int)(t & 1), 0, HA);
+    const uint16_t* bs = slot(smem, (int)(t & 1), 1, HB);
+    const int kbase = KH * 32 + (lane >> 4) * 8;
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm) {
+      int row = wr * 64 + fm * 16 + (lane & 15);
+      int ks = kbase ^ ((row & 7) << 3);
+      fa[S][fm] = *(const short8*)(as + row * PBK + ks);
+    }
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+      int col = wc * 32 + fn * 16 + (lane & 15);
+      int ks = kbase ^ ((col & 7) << 3);
+      fb[S][fn] = *(const short8*)(bs + col * PBK + ks);
+    }
+  };
+  auto mm = [&](int S, auto& accq) {
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+        accq[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            fa[S][fm], fb[S][fn], accq[fm][fn], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+  // prologue: 6 half-tile stages; wait tile0's A0/B0/A1 prefix; read the
+  // first kk-half of Q0(t0) into set 0
+  stage(1, 0, 0);
+  stage(0, 1, 0);
+  stage(0, 0, 0);
+  stage(1, 1, 0);
+  stage(1, 0, 1);
+  stage(0, 1, 1);
+  asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  rd(0, 0, 0, 0, 0);
+
+#define PH8B(waitn)                                            do {                                                           if (tail)                                                      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");           else if (waitn > 0)                                            asm volatile("s_waitcnt vmcnt(" #waitn ")" ::: "memory");     __builtin_amdgcn_s_barrier();                                asm volatile("" ::: "memory");                             } while (0)
+
+  // Body p (quadrant Q, tile tau): on entry set0/1 alternates holding
+  // (p, kk=0); the body issues (p, kk=1) into the other set, runs kk=0
+  // MFMAs, issues (p+1, kk=0) back into the first set, runs kk=1 MFMAs.
+  // Reads of body p+1's data are guarded by the vmcnt BEFORE body p's
+  // barrier (schedule in the header comment); slot reuse is safe because
+  // a stage only runs after the barrier that every wave passed having
+  // consumed its reads of that slot.
+#define PH8_BODY(SETA, SETB, HA, HB, TT, NHA, NHB, NTT, GUARD)          do {                                                                    rd(SETB, 1, HA, HB, TT);                                              mm(SETA, acc[HA][HB]);                                                if (GUARD) rd(SETA, 0, NHA, NHB, NTT);                                mm(SETB, acc[HA][HB]);                                              } while (0)
+
+  for (int64_t t = 0; t < ntiles; t += 2) {
+    const bool tail = (t + 4 >= ntiles);
+    const bool s2 = (t + 2 < ntiles), s3 = (t + 3 < ntiles);
+    const bool last = (t + 2 >= ntiles);
+    // p0: Q(0,0) of t; stage A0(t+1); next reads Q(1,0) of t
+    PH8B(0);
+    stage(0, 0, t + 1);
+    PH8_BODY(0, 1, 0, 0, t, 1, 0, t, true);
+    // p1: Q(1,0) of t; stage B1(t+1); next Q(1,1) of t (needs B1(t): vmcnt)
+    PH8B(6);
+    stage(1, 1, t + 1);
+    PH8_BODY(0, 1, 1, 0, t, 1, 1, t, true);
+    // p2: Q(1,1) of t; stage B0(t+2); next Q(0,1) of t
+    PH8B(0);
+    if (s2) stage(1, 0, t + 2);
+    PH8_BODY(0, 1, 1, 1, t, 0, 1, t, true);
+    // p3: Q(0,1) of t; stage A1(t+2); next Q(0,0) of t+1 (needs A0(t+1))
+    PH8B(4);
+    if (s2) stage(0, 1, t + 2);
+    PH8_BODY(0, 1, 0, 1, t, 0, 0, t + 1, true);
+    // p4: Q(0,0) of t+1; stage A0(t+2); next Q(1,0) of t+1
+    PH8B(0);
+    if (s2) stage(0, 0, t + 2);
+    PH8_BODY(0, 1, 0, 0, t + 1, 1, 0, t + 1, true);
+    // p5: Q(1,0) of t+1; stage B1(t+2); next Q(1,1) of t+1 (needs B1(t+1))
+    PH8B(6);
+    if (s2) stage(1, 1, t + 2);
+    PH8_BODY(0, 1, 1, 0, t + 1, 1, 1, t + 1, true);
+    // p6: Q(1,1) of t+1; stage B0(t+3); next Q(0,1) of t+1
+    PH8B(0);
+    if (s3) stage(1, 0, t + 3);
+    PH8_BODY(0, 1, 1, 1, t + 1, 0, 1, t + 1, true);
+    // p7: Q(0,1) of t+1; stage A1(t+3); next Q(0,0) of t+2 (needs A0(t+2))
+    PH8B(4);
+    if (s3) stage(0, 1, t + 3);
+    PH8_BODY(0, 1, 0, 1, t + 1, 0, 0, t + 2, !last);
+  }
+#undef PH8_BODY
+#undef PH8B
+
+  // LDS-staged vector epilogue (same as variant 2)
+  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+#pragma unroll
+  for (int ha = 0; ha < 2; ++ha)
+#pragma unroll
+    for (int hb = 0; hb < 2; ++hb)
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+          for (int rI = 0; rI < 4; ++rI) {
+            int lrow = ha * 128 + wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+            int lcol = hb * 128 + wc * 32 + fn * 16 + (lane & 15);
+            smem[lrow * 256 + lcol] = f2b(args.alpha * acc[ha][hb][fm][fn][rI]);
+          }
+  __syncthreads();
+  const bool interior = (m0 + PBM <= args.M) && (n0 + PBN <= args.N);
+#pragma unroll 4
+  for (int c = 0; c < 16; ++c) {
+    int e = (c * 512 + (int)threadIdx.x) * 8;
+    int row = e >> 8, col = e & 255;
+    int64_t grow = m0 + row;
+    int64_t gcol = n0 + col;
+    short8 vv = *(const short8*)(smem + e);
+    if (interior) {
+      int64_t idx = grow * args.N + gcol;
+      short8 zout, yout;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        zout[i] = (short)f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        yout[i] = (short)f2b(v);
+      }
+      if (save_pre) *(short8*)(save_pre + idx) = zout;
+      *(short8*)(C + idx) = yout;
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        if (grow >= args.M || gcol + i >= args.N) continue;
+        int64_t idx = grow * args.N + gcol + i;
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        if (save_pre) save_pre[idx] = f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        C[idx] = f2b(v);
+      }
+    }
+  }
+}
+
 void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
                  hipStream_t s) {
   NZ_CHECK(args.K % BK == 0, "hip gemm requires K % 64 == 0, got K=" +
@@ -711,13 +928,28 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     const char* v = getenv("NEZHA_GEMM_VARIANT");
     return v ? atoi(v) : 2;  // 0=2-buf 128x128, 1=3-buf deep, 2=8-phase 256^2
   }();
-  if (!transA && transB && batch == 1 && kNtVariant == 2 &&
+  static const int kNtMinWG = [] {
+    const char* v = getenv("NEZHA_GEMM_MIN_WG");  // test hook: force 8-phase
+    return v ? atoi(v) : 224;
+  }();
+  if (!transA && transB && batch == 1 && kNtVariant == 3 &&
+      args.K % 128 == 0 && args.K / 64 >= 6) {
+    int nwg_m = (int)((args.M + 255) / 256);
+    int nwg_n = (int)((args.N + 255) / 256);
+    if (nwg_m * (int64_t)nwg_n >= kNtMinWG) {
+      dim3 grid(nwg_m * nwg_n);
+      gemm_bf16_nt_8phase2_k<<<grid, dim3(512), 0, s>>>(args, nwg_n);
+      NZ_KERNEL_CHECK();
+      return;
+    }
+  }
+  if (!transA && transB && batch == 1 && kNtVariant >= 2 &&
       args.K % 128 == 0 && args.K / 64 >= 6) {
     int nwg_m = (int)((args.M + 255) / 256);
     int nwg_n = (int)((args.N + 255) / 256);
     // 256^2 tiles only when the grid fills the chip (1 block/CU, 256 CUs);
     // small-N/M shapes run faster on the 128^2 2-buffer kernel below.
-    if (nwg_m * (int64_t)nwg_n >= 224) {
+    if (nwg_m * (int64_t)nwg_n >= kNtMinWG) {
       dim3 grid(nwg_m * nwg_n);
       gemm_bf16_nt_8phase_k<<<grid, dim3(512), 0, s>>>(args, nwg_n);
       NZ_KERNEL_CHECK();
