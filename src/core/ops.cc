@@ -52,8 +52,8 @@ void softmax_bwd(const Tensor& dy, const Tensor& y, Tensor& dx, float scale) {
 }
 void adam_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor& m,
                Tensor& v, float lr, float b1, float b2, float eps, int step,
-               float grad_scale, float wd) {
-  NZ_DISPATCH(adam_step, param, param, grad, master, m, v, lr, b1, b2, eps, step, grad_scale, wd);
+               float grad_scale, float wd, const Tensor* bc_dev) {
+  NZ_DISPATCH(adam_step, param, param, grad, master, m, v, lr, b1, b2, eps, step, grad_scale, wd, bc_dev);
 }
 void sgd_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor* momentum,
               float lr, float mu, float grad_scale) {

@@ -47,9 +47,13 @@ void softmax_bwd(const Tensor& dy, const Tensor& y, Tensor& dx, float scale);
 // ---- K6/K7: optimizer steps on flat buckets.
 // param bf16 (HIP) / f32 (CPU); master/m/v f32. grad_scale multiplies the
 // raw gradient (1/world_size fused here for DP averaging).
+// bc_dev (optional, HIP): f32[2] device tensor holding the bias-correction
+// pair {1-b1^t, 1-b2^t}; when given it overrides `step` so the launch can
+// be captured in a hipGraph and replayed with fresh corrections.
 void adam_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor& m,
                Tensor& v, float lr, float beta1, float beta2, float eps,
-               int step, float grad_scale, float weight_decay);
+               int step, float grad_scale, float weight_decay,
+               const Tensor* bc_dev = nullptr);
 void sgd_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor* momentum,
               float lr, float mu, float grad_scale);
 

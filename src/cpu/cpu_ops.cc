@@ -249,7 +249,8 @@ void softmax_bwd(const Tensor& dy, const Tensor& y, Tensor& dx, float scale) {
 // ----------------------------------------------------------- k6/k7 optim ---
 void adam_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor& m,
                Tensor& v, float lr, float b1, float b2, float eps, int step,
-               float grad_scale, float wd) {
+               float grad_scale, float wd, const Tensor* bc_dev) {
+  (void)bc_dev;  // CPU path has no graph capture; `step` is authoritative
   float* mast = f32p(master);
   float* mp = f32p(m);
   float* vp = f32p(v);

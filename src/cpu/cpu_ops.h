@@ -25,7 +25,7 @@ void softmax_fwd(const Tensor& x, Tensor& y, float scale);
 void softmax_bwd(const Tensor& dy, const Tensor& y, Tensor& dx, float scale);
 void adam_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor& m,
                Tensor& v, float lr, float b1, float b2, float eps, int step,
-               float grad_scale, float wd);
+               float grad_scale, float wd, const Tensor* bc_dev = nullptr);
 void sgd_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor* momentum,
               float lr, float mu, float grad_scale);
 void fill(Tensor& t, float value);

@@ -699,7 +699,8 @@ __global__ void adam_vec_k(ushort4* __restrict__ param,
                            float4* __restrict__ master, float4* __restrict__ m,
                            float4* __restrict__ v, int64_t n4, float lr,
                            float b1, float b2, float eps, float bc1, float bc2,
-                           float gs, float wd) {
+                           float gs, float wd, const float* __restrict__ bc_dev) {
+  if (bc_dev) { bc1 = bc_dev[0]; bc2 = bc_dev[1]; }  // graph-replay path
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n4; i += stride) {
@@ -753,7 +754,7 @@ __global__ void adam_k(TP* __restrict__ param, const TG* __restrict__ grad,
 }
 void adam_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor& m,
                Tensor& v, float lr, float b1, float b2, float eps, int step,
-               float gs, float wd) {
+               float gs, float wd, const Tensor* bc_dev) {
   hipStream_t s = stream_of(param);
   int64_t n = param.numel();
   float bc1 = 1.f - powf(b1, (float)step), bc2 = 1.f - powf(b2, (float)step);
@@ -762,8 +763,10 @@ void adam_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor& m,
     adam_vec_k<<<g, b, 0, s>>>(
         (ushort4*)param.data(), (const ushort4*)grad.data(),
         (float4*)master.data(), (float4*)m.data(), (float4*)v.data(), n / 4, lr,
-        b1, b2, eps, bc1, bc2, gs, wd);
+        b1, b2, eps, bc1, bc2, gs, wd,
+        bc_dev ? (const float*)bc_dev->data() : nullptr);
   else if (param.dtype() == DType::BF16)
+    // scalar fallback ignores bc_dev (not used on the graph path)
     adam_k<uint16_t, uint16_t><<<g, b, 0, s>>>(
         (uint16_t*)param.data(), (const uint16_t*)grad.data(),
         (float*)master.data(), (float*)m.data(), (float*)v.data(), n, lr, b1, b2,

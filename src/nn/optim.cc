@@ -62,12 +62,12 @@ Optimizer::Optimizer(Context* ctx, std::vector<Param*> params, Kind k, float lr_
   }
 }
 
-void Optimizer::step(float grad_scale) {
+void Optimizer::step(float grad_scale, const Tensor* bc_dev) {
   ++step_;
   for (auto& b : buckets_) {
     if (kind == Kind::ADAM) {
       ops::adam_step(b.data_flat, b.grad_flat, b.master, b.m, b.v, lr, beta1,
-                     beta2, eps, step_, grad_scale, weight_decay);
+                     beta2, eps, step_, grad_scale, weight_decay, bc_dev);
     } else {
       if (sgd_momentum != 0.f && !b.momentum.defined()) {
         b.momentum = Tensor(ctx_, DType::F32, Shape{b.numel});

@@ -29,7 +29,7 @@ class Optimizer {
   enum class Kind { ADAM, SGD };
   Optimizer(Context* ctx, std::vector<Param*> params, Kind kind, float lr,
             int64_t bucket_bytes);
-  void step(float grad_scale);
+  void step(float grad_scale, const Tensor* bc_dev = nullptr);
   std::vector<Bucket>& buckets() { return buckets_; }
   const std::vector<Param*>& params() const { return params_; }
   int step_count() const { return step_; }

@@ -1,5 +1,7 @@
 #include "src/train/trainer.h"
 
+#include <hip/hip_runtime.h>
+
 #include "src/ckpt/npz.h"
 #include "src/util/util.h"
 
@@ -19,6 +21,8 @@ Trainer::Trainer(Context* ctx, std::unique_ptr<Model> model,
   ops::rand_normal(target_, data_seed + 1, 0.f, 1.f);
   if (const char* mp = getenv("NEZHA_METRICS"))
     metrics_ = std::make_unique<MetricsWriter>(mp);
+  const char* g = getenv("NEZHA_GRAPH");
+  graph_mode_ = g && g[0] == '1' && ctx->device() == Device::HIP;
 }
 
 void Trainer::reseed_data(uint64_t seed) {
@@ -31,25 +35,10 @@ void Trainer::attach_ddp(Comm* comm) {
   ddp_->sync_params();
 }
 
-void Trainer::step() {
-  // Fault injection: NEZHA_FAULT_STEP=k (optionally
-  // NEZHA_FAULT_RANK=r) kills this process at step k, exercising the
-  // watchdog + restart-from-checkpoint path in tests.
-  static const long fault_step = [] {
-    const char* v = getenv("NEZHA_FAULT_STEP");
-    return v ? atol(v) : -1L;
-  }();
-  if (fault_step >= 0 && opt_->step_count() + 1 == fault_step) {
-    const char* fr = getenv("NEZHA_FAULT_RANK");
-    int frank = fr ? atoi(fr) : 0;
-    if (!ddp_ || ddp_->comm()->rank() == frank) {
-      NZ_LOG_ERROR("NEZHA_FAULT_STEP: injected fault at step %ld", fault_step);
-      _exit(42);
-    }
-  }
+void Trainer::step_body(const Tensor* bc_dev) {
   if (ddp_) ddp_->pre_step();
   tape_.clear();
-  const bool mx = metrics_ && ctx_->device() == Device::HIP;
+  const bool mx = metrics_ && ctx_->device() == Device::HIP && !graph_mode_;
   if (mx) ev_start_.record(*ctx_->compute_stream());
   VarPtr x = make_var(input_, /*needs_grad=*/false);
   VarPtr y = model_->forward(tape_, x);
@@ -58,7 +47,7 @@ void Trainer::step() {
   tape_.backward();
   if (ddp_) ddp_->finish_grads();
   if (mx) ev_bwd_.record(*ctx_->compute_stream());
-  opt_->step(ddp_ ? ddp_->grad_scale() : 1.0f);
+  opt_->step(ddp_ ? ddp_->grad_scale() : 1.0f, bc_dev);
   if (mx) {
     ev_opt_.record(*ctx_->compute_stream());
     ev_opt_.sync();
@@ -72,6 +61,74 @@ void Trainer::step() {
     metrics_->write({{"step", (double)opt_->step_count()},
                      {"loss", last_loss()}});
   }
+}
+
+// NEZHA_FAULT_STEP handling + graph capture/replay wrapper.
+void Trainer::step() {
+  static const long fault_step = [] {
+    const char* v = getenv("NEZHA_FAULT_STEP");
+    return v ? atol(v) : -1L;
+  }();
+  if (fault_step >= 0 && opt_->step_count() + 1 == fault_step) {
+    const char* fr = getenv("NEZHA_FAULT_RANK");
+    int frank = fr ? atoi(fr) : 0;
+    if (!ddp_ || ddp_->comm()->rank() == frank) {
+      NZ_LOG_ERROR("NEZHA_FAULT_STEP: injected fault at step %ld", fault_step);
+      _exit(42);
+    }
+  }
+
+  if (!graph_mode_ || ddp_) {  // DDP path stays eager (RCCL + capture N/A)
+    step_body(nullptr);
+    return;
+  }
+
+  auto upload_bc = [&](int stepno) {
+    float bc[2] = {1.0f - powf(opt_->beta1, (float)stepno),
+                   1.0f - powf(opt_->beta2, (float)stepno)};
+    hipStream_t st = ctx_->compute_stream()->handle();
+    NZ_CHECK(hipMemcpyAsync(bc_dev_.data(), bc, 8, hipMemcpyHostToDevice, st) ==
+                 hipSuccess,
+             "bc upload failed");
+  };
+
+  if (!graph_captured_) {
+    if (opt_->step_count() == 0) {
+      // one eager step warms the arena so capture never calls hipMalloc
+      step_body(nullptr);
+      return;
+    }
+    ctx_->make_current();
+    bc_dev_ = Tensor(ctx_, DType::F32, Shape{2});
+    int next = opt_->step_count() + 1;
+    upload_bc(next);
+    ctx_->compute_stream()->sync();
+    hipStream_t st = ctx_->compute_stream()->handle();
+    NZ_CHECK(hipStreamBeginCapture(st, hipStreamCaptureModeThreadLocal) ==
+                 hipSuccess,
+             "graph capture begin failed");
+    step_body(&bc_dev_);  // increments step_count to `next`
+    hipGraph_t graph = nullptr;
+    NZ_CHECK(hipStreamEndCapture(st, &graph) == hipSuccess,
+             "graph capture end failed");
+    hipGraphExec_t exec = nullptr;
+    NZ_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0) ==
+                 hipSuccess,
+             "graph instantiate failed");
+    (void)hipGraphDestroy(graph);
+    graph_exec_ = (void*)exec;
+    graph_captured_ = true;
+    // the captured work did NOT run; launch it for this step
+    NZ_CHECK(hipGraphLaunch(exec, st) == hipSuccess, "graph launch failed");
+    NZ_LOG_INFO("hipGraph captured: training step replays as one graph");
+    return;
+  }
+  int next = opt_->step_count() + 1;
+  upload_bc(next);
+  NZ_CHECK(hipGraphLaunch((hipGraphExec_t)graph_exec_,
+                          ctx_->compute_stream()->handle()) == hipSuccess,
+           "graph launch failed");
+  opt_->set_step_count(next);
 }
 
 float Trainer::last_loss() {

@@ -50,6 +50,14 @@ class Trainer {
   // mode, off on the bench path).
   std::unique_ptr<MetricsWriter> metrics_;
   Event ev_start_, ev_fwd_, ev_bwd_, ev_opt_;
+  // hipGraph capture (NEZHA_GRAPH=1, HIP + no DDP): the launch-bound step
+  // is captured once and replayed; Adam bias corrections flow through a
+  // 2-float device buffer refreshed before each replay.
+  void step_body(const Tensor* bc_dev);
+  bool graph_mode_ = false;
+  bool graph_captured_ = false;
+  void* graph_exec_ = nullptr;  // hipGraphExec_t (opaque here)
+  Tensor bc_dev_;
 };
 
 }  // namespace nz
