@@ -104,3 +104,31 @@ def test_ddp_single_rank_equivalence(hip_ctx, tmp_path):
     ca, cb = np.load(pa), np.load(pb)
     for k in ca.files:
         np.testing.assert_array_equal(ca[k], cb[k], err_msg=k)
+
+
+def test_graph_capture_matches_eager(tmp_path):
+    """NEZHA_GRAPH=1 replays the captured step; losses must match the eager
+    path exactly (same kernels, same order, bias corrections via device
+    buffer). Run in subprocesses since the env is read at trainer ctor."""
+    import subprocess, sys, os, json
+    script = r"""
+import os, sys, json
+sys.path.insert(0, os.getcwd())
+import nezha_amd as nz
+ctx = nz.Ctx(nz.HIP, 0)
+tr = nz.Trainer.mlp(ctx, layers=2, dim=512, batch=256, lr=1e-3, seed=44)
+losses = []
+for _ in range(6):
+    tr.step()
+    losses.append(tr.loss())
+print(json.dumps(losses))
+"""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    outs = []
+    for graph in ("0", "1"):
+        env = dict(os.environ, NEZHA_GRAPH=graph)
+        r = subprocess.run([sys.executable, "-c", script], env=env, cwd=repo,
+                           capture_output=True, timeout=300)
+        assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
+        outs.append(json.loads(r.stdout.decode().strip().splitlines()[-1]))
+    assert outs[0] == outs[1], f"eager {outs[0]} vs graph {outs[1]}"
