@@ -93,8 +93,11 @@ void Trainer::step() {
   };
 
   if (!graph_captured_) {
-    if (opt_->step_count() == 0) {
-      // one eager step warms the arena so capture never calls hipMalloc
+    if (opt_->step_count() < 2) {
+      // TWO eager steps warm the arena so capture never calls hipMalloc:
+      // step 1 sizes every temporary; step 2 frees step 1's loss scalar
+      // (held alive across steps by loss_) back into the cache so the
+      // captured step's allocation of it is a cache hit.
       step_body(nullptr);
       return;
     }
