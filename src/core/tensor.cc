@@ -32,9 +32,18 @@ void* Arena::raw_alloc(size_t bytes) {
 }
 
 void* Arena::alloc(size_t bytes) {
+  static const bool dbg = [] {
+    const char* v = getenv("NEZHA_DEBUG_ALLOC");
+    return v && v[0] == '1';
+  }();
   if (bytes == 0) bytes = 16;
   bytes = (bytes + 255) & ~size_t(255);  // 256B granularity -> reuse buckets
   std::lock_guard<std::mutex> lk(mu_);
+  if (dbg) {
+    auto itd = free_lists_.find(bytes);
+    bool hit = itd != free_lists_.end() && !itd->second.empty();
+    fprintf(stderr, "[arena] alloc %zu %s\n", bytes, hit ? "hit" : "MISS");
+  }
   auto it = free_lists_.find(bytes);
   if (it != free_lists_.end() && !it->second.empty()) {
     void* p = it->second.back();

@@ -23,6 +23,9 @@ Trainer::Trainer(Context* ctx, std::unique_ptr<Model> model,
     metrics_ = std::make_unique<MetricsWriter>(mp);
   const char* g = getenv("NEZHA_GRAPH");
   graph_mode_ = g && g[0] == '1' && ctx->device() == Device::HIP;
+  // allocate the bias-correction buffer up front: taking a cached block
+  // at capture time would starve the captured step's own allocations
+  if (graph_mode_) bc_dev_ = Tensor(ctx, DType::F32, Shape{2});
 }
 
 void Trainer::reseed_data(uint64_t seed) {
@@ -102,7 +105,6 @@ void Trainer::step() {
       return;
     }
     ctx_->make_current();
-    bc_dev_ = Tensor(ctx_, DType::F32, Shape{2});
     int next = opt_->step_count() + 1;
     upload_bc(next);
     ctx_->compute_stream()->sync();
