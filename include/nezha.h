@@ -91,6 +91,10 @@ nz_status nz_op_l2norm_sq(nz_tensor* src, nz_tensor* out);
 nz_status nz_op_attn_split_qkv(nz_tensor* qkv, nz_tensor* q, nz_tensor* k,
                                nz_tensor* v, int64_t B, int64_t S, int64_t heads,
                                int64_t dh);
+nz_status nz_op_attn_fused_fwd(nz_tensor* q, nz_tensor* k, nz_tensor* v,
+                               nz_tensor* probs, nz_tensor* ctx_out,
+                               int64_t batch_heads, int64_t S, int64_t dh,
+                               float scale);
 nz_status nz_op_attn_merge(nz_tensor* ctx, nz_tensor* out, int64_t B, int64_t S,
                            int64_t heads, int64_t dh);
 

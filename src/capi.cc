@@ -261,6 +261,16 @@ extern "C" nz_status nz_op_attn_split_qkv(nz_tensor* qkv, nz_tensor* q,
   ops::attn_split_qkv(qkv->t, q->t, k->t, v->t, B, S, heads, dh);
   NZ_API_END
 }
+extern "C" nz_status nz_op_attn_fused_fwd(nz_tensor* q, nz_tensor* k,
+                                          nz_tensor* v, nz_tensor* probs,
+                                          nz_tensor* ctx_out,
+                                          int64_t batch_heads, int64_t S,
+                                          int64_t dh, float scale) {
+  NZ_API_BEGIN
+  ops::attn_fused_fwd(q->t, k->t, v->t, probs->t, ctx_out->t, batch_heads, S,
+                      dh, scale);
+  NZ_API_END
+}
 extern "C" nz_status nz_op_attn_merge(nz_tensor* ctx, nz_tensor* out, int64_t B,
                                       int64_t S, int64_t heads, int64_t dh) {
   NZ_API_BEGIN

@@ -104,6 +104,23 @@ void attn_split_qkv_bwd(const Tensor& dq, const Tensor& dk, const Tensor& dv,
                         Tensor& dqkv, int64_t B, int64_t S, int64_t heads, int64_t dh) {
   NZ_DISPATCH(attn_split_qkv_bwd, dq, dq, dk, dv, dqkv, B, S, heads, dh);
 }
+void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
+                    Tensor& probs, Tensor& ctx_out, int64_t batch_heads,
+                    int64_t S, int64_t dh, float scale) {
+  prof::OpScope nz_scope_("attn_fused_fwd", q.ctx());
+  if (q.device() == Device::HIP && hip::attn_fused_fwd_supported(S, dh)) {
+    hip::attn_fused_fwd(q, k, v, probs, ctx_out, batch_heads, scale);
+    return;
+  }
+  // composed fallback (CPU oracle + non-contract shapes): K1 batched + K4
+  Tensor scores(q.ctx(), q.dtype(), Shape{batch_heads * S, S});
+  gemm_batched(q, k, scores, false, true, 1.0f, batch_heads, S, S, dh, S * dh,
+               S * dh, S * S);
+  softmax_fwd(scores, probs, scale);
+  gemm_batched(probs, v, ctx_out, false, false, 1.0f, batch_heads, S, dh, S,
+               S * S, S * dh, S * dh);
+}
+
 void attn_merge(const Tensor& ctx, Tensor& out, int64_t B, int64_t S,
                 int64_t heads, int64_t dh) {
   NZ_DISPATCH(attn_merge, ctx, ctx, out, B, S, heads, dh);

@@ -94,6 +94,13 @@ void attn_split_qkv(const Tensor& qkv, Tensor& q, Tensor& k, Tensor& v,
 // inverse: dq,dk,dv [B,heads,S,dh] -> dqkv [B, S, 3*H]
 void attn_split_qkv_bwd(const Tensor& dq, const Tensor& dk, const Tensor& dv,
                         Tensor& dqkv, int64_t B, int64_t S, int64_t heads, int64_t dh);
+// Fused attention forward (K5): probs = softmax(scale * q.k^T) saved for
+// backward, ctx_out = probs.v. Single-kernel fusion on HIP for the BERT
+// contract shape (S=128, dh=64); composed from K1 batched + K4 otherwise.
+void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
+                    Tensor& probs, Tensor& ctx_out, int64_t batch_heads,
+                    int64_t S, int64_t dh, float scale);
+
 // ctx [B,heads,S,dh] -> out [B, S, H]
 void attn_merge(const Tensor& ctx, Tensor& out, int64_t B, int64_t S,
                 int64_t heads, int64_t dh);

@@ -49,6 +49,14 @@ void l2norm_sq(const Tensor& src, Tensor& out);
 }  // namespace hip
 }  // namespace nz
 
+// fused attention forward (src/hip/attention.hip)
+namespace nz { namespace hip {
+bool attn_fused_fwd_supported(int64_t S, int64_t dh);
+void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
+                    Tensor& probs, Tensor& ctx_out, int64_t batch_heads,
+                    float scale);
+}}
+
 // appended: colsum + attention layout ops (decls mirror core/ops.h)
 namespace nz { namespace hip {
 void colsum(const Tensor& src, Tensor& dst, float beta);

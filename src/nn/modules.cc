@@ -158,14 +158,9 @@ VarPtr MultiHeadAttention::forward(Tape& tape, VarPtr x) {
   Tensor q(ctx_, pd, Shape{B, Hh, S, D}), k(ctx_, pd, Shape{B, Hh, S, D}),
       v(ctx_, pd, Shape{B, Hh, S, D});
   ops::attn_split_qkv(qkv->value, q, k, v, B, S, Hh, D);
-  Tensor scores(ctx_, pd, Shape{B * Hh * S, S});
-  ops::gemm_batched(q, k, scores, false, true, 1.0f, B * Hh, S, S, D,
-                    S * D, S * D, S * S);
   Tensor probs(ctx_, pd, Shape{B * Hh * S, S});
-  ops::softmax_fwd(scores, probs, scale);
   Tensor ctxt(ctx_, pd, Shape{B, Hh, S, D});
-  ops::gemm_batched(probs, v, ctxt, false, false, 1.0f, B * Hh, S, D, S,
-                    S * S, S * D, S * D);
+  ops::attn_fused_fwd(q, k, v, probs, ctxt, B * Hh, S, D, scale);
   auto merged = make_var(Tensor(ctx_, pd, Shape{B * S, hidden}));
   ops::attn_merge(ctxt, merged->value, B, S, Hh, D);
 

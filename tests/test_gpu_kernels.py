@@ -280,3 +280,39 @@ def test_dropout_gpu(hip_ctx):
     yc, mc = cpu.tensor((n,)), cpu.tensor((n,))
     nz.dropout_fwd(xc, yc, mc, 0.25, seed=3)
     np.testing.assert_array_equal(m, mc.numpy())
+
+
+def test_attn_fused_fwd(hip_ctx):
+    """Fused attention kernel vs a torch reference on the contract shape."""
+    import torch
+    BH, S, D = 24, 128, 64
+    scale = 1.0 / np.sqrt(D)
+    Q, K, V = randb(BH, S, D), randb(BH, S, D), randb(BH, S, D)
+    q, k, v = (hip_ctx.from_numpy_bf16(Q), hip_ctx.from_numpy_bf16(K),
+               hip_ctx.from_numpy_bf16(V))
+    probs = hip_ctx.tensor((BH * S, S), nz.BF16)
+    ctx_out = hip_ctx.tensor((BH, S, D), nz.BF16)
+    nz.attn_fused_fwd(q, k, v, probs, ctx_out, BH, S, D, scale)
+    qt, kt, vt = map(torch.from_numpy, (Q, K, V))
+    # reference mirrors engine precision: scores rounded to bf16 pre-softmax
+    scores = torch.from_numpy(
+        nz.bf16_to_f32(nz.f32_to_bf16((qt @ kt.transpose(-1, -2)).numpy())))
+    p_ref = torch.softmax(scale * scores, dim=-1)
+    o_ref = p_ref.to(torch.float32) @ vt
+    check(probs.numpy().reshape(BH, S, S), p_ref.numpy(), rtol=0.02)
+    check(ctx_out.numpy(), o_ref.numpy(), rtol=0.03)
+
+
+def test_attn_fused_determinism(hip_ctx):
+    BH, S, D = 48, 128, 64
+    q = hip_ctx.tensor((BH, S, D), nz.BF16).rand_normal(5)
+    k = hip_ctx.tensor((BH, S, D), nz.BF16).rand_normal(6)
+    v = hip_ctx.tensor((BH, S, D), nz.BF16).rand_normal(7)
+    outs = []
+    for _ in range(2):
+        probs = hip_ctx.tensor((BH * S, S), nz.BF16)
+        o = hip_ctx.tensor((BH, S, D), nz.BF16)
+        nz.attn_fused_fwd(q, k, v, probs, o, BH, S, D, 0.125)
+        outs.append((probs.raw(), o.raw()))
+    np.testing.assert_array_equal(outs[0][0], outs[1][0])
+    np.testing.assert_array_equal(outs[0][1], outs[1][1])

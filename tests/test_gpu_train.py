@@ -132,3 +132,18 @@ print(json.dumps(losses))
         assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
         outs.append(json.loads(r.stdout.decode().strip().splitlines()[-1]))
     assert outs[0] == outs[1], f"eager {outs[0]} vs graph {outs[1]}"
+
+
+def test_bert_cpu_gpu_agreement():
+    """BERT with the fused attention path (GPU) vs the composed CPU oracle:
+    first-step losses agree to bf16 tolerance."""
+    cpu = nz.Ctx(nz.CPU)
+    gpu = nz.Ctx(nz.HIP, 0)
+    kw = dict(layers=2, hidden=256, ffn=512, heads=4, seq=128, batch=4,
+              lr=1e-3, seed=66)
+    a = nz.Trainer.bert(cpu, **kw)
+    b = nz.Trainer.bert(gpu, **kw)
+    a.step()
+    b.step()
+    la, lb = a.loss(), b.loss()
+    assert abs(la - lb) / la < 0.05, f"cpu {la} vs gpu {lb}"
