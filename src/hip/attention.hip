@@ -40,12 +40,13 @@ __launch_bounds__(256) __global__ void attn_fwd_fused_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ probs,
     uint16_t* __restrict__ out, float scale) {
-  // LDS: [Q 128*64][K 128*64][VT 64*136]; P overlays Q+K as [128][136]
-  __shared__ __attribute__((aligned(16))) uint16_t smem[AS * AD * 2 + AD * PSTRIDE];
+  // LDS: [P 128*136 | VT 64*136]; Q and K overlay P's first 32 KB (the P
+  // region is 34 KB — VT must live BEYOND it, clobbering it was a bug).
+  __shared__ __attribute__((aligned(16))) uint16_t smem[(AS + AD) * PSTRIDE];
   uint16_t* Qs = smem;
   uint16_t* Ks = smem + AS * AD;
   uint16_t* Ps = smem;                      // overlays Q..K after QK^T
-  uint16_t* VTs = smem + 2 * AS * AD;
+  uint16_t* VTs = smem + AS * PSTRIDE;
 
   const int64_t bh = blockIdx.x;            // batch*heads index
   const uint16_t* qg = q + bh * AS * AD;
