@@ -12,6 +12,8 @@
 namespace nz {
 namespace hip {
 
+using ushort8_t = __attribute__((ext_vector_type(8))) unsigned short;
+
 // ------------------------------------------------------------------ fill ---
 template <typename T>
 __global__ void fill_k(T* p, int64_t n, float v) {
@@ -264,6 +266,26 @@ __global__ void colsum_stage1_k(const T* __restrict__ src,
   for (int64_t i = r0; i < r1; ++i) acc += ld(src, i * N + j);
   partials[blockIdx.y * N + j] = acc;
 }
+
+// bf16 fast path: 8 columns per thread via ushort8 row sweeps (the scalar
+// version ran at ~0.8 TB/s on the BERT bias grads).
+__global__ void colsum_stage1_vec_k(const ushort8_t* __restrict__ src8,
+                                    float* __restrict__ partials, int64_t M,
+                                    int64_t N8, int64_t rows_per_chunk) {
+  int64_t j8 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  if (j8 >= N8) return;
+  int64_t r0 = blockIdx.y * rows_per_chunk;
+  int64_t r1 = min(M, r0 + rows_per_chunk);
+  float acc[8] = {};
+  for (int64_t i = r0; i < r1; ++i) {
+    ushort8_t v = src8[i * N8 + j8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) acc[u] += b2f(((const uint16_t*)&v)[u]);
+  }
+  float* out = partials + blockIdx.y * N8 * 8 + j8 * 8;
+#pragma unroll
+  for (int u = 0; u < 8; ++u) out[u] = acc[u];
+}
 template <typename TD>
 __global__ void colsum_stage2_k(const float* __restrict__ partials,
                                 TD* __restrict__ dst, int64_t N, int chunks,
@@ -286,6 +308,26 @@ void colsum(const Tensor& src, Tensor& dst, float beta) {
       (M + 63) / 64,
       std::min<int64_t>(64, std::max<int64_t>(1, 512 / colblocks)));
   int64_t rows_per_chunk = (M + chunks - 1) / chunks;
+  if (src.dtype() == DType::BF16 && N % 8 == 0) {
+    // vectorized: 8 cols/thread; coarser grid so re-derive the chunking
+    int64_t cb8 = (N / 8 + 255) / 256;
+    int ch8 = (int)std::min<int64_t>(
+        (M + 63) / 64,
+        std::min<int64_t>(128, std::max<int64_t>(1, 1024 / cb8)));
+    int64_t rpc8 = (M + ch8 - 1) / ch8;
+    Tensor partials8(src.ctx(), DType::F32, Shape{(int64_t)ch8, N});
+    colsum_stage1_vec_k<<<dim3(cb8, ch8), dim3(256), 0, s>>>(
+        (const ushort8_t*)src.data(), (float*)partials8.data(), M, N / 8, rpc8);
+    dim3 g2v((N + 255) / 256);
+    if (dst.dtype() == DType::BF16)
+      colsum_stage2_k<uint16_t><<<g2v, dim3(256), 0, s>>>(
+          (const float*)partials8.data(), (uint16_t*)dst.data(), N, ch8, beta);
+    else
+      colsum_stage2_k<float><<<g2v, dim3(256), 0, s>>>(
+          (const float*)partials8.data(), (float*)dst.data(), N, ch8, beta);
+    NZ_KERNEL_CHECK();
+    return;
+  }
   Tensor partials(src.ctx(), DType::F32, Shape{(int64_t)chunks, N});
   dim3 g1(colblocks, chunks), b(256);
   if (src.dtype() == DType::BF16)

@@ -29,6 +29,7 @@ struct GemmArgs {
   const uint16_t* A;
   const uint16_t* B;
   uint16_t* C;
+  float* out_f32 = nullptr;  // split-K partial mode: raw f32, no epilogue
   const uint16_t* bias;   // null or [N]
   uint16_t* save_pre;     // null or [M,N]
   int64_t M, N, K;
@@ -304,6 +305,7 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_nt_k(GemmArgs args,
     cur ^= 1;
   }
 
+  float* out32 = args.out_f32 ? args.out_f32 + batch * args.sC : nullptr;
 #pragma unroll
   for (int fm = 0; fm < 4; ++fm) {
 #pragma unroll
@@ -314,6 +316,10 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_nt_k(GemmArgs args,
         int64_t gcol = n0 + wc * 64 + fn * 16 + (lane & 15);
         if (grow >= args.M || gcol >= args.N) continue;
         int64_t idx = grow * args.N + gcol;
+        if (out32) {
+          out32[idx] = acc[fm][fn][rI];
+          continue;
+        }
         float v = args.alpha * acc[fm][fn][rI];
         if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
         if (args.bias) v += b2f(args.bias[gcol]);
@@ -442,6 +448,7 @@ __launch_bounds__(512) __global__ void gemm_bf16_nt_deep_k(GemmArgs args,
     asm volatile("" ::: "memory");
   }
 
+  float* out32 = args.out_f32;  // deep kernel: single-batch
 #pragma unroll
   for (int fm = 0; fm < 4; ++fm) {
 #pragma unroll
@@ -452,6 +459,10 @@ __launch_bounds__(512) __global__ void gemm_bf16_nt_deep_k(GemmArgs args,
         int64_t gcol = n0 + wc * 64 + fn * 16 + (lane & 15);
         if (grow >= args.M || gcol >= args.N) continue;
         int64_t idx = grow * args.N + gcol;
+        if (out32) {
+          out32[idx] = acc[fm][fn][rI];
+          continue;
+        }
         float v = args.alpha * acc[fm][fn][rI];
         if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
         if (args.bias) v += b2f(args.bias[gcol]);
@@ -920,6 +931,26 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase2_k(GemmArgs args,
   }
 }
 
+// split-K reduce: bf16 C with the full epilogue from S f32 partial slabs
+__global__ void splitk_reduce_k(const float* __restrict__ parts, GemmArgs args,
+                                int slices) {
+  int64_t total = args.M * args.N;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    float v = 0.f;
+    for (int s_ = 0; s_ < slices; ++s_) v += parts[s_ * total + i];
+    v *= args.alpha;
+    if (args.beta != 0.f) v += args.beta * b2f(args.C[i]);
+    if (args.bias) v += b2f(args.bias[i % args.N]);
+    if (args.save_pre) args.save_pre[i] = f2b(v);
+    if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+    args.C[i] = f2b(v);
+  }
+}
+
+thread_local Context* g_splitk_ctx = nullptr;
+
 void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
                  hipStream_t s) {
   NZ_CHECK(args.K % BK == 0, "hip gemm requires K % 64 == 0, got K=" +
@@ -969,7 +1000,34 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     // fast NT path: glds-staged, XCD-swizzled (see gemm_bf16_nt_k)
     int nwg_m = (int)((args.M + BM - 1) / BM);
     int nwg_n = (int)((args.N + BN - 1) / BN);
-    dim3 grid(nwg_m * nwg_n, batch);
+    int nwg = nwg_m * nwg_n;
+    // Split-K (BERT wgrads: tiny MxN against K=32k left 85% of the chip
+    // idle): run K-slices as the batch dim into f32 partial slabs, then a
+    // fused reduce applies the real epilogue.
+    if (batch == 1 && nwg < 224 && args.K >= 4096 && !args.out_f32 &&
+        g_splitk_ctx) {
+      int slices = (int)std::min<int64_t>(
+          16, std::min<int64_t>(args.K / 2048, (448 + nwg - 1) / nwg));
+      if (slices > 1 && args.K % ((int64_t)slices * 64) == 0) {
+        int64_t ksl = args.K / slices;
+        Tensor parts(g_splitk_ctx, DType::F32,
+                     Shape{(int64_t)slices, args.M, args.N});
+        GemmArgs pa = args;
+        pa.K = ksl;
+        pa.sA = ksl;  // lda stays K: slice s starts at column s*ksl
+        pa.sB = ksl;
+        pa.sC = args.M * args.N;
+        pa.out_f32 = (float*)parts.data();
+        dim3 gridk(nwg, slices);
+        gemm_bf16_nt_k<<<gridk, dim3(THREADS), 0, s>>>(pa, nwg_n);
+        NZ_KERNEL_CHECK();
+        splitk_reduce_k<<<dim3(grid_1d(args.M * args.N, 256)), dim3(256), 0,
+                          s>>>((const float*)parts.data(), args, slices);
+        NZ_KERNEL_CHECK();
+        return;
+      }
+    }
+    dim3 grid(nwg, batch);
     gemm_bf16_nt_k<<<grid, dim3(THREADS), 0, s>>>(args, nwg_n);
     NZ_KERNEL_CHECK();
     return;
@@ -1046,6 +1104,7 @@ void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
       transB = true;
     }
   }
+  g_splitk_ctx = C.ctx();
   launch_gemm(args, transA, transB, 1, stream_of(C));
 }
 
@@ -1094,6 +1153,7 @@ void gemm_batched(const Tensor& A, const Tensor& B, Tensor& C, bool transA,
       transB = true;
     }
   }
+  g_splitk_ctx = C.ctx();
   launch_gemm(args, transA, transB, batch, stream_of(C));
 }
 
