@@ -24,15 +24,20 @@ BASELINE_BERT = {1: 2.1e3, 8: 1.5e4}
 
 
 def exchange_uid(rank, world, key):
-    """File-based ncclUniqueId exchange on one node."""
-    path = f"/tmp/nezha_uid_{key}"
+    """File-based ncclUniqueId exchange on one node.
+
+    The file name is keyed by (MASTER_PORT, parent pid, world): torchrun is
+    the common parent of all ranks, so the key is shared within one launch
+    and unique across back-to-back launches — a stale file from a previous
+    run (same port) can never be picked up."""
+    path = f"/tmp/nezha_uid_{key}_{os.getppid()}_{world}"
     if rank == 0:
         uid = nz.Comm.get_uid()
         with open(path + ".tmp", "wb") as f:
             f.write(uid)
         os.rename(path + ".tmp", path)
         return uid
-    deadline = time.time() + 120
+    deadline = time.time() + 180
     while time.time() < deadline:
         if os.path.exists(path):
             uid = open(path, "rb").read()
@@ -110,6 +115,11 @@ def main():
     value = samples / elapsed
     ms_per_step = elapsed / args.steps * 1e3
 
+    if comm and rank == 0:
+        try:
+            os.remove(f"/tmp/nezha_uid_{port_key}_{os.getppid()}_{world}")
+        except OSError:
+            pass
     if rank == 0:
         out = {
             "metric": "samples/sec (whole node), 4-layer 4096-dim MLP bf16"
