@@ -931,6 +931,188 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase2_k(GemmArgs args,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Variant 4 (template-faithful phase shape): single fragment set, the
+// ds_reads of phase p issue BEFORE the phase barrier — in-wave they hide
+// under the in-flight MFMAs of phase p-1, cross-wave under the barrier
+// wait. Phase p body:
+//     reads_p ; stage S(p) ; BAR_A ; MFMA_p ; [vmcnt?] ; BAR_B
+// Slot safety: stage S(p) writes the slot last read by reads_{p-1}, which
+// every wave completed before its MFMA_{p-1} (compiler lgkm), i.e. before
+// BAR_B of p-1 — and stage S(p) issues after that barrier. The vmcnt
+// guarding reads_{p+1} sits before BAR_B of phase p:
+//   vmcnt(8) at p1/p5 (guards Q(1,1)'s B1), vmcnt(6) at p3/p7 (guards the
+//   next tile's Q(0,0) A0). Same stage schedule as variant 2.
+// ---------------------------------------------------------------------------
+__launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase4_k(GemmArgs args,
+                                                                 int nwg_n) {
+  using namespace ph8;
+  __shared__ __attribute__((aligned(16))) uint16_t smem[2 * BUF_E];
+
+  const uint16_t* __restrict__ A = args.A;
+  const uint16_t* __restrict__ B = args.B;
+  uint16_t* C = args.C;
+  uint16_t* save_pre = args.save_pre;
+
+  int nwg = gridDim.x;
+  int orig = blockIdx.x;
+  int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+  int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+  const int64_t m0 = (int64_t)(wgid / nwg_n) * PBM;
+  const int64_t n0 = (int64_t)(wgid % nwg_n) * PBN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;
+
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  const int64_t ntiles = args.K / PBK;
+  const int ea = (int)threadIdx.x * 8;
+
+  auto stage = [&](int isB, int h, int64_t t) {
+    int64_t k0 = t * PBK;
+    uint16_t* dst = slot(smem, (int)(t & 1), isB, h);
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int e = ea + c * 4096;
+      int row = e >> 6;
+      int kx = (e & 63) ^ ((row & 7) << 3);
+      if (!isB) {
+        int64_t ar = m0 + h * 128 + row;
+        if (ar > Mc) ar = Mc;
+        glds16(A + ar * lda + k0 + kx, dst + e);
+      } else {
+        int64_t br = n0 + h * 128 + row;
+        if (br > Nc) br = Nc;
+        glds16(B + br * ldb + k0 + kx, dst + e);
+      }
+    }
+  };
+
+  f32x4 acc[2][2][4][2] = {};
+  short8 fa[2][4], fb[2][2];  // [kk][frag], ONE phase's fragments
+
+  auto rd = [&](int HA, int HB, int64_t t) {
+    const uint16_t* as = slot(smem, (int)(t & 1), 0, HA);
+    const uint16_t* bs = slot(smem, (int)(t & 1), 1, HB);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int kbase = kk * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        int row = wr * 64 + fm * 16 + (lane & 15);
+        int ks = kbase ^ ((row & 7) << 3);
+        fa[kk][fm] = *(const short8*)(as + row * PBK + ks);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = wc * 32 + fn * 16 + (lane & 15);
+        int ks = kbase ^ ((col & 7) << 3);
+        fb[kk][fn] = *(const short8*)(bs + col * PBK + ks);
+      }
+    }
+  };
+  auto mm = [&](auto& accq) {
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn)
+          accq[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              fa[kk][fm], fb[kk][fn], accq[fm][fn], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+  stage(1, 0, 0);
+  stage(0, 1, 0);
+  stage(0, 0, 0);
+  stage(1, 1, 0);
+  stage(1, 0, 1);
+  stage(0, 1, 1);
+  asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+#define P4BAR()                                    do {                                               asm volatile("" ::: "memory");                   __builtin_amdgcn_s_barrier();                    asm volatile("" ::: "memory");                 } while (0)
+#define P4WAIT(n)                                             do {                                                          if (tail)                                                     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");          else                                                          asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory");   } while (0)
+#define P4PHASE(HA, HB, TT, SGUARD, SB, SH, ST, WAITC)    do {                                                      rd(HA, HB, TT);                                         if (SGUARD) stage(SB, SH, ST);                          P4BAR();                                                mm(acc[HA][HB]);                                        WAITC;                                                  P4BAR();                                              } while (0)
+
+  for (int64_t t = 0; t < ntiles; t += 2) {
+    const bool tail = (t + 4 >= ntiles);
+    const bool s2 = (t + 2 < ntiles), s3 = (t + 3 < ntiles);
+    P4PHASE(0, 0, t, true, 0, 0, t + 1, (void)0);            // p0 +A0(t+1)
+    P4PHASE(1, 0, t, true, 1, 1, t + 1, P4WAIT(8));          // p1 +B1(t+1)
+    P4PHASE(1, 1, t, s2, 1, 0, t + 2, (void)0);              // p2 +B0(t+2)
+    P4PHASE(0, 1, t, s2, 0, 1, t + 2, P4WAIT(6));            // p3 +A1(t+2)
+    P4PHASE(0, 0, t + 1, s2, 0, 0, t + 2, (void)0);          // p4 +A0(t+2)
+    P4PHASE(1, 0, t + 1, s2, 1, 1, t + 2, P4WAIT(8));        // p5 +B1(t+2)
+    P4PHASE(1, 1, t + 1, s3, 1, 0, t + 3, (void)0);          // p6 +B0(t+3)
+    P4PHASE(0, 1, t + 1, s3, 0, 1, t + 3, P4WAIT(6));        // p7 +A1(t+3)
+  }
+#undef P4PHASE
+#undef P4WAIT
+#undef P4BAR
+
+  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+#pragma unroll
+  for (int ha = 0; ha < 2; ++ha)
+#pragma unroll
+    for (int hb = 0; hb < 2; ++hb)
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+          for (int rI = 0; rI < 4; ++rI) {
+            int lrow = ha * 128 + wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+            int lcol = hb * 128 + wc * 32 + fn * 16 + (lane & 15);
+            smem[lrow * 256 + lcol] = f2b(args.alpha * acc[ha][hb][fm][fn][rI]);
+          }
+  __syncthreads();
+  const bool interior = (m0 + PBM <= args.M) && (n0 + PBN <= args.N);
+#pragma unroll 4
+  for (int c = 0; c < 16; ++c) {
+    int e = (c * 512 + (int)threadIdx.x) * 8;
+    int row = e >> 8, col = e & 255;
+    int64_t grow = m0 + row;
+    int64_t gcol = n0 + col;
+    short8 vv = *(const short8*)(smem + e);
+    if (interior) {
+      int64_t idx = grow * args.N + gcol;
+      short8 zout, yout;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        zout[i] = (short)f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        yout[i] = (short)f2b(v);
+      }
+      if (save_pre) *(short8*)(save_pre + idx) = zout;
+      *(short8*)(C + idx) = yout;
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        if (grow >= args.M || gcol + i >= args.N) continue;
+        int64_t idx = grow * args.N + gcol + i;
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        if (save_pre) save_pre[idx] = f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        C[idx] = f2b(v);
+      }
+    }
+  }
+}
+
 // split-K reduce: bf16 C with the full epilogue from S f32 partial slabs
 __global__ void splitk_reduce_k(const float* __restrict__ parts, GemmArgs args,
                                 int slices) {
@@ -963,6 +1145,17 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     const char* v = getenv("NEZHA_GEMM_MIN_WG");  // test hook: force 8-phase
     return v ? atoi(v) : 224;
   }();
+  if (!transA && transB && batch == 1 && kNtVariant == 4 &&
+      args.K % 128 == 0 && args.K / 64 >= 6) {
+    int nwg_m = (int)((args.M + 255) / 256);
+    int nwg_n = (int)((args.N + 255) / 256);
+    if (nwg_m * (int64_t)nwg_n >= kNtMinWG) {
+      dim3 grid(nwg_m * nwg_n);
+      gemm_bf16_nt_8phase4_k<<<grid, dim3(512), 0, s>>>(args, nwg_n);
+      NZ_KERNEL_CHECK();
+      return;
+    }
+  }
   if (!transA && transB && batch == 1 && kNtVariant == 3 &&
       args.K % 128 == 0 && args.K / 64 >= 6) {
     int nwg_m = (int)((args.M + 255) / 256);
