@@ -58,11 +58,7 @@ static int run_rank(const Flags& flags, int rank, int nranks, int uid_fd) {
   std::unique_ptr<Comm> comm;
   if (nranks > 1) {
     std::vector<char> uid(comm_uid_bytes());
-    if (rank == 0) {
-      comm_get_uid(uid.data());
-      // parent wrote to all children including us via pipe; we are child 0:
-    }
-    // every rank reads the uid from the inherited pipe
+    // every rank (incl. 0, which generated it) reads the uid from the pipe
     size_t got = 0;
     while (got < uid.size()) {
       ssize_t k = read(uid_fd, uid.data() + got, uid.size() - got);
@@ -111,34 +107,54 @@ int main(int argc, const char** argv) {
   int nranks = (int)flags.get_int("gpus", 1);
   if (nranks <= 1) return run_rank(flags, 0, 1, -1);
 
-  // Launcher: fork N ranks, share ncclUniqueId via pipes.
-  NZ_CHECK(hip_device_count() >= nranks,
-           "need " + std::to_string(nranks) + " GPUs, have " +
-               std::to_string(hip_device_count()));
-  std::vector<char> uid(comm_uid_bytes());
-  comm_get_uid(uid.data());
+  // Launcher: fork N ranks, share the ncclUniqueId via
+  // pipes. The PARENT must never touch HIP/RCCL (forking after runtime
+  // init is undefined), so rank 0 generates the uid in its own process
+  // and sends it up; the parent only relays bytes.
+  int uid_bytes = comm_uid_bytes();  // sizeof only, no runtime init
+  int up[2];
+  NZ_CHECK(pipe(up) == 0, "pipe failed");
   std::vector<pid_t> pids;
-  std::vector<int> write_fds;
+  std::vector<int> down_fds;
   for (int r = 0; r < nranks; ++r) {
     int fds[2];
     NZ_CHECK(pipe(fds) == 0, "pipe failed");
     pid_t pid = fork();
     if (pid == 0) {
       close(fds[1]);
+      close(up[0]);
+      if (r == 0) {
+        std::vector<char> uid(uid_bytes);
+        comm_get_uid(uid.data());  // rank 0's own process inits RCCL/HIP
+        NZ_CHECK(write(up[1], uid.data(), uid.size()) == (ssize_t)uid.size(),
+                 "uid up-pipe write failed");
+      }
+      close(up[1]);
       int rc = run_rank(flags, r, nranks, fds[0]);
       _exit(rc);
     }
     close(fds[0]);
-    (void)!write(fds[1], uid.data(), uid.size());
-    write_fds.push_back(fds[1]);
+    down_fds.push_back(fds[1]);
     pids.push_back(pid);
   }
+  close(up[1]);
+  std::vector<char> uid(uid_bytes);
+  {
+    size_t got = 0;
+    while (got < uid.size()) {
+      ssize_t k = read(up[0], uid.data() + got, uid.size() - got);
+      NZ_CHECK(k > 0, "uid up-pipe read failed");
+      got += (size_t)k;
+    }
+  }
+  close(up[0]);
+  for (int fd : down_fds) (void)!write(fd, uid.data(), uid.size());
   int fail = 0;
   for (size_t i = 0; i < pids.size(); ++i) {
     int status = 0;
     waitpid(pids[i], &status, 0);
     if (!WIFEXITED(status) || WEXITSTATUS(status) != 0) fail = 1;
-    close(write_fds[i]);
+    close(down_fds[i]);
   }
   return fail;
 }

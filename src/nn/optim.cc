@@ -62,9 +62,12 @@ Optimizer::Optimizer(Context* ctx, std::vector<Param*> params, Kind k, float lr_
   }
 }
 
-void Optimizer::step(float grad_scale, const Tensor* bc_dev) {
+void Optimizer::step(float grad_scale, const Tensor* bc_dev,
+                     const std::function<void(int)>& pre_bucket) {
   ++step_;
-  for (auto& b : buckets_) {
+  for (size_t bi = 0; bi < buckets_.size(); ++bi) {
+    auto& b = buckets_[bi];
+    if (pre_bucket) pre_bucket((int)bi);
     if (kind == Kind::ADAM) {
       ops::adam_step(b.data_flat, b.grad_flat, b.master, b.m, b.v, lr, beta1,
                      beta2, eps, step_, grad_scale, weight_decay, bc_dev);

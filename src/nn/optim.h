@@ -5,6 +5,7 @@
 // (C6: default 64 MB, chosen for the xGMI ring — SURVEY.md §2.4).
 #pragma once
 
+#include <functional>
 #include <vector>
 
 #include "src/autograd/tape.h"
@@ -29,7 +30,12 @@ class Optimizer {
   enum class Kind { ADAM, SGD };
   Optimizer(Context* ctx, std::vector<Param*> params, Kind kind, float lr,
             int64_t bucket_bytes);
-  void step(float grad_scale, const Tensor* bc_dev = nullptr);
+  // pre_bucket (optional): called right before each bucket's update kernel
+  // is enqueued — the DDP engine uses it to make the compute stream wait
+  // only THAT bucket's all-reduce, so earlier buckets' optimizer kernels
+  // overlap later buckets' comm (C14).
+  void step(float grad_scale, const Tensor* bc_dev = nullptr,
+            const std::function<void(int)>& pre_bucket = nullptr);
   std::vector<Bucket>& buckets() { return buckets_; }
   const std::vector<Param*>& params() const { return params_; }
   int step_count() const { return step_; }

@@ -57,6 +57,16 @@ void Ddp::launch_allreduce(Bucket& b) {
   }
 }
 
+void Ddp::wait_bucket(int bi) {
+  if (ctx_->device() != Device::HIP) return;
+  Bucket& b = opt_->buckets()[bi];
+  NZ_CHECK(b.pending == 0, "wait_bucket before bucket ready");
+  if (b.comm_inflight) {
+    b.comm_ev.wait(*ctx_->compute_stream());
+    b.comm_inflight = false;
+  }
+}
+
 void Ddp::finish_grads() {
   if (ctx_->device() != Device::HIP) return;
   for (auto& b : opt_->buckets()) {

@@ -20,6 +20,9 @@ class Ddp {
   void sync_params();
   void pre_step();      // reset per-bucket pending counters
   void finish_grads();  // compute stream waits for outstanding all-reduces
+  // make the compute stream wait only bucket bi's all-reduce (per-bucket
+  // optimizer chaining: earlier buckets' Adam overlaps later comm)
+  void wait_bucket(int bi);
   float grad_scale() const { return 1.0f / float(comm_->nranks()); }
   Comm* comm() const { return comm_; }
   int64_t comm_launches() const { return comm_launches_; }

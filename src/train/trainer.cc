@@ -48,9 +48,15 @@ void Trainer::step_body(const Tensor* bc_dev) {
   loss_ = mse_loss(tape_, y, target_);
   if (mx) ev_fwd_.record(*ctx_->compute_stream());
   tape_.backward();
-  if (ddp_) ddp_->finish_grads();
   if (mx) ev_bwd_.record(*ctx_->compute_stream());
-  opt_->step(ddp_ ? ddp_->grad_scale() : 1.0f, bc_dev);
+  if (ddp_ && ctx_->device() == Device::HIP) {
+    // per-bucket chaining: bucket i's Adam waits only bucket i's comm
+    opt_->step(ddp_->grad_scale(), bc_dev,
+               [this](int bi) { ddp_->wait_bucket(bi); });
+  } else {
+    if (ddp_) ddp_->finish_grads();
+    opt_->step(ddp_ ? ddp_->grad_scale() : 1.0f, bc_dev);
+  }
   if (mx) {
     ev_opt_.record(*ctx_->compute_stream());
     ev_opt_.sync();
