@@ -75,10 +75,20 @@ void* Arena::alloc(size_t bytes) {
 
 void Arena::free(void* p) {
   if (!p) return;
+  static const bool poison = [] {  // surface use-after-free
+    const char* v = getenv("NEZHA_DEBUG_POISON");
+    return v && v[0] == '1';
+  }();
   std::lock_guard<std::mutex> lk(mu_);
   auto it = sizes_.find(p);
   NZ_CHECK(it != sizes_.end(), "free of unknown pointer");
   size_t bytes = it->second;
+  if (poison) {
+    if (device_index_ < 0)
+      std::memset(p, 0xCD, bytes);  // bf16 0xCDCD / f32 0xCDCDCDCD = huge negatives
+    else
+      (void)hipMemset(p, 0xCD, bytes);
+  }
   in_use_ -= bytes;
   cached_ += bytes;
   free_lists_[bytes].push_back(p);

@@ -5,7 +5,7 @@ namespace nz {
 // ------------------------------------------------------------------- MLP ---
 MlpModel::MlpModel(Context* ctx, int64_t layers, int64_t dim, int64_t batch,
                    uint64_t seed)
-    : ctx_(ctx), dim_(dim), batch_(batch) {
+    : dim_(dim), batch_(batch) {
   for (int64_t i = 0; i < layers; ++i)
     layers_.push_back(std::make_unique<Linear>(
         ctx, "mlp.l" + std::to_string(i), dim, dim, /*bias=*/true,
@@ -24,7 +24,7 @@ void MlpModel::collect_params(std::vector<Param*>& out) {
 
 // ------------------------------------------------------------------ BERT ---
 BertEncoderModel::BertEncoderModel(Context* ctx, BertConfig cfg, uint64_t seed)
-    : ctx_(ctx), cfg_(cfg) {
+    : cfg_(cfg) {
   for (int64_t i = 0; i < cfg.layers; ++i) {
     EncoderLayer L;
     std::string nm = "bert.l" + std::to_string(i);

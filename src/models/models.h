@@ -46,7 +46,6 @@ class MlpModel : public Model {
   }
 
  private:
-  Context* ctx_;
   int64_t dim_, batch_;
   std::vector<std::unique_ptr<Linear>> layers_;
 };
@@ -73,7 +72,6 @@ class BertEncoderModel : public Model {
     std::unique_ptr<LayerNorm> ln1, ln2;
     std::unique_ptr<Linear> ffn1, ffn2;
   };
-  Context* ctx_;
   BertConfig cfg_;
   std::vector<EncoderLayer> layers_;
 };
