@@ -1167,8 +1167,10 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
       return;
     }
   }
+  // short-K shapes (BERT K=768) amortize the 256^2 pipeline poorly; the
+  // 128^2 kernel measures faster there (more workgroups, shorter ramp).
   if (!transA && transB && batch == 1 && kNtVariant >= 2 &&
-      args.K % 128 == 0 && args.K / 64 >= 6) {
+      args.K % 128 == 0 && args.K >= 1536) {
     int nwg_m = (int)((args.M + 255) / 256);
     int nwg_n = (int)((args.N + 255) / 256);
     // 256^2 tiles only when the grid fills the chip (1 block/CU, 256 CUs);
