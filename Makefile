@@ -50,5 +50,31 @@ $(BUILD)/test_%: tests/cpp/test_%.cc $(LIB)
 
 -include $(OBJS:.o=.d)
 
+# Sanitizer tier: host-side ASan+UBSan build of the
+# library + C++ tests in build-asan/ (device code is uninstrumented —
+# gfx950 has no xnack+ ASan here; the -Woption-ignored warning is expected).
+SAN_BUILD := build-asan
+SAN_FLAGS := -fsanitize=address,undefined -fno-omit-frame-pointer -Wno-option-ignored
+SAN_OBJS := $(HOST_SRCS:%.cc=$(SAN_BUILD)/%.o) $(HIP_SRCS:%.hip=$(SAN_BUILD)/%.o)
+
+$(SAN_BUILD)/%.o: %.cc
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) $(SAN_FLAGS) -c $< -o $@
+
+$(SAN_BUILD)/%.o: %.hip
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) $(SAN_FLAGS) -c $< -o $@
+
+$(SAN_BUILD)/libnezha.so: $(SAN_OBJS)
+	$(HIPCC) -shared -fPIC $(SAN_FLAGS) $(SAN_OBJS) $(LDFLAGS) -o $@
+
+$(SAN_BUILD)/test_%: tests/cpp/test_%.cc $(SAN_BUILD)/libnezha.so
+	$(HIPCC) $(CXXFLAGS) $(SAN_FLAGS) $< -o $@ -L$(SAN_BUILD) -lnezha $(LDFLAGS) -Wl,-rpath,'$$ORIGIN'
+
+.PHONY: tests-asan
+tests-asan: $(SAN_BUILD)/test_core $(SAN_BUILD)/test_ddp_fake
+	$(SAN_BUILD)/test_core
+	$(SAN_BUILD)/test_ddp_fake
+
 clean:
-	rm -rf $(BUILD)
+	rm -rf $(BUILD) $(SAN_BUILD)

@@ -42,6 +42,13 @@ static void bench_mem(Context& ctx, const char* name, int64_t n, double bytes,
 int main(int argc, const char** argv) {
   Flags flags;
   flags.parse(argc - 1, argv + 1);
+  if (flags.get_bool("help", false)) {
+    puts("nezha-bench — kernel microbenchmarks on the contract shapes\n"
+         "  --op=all|gemm|mem|norm   which suite (default all)\n"
+         "  --iters=N                timed iterations (20)\n"
+         "  --batch=N                GEMM M dimension (8192)");
+    return 0;
+  }
   int iters = (int)flags.get_int("iters", 20);
   std::string which = flags.get_str("op", "all");
   NZ_CHECK(hip_device_count() > 0, "nezha-bench requires a GPU");

@@ -101,9 +101,28 @@ static int run_rank(const Flags& flags, int rank, int nranks, int uid_fd) {
   return 0;
 }
 
+static const char* kHelp =
+    "nezha-train — synthetic-data training CLI (MI355X-native engine)\n"
+    "  --model=mlp|bert      model family (default mlp)\n"
+    "  --device=hip|cpu      backend (default: hip when a GPU is present)\n"
+    "  --gpus=N              fork N DP ranks over RCCL/xGMI (default 1)\n"
+    "  --layers --dim        model size (mlp: 4x4096; bert: 12x768)\n"
+    "  --ffn --heads --seq   bert geometry (3072 / 12 / 128)\n"
+    "  --batch=N             per-rank batch (mlp 8192, bert 256)\n"
+    "  --optim=adam|sgd --lr=F --bucket-mb=N (64)\n"
+    "  --steps=N --warmup=N  timed window (50 / 10)\n"
+    "  --save=path --load=path  .npz checkpoint\n"
+    "  --metrics=path        per-run JSONL summary\n"
+    "  --seed=N              init seed (1234)\n"
+    "  env: NEZHA_* (see docs/api.md)\n";
+
 int main(int argc, const char** argv) {
   Flags flags;
   flags.parse(argc - 1, argv + 1);
+  if (flags.get_bool("help", false)) {
+    fputs(kHelp, stdout);
+    return 0;
+  }
   int nranks = (int)flags.get_int("gpus", 1);
   if (nranks <= 1) return run_rank(flags, 0, 1, -1);
 

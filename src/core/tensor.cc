@@ -16,7 +16,19 @@ namespace nz {
 Arena::Arena(int device_index) : device_index_(device_index) {}
 
 Arena::~Arena() {
-  // Process-lifetime arena; leave final cleanup to the HIP runtime teardown.
+  // Free the cached (unreferenced) blocks; live blocks are owned by
+  // still-alive tensors, whose Buf frees into a dead arena otherwise —
+  // contexts must outlive their tensors.
+  std::lock_guard<std::mutex> lk(mu_);
+  for (auto& [sz, lst] : free_lists_) {
+    for (void* q : lst) {
+      if (device_index_ < 0) ::free(q);
+      else (void)hipFree(q);
+      sizes_.erase(q);
+    }
+    lst.clear();
+  }
+  cached_ = 0;
 }
 
 void* Arena::raw_alloc(size_t bytes) {

@@ -17,6 +17,7 @@ Trainer::Trainer(Context* ctx, std::unique_ptr<Model> model,
   DType dt = ctx->device() == Device::HIP ? DType::BF16 : DType::F32;
   input_ = Tensor(ctx, dt, model_->input_shape());
   target_ = Tensor(ctx, dt, model_->output_shape());
+  data_seed_ = data_seed;
   ops::rand_normal(input_, data_seed, 0.f, 1.f);
   ops::rand_normal(target_, data_seed + 1, 0.f, 1.f);
   if (const char* mp = getenv("NEZHA_METRICS"))
@@ -29,6 +30,7 @@ Trainer::Trainer(Context* ctx, std::unique_ptr<Model> model,
 }
 
 void Trainer::reseed_data(uint64_t seed) {
+  data_seed_ = seed;
   ops::rand_normal(input_, seed, 0.f, 1.f);
   ops::rand_normal(target_, seed + 1, 0.f, 1.f);
 }
@@ -201,6 +203,16 @@ void Trainer::save_checkpoint(const std::string& path) {
   stepa.data.resize(4);
   std::memcpy(stepa.data.data(), &st, 4);
   arrays["meta/step"] = stepa;
+  // RNG state (SURVEY.md §5.4 meta/rng): the generator is counter-based,
+  // so the data seed IS the full state.
+  NpyArray rnga;
+  rnga.dtype = DType::F32;
+  rnga.shape = {2};
+  float sd[2] = {(float)(data_seed_ & 0xffffffu),
+                 (float)((data_seed_ >> 24) & 0xffffffu)};
+  rnga.data.resize(8);
+  std::memcpy(rnga.data.data(), sd, 8);
+  arrays["meta/data_seed"] = rnga;
   npz_save(path, arrays);
   if (ddp_) ddp_->comm()->barrier();
 }
@@ -247,6 +259,13 @@ void Trainer::load_checkpoint(const std::string& path) {
       float st;
       std::memcpy(&st, it->second.data.data(), 4);
       opt_->set_step_count((int)st);
+    }
+    auto itr = arrays.find("meta/data_seed");
+    if (itr != arrays.end() && itr->second.data.size() == 8) {
+      float sd[2];
+      std::memcpy(sd, itr->second.data.data(), 8);
+      uint64_t seed = (uint64_t)sd[0] | ((uint64_t)sd[1] << 24);
+      reseed_data(seed);
     }
   }
   if (ddp_) {
