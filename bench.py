@@ -60,6 +60,12 @@ def main():
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    if world > 1 and "RANK" not in os.environ:
+        raise SystemExit(
+            "--gpus > 1 needs one process per GPU: launch via
+"
+            "  python -m torch.distributed.run --nnodes=1 --nproc-per-node "
+            f"{world} --master-addr 127.0.0.1 bench.py --gpus {world} ...")
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     port_key = os.environ.get("MASTER_PORT", "0")
 
