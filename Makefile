@@ -71,6 +71,14 @@ $(SAN_BUILD)/libnezha.so: $(SAN_OBJS)
 $(SAN_BUILD)/test_%: tests/cpp/test_%.cc $(SAN_BUILD)/libnezha.so
 	$(HIPCC) $(CXXFLAGS) $(SAN_FLAGS) $< -o $@ -L$(SAN_BUILD) -lnezha $(LDFLAGS) -Wl,-rpath,'$$ORIGIN'
 
+# perf yardstick (vendor lib, never linked into the product lib — C7)
+$(BUILD)/yardstick: tools/yardstick.cc
+	@mkdir -p $(BUILD)
+	$(HIPCC) -O2 -std=c++17 -x hip --offload-arch=$(ARCH) $< -o $@ -L/opt/rocm/lib -lhipblaslt
+
+.PHONY: yardstick
+yardstick: $(BUILD)/yardstick
+
 .PHONY: tests-asan
 tests-asan: $(SAN_BUILD)/test_core $(SAN_BUILD)/test_ddp_fake
 	$(SAN_BUILD)/test_core

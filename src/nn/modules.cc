@@ -191,6 +191,48 @@ VarPtr MultiHeadAttention::forward(Tape& tape, VarPtr x) {
   return out_proj.forward(tape, merged);
 }
 
+// ------------------------------------------------- standalone activations ---
+VarPtr Gelu::forward(Tape& tape, VarPtr x) {
+  auto y = make_var(Tensor(ctx_, x->value.dtype(), x->value.shape()));
+  Tensor z = x->value;  // pre-activation IS the input (no bias)
+  ops::bias_gelu_fwd(x->value, nullptr, y->value, nullptr);
+  tape.push([this, x, y, z]() {
+    NZ_CHECK(y->grad.defined(), "gelu backward without upstream grad");
+    if (!x->needs_grad) return;
+    x->ensure_grad();
+    if (!x->grad_init) {
+      ops::bias_gelu_bwd(y->grad, z, x->grad, nullptr);
+      x->grad_init = true;
+    } else {
+      Tensor dz(ctx_, z.dtype(), z.shape());
+      ops::bias_gelu_bwd(y->grad, z, dz, nullptr);
+      ops::add_inplace(x->grad, dz, 1.0f);
+    }
+  });
+  return y;
+}
+
+VarPtr Softmax::forward(Tape& tape, VarPtr x) {
+  auto y = make_var(Tensor(ctx_, x->value.dtype(), x->value.shape()));
+  ops::softmax_fwd(x->value, y->value, scale_);
+  float sc = scale_;
+  Context* ctx = ctx_;
+  tape.push([ctx, sc, x, y]() {
+    NZ_CHECK(y->grad.defined(), "softmax backward without upstream grad");
+    if (!x->needs_grad) return;
+    x->ensure_grad();
+    if (!x->grad_init) {
+      ops::softmax_bwd(y->grad, y->value, x->grad, sc);
+      x->grad_init = true;
+    } else {
+      Tensor dx(ctx, x->value.dtype(), x->value.shape());
+      ops::softmax_bwd(y->grad, y->value, dx, sc);
+      ops::add_inplace(x->grad, dx, 1.0f);
+    }
+  });
+  return y;
+}
+
 // ---------------------------------------------------------- residual add ---
 VarPtr residual_add(Tape& tape, VarPtr a, VarPtr b) {
   Context* ctx = a->value.ctx();
@@ -206,6 +248,17 @@ VarPtr residual_add(Tape& tape, VarPtr a, VarPtr b) {
 }
 
 // ------------------------------------------------------------------ loss ---
+Tensor ce_loss(Tape& tape, VarPtr logits, const Tensor& labels) {
+  Context* ctx = logits->value.ctx();
+  Tensor loss(ctx, DType::F32, Shape{1});
+  NZ_CHECK(!logits->grad_init, "loss applied to var with existing grad");
+  logits->ensure_grad();
+  logits->grad_init = true;
+  ops::ce_loss(logits->value, labels, loss, &logits->grad, 1.0f);
+  (void)tape;
+  return loss;
+}
+
 Tensor mse_loss(Tape& tape, VarPtr y, const Tensor& target) {
   Context* ctx = y->value.ctx();
   Tensor loss(ctx, DType::F32, Shape{1});

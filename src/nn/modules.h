@@ -62,10 +62,36 @@ class MultiHeadAttention : public Module {
   Context* ctx_;
 };
 
+// Standalone activation/softmax modules (the hot path
+// fuses GELU into the producing GEMM — these exist for model-building
+// completeness and the unfused clarity path).
+class Gelu : public Module {
+ public:
+  explicit Gelu(Context* ctx) : ctx_(ctx) {}
+  VarPtr forward(Tape& tape, VarPtr x) override;
+  void collect_params(std::vector<Param*>&) override {}
+
+ private:
+  Context* ctx_;
+};
+
+class Softmax : public Module {
+ public:
+  Softmax(Context* ctx, float scale = 1.0f) : ctx_(ctx), scale_(scale) {}
+  VarPtr forward(Tape& tape, VarPtr x) override;
+  void collect_params(std::vector<Param*>&) override {}
+
+ private:
+  Context* ctx_;
+  float scale_;
+};
+
 // Residual add: out = a + b (both tracked).
 VarPtr residual_add(Tape& tape, VarPtr a, VarPtr b);
 
 // Losses: return a device f32 scalar tensor; seed dy into y's grad.
 Tensor mse_loss(Tape& tape, VarPtr y, const Tensor& target);
+// Cross-entropy over the last dim; labels hold integral class ids (f32).
+Tensor ce_loss(Tape& tape, VarPtr logits, const Tensor& labels);
 
 }  // namespace nz

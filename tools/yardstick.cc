@@ -1,0 +1,113 @@
+// hipBLASLt GEMM yardstick: measures vendor-library
+// bf16 GEMM on the MLP contract shapes so docs/perf can record the gap to
+// our hand-written kernel. PERF YARDSTICK ONLY — never on the product path
+// (C7); this binary is not linked into libnezha.
+#include <hip/hip_runtime.h>
+#include <hipblaslt/hipblaslt.h>
+
+#include <chrono>
+#include <cstdio>
+#include <cstdlib>
+#include <vector>
+
+#define CK(x)                                                      \
+  do {                                                             \
+    auto e = (x);                                                  \
+    if (e != 0) {                                                  \
+      fprintf(stderr, "error %d at %s:%d\n", (int)e, __FILE__, __LINE__); \
+      exit(1);                                                     \
+    }                                                              \
+  } while (0)
+
+static double now_sec() {
+  using namespace std::chrono;
+  return duration<double>(steady_clock::now().time_since_epoch()).count();
+}
+
+// col-major D[m,n] = A[m,k] x B[k,n], bf16 in, f32 compute, bf16 out
+static void bench_shape(hipblasLtHandle_t h, int64_t m, int64_t n, int64_t k,
+                        const char* tag, int iters) {
+  void *dA, *dB, *dD, *dWs;
+  size_t ws_size = 64u << 20;
+  CK(hipMalloc(&dA, m * k * 2));
+  CK(hipMalloc(&dB, k * n * 2));
+  CK(hipMalloc(&dD, m * n * 2));
+  CK(hipMalloc(&dWs, ws_size));
+  // random-ish fill (bf16 bit patterns; magnitude irrelevant for timing,
+  // avoid zero-fill DVFS inflation — guide §5.4 rule 25)
+  CK(hipMemset(dA, 0x3f, m * k * 2));
+  CK(hipMemset(dB, 0x3e, k * n * 2));
+
+  hipblasLtMatmulDesc_t op;
+  CK(hipblasLtMatmulDescCreate(&op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
+  hipblasOperation_t tn = HIPBLAS_OP_N;
+  CK(hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_TRANSA, &tn,
+                                     sizeof(tn)));
+  CK(hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_TRANSB, &tn,
+                                     sizeof(tn)));
+  hipblasLtMatrixLayout_t la, lb, ld;
+  CK(hipblasLtMatrixLayoutCreate(&la, HIP_R_16BF, m, k, m));
+  CK(hipblasLtMatrixLayoutCreate(&lb, HIP_R_16BF, k, n, k));
+  CK(hipblasLtMatrixLayoutCreate(&ld, HIP_R_16BF, m, n, m));
+
+  hipblasLtMatmulPreference_t pref;
+  CK(hipblasLtMatmulPreferenceCreate(&pref));
+  CK(hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws_size,
+      sizeof(ws_size)));
+  hipblasLtMatmulHeuristicResult_t heur[8];
+  int nheur = 0;
+  CK(hipblasLtMatmulAlgoGetHeuristic(h, op, la, lb, ld, ld, pref, 8, heur,
+                                     &nheur));
+  if (nheur == 0) {
+    printf("yardstick %-18s M=%-6lld N=%-6lld K=%-6lld  NO ALGO\n", tag,
+           (long long)m, (long long)n, (long long)k);
+    return;
+  }
+  float alpha = 1.f, beta = 0.f;
+  auto run = [&](const hipblasLtMatmulAlgo_t* algo) {
+    CK(hipblasLtMatmul(h, op, &alpha, dA, la, dB, lb, &beta, dD, ld, dD, ld,
+                       algo, dWs, ws_size, 0));
+  };
+  // pick the fastest of the returned heuristics (best vendor foot forward)
+  double best = 1e30;
+  int best_i = 0;
+  for (int i = 0; i < nheur; ++i) {
+    run(&heur[i].algo);
+    CK(hipDeviceSynchronize());
+    double t0 = now_sec();
+    for (int r = 0; r < 5; ++r) run(&heur[i].algo);
+    CK(hipDeviceSynchronize());
+    double t = (now_sec() - t0) / 5;
+    if (t < best) { best = t; best_i = i; }
+  }
+  run(&heur[best_i].algo);
+  CK(hipDeviceSynchronize());
+  double t0 = now_sec();
+  for (int r = 0; r < iters; ++r) run(&heur[best_i].algo);
+  CK(hipDeviceSynchronize());
+  double sec = (now_sec() - t0) / iters;
+  printf("yardstick %-18s M=%-6lld N=%-6lld K=%-6lld  %8.3f ms  %7.1f TF/s\n",
+         tag, (long long)m, (long long)n, (long long)k, sec * 1e3,
+         2.0 * m * n * k / sec / 1e12);
+  hipblasLtMatmulPreferenceDestroy(pref);
+  hipblasLtMatrixLayoutDestroy(la);
+  hipblasLtMatrixLayoutDestroy(lb);
+  hipblasLtMatrixLayoutDestroy(ld);
+  hipblasLtMatmulDescDestroy(op);
+  hipFree(dA); hipFree(dB); hipFree(dD); hipFree(dWs);
+}
+
+int main() {
+  hipblasLtHandle_t h;
+  CK(hipblasLtCreate(&h));
+  // the three MLP contract shapes + squares (shape triples; layout
+  // col-major here vs our row-major — FLOP count and regime identical)
+  bench_shape(h, 8192, 4096, 4096, "mlp-fwd", 10);
+  bench_shape(h, 8192, 4096, 4096, "mlp-dgrad", 10);
+  bench_shape(h, 4096, 4096, 8192, "mlp-wgrad", 10);
+  bench_shape(h, 4096, 4096, 4096, "square-4k", 10);
+  bench_shape(h, 8192, 8192, 8192, "square-8k", 5);
+  hipblasLtDestroy(h);
+  return 0;
+}
