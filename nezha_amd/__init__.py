@@ -44,6 +44,7 @@ _lib.nz_ctx_new.restype = ctypes.c_void_p
 _lib.nz_ctx_new.argtypes = [ctypes.c_int, ctypes.c_int]
 _lib.nz_ctx_free.argtypes = [ctypes.c_void_p]
 _lib.nz_ctx_sync.argtypes = [ctypes.c_void_p]
+_lib.nz_prof_flush.argtypes = [ctypes.c_void_p]
 _lib.nz_ctx_mem_high_water.restype = ctypes.c_int64
 _lib.nz_ctx_mem_high_water.argtypes = [ctypes.c_void_p]
 _lib.nz_tensor_new.restype = ctypes.c_void_p
@@ -168,6 +169,10 @@ class Ctx:
 
     def mem_high_water(self):
         return _lib.nz_ctx_mem_high_water(self._h)
+
+    def prof_flush(self):
+        """write the NEZHA_TRACE chrome-trace file (no-op when disabled)"""
+        _ck(_lib.nz_prof_flush(self._h))
 
     def __del__(self):
         if getattr(self, "_h", None) and _lib:

@@ -8,6 +8,7 @@
 #include <chrono>
 #include <cstdio>
 #include <cstdlib>
+#include <cstring>
 #include <vector>
 
 #define CK(x)                                                      \
@@ -33,10 +34,21 @@ static void bench_shape(hipblasLtHandle_t h, int64_t m, int64_t n, int64_t k,
   CK(hipMalloc(&dB, k * n * 2));
   CK(hipMalloc(&dD, m * n * 2));
   CK(hipMalloc(&dWs, ws_size));
-  // random-ish fill (bf16 bit patterns; magnitude irrelevant for timing,
-  // avoid zero-fill DVFS inflation — guide §5.4 rule 25)
-  CK(hipMemset(dA, 0x3f, m * k * 2));
-  CK(hipMemset(dB, 0x3e, k * n * 2));
+  // genuine random fill (guide §5.4 rule 25: constant/zero data inflates
+  // TF via DVFS — quote the random-data number)
+  {
+    std::vector<unsigned short> host(std::max(m * k, k * n));
+    unsigned long long st = 0x12345678ULL;
+    for (auto& v : host) {
+      st = st * 6364136223846793005ULL + 1442695040888963407ULL;
+      float f = float((st >> 40) & 0xFFFFFF) / 8388608.0f - 1.0f;  // [-1,1)
+      unsigned u;
+      memcpy(&u, &f, 4);
+      v = (unsigned short)(u >> 16);
+    }
+    CK(hipMemcpy(dA, host.data(), m * k * 2, hipMemcpyHostToDevice));
+    CK(hipMemcpy(dB, host.data(), k * n * 2, hipMemcpyHostToDevice));
+  }
 
   hipblasLtMatmulDesc_t op;
   CK(hipblasLtMatmulDescCreate(&op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
