@@ -1113,6 +1113,195 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase4_k(GemmArgs args,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Variant 5: the variant-2 phase structure on v_mfma_f32_32x32x16_bf16.
+// Same half-tile DMA schedule and waits; each phase's quadrant is computed
+// as 2x1 fragments of 32x32 over four K=16 steps — half the MFMA
+// instruction slots at the higher 32x32 pipe ceiling (2382 vs 2075 TF
+// µbench), with identical ds_read count. C/D fragment map (guide §3):
+// col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5), reg in [0,16).
+// ---------------------------------------------------------------------------
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+__launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase32_k(
+    GemmArgs args, int nwg_n) {
+  using namespace ph8;
+  __shared__ __attribute__((aligned(16))) uint16_t smem[2 * BUF_E];
+
+  const uint16_t* __restrict__ A = args.A;
+  const uint16_t* __restrict__ B = args.B;
+  uint16_t* C = args.C;
+  uint16_t* save_pre = args.save_pre;
+
+  int nwg = gridDim.x;
+  int orig = blockIdx.x;
+  int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+  int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+  const int64_t m0 = (int64_t)(wgid / nwg_n) * PBM;
+  const int64_t n0 = (int64_t)(wgid % nwg_n) * PBN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;  // 2x4 wave grid, 64x32 per quadrant
+
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  const int64_t ntiles = args.K / PBK;
+  const int ea = (int)threadIdx.x * 8;
+
+  auto stage = [&](int isB, int h, int64_t t) {
+    int64_t k0 = t * PBK;
+    uint16_t* dst = slot(smem, (int)(t & 1), isB, h);
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int e = ea + c * 4096;
+      int row = e >> 6;
+      int kx = (e & 63) ^ ((row & 7) << 3);
+      if (!isB) {
+        int64_t ar = m0 + h * 128 + row;
+        if (ar > Mc) ar = Mc;
+        glds16(A + ar * lda + k0 + kx, dst + e);
+      } else {
+        int64_t br = n0 + h * 128 + row;
+        if (br > Nc) br = Nc;
+        glds16(B + br * ldb + k0 + kx, dst + e);
+      }
+    }
+  };
+
+  // acc[ha][hb]: one 32-col fragment pair (2 fm of 32 rows) per quadrant
+  f32x16 acc[2][2][2] = {};
+
+  auto compute = [&](int HA, int HB, int64_t t, auto& accq) {
+    const uint16_t* as = slot(smem, (int)(t & 1), 0, HA);
+    const uint16_t* bs = slot(smem, (int)(t & 1), 1, HB);
+#pragma unroll
+    for (int kk = 0; kk < PBK; kk += 16) {
+      short8 a_frag[2], b_frag;
+      const int kbase = kk + (lane >> 5) * 8;  // 32x32x16: k = (lane>>5)*8+j
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        int row = wr * 64 + fm * 32 + (lane & 31);
+        int ks = kbase ^ ((row & 7) << 3);
+        a_frag[fm] = *(const short8*)(as + row * PBK + ks);
+      }
+      {
+        int col = wc * 32 + (lane & 31);
+        int ks = kbase ^ ((col & 7) << 3);
+        b_frag = *(const short8*)(bs + col * PBK + ks);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+        accq[fm] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_frag[fm], b_frag,
+                                                           accq[fm], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  };
+
+  stage(1, 0, 0);
+  stage(0, 1, 0);
+  stage(0, 0, 0);
+  stage(1, 1, 0);
+  stage(1, 0, 1);
+  stage(0, 1, 1);
+
+#define PH32_WAIT(n)                                           do {                                                           if (tail)                                                      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");           else                                                           asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory");      __builtin_amdgcn_s_barrier();                                asm volatile("" ::: "memory");                             } while (0)
+#define PH32_BAR()                                             do {                                                           asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");           __builtin_amdgcn_s_barrier();                                asm volatile("" ::: "memory");                             } while (0)
+
+  for (int64_t t = 0; t < ntiles; t += 2) {
+    const bool tail = (t + 4 >= ntiles);
+    const bool s2 = (t + 2 < ntiles), s3 = (t + 3 < ntiles);
+    PH32_WAIT(6);
+    stage(0, 0, t + 1);
+    compute(0, 0, t, acc[0][0]);
+    PH32_BAR();
+    stage(1, 1, t + 1);
+    compute(1, 0, t, acc[1][0]);
+    PH32_BAR();
+    PH32_WAIT(8);
+    if (s2) stage(1, 0, t + 2);
+    compute(1, 1, t, acc[1][1]);
+    PH32_BAR();
+    if (s2) stage(0, 1, t + 2);
+    compute(0, 1, t, acc[0][1]);
+    PH32_BAR();
+    PH32_WAIT(6);
+    if (s2) stage(0, 0, t + 2);
+    compute(0, 0, t + 1, acc[0][0]);
+    PH32_BAR();
+    if (s2) stage(1, 1, t + 2);
+    compute(1, 0, t + 1, acc[1][0]);
+    PH32_BAR();
+    PH32_WAIT(8);
+    if (s3) stage(1, 0, t + 3);
+    compute(1, 1, t + 1, acc[1][1]);
+    PH32_BAR();
+    if (s3) stage(0, 1, t + 3);
+    compute(0, 1, t + 1, acc[0][1]);
+    PH32_BAR();
+  }
+#undef PH32_WAIT
+#undef PH32_BAR
+
+  // epilogue via LDS staging; 32x32 C/D map: col=lane&31,
+  // row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+#pragma unroll
+  for (int ha = 0; ha < 2; ++ha)
+#pragma unroll
+    for (int hb = 0; hb < 2; ++hb)
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+          int lrow = ha * 128 + wr * 64 + fm * 32 + (reg & 3) +
+                     8 * (reg >> 2) + 4 * (lane >> 5);
+          int lcol = hb * 128 + wc * 32 + (lane & 31);
+          smem[lrow * 256 + lcol] = f2b(args.alpha * acc[ha][hb][fm][reg]);
+        }
+  __syncthreads();
+  const bool interior = (m0 + PBM <= args.M) && (n0 + PBN <= args.N);
+#pragma unroll 4
+  for (int c = 0; c < 16; ++c) {
+    int e = (c * 512 + (int)threadIdx.x) * 8;
+    int row = e >> 8, col = e & 255;
+    int64_t grow = m0 + row;
+    int64_t gcol = n0 + col;
+    short8 vv = *(const short8*)(smem + e);
+    if (interior) {
+      int64_t idx = grow * args.N + gcol;
+      short8 zout, yout;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        zout[i] = (short)f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        yout[i] = (short)f2b(v);
+      }
+      if (save_pre) *(short8*)(save_pre + idx) = zout;
+      *(short8*)(C + idx) = yout;
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        if (grow >= args.M || gcol + i >= args.N) continue;
+        int64_t idx = grow * args.N + gcol + i;
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        if (save_pre) save_pre[idx] = f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        C[idx] = f2b(v);
+      }
+    }
+  }
+}
+
 // split-K reduce: bf16 C with the full epilogue from S f32 partial slabs
 __global__ void splitk_reduce_k(const float* __restrict__ parts, GemmArgs args,
                                 int slices) {
@@ -1145,6 +1334,17 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     const char* v = getenv("NEZHA_GEMM_MIN_WG");  // test hook: force 8-phase
     return v ? atoi(v) : 224;
   }();
+  if (!transA && transB && batch == 1 && kNtVariant == 5 &&
+      args.K % 128 == 0 && args.K >= 1536) {
+    int nwg_m = (int)((args.M + 255) / 256);
+    int nwg_n = (int)((args.N + 255) / 256);
+    if (nwg_m * (int64_t)nwg_n >= kNtMinWG) {
+      dim3 grid(nwg_m * nwg_n);
+      gemm_bf16_nt_8phase32_k<<<grid, dim3(512), 0, s>>>(args, nwg_n);
+      NZ_KERNEL_CHECK();
+      return;
+    }
+  }
   if (!transA && transB && batch == 1 && kNtVariant == 4 &&
       args.K % 128 == 0 && args.K / 64 >= 6) {
     int nwg_m = (int)((args.M + 255) / 256);
