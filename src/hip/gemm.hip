@@ -1,15 +1,16 @@
 // K1: hand-written CDNA4 bf16 GEMM (MFMA v_mfma_f32_16x16x32_bf16, f32
 // accumulate) with fused bias(+GELU) epilogue and strided-batch mode.
 //
-// Tier A structure ("correct-first", guide §5 anatomy): 128x128 macro-tile,
-// BK=64, 256 threads = 4 waves in a 2x2 grid, each wave owns a 64x64
-// sub-tile = 4x4 fragments of 16x16. A staged in LDS as [BM][BK], B staged
-// TRANSPOSED as [BN][BK] so both fragment reads are contiguous 16-byte
-// ds_read_b128 along k. Rows padded +8 bf16 (16 B) to break bank conflicts
-// while keeping 16-B alignment (guide §6 Guideline 4/17). Double-buffered
-// LDS with global loads issued before compute (T14 issue-early/write-late).
-// Tier B (glds + XCD swizzle, guide §5 ladder step 3) comes after this is
-// proven correct on hardware.
+// Kernel ladder in this file (docs/perf has the measured TF/s per rung):
+//   gemm_bf16_k           generic trans-variant tiled kernel (TT + ragged)
+//   gemm_bf16_nt_k        2-buffer glds NT, 128^2 tile — batched/small/
+//                         split-K slices
+//   gemm_bf16_nt_deep_k   3-buffer counted-vmcnt experiment (variant 1)
+//   gemm_bf16_nt_8phase_k 256^2 half-tile 8-phase — THE default (variant 2)
+//   gemm_bf16_nt_8phase2/4/32_k  measured alternatives (variants 3/4/5)
+// NN and TN lower onto NT via the vectorized transpose (design rationale:
+// docs/design.md "why NT is the canonical form"); tiny-output large-K
+// shapes split K into f32 slabs + a fused reduce.
 #include "src/hip/hip_common.h"
 #include "src/hip/hip_ops.h"
 

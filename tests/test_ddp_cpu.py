@@ -86,6 +86,15 @@ def test_fake_comm_ddp_cpp():
     assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
 
 
+def test_modules_cpp():
+    binp = os.path.join(REPO, "build", "test_modules")
+    if not os.path.exists(binp):
+        subprocess.run(["make", "tests", "-j4"], cwd=REPO, check=True,
+                       capture_output=True)
+    r = subprocess.run([binp], capture_output=True, timeout=300)
+    assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
+
+
 def test_core_cpp():
     binp = os.path.join(REPO, "build", "test_core")
     if not os.path.exists(binp):
