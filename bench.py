@@ -62,8 +62,7 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
     if world > 1 and "RANK" not in os.environ:
         raise SystemExit(
-            "--gpus > 1 needs one process per GPU: launch via
-"
+            "--gpus > 1 needs one process per GPU: launch via\n"
             "  python -m torch.distributed.run --nnodes=1 --nproc-per-node "
             f"{world} --master-addr 127.0.0.1 bench.py --gpus {world} ...")
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
