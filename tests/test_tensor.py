@@ -46,3 +46,11 @@ def test_rand_uniform_range(cpu_ctx):
     a = cpu_ctx.tensor((5000,), nz.F32).rand_uniform(7, -2.0, 3.0).numpy()
     assert a.min() >= -2.0 and a.max() < 3.0
     assert abs(a.mean() - 0.5) < 0.2
+
+
+def test_bench_script_parses():
+    """bench.py is the driver contract — it must always at least parse."""
+    import ast, os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for f in ("bench.py", "__graft_entry__.py"):
+        ast.parse(open(os.path.join(repo, f)).read())
