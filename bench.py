@@ -73,7 +73,9 @@ def main():
     ctx = nz.Ctx(nz.HIP, local_rank % navail)
 
     if args.model == "mlp":
-        batch = args.batch or 8192
+        # 16384/GPU measured ~7% faster than 8192 (larger GEMM M amortizes
+        # fixed costs) and halves the comm:compute ratio for DP weak scaling
+        batch = args.batch or 16384
         tr = nz.Trainer.mlp(ctx, layers=4, dim=4096, batch=batch, optim="adam",
                             lr=1e-4, bucket_bytes=args.bucket_mb << 20,
                             seed=1234)
