@@ -167,7 +167,9 @@ int main(int argc, const char** argv) {
     }
   }
   close(up[0]);
-  for (int fd : down_fds) (void)!write(fd, uid.data(), uid.size());
+  for (int fd : down_fds)
+    NZ_CHECK(write(fd, uid.data(), uid.size()) == (ssize_t)uid.size(),
+             "uid down-pipe write failed");
   int fail = 0;
   for (size_t i = 0; i < pids.size(); ++i) {
     int status = 0;

@@ -30,7 +30,6 @@ void gemm_batched(const Tensor& A, const Tensor& B, Tensor& C, bool transA,
 void bias_gelu_bwd(const Tensor& dy, const Tensor& z, Tensor& dz, Tensor* dbias);
 // Standalone forward variant (unfused clarity path): y = gelu(x + bias).
 void bias_gelu_fwd(const Tensor& x, const Tensor* bias, Tensor& y, Tensor* save_pre);
-// GELU grad w.r.t. input only: dx = dy * gelu'(z)  (dz may alias dy? no: distinct)
 
 // ---- K3: LayerNorm over the last dim.
 void layernorm_fwd(const Tensor& x, const Tensor& gamma, const Tensor& beta,

@@ -81,6 +81,17 @@ class Event {
  public:
   Event();
   ~Event();
+  Event(const Event&) = delete;             // owns a hipEvent_t
+  Event& operator=(const Event&) = delete;
+  Event(Event&& o) noexcept : e_(o.e_) { o.e_ = nullptr; }
+  Event& operator=(Event&& o) noexcept {
+    if (this != &o) {
+      this->~Event();
+      e_ = o.e_;
+      o.e_ = nullptr;
+    }
+    return *this;
+  }
   void record(Stream& s);
   void wait(Stream& s) const;  // makes s wait on this event (device-side)
   void sync() const;
