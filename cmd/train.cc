@@ -54,6 +54,8 @@ static int run_rank(const Flags& flags, int rank, int nranks, int uid_fd) {
   int64_t samples = m->samples_per_batch();
   double flops = m->flops_per_step();
   Trainer trainer(ctx.get(), std::move(m), kind, lr, bucket, seed + 12345);
+  trainer.optimizer()->weight_decay = (float)flags.get_float("wd", 0.0);
+  trainer.optimizer()->sgd_momentum = (float)flags.get_float("momentum", 0.0);
 
   std::unique_ptr<Comm> comm;
   if (nranks > 1) {
@@ -109,7 +111,7 @@ static const char* kHelp =
     "  --layers --dim        model size (mlp: 4x4096; bert: 12x768)\n"
     "  --ffn --heads --seq   bert geometry (3072 / 12 / 128)\n"
     "  --batch=N             per-rank batch (mlp 8192, bert 256)\n"
-    "  --optim=adam|sgd --lr=F --bucket-mb=N (64)\n"
+    "  --optim=adam|sgd --lr=F --wd=F --momentum=F --bucket-mb=N (64)\n"
     "  --steps=N --warmup=N  timed window (50 / 10)\n"
     "  --save=path --load=path  .npz checkpoint\n"
     "  --metrics=path        per-run JSONL summary\n"

@@ -29,13 +29,19 @@ class Ddp {
 
  private:
   void on_ready(Param& p);
-  void launch_allreduce(Bucket& b);
+  void launch_allreduce(Bucket& b, int bi);
 
   Context* ctx_;
   Optimizer* opt_;
   Comm* comm_;
   std::unordered_map<Param*, int> bucket_of_;
   int64_t comm_launches_ = 0;
+  // --allreduce-dtype=f32 fallback: grads are all-reduced
+  // in bf16 by default (half the ring bytes on the per-link-bound xGMI
+  // ring); NEZHA_ALLREDUCE_DTYPE=f32 upcasts each bucket to an f32 scratch
+  // around the collective for full-precision reduction.
+  bool allreduce_f32_ = false;
+  std::vector<Tensor> f32_scratch_;
 };
 
 }  // namespace nz

@@ -123,3 +123,29 @@ print("OK")
     for p in ps:
         o, _ = p.communicate(timeout=60)
         assert p.returncode == 0 and b"OK" in o, o.decode()
+
+
+def test_tcp_ddp_f32_allreduce(tmp_path):
+    """--allreduce-dtype=f32 fallback: 2-rank CPU DP with the
+    f32 upcast path trains and stays rank-synchronized."""
+    out = str(tmp_path / "f32.npz")
+    env = {"N": "2", "P": "29831", "OUT": out, "SAME_DATA": "1",
+           "NEZHA_ALLREDUCE_DTYPE": "f32"}
+    ps = [run_worker(WORKER, {**env, "R": str(r)}) for r in range(2)]
+    for p in ps:
+        o, _ = p.communicate(timeout=120)
+        assert p.returncode == 0, o.decode()
+    c = np.load(out)
+    # same-data DP == single-rank training (f32 round trip is exact for
+    # values that came from bf16 grads)
+    import nezha_amd as nz
+    ctx = nz.Ctx(nz.CPU)
+    tr = nz.Trainer.mlp(ctx, layers=2, dim=32, batch=8, lr=1e-3, seed=5,
+                        bucket_bytes=4096)
+    for _ in range(4):
+        tr.step()
+    single = str(tmp_path / "single.npz")
+    tr.save(single)
+    cs = np.load(single)
+    for k in cs.files:
+        np.testing.assert_allclose(c[k], cs[k], rtol=1e-6, atol=1e-7, err_msg=k)
