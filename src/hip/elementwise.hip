@@ -518,6 +518,39 @@ __global__ void layernorm_dgb1_k(const T* __restrict__ dy, const T* __restrict__
   partials[(blockIdx.y * 2 + 0) * N + j] = sg;
   partials[(blockIdx.y * 2 + 1) * N + j] = sb;
 }
+// bf16 fast path: 8 columns per thread, ushort8 sweeps over dy and x
+__global__ void layernorm_dgb1_vec_k(const ushort8_t* __restrict__ dy8,
+                                     const ushort8_t* __restrict__ x8,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ rstd,
+                                     float* __restrict__ partials, int64_t R,
+                                     int64_t N8, int64_t rows_per_chunk) {
+  int64_t j8 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  if (j8 >= N8) return;
+  int64_t r0 = blockIdx.y * rows_per_chunk;
+  int64_t r1 = min(R, r0 + rows_per_chunk);
+  float sg[8] = {}, sb[8] = {};
+  for (int64_t i = r0; i < r1; ++i) {
+    ushort8_t dv = dy8[i * N8 + j8];
+    ushort8_t xv = x8[i * N8 + j8];
+    float mu = mean[i], rs = rstd[i];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      float dyv = b2f(((const uint16_t*)&dv)[u]);
+      float xh = (b2f(((const uint16_t*)&xv)[u]) - mu) * rs;
+      sg[u] += dyv * xh;
+      sb[u] += dyv;
+    }
+  }
+  float* og = partials + (blockIdx.y * 2 + 0) * N8 * 8 + j8 * 8;
+  float* ob = partials + (blockIdx.y * 2 + 1) * N8 * 8 + j8 * 8;
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    og[u] = sg[u];
+    ob[u] = sb[u];
+  }
+}
+
 template <typename T>
 __global__ void layernorm_dgb2_k(const float* __restrict__ partials,
                                  T* __restrict__ dgamma, T* __restrict__ dbeta,
@@ -550,13 +583,29 @@ void layernorm_bwd(const Tensor& dy, const Tensor& x, const Tensor& gamma,
         (const uint16_t*)dy.data(), (const uint16_t*)x.data(),
         (const uint16_t*)gamma.data(), (const float*)mean.data(),
         (const float*)rstd.data(), (uint16_t*)dx.data(), R, N, beta_dx);
-    layernorm_dgb1_k<uint16_t><<<g1, b, 0, s>>>(
-        (const uint16_t*)dy.data(), (const uint16_t*)x.data(),
-        (const float*)mean.data(), (const float*)rstd.data(),
-        (float*)partials.data(), R, N, rows_per_chunk);
-    layernorm_dgb2_k<uint16_t><<<g2, b, 0, s>>>(
-        (const float*)partials.data(), (uint16_t*)dgamma.data(),
-        (uint16_t*)dbeta.data(), N, chunks);
+    if (N % 8 == 0) {
+      int64_t cb8 = (N / 8 + 255) / 256;
+      int ch8 = (int)std::min<int64_t>(
+          (R + 63) / 64,
+          std::min<int64_t>(128, std::max<int64_t>(1, 1024 / cb8)));
+      int64_t rpc8 = (R + ch8 - 1) / ch8;
+      Tensor p8(x.ctx(), DType::F32, Shape{(int64_t)ch8 * 2, N});
+      layernorm_dgb1_vec_k<<<dim3(cb8, ch8), b, 0, s>>>(
+          (const ushort8_t*)dy.data(), (const ushort8_t*)x.data(),
+          (const float*)mean.data(), (const float*)rstd.data(),
+          (float*)p8.data(), R, N / 8, rpc8);
+      layernorm_dgb2_k<uint16_t><<<g2, b, 0, s>>>(
+          (const float*)p8.data(), (uint16_t*)dgamma.data(),
+          (uint16_t*)dbeta.data(), N, ch8);
+    } else {
+      layernorm_dgb1_k<uint16_t><<<g1, b, 0, s>>>(
+          (const uint16_t*)dy.data(), (const uint16_t*)x.data(),
+          (const float*)mean.data(), (const float*)rstd.data(),
+          (float*)partials.data(), R, N, rows_per_chunk);
+      layernorm_dgb2_k<uint16_t><<<g2, b, 0, s>>>(
+          (const float*)partials.data(), (uint16_t*)dgamma.data(),
+          (uint16_t*)dbeta.data(), N, chunks);
+    }
   } else {
     layernorm_dx_k<float><<<dim3(grid_1d(R, 1, 2048)), b, 0, s>>>(
         (const float*)dy.data(), (const float*)x.data(), (const float*)gamma.data(),
