@@ -16,7 +16,7 @@ HOST_SRCS := src/core/tensor.cc src/core/ops.cc src/cpu/cpu_ops.cc \
              src/models/models.cc src/train/trainer.cc src/train/ddp.cc \
              src/comm/rccl_comm.cc src/comm/fake_comm.cc src/comm/tcp_comm.cc \
              src/capi.cc
-HIP_SRCS  := src/hip/elementwise.hip src/hip/gemm.hip src/hip/attention.hip
+HIP_SRCS  := src/hip/elementwise.hip src/hip/gemm.hip src/hip/gemm_f32.hip src/hip/attention.hip
 
 OBJS := $(HOST_SRCS:%.cc=$(BUILD)/%.o) $(HIP_SRCS:%.hip=$(BUILD)/%.o)
 

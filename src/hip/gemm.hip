@@ -1453,6 +1453,12 @@ static bool can_fast_transpose(int64_t R, int64_t C) {
 
 void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
           float alpha, float beta, const ops::GemmEpilogue* epi) {
+  if (A.dtype() == DType::F32) {
+    NZ_CHECK(B.dtype() == DType::F32 && C.dtype() == DType::F32,
+             "hip gemm: mixed dtypes");
+    gemm_f32(A, B, C, transA, transB, alpha, beta, epi);
+    return;
+  }
   NZ_CHECK(A.dtype() == DType::BF16 && B.dtype() == DType::BF16 &&
                C.dtype() == DType::BF16,
            "hip gemm: bf16 tensors required");
@@ -1507,6 +1513,10 @@ void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
 void gemm_batched(const Tensor& A, const Tensor& B, Tensor& C, bool transA,
                   bool transB, float alpha, int64_t batch, int64_t M, int64_t N,
                   int64_t K, int64_t sA, int64_t sB, int64_t sC) {
+  if (A.dtype() == DType::F32) {
+    gemm_f32_batched(A, B, C, transA, transB, alpha, batch, M, N, K, sA, sB, sC);
+    return;
+  }
   NZ_CHECK(A.dtype() == DType::BF16, "hip gemm_batched: bf16 required");
   GemmArgs args{};
   args.A = (const uint16_t*)A.data();

@@ -49,6 +49,15 @@ void l2norm_sq(const Tensor& src, Tensor& out);
 }  // namespace hip
 }  // namespace nz
 
+// f32 GEMM on the exact-f32 MFMA (src/hip/gemm_f32.hip)
+namespace nz { namespace hip {
+void gemm_f32(const Tensor& A, const Tensor& B, Tensor& C, bool transA,
+              bool transB, float alpha, float beta, const ops::GemmEpilogue* epi);
+void gemm_f32_batched(const Tensor& A, const Tensor& B, Tensor& C, bool transA,
+                      bool transB, float alpha, int64_t batch, int64_t M,
+                      int64_t N, int64_t K, int64_t sA, int64_t sB, int64_t sC);
+}}
+
 // fused attention forward (src/hip/attention.hip)
 namespace nz { namespace hip {
 bool attn_fused_fwd_supported(int64_t S, int64_t dh);

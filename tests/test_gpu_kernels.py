@@ -316,3 +316,35 @@ def test_attn_fused_determinism(hip_ctx):
         outs.append((probs.raw(), o.raw()))
     np.testing.assert_array_equal(outs[0][0], outs[1][0])
     np.testing.assert_array_equal(outs[0][1], outs[1][1])
+
+
+@pytest.mark.parametrize("transA,transB", [(False, False), (False, True),
+                                           (True, False), (True, True)])
+def test_gemm_f32_gpu(hip_ctx, transA, transB):
+    """f32 GEMM path (exact-f32 MFMA): tight tolerance vs numpy f32."""
+    M, N, K = 96, 80, 64
+    A = RNG.normal(size=(K, M) if transA else (M, K)).astype(np.float32)
+    B = RNG.normal(size=(N, K) if transB else (K, N)).astype(np.float32)
+    ref = (A.T if transA else A) @ (B.T if transB else B)
+    a, b = hip_ctx.from_numpy(A), hip_ctx.from_numpy(B)
+    c = hip_ctx.tensor((M, N), nz.F32)
+    nz.gemm(a, b, c, transA, transB)
+    np.testing.assert_allclose(c.numpy(), ref, rtol=1e-5, atol=1e-4)
+
+
+def test_gemm_f32_epilogue_gpu(hip_ctx):
+    import torch
+    M, N, K = 64, 64, 32
+    A = RNG.normal(size=(M, K)).astype(np.float32)
+    B = RNG.normal(size=(K, N)).astype(np.float32)
+    bias = RNG.normal(size=N).astype(np.float32)
+    a, b = hip_ctx.from_numpy(A), hip_ctx.from_numpy(B)
+    bi = hip_ctx.from_numpy(bias)
+    c = hip_ctx.tensor((M, N), nz.F32)
+    pre = hip_ctx.tensor((M, N), nz.F32)
+    nz.gemm(a, b, c, bias=bi, gelu=True, save_pre=pre)
+    z_ref = A @ B + bias
+    y_ref = torch.nn.functional.gelu(torch.from_numpy(z_ref),
+                                     approximate="tanh").numpy()
+    np.testing.assert_allclose(pre.numpy(), z_ref, rtol=1e-5, atol=1e-4)
+    np.testing.assert_allclose(c.numpy(), y_ref, rtol=1e-4, atol=1e-4)
