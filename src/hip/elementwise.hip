@@ -1038,31 +1038,33 @@ void mse_loss(const Tensor& y, const Tensor& target, Tensor& loss_out, Tensor* d
   NZ_KERNEL_CHECK();
 }
 
-// CE over rows (f32 logits; labels hold integral class ids in f32).
-__global__ void ce_loss_k(const float* __restrict__ logits,
+// CE over rows (f32 or bf16 logits; labels hold integral class ids in f32).
+template <typename T>
+__global__ void ce_loss_k(const T* __restrict__ logits,
                           const float* __restrict__ labels,
-                          float* __restrict__ dlogits,
+                          T* __restrict__ dlogits,
                           float* __restrict__ partials, int64_t R, int64_t N,
                           float gs) {
   __shared__ float lds4[4];
   float row_losses = 0.f;
   for (int64_t row = blockIdx.x; row < R; row += gridDim.x) {
-    const float* lr = logits + row * N;
+    const T* lr = logits + row * N;
     float m = -INFINITY;
-    for (int64_t j = threadIdx.x; j < N; j += blockDim.x) m = fmaxf(m, lr[j]);
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x)
+      m = fmaxf(m, ld(lr, j));
     m = block_max<256>(m, lds4);
     __syncthreads();
     float z = 0.f;
-    for (int64_t j = threadIdx.x; j < N; j += blockDim.x) z += __expf(lr[j] - m);
+    for (int64_t j = threadIdx.x; j < N; j += blockDim.x)
+      z += __expf(ld(lr, j) - m);
     z = block_sum<256>(z, lds4);
     int64_t y = (int64_t)labels[row];
-    if (threadIdx.x == 0)
-      row_losses += -(lr[y] - m - __logf(z));
+    if (threadIdx.x == 0) row_losses += -(ld(lr, y) - m - __logf(z));
     if (dlogits) {
-      float* dlr = dlogits + row * N;
+      T* dlr = dlogits + row * N;
       for (int64_t j = threadIdx.x; j < N; j += blockDim.x) {
-        float p = __expf(lr[j] - m) / z;
-        dlr[j] = (p - (j == y ? 1.f : 0.f)) / float(R) * gs;
+        float p = __expf(ld(lr, j) - m) / z;
+        st(dlr, j, (p - (j == y ? 1.f : 0.f)) / float(R) * gs);
       }
     }
     __syncthreads();
@@ -1071,14 +1073,20 @@ __global__ void ce_loss_k(const float* __restrict__ logits,
 }
 void ce_loss(const Tensor& logits, const Tensor& labels, Tensor& loss_out,
              Tensor* dlogits, float gs) {
-  NZ_CHECK(logits.dtype() == DType::F32, "hip ce_loss expects f32 logits");
   hipStream_t s = stream_of(logits);
   int64_t R = logits.rows(), N = logits.cols();
   int nb = (int)grid_1d(R, 1, 1024);
   Tensor partials(logits.ctx(), DType::F32, Shape{nb});
-  ce_loss_k<<<dim3(nb), dim3(256), 0, s>>>(
-      (const float*)logits.data(), (const float*)labels.data(),
-      dlogits ? (float*)dlogits->data() : nullptr, (float*)partials.data(), R, N, gs);
+  if (logits.dtype() == DType::BF16)
+    ce_loss_k<uint16_t><<<dim3(nb), dim3(256), 0, s>>>(
+        (const uint16_t*)logits.data(), (const float*)labels.data(),
+        dlogits ? (uint16_t*)dlogits->data() : nullptr,
+        (float*)partials.data(), R, N, gs);
+  else
+    ce_loss_k<float><<<dim3(nb), dim3(256), 0, s>>>(
+        (const float*)logits.data(), (const float*)labels.data(),
+        dlogits ? (float*)dlogits->data() : nullptr, (float*)partials.data(),
+        R, N, gs);
   combine_partials_k<<<dim3(1), dim3(256), 0, s>>>(
       (const float*)partials.data(), nb, (float*)loss_out.data(), 1.0f / R);
   NZ_KERNEL_CHECK();

@@ -348,3 +348,20 @@ def test_gemm_f32_epilogue_gpu(hip_ctx):
                                      approximate="tanh").numpy()
     np.testing.assert_allclose(pre.numpy(), z_ref, rtol=1e-5, atol=1e-4)
     np.testing.assert_allclose(c.numpy(), y_ref, rtol=1e-4, atol=1e-4)
+
+
+def test_ce_loss_bf16_gpu(hip_ctx):
+    import torch
+    R, N = 64, 128
+    L = randb(R, N)
+    labels = RNG.integers(0, N, size=R).astype(np.float32)
+    lt = torch.from_numpy(L).requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(lt, torch.from_numpy(labels).long())
+    ref.backward()
+    lo = hip_ctx.from_numpy_bf16(L)
+    lab = hip_ctx.from_numpy(labels)
+    loss = hip_ctx.tensor((1,))
+    dl = hip_ctx.tensor((R, N), nz.BF16)
+    nz.ce_loss(lo, lab, loss, dl)
+    assert abs(loss.numpy()[0] - ref.item()) < 2e-2
+    check(dl.numpy(), lt.grad.numpy(), rtol=0.03)
