@@ -22,8 +22,11 @@ Trainer::Trainer(Context* ctx, std::unique_ptr<Model> model,
   ops::rand_normal(target_, data_seed + 1, 0.f, 1.f);
   if (const char* mp = getenv("NEZHA_METRICS"))
     metrics_ = std::make_unique<MetricsWriter>(mp);
+  const char* dm = getenv("NEZHA_DATA");
+  data_stream_ = dm && std::string(dm) == "stream";
   const char* g = getenv("NEZHA_GRAPH");
-  graph_mode_ = g && g[0] == '1' && ctx->device() == Device::HIP;
+  graph_mode_ = g && g[0] == '1' && ctx->device() == Device::HIP &&
+                !data_stream_;  // stream mode regenerates inputs per step
   // allocate the bias-correction buffer up front: taking a cached block
   // at capture time would starve the captured step's own allocations
   if (graph_mode_) bc_dev_ = Tensor(ctx, DType::F32, Shape{2});
@@ -41,6 +44,13 @@ void Trainer::attach_ddp(Comm* comm) {
 }
 
 void Trainer::step_body(const Tensor* bc_dev) {
+  if (data_stream_) {
+    // streaming synthetic data: a fresh deterministic batch every step
+    // (counter-based RNG keyed by data_seed + step index)
+    uint64_t s = data_seed_ + 2654435761ull * (uint64_t)(opt_->step_count() + 1);
+    ops::rand_normal(input_, s, 0.f, 1.f);
+    ops::rand_normal(target_, s + 1, 0.f, 1.f);
+  }
   if (ddp_) ddp_->pre_step();
   tape_.clear();
   const bool mx = metrics_ && ctx_->device() == Device::HIP && !graph_mode_;

@@ -46,6 +46,7 @@ class Trainer {
   Tape tape_;
   Tensor input_, target_, loss_;
   uint64_t data_seed_ = 0;  // persisted as meta/data_seed; reseed on load
+  bool data_stream_ = false;  // NEZHA_DATA=stream: fresh batch every step
   // per-step metrics: enabled by NEZHA_METRICS=<path>;
   // fwd/bwd/opt ms splits via hipEvents (syncs each step — observability
   // mode, off on the bench path).
