@@ -23,7 +23,7 @@ OBJS := $(HOST_SRCS:%.cc=$(BUILD)/%.o) $(HIP_SRCS:%.hip=$(BUILD)/%.o)
 LIB := $(BUILD)/libnezha.so
 
 CMDS := $(BUILD)/nezha-train $(BUILD)/nezha-bench $(BUILD)/nezha-info
-TESTS_CPP := $(BUILD)/test_core $(BUILD)/test_ddp_fake $(BUILD)/test_modules
+TESTS_CPP := $(BUILD)/test_core $(BUILD)/test_ddp_fake $(BUILD)/test_modules $(BUILD)/test_ddp_fake_gpu
 
 .PHONY: all lib cmds tests clean
 all: lib cmds tests

@@ -39,7 +39,11 @@ std::unique_ptr<Comm> make_rccl_comm(Context* ctx, int nranks, int rank,
                                      const void* uid);
 
 // In-process fake comm: returns n connected instances (threads share state).
-std::vector<std::unique_ptr<Comm>> make_fake_comm_group(int nranks);
+// device_buffers=true stages device pointers through the host around the
+// reduce (synchronous w.r.t. the passed stream) — lets N simulated ranks
+// run REAL GPU DDP on one device (test tier T4 on hardware).
+std::vector<std::unique_ptr<Comm>> make_fake_comm_group(int nranks,
+                                                        bool device_buffers = false);
 
 // TCP star over localhost (CPU tensors / host scalars only).
 std::unique_ptr<Comm> make_tcp_comm(int nranks, int rank, const char* host, int port);

@@ -2,6 +2,8 @@
 // ranks running as threads over host (CPU) buffers. The DDP engine sees
 // only the Comm interface, so this exercises real 8-rank bucketing/overlap
 // logic on the 8-CPU host with no cluster.
+#include <hip/hip_runtime.h>
+
 #include <condition_variable>
 #include <mutex>
 
@@ -61,25 +63,50 @@ void reduce_buffers(std::vector<void*>& ptrs, int64_t count, ReduceOp op,
 
 class FakeComm : public Comm {
  public:
-  FakeComm(std::shared_ptr<FakeState> st, int rank) : st_(std::move(st)), rank_(rank) {}
+  FakeComm(std::shared_ptr<FakeState> st, int rank, bool device)
+      : st_(std::move(st)), rank_(rank), device_(device) {}
   int rank() const override { return rank_; }
   int nranks() const override { return st_->nranks; }
 
   void all_reduce(void* data, int64_t count, DType dt, ReduceOp op,
-                  Stream*) override {
+                  Stream* stream) override {
+    if (device_ && stream) stream->sync();  // grads ready on the device
     st_->rendezvous(rank_, data, [&](std::vector<void*>& ptrs) {
-      reduce_buffers<float>(ptrs, count, op, dt == DType::BF16);
+      size_t bytes = size_t(count) * dtype_size(dt);
+      if (!device_) {
+        reduce_buffers<float>(ptrs, count, op, dt == DType::BF16);
+        return;
+      }
+      // stage every rank's device buffer through the host, reduce, restore
+      std::vector<std::vector<char>> host(ptrs.size());
+      std::vector<void*> hptrs(ptrs.size());
+      for (size_t i = 0; i < ptrs.size(); ++i) {
+        host[i].resize(bytes);
+        (void)hipMemcpy(host[i].data(), ptrs[i], bytes, hipMemcpyDeviceToHost);
+        hptrs[i] = host[i].data();
+      }
+      reduce_buffers<float>(hptrs, count, op, dt == DType::BF16);
+      for (size_t i = 0; i < ptrs.size(); ++i)
+        (void)hipMemcpy(ptrs[i], host[i].data(), bytes, hipMemcpyHostToDevice);
     });
   }
 
-  void broadcast(void* data, int64_t count, DType dt, int root, Stream*) override {
+  void broadcast(void* data, int64_t count, DType dt, int root, Stream* stream) override {
+    if (device_ && stream) stream->sync();
     size_t bytes = size_t(count) * dtype_size(dt);
-    int r = rank_;
-    st_->rendezvous(rank_, data, [&, root, bytes](std::vector<void*>& ptrs) {
+    bool dev = device_;
+    st_->rendezvous(rank_, data, [&, root, bytes, dev](std::vector<void*>& ptrs) {
+      if (!dev) {
+        for (int i = 0; i < (int)ptrs.size(); ++i)
+          if (i != root) std::memcpy(ptrs[i], ptrs[root], bytes);
+        return;
+      }
+      std::vector<char> host(bytes);
+      (void)hipMemcpy(host.data(), ptrs[root], bytes, hipMemcpyDeviceToHost);
       for (int i = 0; i < (int)ptrs.size(); ++i)
-        if (i != root) std::memcpy(ptrs[i], ptrs[root], bytes);
+        if (i != root)
+          (void)hipMemcpy(ptrs[i], host.data(), bytes, hipMemcpyHostToDevice);
     });
-    (void)r;
   }
 
   void barrier() override {
@@ -95,15 +122,17 @@ class FakeComm : public Comm {
  private:
   std::shared_ptr<FakeState> st_;
   int rank_;
+  bool device_;
 };
 
 }  // namespace
 
-std::vector<std::unique_ptr<Comm>> make_fake_comm_group(int nranks) {
+std::vector<std::unique_ptr<Comm>> make_fake_comm_group(int nranks,
+                                                         bool device_buffers) {
   auto st = std::make_shared<FakeState>(nranks);
   std::vector<std::unique_ptr<Comm>> out;
   for (int r = 0; r < nranks; ++r)
-    out.push_back(std::make_unique<FakeComm>(st, r));
+    out.push_back(std::make_unique<FakeComm>(st, r, device_buffers));
   return out;
 }
 

@@ -172,3 +172,13 @@ def test_gpu_resume_bitwise(hip_ctx, tmp_path):
     ca, cb = np.load(fa), np.load(fb)
     for k in ca.files:
         np.testing.assert_array_equal(ca[k], cb[k], err_msg=k)
+
+
+def test_ddp_8rank_one_device():
+    """8 simulated DP ranks (threads, own streams) on one GPU via the
+    device-staging FakeComm — real event/comm-stream/chaining path."""
+    import subprocess, os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    binp = os.path.join(repo, "build", "test_ddp_fake_gpu")
+    r = subprocess.run([binp], capture_output=True, timeout=600)
+    assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
