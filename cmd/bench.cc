@@ -55,6 +55,29 @@ int main(int argc, const char** argv) {
   Context ctx(Device::HIP, 0);
 
   int64_t B = flags.get_int("batch", 8192);
+  if (which == "shape") {
+    // one arbitrary GEMM shape: --m= --n= --k= [--ta=1] [--tb=1]
+    bench_gemm(ctx, flags.get_int("m", 4096), flags.get_int("n", 4096),
+               flags.get_int("k", 4096), flags.get_bool("ta", false),
+               flags.get_bool("tb", false), "custom", iters);
+    return 0;
+  }
+  if (which == "bert32k") {
+    // the per-layer BERT-base linear GEMMs at the contract batch
+    // (B=256, S=128 -> M=32768) — the shapes that dominate the step.
+    bench_gemm(ctx, 32768, 2304, 768, false, false, "qkv-fwd (NN)", iters);
+    bench_gemm(ctx, 32768, 768, 768, false, false, "out-fwd (NN)", iters);
+    bench_gemm(ctx, 32768, 3072, 768, false, false, "ffn1-fwd (NN)", iters);
+    bench_gemm(ctx, 32768, 768, 3072, false, false, "ffn2-fwd (NN)", iters);
+    bench_gemm(ctx, 32768, 768, 2304, false, true, "qkv-dgrad (NT)", iters);
+    bench_gemm(ctx, 32768, 768, 768, false, true, "out-dgrad (NT)", iters);
+    bench_gemm(ctx, 32768, 3072, 768, false, true, "ffn1-dgrad (NT)", iters);
+    bench_gemm(ctx, 32768, 768, 3072, false, true, "ffn2-dgrad (NT)", iters);
+    bench_gemm(ctx, 768, 2304, 32768, true, false, "qkv-wgrad (TN)", iters);
+    bench_gemm(ctx, 768, 3072, 32768, true, false, "ffn1-wgrad (TN)", iters);
+    bench_gemm(ctx, 3072, 768, 32768, true, false, "ffn2-wgrad (TN)", iters);
+    return 0;
+  }
   if (which == "all" || which == "gemm") {
     // the MLP contract shapes (C11): fwd, dgrad, wgrad at batch 8192
     bench_gemm(ctx, B, 4096, 4096, false, false, "mlp-fwd (NN)", iters);
