@@ -333,6 +333,160 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_nt_k(GemmArgs args,
 }
 
 // ---------------------------------------------------------------------------
+// Persistent NT kernel: fixed 512-WG grid (2 blocks/CU), each WG sweeps a
+// strided run of (batch, m, n) tiles with the SAME 2-buffer glds pipeline as
+// gemm_bf16_nt_k, staging tile t+1 / the NEXT tile's first k-block during the
+// current compute. Motivation (measured): at K=768 a 128^2 tile is only 12
+// k-iterations (~5 us of MFMA) against a ~2 us stage prologue + drain, so the
+// one-tile-per-WG kernel tops out near 60% of its long-K rate (536 vs 880+
+// TF/s on the BERT qkv-fwd shape); tiny batched tiles (attention backward,
+// K=64, ONE k-iteration) are almost pure prologue. Amortizing one prologue
+// over cnt/64 tiles per WG and making every WG process the same tile count
+// removes both the ramp and the grid-quantization waste (e.g. N=768 ->
+// 1536 WGs = exactly 3 rounds, but 576 -> 1.125 rounds = 12.5% idle).
+// Tile order: XCD-chunked like the block swizzle (WG w = slot w/8 on XCD
+// w%8 walks chunk tiles slot, slot+64, ... so the 64 co-resident WGs of an
+// XCD always cover ~64 consecutive m-major tiles -> A rows and B panels
+// stay hot in that XCD's L2).
+// ---------------------------------------------------------------------------
+__launch_bounds__(THREADS) __global__ void gemm_bf16_nt_pers_k(
+    GemmArgs args, int nwg_n, int64_t per_b, int64_t total) {
+  constexpr int NT_BK = 64;
+  __shared__ __attribute__((aligned(16))) uint16_t smem[2 * (BM + BN) * NT_BK];
+  constexpr int BUF_STRIDE = (BM + BN) * NT_BK;
+  constexpr int B_OFF = BM * NT_BK;
+
+  const int orig = blockIdx.x;
+  const int xcd = orig & 7, slot = orig >> 3;  // 64 slots per XCD
+  const int64_t q = total / 8, r8 = total % 8;
+  const int64_t lo = xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q;
+  const int64_t cnt = xcd < r8 ? q + 1 : q;
+  if (slot >= cnt) return;
+  constexpr int S = 64;  // slot stride inside the chunk
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  int e0 = (wid * 2048) + lane * 8;
+
+  f32x4 acc[4][4] = {};
+  const int64_t ntiles = args.K / NT_BK;
+
+  // Staging-tile state, advanced ONCE per tile: the two 64-bit div/mods and
+  // the 8 address rebuilds must stay out of the per-k-iteration stage call
+  // (first version recomputed them every iteration and lost ~15% on every
+  // shape). The per-lane A/B source addresses for the 4 glds chunks are
+  // fully precomputed (clamp included); stage() only adds k0.
+  const uint16_t* asrc[4];
+  const uint16_t* bsrc[4];
+  auto set_tile = [&](int64_t t_l) {
+    const int64_t b = t_l / per_b, rtl = t_l % per_b;
+    const uint16_t* __restrict__ Ab = args.A + b * args.sA;
+    const uint16_t* __restrict__ Bb = args.B + b * args.sB;
+    const int64_t sm0 = (rtl / nwg_n) * BM;
+    const int64_t sn0 = (rtl % nwg_n) * BN;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int e = e0 + c * 512;
+      int row = e >> 6;
+      int k = (e & 63) ^ ((row & 7) << 3);  // XOR involution, source side
+      int64_t ar = sm0 + row;
+      if (ar > Mc) ar = Mc;  // clamp: masked at the epilogue
+      asrc[c] = Ab + ar * lda + k;
+      int64_t br = sn0 + row;
+      if (br > Nc) br = Nc;
+      bsrc[c] = Bb + br * ldb + k;
+    }
+  };
+  auto stage = [&](int buf, int64_t k0) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int e = e0 + c * 512;
+      glds16(asrc[c] + k0, smem + buf * BUF_STRIDE + e);
+      glds16(bsrc[c] + k0, smem + buf * BUF_STRIDE + B_OFF + e);
+    }
+  };
+
+  auto epilogue = [&](int64_t tl) {
+    const int64_t b = tl / per_b, rtl = tl % per_b;
+    const int64_t em0 = (rtl / nwg_n) * BM;
+    const int64_t en0 = (rtl % nwg_n) * BN;
+    uint16_t* C = args.C + b * args.sC;
+    uint16_t* save_pre = args.save_pre ? args.save_pre + b * args.sC : nullptr;
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+#pragma unroll
+        for (int rI = 0; rI < 4; ++rI) {
+          int64_t grow = em0 + wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+          int64_t gcol = en0 + wc * 64 + fn * 16 + (lane & 15);
+          if (grow < args.M && gcol < args.N) {
+            int64_t idx = grow * args.N + gcol;
+            float v = args.alpha * acc[fm][fn][rI];
+            if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+            if (args.bias) v += b2f(args.bias[gcol]);
+            if (save_pre) save_pre[idx] = f2b(v);
+            if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+            C[idx] = f2b(v);
+          }
+        }
+        acc[fm][fn] = f32x4{};
+      }
+  };
+
+  int64_t tl = lo + slot;
+  set_tile(tl);
+  stage(0, 0);
+  __syncthreads();  // vmcnt(0): first tile landed
+
+  int cur = 0;
+  for (; tl < lo + cnt; tl += S) {
+    for (int64_t t = 0; t < ntiles; ++t) {
+      // prefetch: next k-block of this tile, else the next tile's block 0
+      if (t + 1 < ntiles) {
+        stage(cur ^ 1, (t + 1) * NT_BK);
+      } else if (tl + S < lo + cnt) {
+        set_tile(tl + S);  // once per tile, off the steady-state path
+        stage(cur ^ 1, 0);
+      }
+#pragma unroll
+      for (int kk = 0; kk < NT_BK; kk += 32) {
+        short8 a_frag[4], b_frag[4];
+        const int kbase = kk + (lane >> 4) * 8;
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm) {
+          int row = wr * 64 + fm * 16 + (lane & 15);
+          int ks = kbase ^ ((row & 7) << 3);
+          a_frag[fm] =
+              *(const short8*)(smem + cur * BUF_STRIDE + row * NT_BK + ks);
+        }
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int col = wc * 64 + fn * 16 + (lane & 15);
+          int ks = kbase ^ ((col & 7) << 3);
+          b_frag[fn] = *(const short8*)(smem + cur * BUF_STRIDE + B_OFF +
+                                        col * NT_BK + ks);
+        }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < 4; ++fn)
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[fm], b_frag[fn], acc[fm][fn], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
+      if (t + 1 == ntiles) epilogue(tl);  // acc complete; stores overlap DMA
+      __syncthreads();
+      cur ^= 1;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Tier B+: deep-pipelined NT kernel. 256x128 tile, 512 threads (8 waves as
 // 4Mx2N, 64x64 per wave), BK=64, THREE LDS buffers (144 KB, 1 block/CU)
 // with counted s_waitcnt vmcnt(N) and raw s_barriers so two tiles of glds
@@ -1335,6 +1489,13 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     const char* v = getenv("NEZHA_GEMM_MIN_WG");  // test hook: force 8-phase
     return v ? atoi(v) : 224;
   }();
+  static const bool kMinWgForced = getenv("NEZHA_GEMM_MIN_WG") != nullptr;
+  // The 256^2 kernel runs ~1 workgroup per CU, so a grid that is not close
+  // to a multiple of 256 leaves whole CUs idle for a full round (N=768 ->
+  // 384 WGs = 25% idle; measured 679 vs 884 TF against the 128^2 kernel).
+  auto balanced_256 = [&](int nwg) {
+    return kMinWgForced || nwg % 256 == 0 || nwg >= 1024;
+  };
   if (!transA && transB && batch == 1 && kNtVariant == 5 &&
       args.K % 128 == 0 && args.K >= 1536) {
     int nwg_m = (int)((args.M + 255) / 256);
@@ -1376,7 +1537,8 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     int nwg_n = (int)((args.N + 255) / 256);
     // 256^2 tiles only when the grid fills the chip (1 block/CU, 256 CUs);
     // small-N/M shapes run faster on the 128^2 2-buffer kernel below.
-    if (nwg_m * (int64_t)nwg_n >= kNtMinWG) {
+    if (nwg_m * (int64_t)nwg_n >= kNtMinWG &&
+        balanced_256(nwg_m * nwg_n)) {
       dim3 grid(nwg_m * nwg_n);
       gemm_bf16_nt_8phase_k<<<grid, dim3(512), 0, s>>>(args, nwg_n);
       NZ_KERNEL_CHECK();
@@ -1402,8 +1564,20 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     // fused reduce applies the real epilogue.
     if (batch == 1 && nwg < 224 && args.K >= 4096 && !args.out_f32 &&
         g_splitk_ctx) {
-      int slices = (int)std::min<int64_t>(
-          16, std::min<int64_t>(args.K / 2048, (448 + nwg - 1) / nwg));
+      // Power-of-two slice count: double until the grid reaches ~2 WGs/CU
+      // (the 128^2 kernel's occupancy), keeping K/slices a >=2048 multiple
+      // of 64. A non-power-of-two count (the old (448+nwg-1)/nwg) usually
+      // failed the divisibility test and silently disabled split-K — the
+      // BERT qkv wgrad then ran 108 WGs on 256 CUs at 227 TF.
+      static const int kSplitkTarget = [] {
+        const char* v = getenv("NEZHA_SPLITK_TARGET");
+        return v ? atoi(v) : 512;
+      }();
+      int slices = 1;
+      while (slices < 16 && nwg * slices < kSplitkTarget &&
+             args.K % ((int64_t)slices * 128) == 0 &&
+             args.K / (slices * 2) >= 2048)
+        slices *= 2;
       if (slices > 1 && args.K % ((int64_t)slices * 64) == 0) {
         int64_t ksl = args.K / slices;
         Tensor parts(g_splitk_ctx, DType::F32,
@@ -1422,6 +1596,19 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
         NZ_KERNEL_CHECK();
         return;
       }
+    }
+    // Persistent sweep: amortizes the stage prologue over many tiles and
+    // equalizes per-WG work (see gemm_bf16_nt_pers_k). Wins for short-K
+    // shapes and small batched tiles; neutral for long K.
+    static const bool kPersOff =
+        getenv("NEZHA_GEMM_PERS") && atoi(getenv("NEZHA_GEMM_PERS")) == 0;
+    int64_t per_b = (int64_t)nwg_m * nwg_n;
+    int64_t total = per_b * batch;
+    if (!kPersOff && total >= 512 && !args.out_f32) {
+      gemm_bf16_nt_pers_k<<<dim3(512), dim3(THREADS), 0, s>>>(args, nwg_n,
+                                                              per_b, total);
+      NZ_KERNEL_CHECK();
+      return;
     }
     dim3 grid(nwg, batch);
     gemm_bf16_nt_k<<<grid, dim3(THREADS), 0, s>>>(args, nwg_n);
