@@ -267,24 +267,42 @@ __global__ void colsum_stage1_k(const T* __restrict__ src,
   partials[blockIdx.y * N + j] = acc;
 }
 
-// bf16 fast path: 8 columns per thread via ushort8 row sweeps (the scalar
-// version ran at ~0.8 TB/s on the BERT bias grads).
+// bf16 fast path. One 256-thread WG covers 128 columns (16 ushort8 lanes)
+// x 16 row-subchunks: thread (jj, sub) sweeps rows sub, sub+16, ... of the
+// chunk, then a fixed-order LDS tree folds the 16 sub-partials. The earlier
+// one-thread-per-column-strip version was LATENCY bound on the narrow BERT
+// bias grads (N=2304 -> 288 active lanes per chunk, one dependent load in
+// flight, 1.5 TB/s measured); this layout gives 8x the wave count and
+// fully-coalesced 256 B quarter-wave segments. Deterministic: serial row
+// order and a fixed 16-way fold order per column.
 __global__ void colsum_stage1_vec_k(const ushort8_t* __restrict__ src8,
                                     float* __restrict__ partials, int64_t M,
                                     int64_t N8, int64_t rows_per_chunk) {
-  int64_t j8 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-  if (j8 >= N8) return;
-  int64_t r0 = blockIdx.y * rows_per_chunk;
-  int64_t r1 = min(M, r0 + rows_per_chunk);
+  __shared__ float sm[16 * 128];
+  const int jj = threadIdx.x & 15;         // ushort8 column within the block
+  const int sub = threadIdx.x >> 4;        // row subchunk 0..15
+  const int64_t j8 = (int64_t)blockIdx.x * 16 + jj;
+  const int64_t r0 = blockIdx.y * rows_per_chunk;
+  const int64_t r1 = min(M, r0 + rows_per_chunk);
   float acc[8] = {};
-  for (int64_t i = r0; i < r1; ++i) {
-    ushort8_t v = src8[i * N8 + j8];
+  if (j8 < N8) {
+    for (int64_t i = r0 + sub; i < r1; i += 16) {
+      ushort8_t v = src8[i * N8 + j8];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) acc[u] += b2f(((const uint16_t*)&v)[u]);
+      for (int u = 0; u < 8; ++u) acc[u] += b2f(((const uint16_t*)&v)[u]);
+    }
   }
-  float* out = partials + blockIdx.y * N8 * 8 + j8 * 8;
 #pragma unroll
-  for (int u = 0; u < 8; ++u) out[u] = acc[u];
+  for (int u = 0; u < 8; ++u) sm[sub * 128 + jj * 8 + u] = acc[u];
+  __syncthreads();
+  if (threadIdx.x < 128) {
+    const int col = threadIdx.x;
+    float v = 0.f;
+#pragma unroll
+    for (int s_ = 0; s_ < 16; ++s_) v += sm[s_ * 128 + col];
+    int64_t gcol = (int64_t)blockIdx.x * 128 + col;
+    if (gcol < N8 * 8) partials[blockIdx.y * N8 * 8 + gcol] = v;
+  }
 }
 template <typename TD>
 __global__ void colsum_stage2_k(const float* __restrict__ partials,
@@ -292,8 +310,18 @@ __global__ void colsum_stage2_k(const float* __restrict__ partials,
                                 float beta) {
   int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   if (j >= N) return;
-  float acc = 0.f;
-  for (int c = 0; c < chunks; ++c) acc += partials[c * N + j];
+  // 4 independent accumulators: the fold is latency-bound at high chunk
+  // counts (few waves, serial dependent adds); fixed order -> deterministic
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  int c = 0;
+  for (; c + 4 <= chunks; c += 4) {
+    a0 += partials[(int64_t)c * N + j];
+    a1 += partials[(int64_t)(c + 1) * N + j];
+    a2 += partials[(int64_t)(c + 2) * N + j];
+    a3 += partials[(int64_t)(c + 3) * N + j];
+  }
+  for (; c < chunks; ++c) a0 += partials[(int64_t)c * N + j];
+  float acc = (a0 + a1) + (a2 + a3);
   float prev = beta != 0.f ? beta * ld(dst, j) : 0.f;
   st(dst, j, prev + acc);
 }
@@ -309,11 +337,12 @@ void colsum(const Tensor& src, Tensor& dst, float beta) {
       std::min<int64_t>(64, std::max<int64_t>(1, 512 / colblocks)));
   int64_t rows_per_chunk = (M + chunks - 1) / chunks;
   if (src.dtype() == DType::BF16 && N % 8 == 0) {
-    // vectorized: 8 cols/thread; coarser grid so re-derive the chunking
-    int64_t cb8 = (N / 8 + 255) / 256;
+    // vectorized: WG covers 128 columns x 16 row-subchunks; pick the chunk
+    // count so jblocks*chunks ~ 1024 WGs (4 waves/SIMD)
+    int64_t cb8 = (N + 127) / 128;
     int ch8 = (int)std::min<int64_t>(
-        (M + 63) / 64,
-        std::min<int64_t>(128, std::max<int64_t>(1, 1024 / cb8)));
+        (M + 15) / 16,
+        std::min<int64_t>(256, std::max<int64_t>(1, 1024 / cb8)));
     int64_t rpc8 = (M + ch8 - 1) / ch8;
     Tensor partials8(src.ctx(), DType::F32, Shape{(int64_t)ch8, N});
     colsum_stage1_vec_k<<<dim3(cb8, ch8), dim3(256), 0, s>>>(

@@ -414,6 +414,7 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_nt_pers_k(
     const int64_t em0 = (rtl / nwg_n) * BM;
     const int64_t en0 = (rtl % nwg_n) * BN;
     uint16_t* C = args.C + b * args.sC;
+    float* out32 = args.out_f32 ? args.out_f32 + b * args.sC : nullptr;
     uint16_t* save_pre = args.save_pre ? args.save_pre + b * args.sC : nullptr;
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm)
@@ -425,12 +426,16 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_nt_pers_k(
           int64_t gcol = en0 + wc * 64 + fn * 16 + (lane & 15);
           if (grow < args.M && gcol < args.N) {
             int64_t idx = grow * args.N + gcol;
-            float v = args.alpha * acc[fm][fn][rI];
-            if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
-            if (args.bias) v += b2f(args.bias[gcol]);
-            if (save_pre) save_pre[idx] = f2b(v);
-            if (args.fuse_gelu) v = gelu_fwd_scalar(v);
-            C[idx] = f2b(v);
+            if (out32) {
+              out32[idx] = acc[fm][fn][rI];  // split-K partial slab
+            } else {
+              float v = args.alpha * acc[fm][fn][rI];
+              if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+              if (args.bias) v += b2f(args.bias[gcol]);
+              if (save_pre) save_pre[idx] = f2b(v);
+              if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+              C[idx] = f2b(v);
+            }
           }
         }
         acc[fm][fn] = f32x4{};
@@ -1559,6 +1564,8 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     int nwg_m = (int)((args.M + BM - 1) / BM);
     int nwg_n = (int)((args.N + BN - 1) / BN);
     int nwg = nwg_m * nwg_n;
+    static const bool kPersOff =
+        getenv("NEZHA_GEMM_PERS") && atoi(getenv("NEZHA_GEMM_PERS")) == 0;
     // Split-K (BERT wgrads: tiny MxN against K=32k left 85% of the chip
     // idle): run K-slices as the batch dim into f32 partial slabs, then a
     // fused reduce applies the real epilogue.
@@ -1588,8 +1595,16 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
         pa.sB = ksl;
         pa.sC = args.M * args.N;
         pa.out_f32 = (float*)parts.data();
-        dim3 gridk(nwg, slices);
-        gemm_bf16_nt_k<<<gridk, dim3(THREADS), 0, s>>>(pa, nwg_n);
+        int64_t sk_total = (int64_t)nwg * slices;
+        if (!kPersOff && sk_total >= 512) {
+          // slices ride the persistent kernel's batch dimension (per-slice
+          // strides sA/sB/sC are exactly the batched-pointer contract)
+          gemm_bf16_nt_pers_k<<<dim3(512), dim3(THREADS), 0, s>>>(
+              pa, nwg_n, nwg, sk_total);
+        } else {
+          dim3 gridk(nwg, slices);
+          gemm_bf16_nt_k<<<gridk, dim3(THREADS), 0, s>>>(pa, nwg_n);
+        }
         NZ_KERNEL_CHECK();
         splitk_reduce_k<<<dim3(grid_1d(args.M * args.N, 256)), dim3(256), 0,
                           s>>>((const float*)parts.data(), args, slices);
@@ -1600,8 +1615,6 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     // Persistent sweep: amortizes the stage prologue over many tiles and
     // equalizes per-WG work (see gemm_bf16_nt_pers_k). Wins for short-K
     // shapes and small batched tiles; neutral for long K.
-    static const bool kPersOff =
-        getenv("NEZHA_GEMM_PERS") && atoi(getenv("NEZHA_GEMM_PERS")) == 0;
     int64_t per_b = (int64_t)nwg_m * nwg_n;
     int64_t total = per_b * batch;
     if (!kPersOff && total >= 512 && !args.out_f32) {
