@@ -548,35 +548,55 @@ __global__ void layernorm_dgb1_k(const T* __restrict__ dy, const T* __restrict__
   partials[(blockIdx.y * 2 + 1) * N + j] = sb;
 }
 // bf16 fast path: 8 columns per thread, ushort8 sweeps over dy and x
+// Same WG layout as colsum_stage1_vec_k (128 cols x 16 row-subchunks with a
+// fixed-order LDS fold): the one-thread-per-column-strip version was
+// latency-bound on BERT's N=768 (96 active lanes per chunk).
 __global__ void layernorm_dgb1_vec_k(const ushort8_t* __restrict__ dy8,
                                      const ushort8_t* __restrict__ x8,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ rstd,
                                      float* __restrict__ partials, int64_t R,
                                      int64_t N8, int64_t rows_per_chunk) {
-  int64_t j8 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-  if (j8 >= N8) return;
-  int64_t r0 = blockIdx.y * rows_per_chunk;
-  int64_t r1 = min(R, r0 + rows_per_chunk);
+  __shared__ float smg[16 * 128], smb[16 * 128];
+  const int jj = threadIdx.x & 15;
+  const int sub = threadIdx.x >> 4;
+  const int64_t j8 = (int64_t)blockIdx.x * 16 + jj;
+  const int64_t r0 = blockIdx.y * rows_per_chunk;
+  const int64_t r1 = min(R, r0 + rows_per_chunk);
   float sg[8] = {}, sb[8] = {};
-  for (int64_t i = r0; i < r1; ++i) {
-    ushort8_t dv = dy8[i * N8 + j8];
-    ushort8_t xv = x8[i * N8 + j8];
-    float mu = mean[i], rs = rstd[i];
+  if (j8 < N8) {
+    for (int64_t i = r0 + sub; i < r1; i += 16) {
+      ushort8_t dv = dy8[i * N8 + j8];
+      ushort8_t xv = x8[i * N8 + j8];
+      float mu = mean[i], rs = rstd[i];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      float dyv = b2f(((const uint16_t*)&dv)[u]);
-      float xh = (b2f(((const uint16_t*)&xv)[u]) - mu) * rs;
-      sg[u] += dyv * xh;
-      sb[u] += dyv;
+      for (int u = 0; u < 8; ++u) {
+        float dyv = b2f(((const uint16_t*)&dv)[u]);
+        float xh = (b2f(((const uint16_t*)&xv)[u]) - mu) * rs;
+        sg[u] += dyv * xh;
+        sb[u] += dyv;
+      }
     }
   }
-  float* og = partials + (blockIdx.y * 2 + 0) * N8 * 8 + j8 * 8;
-  float* ob = partials + (blockIdx.y * 2 + 1) * N8 * 8 + j8 * 8;
 #pragma unroll
   for (int u = 0; u < 8; ++u) {
-    og[u] = sg[u];
-    ob[u] = sb[u];
+    smg[sub * 128 + jj * 8 + u] = sg[u];
+    smb[sub * 128 + jj * 8 + u] = sb[u];
+  }
+  __syncthreads();
+  if (threadIdx.x < 128) {
+    const int col = threadIdx.x;
+    float vg = 0.f, vb = 0.f;
+#pragma unroll
+    for (int s_ = 0; s_ < 16; ++s_) {
+      vg += smg[s_ * 128 + col];
+      vb += smb[s_ * 128 + col];
+    }
+    int64_t gcol = (int64_t)blockIdx.x * 128 + col;
+    if (gcol < N8 * 8) {
+      partials[(blockIdx.y * 2 + 0) * N8 * 8 + gcol] = vg;
+      partials[(blockIdx.y * 2 + 1) * N8 * 8 + gcol] = vb;
+    }
   }
 }
 
@@ -586,13 +606,21 @@ __global__ void layernorm_dgb2_k(const float* __restrict__ partials,
                                  int64_t N, int chunks) {
   int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   if (j >= N) return;
-  float sg = 0.f, sb = 0.f;
-  for (int c = 0; c < chunks; ++c) {
-    sg += partials[(c * 2 + 0) * N + j];
-    sb += partials[(c * 2 + 1) * N + j];
+  // paired accumulators for load ILP (fold is latency-bound; fixed order)
+  float g0 = 0.f, g1 = 0.f, b0 = 0.f, b1 = 0.f;
+  int c = 0;
+  for (; c + 2 <= chunks; c += 2) {
+    g0 += partials[((int64_t)c * 2 + 0) * N + j];
+    b0 += partials[((int64_t)c * 2 + 1) * N + j];
+    g1 += partials[((int64_t)(c + 1) * 2 + 0) * N + j];
+    b1 += partials[((int64_t)(c + 1) * 2 + 1) * N + j];
   }
-  st(dgamma, j, sg);
-  st(dbeta, j, sb);
+  for (; c < chunks; ++c) {
+    g0 += partials[((int64_t)c * 2 + 0) * N + j];
+    b0 += partials[((int64_t)c * 2 + 1) * N + j];
+  }
+  st(dgamma, j, g0 + g1);
+  st(dbeta, j, b0 + b1);
 }
 void layernorm_bwd(const Tensor& dy, const Tensor& x, const Tensor& gamma,
                    const Tensor& mean, const Tensor& rstd, Tensor& dx,
@@ -613,10 +641,10 @@ void layernorm_bwd(const Tensor& dy, const Tensor& x, const Tensor& gamma,
         (const uint16_t*)gamma.data(), (const float*)mean.data(),
         (const float*)rstd.data(), (uint16_t*)dx.data(), R, N, beta_dx);
     if (N % 8 == 0) {
-      int64_t cb8 = (N / 8 + 255) / 256;
+      int64_t cb8 = (N + 127) / 128;
       int ch8 = (int)std::min<int64_t>(
-          (R + 63) / 64,
-          std::min<int64_t>(128, std::max<int64_t>(1, 1024 / cb8)));
+          (R + 15) / 16,
+          std::min<int64_t>(256, std::max<int64_t>(1, 1024 / cb8)));
       int64_t rpc8 = (R + ch8 - 1) / ch8;
       Tensor p8(x.ctx(), DType::F32, Shape{(int64_t)ch8 * 2, N});
       layernorm_dgb1_vec_k<<<dim3(cb8, ch8), b, 0, s>>>(
@@ -720,6 +748,38 @@ __global__ void softmax_bwd_wave_k(const T* __restrict__ dy,
   const T* dyr = dy + row * N;
   const T* yr = y + row * N;
   float dv[EPL], yv[EPL], dot = 0.f;
+  T* dxr = dx + row * N;
+  if constexpr (sizeof(T) == 2) {
+    if (N == 64 * EPL) {
+      // packed dword loads/stores: the scalar path issued 2-byte strided
+      // transactions and ran 3.5x off the traffic bound on the S=128 probs
+      const uint32_t* dw = (const uint32_t*)dyr + lane * (EPL / 2);
+      const uint32_t* yw = (const uint32_t*)yr + lane * (EPL / 2);
+      uint32_t dpk[EPL / 2], ypk[EPL / 2];
+#pragma unroll
+      for (int w = 0; w < EPL / 2; ++w) {
+        dpk[w] = dw[w];
+        ypk[w] = yw[w];
+      }
+#pragma unroll
+      for (int i = 0; i < EPL; ++i) {
+        uint16_t db = (uint16_t)(dpk[i / 2] >> (16 * (i & 1)));
+        uint16_t yb = (uint16_t)(ypk[i / 2] >> (16 * (i & 1)));
+        dv[i] = b2f(db);
+        yv[i] = b2f(yb);
+        dot += dv[i] * yv[i];
+      }
+      dot = wave_sum(dot);
+      uint32_t* ow = (uint32_t*)dxr + lane * (EPL / 2);
+#pragma unroll
+      for (int w = 0; w < EPL / 2; ++w) {
+        uint32_t lo = f2b(scale * yv[2 * w] * (dv[2 * w] - dot));
+        uint32_t hi = f2b(scale * yv[2 * w + 1] * (dv[2 * w + 1] - dot));
+        ow[w] = lo | (hi << 16);
+      }
+      return;
+    }
+  }
 #pragma unroll
   for (int i = 0; i < EPL; ++i) {
     int64_t j = lane * EPL + i;
@@ -728,7 +788,6 @@ __global__ void softmax_bwd_wave_k(const T* __restrict__ dy,
     dot += dv[i] * yv[i];
   }
   dot = wave_sum(dot);
-  T* dxr = dx + row * N;
 #pragma unroll
   for (int i = 0; i < EPL; ++i) {
     int64_t j = lane * EPL + i;
