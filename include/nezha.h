@@ -95,6 +95,11 @@ nz_status nz_op_attn_fused_fwd(nz_tensor* q, nz_tensor* k, nz_tensor* v,
                                nz_tensor* probs, nz_tensor* ctx_out,
                                int64_t batch_heads, int64_t S, int64_t dh,
                                float scale);
+nz_status nz_op_attn_fused_bwd(nz_tensor* q, nz_tensor* k, nz_tensor* v,
+                               nz_tensor* probs, nz_tensor* dctx, nz_tensor* dq,
+                               nz_tensor* dk, nz_tensor* dv,
+                               int64_t batch_heads, int64_t S, int64_t dh,
+                               float scale);
 nz_status nz_op_attn_merge(nz_tensor* ctx, nz_tensor* out, int64_t B, int64_t S,
                            int64_t heads, int64_t dh);
 

@@ -271,6 +271,17 @@ extern "C" nz_status nz_op_attn_fused_fwd(nz_tensor* q, nz_tensor* k,
                       dh, scale);
   NZ_API_END
 }
+extern "C" nz_status nz_op_attn_fused_bwd(nz_tensor* q, nz_tensor* k,
+                                          nz_tensor* v, nz_tensor* probs,
+                                          nz_tensor* dctx, nz_tensor* dq,
+                                          nz_tensor* dk, nz_tensor* dv,
+                                          int64_t batch_heads, int64_t S,
+                                          int64_t dh, float scale) {
+  NZ_API_BEGIN
+  ops::attn_fused_bwd(q->t, k->t, v->t, probs->t, dctx->t, dq->t, dk->t,
+                      dv->t, batch_heads, S, dh, scale);
+  NZ_API_END
+}
 extern "C" nz_status nz_op_attn_merge(nz_tensor* ctx, nz_tensor* out, int64_t B,
                                       int64_t S, int64_t heads, int64_t dh) {
   NZ_API_BEGIN

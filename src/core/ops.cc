@@ -121,6 +121,29 @@ void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
                S * S, S * dh, S * dh);
 }
 
+void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
+                    const Tensor& probs, const Tensor& dctx, Tensor& dq,
+                    Tensor& dk, Tensor& dv, int64_t batch_heads, int64_t S,
+                    int64_t dh, float scale) {
+  prof::OpScope nz_scope_("attn_fused_bwd", q.ctx());
+  if (q.device() == Device::HIP && hip::attn_fused_fwd_supported(S, dh)) {
+    hip::attn_fused_bwd(q, k, v, probs, dctx, dq, dk, dv, batch_heads, scale);
+    return;
+  }
+  // composed fallback (CPU oracle + non-contract shapes)
+  Tensor dprobs(q.ctx(), q.dtype(), Shape{batch_heads * S, S});
+  gemm_batched(dctx, v, dprobs, false, true, 1.0f, batch_heads, S, S, dh,
+               S * dh, S * dh, S * S);
+  gemm_batched(probs, dctx, dv, true, false, 1.0f, batch_heads, S, dh, S,
+               S * S, S * dh, S * dh);
+  Tensor dscores(q.ctx(), q.dtype(), Shape{batch_heads * S, S});
+  softmax_bwd(dprobs, probs, dscores, scale);
+  gemm_batched(dscores, k, dq, false, false, 1.0f, batch_heads, S, dh, S,
+               S * S, S * dh, S * dh);
+  gemm_batched(dscores, q, dk, true, false, 1.0f, batch_heads, S, dh, S,
+               S * S, S * dh, S * dh);
+}
+
 void attn_merge(const Tensor& ctx, Tensor& out, int64_t B, int64_t S,
                 int64_t heads, int64_t dh) {
   NZ_DISPATCH(attn_merge, ctx, ctx, out, B, S, heads, dh);

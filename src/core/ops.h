@@ -100,6 +100,15 @@ void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
                     Tensor& probs, Tensor& ctx_out, int64_t batch_heads,
                     int64_t S, int64_t dh, float scale);
 
+// Fused attention backward (K5): given saved probs and upstream dctx,
+// produces dq/dk/dv. Single kernel on HIP for the contract shape
+// (src/hip/attention.hip attn_bwd_fused_k); composed from K1 batched +
+// K4 softmax_bwd otherwise (and as the CPU oracle).
+void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
+                    const Tensor& probs, const Tensor& dctx, Tensor& dq,
+                    Tensor& dk, Tensor& dv, int64_t batch_heads, int64_t S,
+                    int64_t dh, float scale);
+
 // ctx [B,heads,S,dh] -> out [B, S, H]
 void attn_merge(const Tensor& ctx, Tensor& out, int64_t B, int64_t S,
                 int64_t heads, int64_t dh);

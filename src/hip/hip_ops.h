@@ -64,6 +64,9 @@ bool attn_fused_fwd_supported(int64_t S, int64_t dh);
 void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
                     Tensor& probs, Tensor& ctx_out, int64_t batch_heads,
                     float scale);
+void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
+                    const Tensor& probs, const Tensor& dout, Tensor& dq,
+                    Tensor& dk, Tensor& dv, int64_t batch_heads, float scale);
 }}
 
 // appended: colsum + attention layout ops (decls mirror core/ops.h)

@@ -169,20 +169,10 @@ VarPtr MultiHeadAttention::forward(Tape& tape, VarPtr x) {
     NZ_CHECK(dmerged.defined(), "attention backward without upstream grad");
     Tensor dctx(ctx_, pd, Shape{B, Hh, S, D});
     ops::attn_merge_bwd(dmerged, dctx, B, S, Hh, D);
-    Tensor dprobs(ctx_, pd, Shape{B * Hh * S, S});
-    ops::gemm_batched(dctx, v, dprobs, false, true, 1.0f, B * Hh, S, S, D,
-                      S * D, S * D, S * S);
-    Tensor dv(ctx_, pd, Shape{B, Hh, S, D});
-    ops::gemm_batched(probs, dctx, dv, true, false, 1.0f, B * Hh, S, D, S,
-                      S * S, S * D, S * D);
-    Tensor dscores(ctx_, pd, Shape{B * Hh * S, S});
-    ops::softmax_bwd(dprobs, probs, dscores, scale);
     Tensor dq(ctx_, pd, Shape{B, Hh, S, D});
-    ops::gemm_batched(dscores, k, dq, false, false, 1.0f, B * Hh, S, D, S,
-                      S * S, S * D, S * D);
     Tensor dk(ctx_, pd, Shape{B, Hh, S, D});
-    ops::gemm_batched(dscores, q, dk, true, false, 1.0f, B * Hh, S, D, S,
-                      S * S, S * D, S * D);
+    Tensor dv(ctx_, pd, Shape{B, Hh, S, D});
+    ops::attn_fused_bwd(q, k, v, probs, dctx, dq, dk, dv, B * Hh, S, D, scale);
     qkv->ensure_grad();
     NZ_CHECK(!qkv->grad_init, "qkv grad written twice");
     ops::attn_split_qkv_bwd(dq, dk, dv, qkv->grad, B, S, Hh, D);

@@ -303,6 +303,57 @@ def test_attn_fused_fwd(hip_ctx):
     check(ctx_out.numpy(), o_ref.numpy(), rtol=0.03)
 
 
+def test_attn_fused_bwd(hip_ctx):
+    """Fused attention backward vs a torch fp32 reference of the same math."""
+    import torch
+    BH, S, D = 24, 128, 64
+    scale = 1.0 / np.sqrt(D)
+    Q, K, V, DO = randb(BH, S, D), randb(BH, S, D), randb(BH, S, D), randb(BH, S, D)
+    q, k, v = (hip_ctx.from_numpy_bf16(Q), hip_ctx.from_numpy_bf16(K),
+               hip_ctx.from_numpy_bf16(V))
+    do = hip_ctx.from_numpy_bf16(DO)
+    probs = hip_ctx.tensor((BH * S, S), nz.BF16)
+    ctx_out = hip_ctx.tensor((BH, S, D), nz.BF16)
+    nz.attn_fused_fwd(q, k, v, probs, ctx_out, BH, S, D, scale)
+    dq = hip_ctx.tensor((BH, S, D), nz.BF16)
+    dk = hip_ctx.tensor((BH, S, D), nz.BF16)
+    dv = hip_ctx.tensor((BH, S, D), nz.BF16)
+    nz.attn_fused_bwd(q, k, v, probs, do, dq, dk, dv, BH, S, D, scale)
+    # reference: P taken from the engine (bf16), everything else fp32
+    P = torch.from_numpy(probs.numpy().reshape(BH, S, S))
+    dOt = torch.from_numpy(DO)
+    Vt, Kt, Qt = map(torch.from_numpy, (V, K, Q))
+    dP = dOt @ Vt.transpose(-1, -2)
+    dV_ref = P.transpose(-1, -2) @ dOt
+    rs = (dP * P).sum(-1, keepdim=True)
+    dS = scale * P * (dP - rs)
+    dQ_ref = dS @ Kt
+    dK_ref = dS.transpose(-1, -2) @ Qt
+    check(dv.numpy(), dV_ref.numpy(), rtol=0.03)
+    check(dq.numpy(), dQ_ref.numpy(), rtol=0.03)
+    check(dk.numpy(), dK_ref.numpy(), rtol=0.03)
+
+
+def test_attn_fused_bwd_determinism(hip_ctx):
+    BH, S, D = 48, 128, 64
+    q = hip_ctx.tensor((BH, S, D), nz.BF16).rand_normal(5)
+    k = hip_ctx.tensor((BH, S, D), nz.BF16).rand_normal(6)
+    v = hip_ctx.tensor((BH, S, D), nz.BF16).rand_normal(7)
+    do = hip_ctx.tensor((BH, S, D), nz.BF16).rand_normal(8)
+    probs = hip_ctx.tensor((BH * S, S), nz.BF16)
+    o = hip_ctx.tensor((BH, S, D), nz.BF16)
+    nz.attn_fused_fwd(q, k, v, probs, o, BH, S, D, 0.125)
+    outs = []
+    for _ in range(2):
+        dq = hip_ctx.tensor((BH, S, D), nz.BF16)
+        dk = hip_ctx.tensor((BH, S, D), nz.BF16)
+        dv = hip_ctx.tensor((BH, S, D), nz.BF16)
+        nz.attn_fused_bwd(q, k, v, probs, do, dq, dk, dv, BH, S, D, 0.125)
+        outs.append((dq.raw(), dk.raw(), dv.raw()))
+    for a, b in zip(outs[0], outs[1]):
+        np.testing.assert_array_equal(a, b)
+
+
 def test_attn_fused_determinism(hip_ctx):
     BH, S, D = 48, 128, 64
     q = hip_ctx.tensor((BH, S, D), nz.BF16).rand_normal(5)
