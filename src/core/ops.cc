@@ -67,6 +67,9 @@ void rand_normal(Tensor& t, uint64_t seed, float mean, float std) {
   NZ_DISPATCH(rand_normal, t, t, seed, mean, std);
 }
 void cast_copy(Tensor& dst, const Tensor& src) { NZ_DISPATCH(cast_copy, dst, dst, src); }
+void add_out(Tensor& out, const Tensor& x, const Tensor& y) {
+  NZ_DISPATCH(add_out, out, out, x, y);
+}
 void add_inplace(Tensor& dst, const Tensor& src, float alpha) {
   NZ_DISPATCH(add_inplace, dst, dst, src, alpha);
 }

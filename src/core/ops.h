@@ -62,6 +62,7 @@ void rand_uniform(Tensor& t, uint64_t seed, float lo, float hi);
 void rand_normal(Tensor& t, uint64_t seed, float mean, float std);
 void cast_copy(Tensor& dst, const Tensor& src);       // dtype conversion or plain copy
 void add_inplace(Tensor& dst, const Tensor& src, float alpha);  // dst += alpha*src
+void add_out(Tensor& out, const Tensor& x, const Tensor& y);     // out = x + y
 void scale_inplace(Tensor& dst, float alpha);
 void transpose2d(const Tensor& src, Tensor& dst);     // dst[j,i] = src[i,j]
 

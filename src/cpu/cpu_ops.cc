@@ -349,6 +349,28 @@ void cast_copy(Tensor& dst, const Tensor& src) {
   }
 }
 
+void add_out(Tensor& out, const Tensor& x, const Tensor& y) {
+  NZ_CHECK(out.numel() == x.numel() && x.numel() == y.numel(),
+           "add_out numel mismatch");
+  int64_t n = out.numel();
+  if (out.dtype() == DType::F32) {
+    float* o = (float*)out.data();
+    const float* a = (const float*)x.data();
+    const float* b = (const float*)y.data();
+    parallel_for(n, 8192, [&](int64_t lo, int64_t hi) {
+      for (int64_t i = lo; i < hi; ++i) o[i] = a[i] + b[i];
+    });
+  } else {
+    uint16_t* o = (uint16_t*)out.data();
+    const uint16_t* a = (const uint16_t*)x.data();
+    const uint16_t* b = (const uint16_t*)y.data();
+    parallel_for(n, 8192, [&](int64_t lo, int64_t hi) {
+      for (int64_t i = lo; i < hi; ++i)
+        o[i] = f32_to_bf16(bf16_to_f32(a[i]) + bf16_to_f32(b[i]));
+    });
+  }
+}
+
 void add_inplace(Tensor& dst, const Tensor& src, float alpha) {
   NZ_CHECK(dst.numel() == src.numel(), "add numel mismatch");
   int64_t n = dst.numel();

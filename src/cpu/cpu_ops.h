@@ -33,6 +33,7 @@ void rand_uniform(Tensor& t, uint64_t seed, float lo, float hi);
 void rand_normal(Tensor& t, uint64_t seed, float mean, float std);
 void cast_copy(Tensor& dst, const Tensor& src);
 void add_inplace(Tensor& dst, const Tensor& src, float alpha);
+void add_out(Tensor& out, const Tensor& x, const Tensor& y);
 void scale_inplace(Tensor& dst, float alpha);
 void transpose2d(const Tensor& src, Tensor& dst);
 void dropout_fwd(const Tensor& x, Tensor& y, Tensor& mask, float p, uint64_t seed);

@@ -145,6 +145,51 @@ void add_inplace(Tensor& dst, const Tensor& src, float alpha) {
   NZ_KERNEL_CHECK();
 }
 
+__global__ void add_out_b16_k(uint16_t* o, const uint16_t* x,
+                              const uint16_t* y, int64_t n) {
+  int64_t i4 = (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (; i4 + 3 < n; i4 += stride) {
+    ushort4 xv = *(const ushort4*)(x + i4);
+    ushort4 yv = *(const ushort4*)(y + i4);
+    ushort4 ov;
+    ov.x = f2b(b2f(xv.x) + b2f(yv.x));
+    ov.y = f2b(b2f(xv.y) + b2f(yv.y));
+    ov.z = f2b(b2f(xv.z) + b2f(yv.z));
+    ov.w = f2b(b2f(xv.w) + b2f(yv.w));
+    *(ushort4*)(o + i4) = ov;
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    for (int64_t i = n & ~int64_t(3); i < n; ++i)
+      o[i] = f2b(b2f(x[i]) + b2f(y[i]));
+}
+template <typename T>
+__global__ void add_out_k(T* o, const T* x, const T* y, int64_t n) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) st(o, i, ld(x, i) + ld(y, i));
+}
+
+// out = x + y (residual forward: the copy+add pair cost an extra full pass)
+void add_out(Tensor& out, const Tensor& x, const Tensor& y) {
+  NZ_CHECK(out.numel() == x.numel() && x.numel() == y.numel(),
+           "add_out numel mismatch");
+  NZ_CHECK(out.dtype() == x.dtype() && x.dtype() == y.dtype(),
+           "add_out dtype mismatch");
+  hipStream_t s = stream_of(out);
+  int64_t n = out.numel();
+  dim3 b(256);
+  if (out.dtype() == DType::BF16) {
+    add_out_b16_k<<<dim3(grid_1d((n + 3) / 4, 256)), b, 0, s>>>(
+        (uint16_t*)out.data(), (const uint16_t*)x.data(),
+        (const uint16_t*)y.data(), n);
+  } else {
+    add_out_k<float><<<dim3(grid_1d(n, 256)), b, 0, s>>>(
+        (float*)out.data(), (const float*)x.data(), (const float*)y.data(), n);
+  }
+  NZ_KERNEL_CHECK();
+}
+
 template <typename T>
 __global__ void scale_k(T* d, int64_t n, float a) {
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;

@@ -32,6 +32,7 @@ void rand_uniform(Tensor& t, uint64_t seed, float lo, float hi);
 void rand_normal(Tensor& t, uint64_t seed, float mean, float std);
 void cast_copy(Tensor& dst, const Tensor& src);
 void add_inplace(Tensor& dst, const Tensor& src, float alpha);
+void add_out(Tensor& out, const Tensor& x, const Tensor& y);
 void scale_inplace(Tensor& dst, float alpha);
 void transpose2d(const Tensor& src, Tensor& dst);
 // batched fast transpose (R,C % 64): dst[b][c][r] = src[b][r][c]

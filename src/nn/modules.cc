@@ -8,17 +8,6 @@ static DType param_dtype(Context* ctx) {
   return ctx->device() == Device::HIP ? DType::BF16 : DType::F32;
 }
 
-// Accumulate gradient tensor g into var's grad (first write = copy).
-static void accum_into(Var& v, const Tensor& g) {
-  if (!v.grad_init) {
-    v.ensure_grad();
-    ops::cast_copy(v.grad, g);
-    v.grad_init = true;
-  } else {
-    ops::add_inplace(v.grad, g, 1.0f);
-  }
-}
-
 // ---------------------------------------------------------------- Linear ---
 Linear::Linear(Context* ctx, std::string name, int64_t in, int64_t out,
                bool bias, bool gelu, uint64_t seed)
@@ -227,12 +216,23 @@ VarPtr Softmax::forward(Tape& tape, VarPtr x) {
 VarPtr residual_add(Tape& tape, VarPtr a, VarPtr b) {
   Context* ctx = a->value.ctx();
   auto out = make_var(Tensor(ctx, a->value.dtype(), a->value.shape()));
-  ops::cast_copy(out->value, a->value);
-  ops::add_inplace(out->value, b->value, 1.0f);
+  ops::add_out(out->value, a->value, b->value);
   tape.push([a, b, out]() {
     NZ_CHECK(out->grad.defined(), "residual backward without upstream grad");
-    accum_into(*a, out->grad);
-    accum_into(*b, out->grad);
+    // ALIAS out's grad into uninitialized inputs instead of copying: out is
+    // dead after this entry, and the replay order guarantees b's grad is
+    // fully consumed (start of the sublayer's reverse chain) before any
+    // accumulation into a's grad mutates the shared buffer.
+    auto attach = [&](Var& v) {
+      if (!v.grad_init) {
+        v.grad = out->grad;
+        v.grad_init = true;
+      } else {
+        ops::add_inplace(v.grad, out->grad, 1.0f);
+      }
+    };
+    attach(*b);
+    attach(*a);
   });
   return out;
 }
