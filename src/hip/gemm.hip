@@ -673,9 +673,13 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase_k(GemmArgs args,
   using namespace ph8;
   __shared__ __attribute__((aligned(16))) uint16_t smem[2 * BUF_E];
 
-  const uint16_t* __restrict__ A = args.A;
-  const uint16_t* __restrict__ B = args.B;
-  uint16_t* C = args.C;
+  // grid.y = split-K slice (sA/sB/sC element strides; out_f32 = raw f32
+  // partial-slab mode, reduced by splitk_reduce_k)
+  const int64_t batch = blockIdx.y;
+  const uint16_t* __restrict__ A = args.A + batch * args.sA;
+  const uint16_t* __restrict__ B = args.B + batch * args.sB;
+  uint16_t* C = args.C + batch * args.sC;
+  float* out32 = args.out_f32 ? args.out_f32 + batch * args.sC : nullptr;
   uint16_t* save_pre = args.save_pre;
 
   int nwg = gridDim.x;
@@ -816,6 +820,28 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase_k(GemmArgs args,
   }
 #undef PH8_WAIT
 #undef PH8_BAR
+
+  if (out32) {
+    // split-K partial mode: scatter raw f32. 128 stores/lane, but the tile
+    // did >= 6 deep-K iterations so the issue cost is amortized.
+#pragma unroll
+    for (int ha = 0; ha < 2; ++ha)
+#pragma unroll
+      for (int hb = 0; hb < 2; ++hb)
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+            for (int rI = 0; rI < 4; ++rI) {
+              int64_t grow = m0 + ha * 128 + wr * 64 + fm * 16 +
+                             (lane >> 4) * 4 + rI;
+              int64_t gcol = n0 + hb * 128 + wc * 32 + fn * 16 + (lane & 15);
+              if (grow < args.M && gcol < args.N)
+                out32[grow * args.N + gcol] = acc[ha][hb][fm][fn][rI];
+            }
+    return;
+  }
 
   // Epilogue: stage the 256x256 bf16 C tile through LDS (exactly reuses the
   // 128 KB smem, DMA already drained), then stream it out with 16-B vector
@@ -1571,6 +1597,38 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     // fused reduce applies the real epilogue.
     if (batch == 1 && nwg < 224 && args.K >= 4096 && !args.out_f32 &&
         g_splitk_ctx) {
+      // Prefer 256^2 tiles per slice (8-phase kernel): half the staging
+      // bytes per flop of the 128^2 kernel, and the >=768-deep per-slice K
+      // amortizes the 8-phase ramp. Measured on the BERT wgrad panels:
+      // qkv 464 -> ~800 TF equivalent (see profiles/).
+      int nwg_m8 = (int)((args.M + 255) / 256);
+      int nwg_n8 = (int)((args.N + 255) / 256);
+      int64_t nwg8 = (int64_t)nwg_m8 * nwg_n8;
+      if (kNtVariant >= 2 && args.K % 256 == 0 && args.K / 2 >= 768 &&
+          nwg8 >= 9) {
+        int sl = 1;
+        while (sl < 16 && nwg8 * sl * 2 <= 288 &&
+               args.K % ((int64_t)sl * 256) == 0 && args.K / (sl * 2) >= 768)
+          sl *= 2;
+        if (sl > 1 && nwg8 * sl >= 96) {
+          int64_t ksl = args.K / sl;
+          Tensor parts(g_splitk_ctx, DType::F32,
+                       Shape{(int64_t)sl, args.M, args.N});
+          GemmArgs pa = args;
+          pa.K = ksl;
+          pa.sA = ksl;
+          pa.sB = ksl;
+          pa.sC = args.M * args.N;
+          pa.out_f32 = (float*)parts.data();
+          dim3 g8((uint32_t)nwg8, sl);
+          gemm_bf16_nt_8phase_k<<<g8, dim3(512), 0, s>>>(pa, nwg_n8);
+          NZ_KERNEL_CHECK();
+          splitk_reduce_k<<<dim3(grid_1d(args.M * args.N, 256)), dim3(256), 0,
+                            s>>>((const float*)parts.data(), args, sl);
+          NZ_KERNEL_CHECK();
+          return;
+        }
+      }
       // Power-of-two slice count: double until the grid reaches ~2 WGs/CU
       // (the 128^2 kernel's occupancy), keeping K/slices a >=2048 multiple
       // of 64. A non-power-of-two count (the old (448+nwg-1)/nwg) usually
