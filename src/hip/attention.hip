@@ -242,35 +242,37 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
     att_glds16(dog + row * AD + kk, dOs + e);
     att_glds16(vg + row * AD + kk, Vs + e);
   }
-  // register loads + transposed scatters (issued while the DMA flies)
+  // register loads + transposed scatters (issued while the DMA flies);
+  // pairs of loads per scatter group keep the live staging registers small
+  // (the all-loads-then-all-scatters version allocated 316 VGPRs -> 1
+  // wave/SIMD; this kernel wants 2)
   {
-    short8 pr[8], dor[4], kr[4], qr[4];
 #pragma unroll
-    for (int c = 0; c < 8; ++c)
-      pr[c] = *(const short8*)(pg + e0 + c * 2048);
+    for (int c = 0; c < 8; c += 2) {
+      short8 p0 = *(const short8*)(pg + e0 + c * 2048);
+      short8 p1 = *(const short8*)(pg + e0 + (c + 1) * 2048);
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      int e = e0 + c * 2048;
-      dor[c] = *(const short8*)(dog + e);
-      kr[c] = *(const short8*)(kg + e);
-      qr[c] = *(const short8*)(qg + e);
-    }
+      for (int u = 0; u < 2; ++u) {
+        int e = e0 + (c + u) * 2048;  // element in P [s1][s2]
+        int s1 = e >> 7, s2 = e & 127;
+        const short8& pv = u ? p1 : p0;
 #pragma unroll
-    for (int c = 0; c < 8; ++c) {
-      int e = e0 + c * 2048;          // element in P [s1][s2]
-      int s1 = e >> 7, s2 = e & 127;
-#pragma unroll
-      for (int i = 0; i < 8; ++i) Pt[(s2 + i) * PSTRIDE + s1] = (uint16_t)pr[c][i];
+        for (int i = 0; i < 8; ++i)
+          Pt[(s2 + i) * PSTRIDE + s1] = (uint16_t)pv[i];
+      }
     }
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       int e = e0 + c * 2048;          // element in [s][d]
+      short8 dor = *(const short8*)(dog + e);
+      short8 kr = *(const short8*)(kg + e);
+      short8 qr = *(const short8*)(qg + e);
       int s_ = e >> 6, d = e & 63;
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        dOt[(d + i) * PSTRIDE + s_] = (uint16_t)dor[c][i];
-        Kt[(d + i) * PSTRIDE + s_] = (uint16_t)kr[c][i];
-        Qt[(d + i) * PSTRIDE + s_] = (uint16_t)qr[c][i];
+        dOt[(d + i) * PSTRIDE + s_] = (uint16_t)dor[i];
+        Kt[(d + i) * PSTRIDE + s_] = (uint16_t)kr[i];
+        Qt[(d + i) * PSTRIDE + s_] = (uint16_t)qr[i];
       }
     }
   }
@@ -302,8 +304,8 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
             a_frag[fm], b_frag[fn], acc[fm][fn], 0, 0, 0);
   }
 
-  // ---- P2: read P into regs; rowsum partials of dP o P --------------------
-  uint16_t preg[4][4][4];
+  // ---- P2: rowsum partials of dP o P (P read from LDS; each (row,col)
+  // element of Pt is read and later overwritten only by its owning lane) ----
 #pragma unroll
   for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
@@ -313,9 +315,7 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
 #pragma unroll
       for (int fn = 0; fn < 4; ++fn) {
         int col = wc * 64 + fn * 16 + (lane & 15);
-        uint16_t pb = Pt[col * PSTRIDE + row];
-        preg[fm][fn][rI] = pb;
-        part += acc[fm][fn][rI] * b2f(pb);
+        part += acc[fm][fn][rI] * b2f(Pt[col * PSTRIDE + row]);
       }
       psum[row * 32 + wc * 16 + (lane & 15)] = part;
     }
@@ -366,7 +366,8 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
   }
   __syncthreads();
 
-  // ---- P4: dS = scale * P o (dP - rs); write dSn (swz) + dSt (transposed) -
+  // ---- P4: dS = scale * P o (dP - rs); write dSn (swz) + dSt (transposed,
+  // overwriting Pt in place: read P just before the store, same lane) ------
 #pragma unroll
   for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
@@ -376,8 +377,8 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
 #pragma unroll
       for (int fn = 0; fn < 4; ++fn) {
         int col = wc * 64 + fn * 16 + (lane & 15);
-        uint16_t ds = f2b(scale * b2f(preg[fm][fn][rI]) *
-                          (acc[fm][fn][rI] - r));
+        float pb = b2f(Pt[col * PSTRIDE + row]);
+        uint16_t ds = f2b(scale * pb * (acc[fm][fn][rI] - r));
         dSn[row * AS + (col ^ ((row & 7) << 3))] = ds;
         dSt[col * PSTRIDE + row] = ds;
       }
