@@ -100,6 +100,13 @@ nz_status nz_op_attn_fused_bwd(nz_tensor* q, nz_tensor* k, nz_tensor* v,
                                nz_tensor* dk, nz_tensor* dv,
                                int64_t batch_heads, int64_t S, int64_t dh,
                                float scale);
+nz_status nz_op_attn_fwd_packed(nz_tensor* qkv, nz_tensor* probs, nz_tensor* out,
+                                int64_t B, int64_t S, int64_t heads, int64_t dh,
+                                float scale);
+nz_status nz_op_attn_bwd_packed(nz_tensor* qkv, nz_tensor* probs,
+                                nz_tensor* dmerged, nz_tensor* dqkv, int64_t B,
+                                int64_t S, int64_t heads, int64_t dh,
+                                float scale);
 nz_status nz_op_attn_merge(nz_tensor* ctx, nz_tensor* out, int64_t B, int64_t S,
                            int64_t heads, int64_t dh);
 

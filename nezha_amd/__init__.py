@@ -88,6 +88,8 @@ _lib.nz_op_l2norm_sq.argtypes = [ctypes.c_void_p] * 2
 _lib.nz_op_attn_split_qkv.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 4
 _lib.nz_op_attn_fused_fwd.argtypes = [ctypes.c_void_p] * 5 + [ctypes.c_int64] * 3 + [ctypes.c_float]
 _lib.nz_op_attn_fused_bwd.argtypes = [ctypes.c_void_p] * 8 + [ctypes.c_int64] * 3 + [ctypes.c_float]
+_lib.nz_op_attn_fwd_packed.argtypes = [ctypes.c_void_p] * 3 + [ctypes.c_int64] * 4 + [ctypes.c_float]
+_lib.nz_op_attn_bwd_packed.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int64] * 4 + [ctypes.c_float]
 _lib.nz_op_attn_merge.argtypes = [ctypes.c_void_p] * 2 + [ctypes.c_int64] * 4
 _lib.nz_comm_uid_bytes.restype = ctypes.c_int
 _lib.nz_comm_get_uid.argtypes = [ctypes.c_void_p]
@@ -371,6 +373,16 @@ def attn_fused_bwd(q, k, v, probs, dctx, dq, dk, dv, batch_heads, S, dh, scale):
     _ck(_lib.nz_op_attn_fused_bwd(_h(q), _h(k), _h(v), _h(probs), _h(dctx),
                                   _h(dq), _h(dk), _h(dv), batch_heads, S, dh,
                                   scale))
+
+
+def attn_fwd_packed(qkv, probs, out, B, S, heads, dh, scale):
+    _ck(_lib.nz_op_attn_fwd_packed(_h(qkv), _h(probs), _h(out), B, S, heads,
+                                   dh, scale))
+
+
+def attn_bwd_packed(qkv, probs, dmerged, dqkv, B, S, heads, dh, scale):
+    _ck(_lib.nz_op_attn_bwd_packed(_h(qkv), _h(probs), _h(dmerged), _h(dqkv),
+                                   B, S, heads, dh, scale))
 
 
 def attn_merge(ctx_t, out, B, S, heads, dh):

@@ -282,6 +282,23 @@ extern "C" nz_status nz_op_attn_fused_bwd(nz_tensor* q, nz_tensor* k,
                       dv->t, batch_heads, S, dh, scale);
   NZ_API_END
 }
+extern "C" nz_status nz_op_attn_fwd_packed(nz_tensor* qkv, nz_tensor* probs,
+                                           nz_tensor* out, int64_t B, int64_t S,
+                                           int64_t heads, int64_t dh,
+                                           float scale) {
+  NZ_API_BEGIN
+  ops::attn_fwd_packed(qkv->t, probs->t, out->t, B, S, heads, dh, scale);
+  NZ_API_END
+}
+extern "C" nz_status nz_op_attn_bwd_packed(nz_tensor* qkv, nz_tensor* probs,
+                                           nz_tensor* dmerged, nz_tensor* dqkv,
+                                           int64_t B, int64_t S, int64_t heads,
+                                           int64_t dh, float scale) {
+  NZ_API_BEGIN
+  ops::attn_bwd_packed(qkv->t, probs->t, dmerged->t, dqkv->t, B, S, heads, dh,
+                       scale);
+  NZ_API_END
+}
 extern "C" nz_status nz_op_attn_merge(nz_tensor* ctx, nz_tensor* out, int64_t B,
                                       int64_t S, int64_t heads, int64_t dh) {
   NZ_API_BEGIN

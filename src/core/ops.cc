@@ -147,6 +147,44 @@ void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
                S * S, S * dh, S * dh);
 }
 
+void attn_fwd_packed(const Tensor& qkv, Tensor& probs, Tensor& out, int64_t B,
+                     int64_t S, int64_t heads, int64_t dh, float scale) {
+  prof::OpScope nz_scope_("attn_fwd_packed", qkv.ctx());
+  if (qkv.device() == Device::HIP && hip::attn_fused_fwd_supported(S, dh)) {
+    hip::attn_fused_fwd_packed(qkv, probs, out, B, S, heads, dh, scale);
+    return;
+  }
+  Tensor q(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});
+  Tensor k(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});
+  Tensor v(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});
+  attn_split_qkv(qkv, q, k, v, B, S, heads, dh);
+  Tensor ctxt(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});
+  attn_fused_fwd(q, k, v, probs, ctxt, B * heads, S, dh, scale);
+  attn_merge(ctxt, out, B, S, heads, dh);
+}
+
+void attn_bwd_packed(const Tensor& qkv, const Tensor& probs,
+                     const Tensor& dmerged, Tensor& dqkv, int64_t B, int64_t S,
+                     int64_t heads, int64_t dh, float scale) {
+  prof::OpScope nz_scope_("attn_bwd_packed", qkv.ctx());
+  if (qkv.device() == Device::HIP && hip::attn_fused_fwd_supported(S, dh)) {
+    hip::attn_fused_bwd_packed(qkv, probs, dmerged, dqkv, B, S, heads, dh,
+                               scale);
+    return;
+  }
+  Tensor q(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});
+  Tensor k(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});
+  Tensor v(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});
+  attn_split_qkv(qkv, q, k, v, B, S, heads, dh);
+  Tensor dctx(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});
+  attn_merge_bwd(dmerged, dctx, B, S, heads, dh);
+  Tensor dq(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});
+  Tensor dk(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});
+  Tensor dv(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});
+  attn_fused_bwd(q, k, v, probs, dctx, dq, dk, dv, B * heads, S, dh, scale);
+  attn_split_qkv_bwd(dq, dk, dv, dqkv, B, S, heads, dh);
+}
+
 void attn_merge(const Tensor& ctx, Tensor& out, int64_t B, int64_t S,
                 int64_t heads, int64_t dh) {
   NZ_DISPATCH(attn_merge, ctx, ctx, out, B, S, heads, dh);

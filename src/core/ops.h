@@ -110,6 +110,16 @@ void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
                     Tensor& dk, Tensor& dv, int64_t batch_heads, int64_t S,
                     int64_t dh, float scale);
 
+// Packed-layout attention (the module's hot path): consumes the qkv
+// projection [B, S, 3H] and produces merged [B, S, H] directly — on HIP at
+// the contract shape the fused kernels absorb the attn_split / attn_merge
+// layout passes; elsewhere this composes split + fused + merge.
+void attn_fwd_packed(const Tensor& qkv, Tensor& probs, Tensor& out, int64_t B,
+                     int64_t S, int64_t heads, int64_t dh, float scale);
+void attn_bwd_packed(const Tensor& qkv, const Tensor& probs,
+                     const Tensor& dmerged, Tensor& dqkv, int64_t B, int64_t S,
+                     int64_t heads, int64_t dh, float scale);
+
 // ctx [B,heads,S,dh] -> out [B, S, H]
 void attn_merge(const Tensor& ctx, Tensor& out, int64_t B, int64_t S,
                 int64_t heads, int64_t dh);

@@ -36,10 +36,16 @@ __device__ __forceinline__ void att_glds16(const uint16_t* gsrc,
       (__attribute__((address_space(3))) uint32_t*)lds_dst, 16, 0, 0);
 }
 
+// Generalized input/output addressing so the kernel can read q/k/v either
+// from per-head-contiguous [BH, S, dh] tensors (qr=64, qh=0, qb=S*64) or
+// straight from the packed qkv projection [B, S, 3H] (qr=3H, qh=64,
+// qb=S*3H, with k/v base pointers offset by H / 2H) — the latter removes
+// the attn_split/attn_merge layout passes entirely. Same for `out`.
 __launch_bounds__(256) __global__ void attn_fwd_fused_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ probs,
-    uint16_t* __restrict__ out, float scale) {
+    uint16_t* __restrict__ out, float scale, int heads, int64_t qb,
+    int64_t qh, int64_t qr, int64_t ob, int64_t oh, int64_t orr) {
   // LDS: [P 128*136 | VT 64*136]; Q and K overlay P's first 32 KB (the P
   // region is 34 KB — VT must live BEYOND it, clobbering it was a bug).
   __shared__ __attribute__((aligned(16))) uint16_t smem[(AS + AD) * PSTRIDE];
@@ -49,11 +55,12 @@ __launch_bounds__(256) __global__ void attn_fwd_fused_k(
   uint16_t* VTs = smem + AS * PSTRIDE;
 
   const int64_t bh = blockIdx.x;            // batch*heads index
-  const uint16_t* qg = q + bh * AS * AD;
-  const uint16_t* kg = k + bh * AS * AD;
-  const uint16_t* vg = v + bh * AS * AD;
+  const int64_t b_ = bh / heads, h_ = bh % heads;
+  const uint16_t* qg = q + b_ * qb + h_ * qh;
+  const uint16_t* kg = k + b_ * qb + h_ * qh;
+  const uint16_t* vg = v + b_ * qb + h_ * qh;
   uint16_t* pg = probs + bh * AS * AS;
-  uint16_t* og = out + bh * AS * AD;
+  uint16_t* og = out + b_ * ob + h_ * oh;
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -65,14 +72,14 @@ __launch_bounds__(256) __global__ void attn_fwd_fused_k(
     int e = e0 + c * 2048;
     int row = e >> 6;
     int kk = (e & 63) ^ ((row & 7) << 3);
-    att_glds16(qg + row * AD + kk, Qs + e);
-    att_glds16(kg + row * AD + kk, Ks + e);
+    att_glds16(qg + row * qr + kk, Qs + e);
+    att_glds16(kg + row * qr + kk, Ks + e);
   }
   short8 vreg[4];
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
     int e = e0 + c * 2048;                  // element in [s][d]
-    vreg[c] = *(const short8*)(vg + e);
+    vreg[c] = *(const short8*)(vg + (e >> 6) * qr + (e & 63));
   }
   // scatter V^T while the DMA flies: (s,d..d+7) -> VT[d+i][s]
 #pragma unroll
@@ -178,7 +185,7 @@ __launch_bounds__(256) __global__ void attn_fwd_fused_k(
         for (int rI = 0; rI < 4; ++rI) {
           int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
           int col = wc * 32 + fn * 16 + (lane & 15);
-          og[row * AD + col] = f2b(acc[fm][fn][rI]);
+          og[row * orr + col] = f2b(acc[fm][fn][rI]);
         }
   }
 }
@@ -205,7 +212,9 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const uint16_t* __restrict__ probs,
     const uint16_t* __restrict__ dout, uint16_t* __restrict__ dq,
-    uint16_t* __restrict__ dk, uint16_t* __restrict__ dv, float scale) {
+    uint16_t* __restrict__ dk, uint16_t* __restrict__ dv, float scale,
+    int heads, int64_t qb, int64_t qh, int64_t qr, int64_t db, int64_t dhh,
+    int64_t dr, int64_t gb, int64_t gh, int64_t gr) {
   __shared__ __attribute__((aligned(16))) uint16_t smem[68352];
   uint16_t* dOs = smem;                    // [128][64] swz
   uint16_t* Vs = smem + AS * AD;           // [128][64] swz
@@ -219,14 +228,15 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
   float* rs = psum + AS * 32;                 // [128]
 
   const int64_t bh = blockIdx.x;
-  const uint16_t* qg = q + bh * AS * AD;
-  const uint16_t* kg = k + bh * AS * AD;
-  const uint16_t* vg = v + bh * AS * AD;
+  const int64_t b_ = bh / heads, h_ = bh % heads;
+  const uint16_t* qg = q + b_ * qb + h_ * qh;
+  const uint16_t* kg = k + b_ * qb + h_ * qh;
+  const uint16_t* vg = v + b_ * qb + h_ * qh;
   const uint16_t* pg = probs + bh * AS * AS;
-  const uint16_t* dog = dout + bh * AS * AD;
-  uint16_t* dqg = dq + bh * AS * AD;
-  uint16_t* dkg = dk + bh * AS * AD;
-  uint16_t* dvg = dv + bh * AS * AD;
+  const uint16_t* dog = dout + b_ * db + h_ * dhh;
+  uint16_t* dqg = dq + b_ * gb + h_ * gh;
+  uint16_t* dkg = dk + b_ * gb + h_ * gh;
+  uint16_t* dvg = dv + b_ * gb + h_ * gh;
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -239,8 +249,8 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
     int e = e0 + c * 2048;
     int row = e >> 6;
     int kk = (e & 63) ^ ((row & 7) << 3);
-    att_glds16(dog + row * AD + kk, dOs + e);
-    att_glds16(vg + row * AD + kk, Vs + e);
+    att_glds16(dog + row * dr + kk, dOs + e);
+    att_glds16(vg + row * qr + kk, Vs + e);
   }
   // register loads + transposed scatters (issued while the DMA flies);
   // pairs of loads per scatter group keep the live staging registers small
@@ -264,15 +274,15 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       int e = e0 + c * 2048;          // element in [s][d]
-      short8 dor = *(const short8*)(dog + e);
-      short8 kr = *(const short8*)(kg + e);
-      short8 qr = *(const short8*)(qg + e);
       int s_ = e >> 6, d = e & 63;
+      short8 dov = *(const short8*)(dog + s_ * dr + d);
+      short8 kv = *(const short8*)(kg + s_ * qr + d);
+      short8 qv = *(const short8*)(qg + s_ * qr + d);
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        dOt[(d + i) * PSTRIDE + s_] = (uint16_t)dor[i];
-        Kt[(d + i) * PSTRIDE + s_] = (uint16_t)kr[i];
-        Qt[(d + i) * PSTRIDE + s_] = (uint16_t)qr[i];
+        dOt[(d + i) * PSTRIDE + s_] = (uint16_t)dov[i];
+        Kt[(d + i) * PSTRIDE + s_] = (uint16_t)kv[i];
+        Qt[(d + i) * PSTRIDE + s_] = (uint16_t)qv[i];
       }
     }
   }
@@ -352,7 +362,7 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
         for (int rI = 0; rI < 4; ++rI) {
           int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
           int col = wc * 32 + fn * 16 + (lane & 15);
-          dvg[row * AD + col] = f2b(acc2[fm][fn][rI]);
+          dvg[row * gr + col] = f2b(acc2[fm][fn][rI]);
         }
   }
   __syncthreads();  // psum complete (and Pt reads done -> dSt may overlay)
@@ -418,7 +428,7 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
         for (int rI = 0; rI < 4; ++rI) {
           int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
           int col = wc * 32 + fn * 16 + (lane & 15);
-          dqg[row * AD + col] = f2b(acc2[fm][fn][rI]);
+          dqg[row * gr + col] = f2b(acc2[fm][fn][rI]);
         }
   }
 
@@ -454,7 +464,7 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
         for (int rI = 0; rI < 4; ++rI) {
           int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
           int col = wc * 32 + fn * 16 + (lane & 15);
-          dkg[row * AD + col] = f2b(acc2[fm][fn][rI]);
+          dkg[row * gr + col] = f2b(acc2[fm][fn][rI]);
         }
   }
 }
@@ -474,7 +484,24 @@ void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
   attn_fwd_fused_k<<<dim3((uint32_t)batch_heads), dim3(256), 0, s>>>(
       (const uint16_t*)q.data(), (const uint16_t*)k.data(),
       (const uint16_t*)v.data(), (uint16_t*)probs.data(),
-      (uint16_t*)ctx_out.data(), scale);
+      (uint16_t*)ctx_out.data(), scale, 1, (int64_t)AS * AD, 0, AD,
+      (int64_t)AS * AD, 0, AD);
+  NZ_KERNEL_CHECK();
+}
+
+// Packed layouts: qkv [B, S, 3H] (q|k|v within each row), out [B, S, H] —
+// the fused kernel absorbs the attn_split / attn_merge passes.
+void attn_fused_fwd_packed(const Tensor& qkv, Tensor& probs, Tensor& out,
+                           int64_t B, int64_t S, int64_t heads, int64_t dh,
+                           float scale) {
+  NZ_CHECK(qkv.dtype() == DType::BF16, "fused attention: bf16 required");
+  hipStream_t s = stream_of(qkv);
+  int64_t H = heads * dh;
+  const uint16_t* base = (const uint16_t*)qkv.data();
+  attn_fwd_fused_k<<<dim3((uint32_t)(B * heads)), dim3(256), 0, s>>>(
+      base, base + H, base + 2 * H, (uint16_t*)probs.data(),
+      (uint16_t*)out.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
+      S * H, dh, H);
   NZ_KERNEL_CHECK();
 }
 
@@ -484,11 +511,31 @@ void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
                     Tensor& dk, Tensor& dv, int64_t batch_heads, float scale) {
   NZ_CHECK(q.dtype() == DType::BF16, "fused attention bwd: bf16 required");
   hipStream_t s = stream_of(q);
+  const int64_t sh = (int64_t)AS * AD;
   attn_bwd_fused_k<<<dim3((uint32_t)batch_heads), dim3(256), 0, s>>>(
       (const uint16_t*)q.data(), (const uint16_t*)k.data(),
       (const uint16_t*)v.data(), (const uint16_t*)probs.data(),
       (const uint16_t*)dout.data(), (uint16_t*)dq.data(),
-      (uint16_t*)dk.data(), (uint16_t*)dv.data(), scale);
+      (uint16_t*)dk.data(), (uint16_t*)dv.data(), scale, 1, sh, 0, AD, sh, 0,
+      AD, sh, 0, AD);
+  NZ_KERNEL_CHECK();
+}
+
+// Packed: qkv [B,S,3H], dmerged [B,S,H], dqkv [B,S,3H] (fully covered by
+// the B*heads workgroups: head h writes columns h*dh..h*dh+dh of each of
+// the q|k|v thirds).
+void attn_fused_bwd_packed(const Tensor& qkv, const Tensor& probs,
+                           const Tensor& dmerged, Tensor& dqkv, int64_t B,
+                           int64_t S, int64_t heads, int64_t dh, float scale) {
+  NZ_CHECK(qkv.dtype() == DType::BF16, "fused attention bwd: bf16 required");
+  hipStream_t s = stream_of(qkv);
+  int64_t H = heads * dh;
+  const uint16_t* base = (const uint16_t*)qkv.data();
+  uint16_t* gbase = (uint16_t*)dqkv.data();
+  attn_bwd_fused_k<<<dim3((uint32_t)(B * heads)), dim3(256), 0, s>>>(
+      base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
+      (const uint16_t*)dmerged.data(), gbase, gbase + H, gbase + 2 * H, scale,
+      (int)heads, S * 3 * H, dh, 3 * H, S * H, dh, H, S * 3 * H, dh, 3 * H);
   NZ_KERNEL_CHECK();
 }
 

@@ -68,6 +68,12 @@ void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
 void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
                     const Tensor& probs, const Tensor& dout, Tensor& dq,
                     Tensor& dk, Tensor& dv, int64_t batch_heads, float scale);
+void attn_fused_fwd_packed(const Tensor& qkv, Tensor& probs, Tensor& out,
+                           int64_t B, int64_t S, int64_t heads, int64_t dh,
+                           float scale);
+void attn_fused_bwd_packed(const Tensor& qkv, const Tensor& probs,
+                           const Tensor& dmerged, Tensor& dqkv, int64_t B,
+                           int64_t S, int64_t heads, int64_t dh, float scale);
 }}
 
 // appended: colsum + attention layout ops (decls mirror core/ops.h)

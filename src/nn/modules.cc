@@ -144,27 +144,17 @@ VarPtr MultiHeadAttention::forward(Tape& tape, VarPtr x) {
   float scale = 1.0f / std::sqrt(float(D));
   VarPtr qkv = qkv_proj.forward(tape, x);  // [B*S, 3H]
 
-  Tensor q(ctx_, pd, Shape{B, Hh, S, D}), k(ctx_, pd, Shape{B, Hh, S, D}),
-      v(ctx_, pd, Shape{B, Hh, S, D});
-  ops::attn_split_qkv(qkv->value, q, k, v, B, S, Hh, D);
   Tensor probs(ctx_, pd, Shape{B * Hh * S, S});
-  Tensor ctxt(ctx_, pd, Shape{B, Hh, S, D});
-  ops::attn_fused_fwd(q, k, v, probs, ctxt, B * Hh, S, D, scale);
   auto merged = make_var(Tensor(ctx_, pd, Shape{B * S, hidden}));
-  ops::attn_merge(ctxt, merged->value, B, S, Hh, D);
+  ops::attn_fwd_packed(qkv->value, probs, merged->value, B, S, Hh, D, scale);
 
-  tape.push([this, qkv, merged, q, k, v, probs, pd, B, S, Hh, D, scale]() {
+  tape.push([qkv, merged, probs, B, S, Hh, D, scale]() {
     Tensor& dmerged = merged->grad;
     NZ_CHECK(dmerged.defined(), "attention backward without upstream grad");
-    Tensor dctx(ctx_, pd, Shape{B, Hh, S, D});
-    ops::attn_merge_bwd(dmerged, dctx, B, S, Hh, D);
-    Tensor dq(ctx_, pd, Shape{B, Hh, S, D});
-    Tensor dk(ctx_, pd, Shape{B, Hh, S, D});
-    Tensor dv(ctx_, pd, Shape{B, Hh, S, D});
-    ops::attn_fused_bwd(q, k, v, probs, dctx, dq, dk, dv, B * Hh, S, D, scale);
     qkv->ensure_grad();
     NZ_CHECK(!qkv->grad_init, "qkv grad written twice");
-    ops::attn_split_qkv_bwd(dq, dk, dv, qkv->grad, B, S, Hh, D);
+    ops::attn_bwd_packed(qkv->value, probs, dmerged, qkv->grad, B, S, Hh, D,
+                         scale);
     qkv->grad_init = true;
   });
   return out_proj.forward(tape, merged);

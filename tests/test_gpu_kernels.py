@@ -354,6 +354,43 @@ def test_attn_fused_bwd_determinism(hip_ctx):
         np.testing.assert_array_equal(a, b)
 
 
+def test_attn_packed_matches_composed(hip_ctx):
+    """Packed-layout fused attention (reads qkv [B,S,3H], writes merged
+    [B,S,H] / dqkv directly) vs the split->fused->merge composition."""
+    B, S, H, D = 4, 128, 6, 64
+    QKV = randb(B * S, 3 * H * D)
+    DM = randb(B * S, H * D)
+    scale = 1.0 / np.sqrt(D)
+    qkv = hip_ctx.from_numpy_bf16(QKV)
+    dm = hip_ctx.from_numpy_bf16(DM)
+    # packed path
+    probs_p = hip_ctx.tensor((B * H * S, S), nz.BF16)
+    out_p = hip_ctx.tensor((B * S, H * D), nz.BF16)
+    nz.attn_fwd_packed(qkv, probs_p, out_p, B, S, H, D, scale)
+    dqkv_p = hip_ctx.tensor((B * S, 3 * H * D), nz.BF16)
+    nz.attn_bwd_packed(qkv, probs_p, dm, dqkv_p, B, S, H, D, scale)
+    # composed path
+    q = hip_ctx.tensor((B, H, S, D), nz.BF16)
+    k = hip_ctx.tensor((B, H, S, D), nz.BF16)
+    v = hip_ctx.tensor((B, H, S, D), nz.BF16)
+    nz.attn_split_qkv(qkv, q, k, v, B, S, H, D)
+    probs_c = hip_ctx.tensor((B * H * S, S), nz.BF16)
+    ctx_c = hip_ctx.tensor((B, H, S, D), nz.BF16)
+    nz.attn_fused_fwd(q, k, v, probs_c, ctx_c, B * H, S, D, scale)
+    out_c = hip_ctx.tensor((B * S, H * D), nz.BF16)
+    nz.attn_merge(ctx_c, out_c, B, S, H, D)
+    np.testing.assert_array_equal(probs_p.raw(), probs_c.raw())
+    np.testing.assert_array_equal(out_p.raw(), out_c.raw())
+    # backward vs the CPU composed fallback (f32 reference, same bf16 values)
+    cpu = nz.Ctx(nz.CPU)
+    qkv_c = cpu.from_numpy(QKV)
+    dm_c = cpu.from_numpy(DM)
+    probs_c32 = cpu.from_numpy(probs_p.numpy())
+    dqkv_ref = cpu.tensor((B * S, 3 * H * D))
+    nz.attn_bwd_packed(qkv_c, probs_c32, dm_c, dqkv_ref, B, S, H, D, scale)
+    check(dqkv_p.numpy(), dqkv_ref.numpy(), rtol=0.04)
+
+
 def test_attn_fused_determinism(hip_ctx):
     BH, S, D = 48, 128, 64
     q = hip_ctx.tensor((BH, S, D), nz.BF16).rand_normal(5)
