@@ -549,8 +549,18 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_nt_pers_k(
                 a_frag[fm], b_frag[fn], acc[fm][fn], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
       }
-      if (t + 1 == ntiles) epilogue(tl);  // acc complete; stores overlap DMA
-      __syncthreads();
+      if (t + 1 == ntiles) {
+        epilogue(tl);  // acc complete; stores issue behind the next stage
+        // Counted wait: outstanding = 8 glds (FIFO-older) + <=64 epilogue
+        // stores. vmcnt(63) guarantees the glds retired (>=9 retirements)
+        // while the stores keep draining under the next tile's compute —
+        // a full __syncthreads here paid the store-drain once per tile.
+        asm volatile("s_waitcnt vmcnt(63)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        asm volatile("" ::: "memory");
+      } else {
+        __syncthreads();
+      }
       cur ^= 1;
     }
   }
