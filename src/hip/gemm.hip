@@ -996,6 +996,303 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase_k(GemmArgs args,
 
 
 // ---------------------------------------------------------------------------
+// PERSISTENT 8-phase kernel: 256 WGs (1/CU), each sweeping a strided run of
+// (batch, m, n) 256^2 tiles with the SAME 8-phase wait schedule as
+// gemm_bf16_nt_8phase_k — and the pipeline flowing ACROSS tile boundaries.
+// The key observation: the s2/s3-guarded stages of a tile's LAST iteration
+// (B0/A1/A0/B1 of t+2 at p2..p5, B0/A1 of t+3 at p6/p7) are, in both ORDER
+// and LDS SLOT, exactly the 6-unit prologue of the next tile (B0(0) A1(0)
+// A0(0) B1(0) B0(1) A1(1), k0 -> buf0, k1 -> buf1 since ntiles is even).
+// Substituting next-tile stages there keeps every per-phase glds count, and
+// therefore the whole counted-vmcnt schedule, unchanged. Only two waits
+// differ from the single-tile kernel:
+//   - phase 0 of each tile waits vmcnt(0): it absorbs the previous tile's
+//     epilogue stores (which share the vmcnt FIFO) and re-establishes the
+//     counting invariant from a clean slate;
+//   - the true tail (last tile's last iteration, nothing left to stage)
+//     keeps the original vmcnt(0)-everywhere behavior.
+// This removes the 8-phase ramp at short K (K=768 = 3 iterations was ramp-
+// dominated) AND the 256-WG round-balance restriction — one prologue and
+// one drain per WG instead of per tile. Epilogue is a direct scatter
+// (interior fast path); the LDS-staged epilogue of the single-tile kernel
+// would clobber the buffers holding the next tile's prefetched halves.
+// ---------------------------------------------------------------------------
+__launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase_pers_k(
+    GemmArgs args, int nwg_n, int64_t per_b, int64_t total) {
+  using namespace ph8;
+  __shared__ __attribute__((aligned(16))) uint16_t smem[2 * BUF_E];
+
+  const int orig = blockIdx.x;  // 256 WGs
+  const int xcd = orig & 7, slotq = orig >> 3;  // 32 slots per XCD
+  const int64_t q = total / 8, r8 = total % 8;
+  const int64_t lo = xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q;
+  const int64_t cnt = xcd < r8 ? q + 1 : q;
+  if (slotq >= cnt) return;
+  constexpr int S = 32;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;  // 2x4 wave grid
+
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  const int64_t ntiles = args.K / PBK;  // even, >= 2 (launch guard)
+  const int ea = (int)threadIdx.x * 8;
+
+  struct TileSt {
+    const uint16_t* A;  // batch-offset bases
+    const uint16_t* B;
+    int32_t m0, n0;     // 32-bit: tile coords fit easily
+  };
+  auto make_state = [&](int64_t tl) {
+    TileSt ts;
+    const int64_t b = tl / per_b;
+    const int64_t rtl = tl % per_b;
+    ts.A = args.A + b * args.sA;
+    ts.B = args.B + b * args.sB;
+    ts.m0 = (int32_t)((rtl / nwg_n) * PBM);
+    ts.n0 = (int32_t)((rtl % nwg_n) * PBN);
+    return ts;
+  };
+
+  auto stage = [&](const TileSt& ts, int isB, int h, int buf, int64_t kt) {
+    int64_t k0 = kt * PBK;
+    uint16_t* dst = slot(smem, buf, isB, h);
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int e = ea + c * 4096;
+      int row = e >> 6;
+      int k = (e & 63) ^ ((row & 7) << 3);
+      if (!isB) {
+        int64_t ar = ts.m0 + h * 128 + row;
+        if (ar > Mc) ar = Mc;
+        glds16(ts.A + ar * lda + k0 + k, dst + e);
+      } else {
+        int64_t br = ts.n0 + h * 128 + row;
+        if (br > Nc) br = Nc;
+        glds16(ts.B + br * ldb + k0 + k, dst + e);
+      }
+    }
+  };
+
+  f32x4 acc[2][2][4][2] = {};
+
+  auto compute = [&](int HA, int HB, int buf, auto& accq) {
+    const uint16_t* as = slot(smem, buf, 0, HA);
+    const uint16_t* bs = slot(smem, buf, 1, HB);
+#pragma unroll
+    for (int kk = 0; kk < PBK; kk += 32) {
+      short8 a_frag[4], b_frag[2];
+      const int kbase = kk + (lane >> 4) * 8;
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        int row = wr * 64 + fm * 16 + (lane & 15);
+        int ks = kbase ^ ((row & 7) << 3);
+        a_frag[fm] = *(const short8*)(as + row * PBK + ks);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = wc * 32 + fn * 16 + (lane & 15);
+        int ks = kbase ^ ((col & 7) << 3);
+        b_frag[fn] = *(const short8*)(bs + col * PBK + ks);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn)
+          accq[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[fm], b_frag[fn], accq[fm][fn], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  };
+
+  // scatter epilogue (interior fast path); stores stay in flight — the next
+  // tile's phase-0 vmcnt(0) drains them
+  auto epilogue = [&](int64_t tl) {
+    const int64_t b = tl / per_b;
+    const int64_t rtl = tl % per_b;
+    uint16_t* C = args.C + b * args.sC;
+    float* out32 = args.out_f32 ? args.out_f32 + b * args.sC : nullptr;
+    uint16_t* save_pre = args.save_pre ? args.save_pre + b * args.sC : nullptr;
+    const int64_t N = args.N;
+    const int64_t m0 = (rtl / nwg_n) * PBM, n0 = (rtl % nwg_n) * PBN;
+    if (m0 + PBM <= args.M && n0 + PBN <= args.N) {
+      const int64_t base =
+          (m0 + wr * 64 + (lane >> 4) * 4) * N + n0 + wc * 32 + (lane & 15);
+      const uint16_t* bias_l =
+          args.bias ? args.bias + n0 + wc * 32 + (lane & 15) : nullptr;
+#pragma unroll
+      for (int ha = 0; ha < 2; ++ha)
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+          for (int rI = 0; rI < 4; ++rI) {
+            const int64_t rowbase =
+                base + (int64_t)(ha * 128 + fm * 16 + rI) * N;
+#pragma unroll
+            for (int hb = 0; hb < 2; ++hb)
+#pragma unroll
+              for (int fn = 0; fn < 2; ++fn) {
+                const int64_t idx = rowbase + hb * 128 + fn * 16;
+                if (out32) {
+                  out32[idx] = acc[ha][hb][fm][fn][rI];
+                } else {
+                  float v = args.alpha * acc[ha][hb][fm][fn][rI];
+                  if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+                  if (bias_l) v += b2f(bias_l[hb * 128 + fn * 16]);
+                  if (save_pre) save_pre[idx] = f2b(v);
+                  if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+                  C[idx] = f2b(v);
+                }
+              }
+          }
+    } else {
+#pragma unroll
+      for (int ha = 0; ha < 2; ++ha)
+#pragma unroll
+        for (int hb = 0; hb < 2; ++hb)
+#pragma unroll
+          for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+            for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+              for (int rI = 0; rI < 4; ++rI) {
+                int64_t grow =
+                    m0 + ha * 128 + wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+                int64_t gcol =
+                    n0 + hb * 128 + wc * 32 + fn * 16 + (lane & 15);
+                if (grow < args.M && gcol < args.N) {
+                  int64_t idx = grow * N + gcol;
+                  if (out32) {
+                    out32[idx] = acc[ha][hb][fm][fn][rI];
+                  } else {
+                    float v = args.alpha * acc[ha][hb][fm][fn][rI];
+                    if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+                    if (args.bias) v += b2f(args.bias[gcol]);
+                    if (save_pre) save_pre[idx] = f2b(v);
+                    if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+                    C[idx] = f2b(v);
+                  }
+                }
+              }
+    }
+#pragma unroll
+    for (int ha = 0; ha < 2; ++ha)
+#pragma unroll
+      for (int hb = 0; hb < 2; ++hb)
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn) acc[ha][hb][fm][fn] = f32x4{};
+  };
+
+#define P8P_WAIT0()                                     \
+  do {                                                  \
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");    \
+    __builtin_amdgcn_s_barrier();                       \
+    asm volatile("" ::: "memory");                      \
+  } while (0)
+#define P8P_WAITN(n)                                    \
+  do {                                                  \
+    if (tail)                                           \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  \
+    else                                                \
+      asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory"); \
+    __builtin_amdgcn_s_barrier();                       \
+    asm volatile("" ::: "memory");                      \
+  } while (0)
+#define P8P_BAR()                                       \
+  do {                                                  \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  \
+    __builtin_amdgcn_s_barrier();                       \
+    asm volatile("" ::: "memory");                      \
+  } while (0)
+
+  int64_t tl = lo + slotq;
+  TileSt cur = make_state(tl);
+  // prologue of the first tile (order the wait schedule assumes)
+  stage(cur, 1, 0, 0, 0);  // B0(0)
+  stage(cur, 0, 1, 0, 0);  // A1(0)
+  stage(cur, 0, 0, 0, 0);  // A0(0)
+  stage(cur, 1, 1, 0, 0);  // B1(0)
+  stage(cur, 1, 0, 1, 1);  // B0(1)
+  stage(cur, 0, 1, 1, 1);  // A1(1)
+
+  for (; tl < lo + cnt; tl += S) {
+    const bool has_next = (tl + S) < (lo + cnt);
+    TileSt nxt = has_next ? make_state(tl + S) : cur;
+    for (int64_t t = 0; t < ntiles; t += 2) {
+      const bool last = (t + 2 >= ntiles);  // note: !last implies t+3 < ntiles
+      const bool tail = last && !has_next;
+      // phase 0: Q(0,0) of t (buf0); stage A0(t+1) -> buf1
+      if (t == 0) {
+        P8P_WAIT0();
+      } else {
+        P8P_WAITN(6);
+      }
+      stage(cur, 0, 0, 1, t + 1);
+      compute(0, 0, 0, acc[0][0]);
+      P8P_BAR();
+      // phase 1: Q(1,0) of t; stage B1(t+1) -> buf1
+      stage(cur, 1, 1, 1, t + 1);
+      compute(1, 0, 0, acc[1][0]);
+      P8P_BAR();
+      // phase 2: Q(1,1) of t; stage B0(t+2) / next-tile B0(0) -> buf0
+      P8P_WAITN(8);
+      if (!last)
+        stage(cur, 1, 0, 0, t + 2);
+      else if (has_next)
+        stage(nxt, 1, 0, 0, 0);
+      compute(1, 1, 0, acc[1][1]);
+      P8P_BAR();
+      // phase 3: Q(0,1) of t; stage A1(t+2) / next A1(0) -> buf0
+      if (!last)
+        stage(cur, 0, 1, 0, t + 2);
+      else if (has_next)
+        stage(nxt, 0, 1, 0, 0);
+      compute(0, 1, 0, acc[0][1]);
+      P8P_BAR();
+      // phase 4: Q(0,0) of t+1 (buf1); stage A0(t+2) / next A0(0) -> buf0
+      P8P_WAITN(6);
+      if (!last)
+        stage(cur, 0, 0, 0, t + 2);
+      else if (has_next)
+        stage(nxt, 0, 0, 0, 0);
+      compute(0, 0, 1, acc[0][0]);
+      P8P_BAR();
+      // phase 5: Q(1,0) of t+1; stage B1(t+2) / next B1(0) -> buf0
+      if (!last)
+        stage(cur, 1, 1, 0, t + 2);
+      else if (has_next)
+        stage(nxt, 1, 1, 0, 0);
+      compute(1, 0, 1, acc[1][0]);
+      P8P_BAR();
+      // phase 6: Q(1,1) of t+1; stage B0(t+3) / next B0(1) -> buf1
+      P8P_WAITN(8);
+      if (!last)
+        stage(cur, 1, 0, 1, t + 3);
+      else if (has_next)
+        stage(nxt, 1, 0, 1, 1);
+      compute(1, 1, 1, acc[1][1]);
+      P8P_BAR();
+      // phase 7: Q(0,1) of t+1; stage A1(t+3) / next A1(1) -> buf1
+      if (!last)
+        stage(cur, 0, 1, 1, t + 3);
+      else if (has_next)
+        stage(nxt, 0, 1, 1, 1);
+      compute(0, 1, 1, acc[0][1]);
+      P8P_BAR();
+    }
+    epilogue(tl);
+    cur = nxt;
+  }
+#undef P8P_WAIT0
+#undef P8P_WAITN
+#undef P8P_BAR
+}
+
+// ---------------------------------------------------------------------------
 // Variant 3: 8-phase kernel with fragment reads PIPELINED one phase ahead
 // (the guide's "per-phase interleave is the lever"): body p issues the
 // ds_reads of phase p+1 right before the MFMAs of phase p, so LDS reads,
@@ -1650,6 +1947,28 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     if (nwg_m * (int64_t)nwg_n >= kNtMinWG) {
       dim3 grid(nwg_m * nwg_n);
       gemm_bf16_nt_8phase2_k<<<grid, dim3(512), 0, s>>>(args, nwg_n);
+      NZ_KERNEL_CHECK();
+      return;
+    }
+  }
+  // Persistent 8-phase (NEZHA_GEMM_PERS8=1, measurement opt-in): pipeline
+  // flows across tiles via the prologue/s2-s3 stage correspondence — but at
+  // 1 WG/CU the per-tile phase-0 drain and scatter epilogue have no
+  // co-resident WG to hide under, and it measures BELOW the 2-WG/CU 128^2
+  // persistent kernel on every contract shape (qkv-fwd 504 vs 566 TF,
+  // ffn2-fwd 712 vs 874; square-4k a wash at 927 vs 908). Kept for A/B
+  // evidence; see docs/perf/README.md.
+  if (!transA && transB && kNtVariant >= 2 && args.K % 128 == 0 &&
+      args.M >= 256 && args.N >= 256) {
+    static const bool kP8On =
+        getenv("NEZHA_GEMM_PERS8") && atoi(getenv("NEZHA_GEMM_PERS8")) == 1;
+    int nwg_m8 = (int)((args.M + 255) / 256);
+    int nwg_n8 = (int)((args.N + 255) / 256);
+    int64_t per_b8 = (int64_t)nwg_m8 * nwg_n8;
+    int64_t total8 = per_b8 * batch;
+    if (kP8On && total8 >= 256) {
+      gemm_bf16_nt_8phase_pers_k<<<dim3(256), dim3(512), 0, s>>>(
+          args, nwg_n8, per_b8, total8);
       NZ_KERNEL_CHECK();
       return;
     }
