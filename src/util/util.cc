@@ -108,7 +108,19 @@ void MetricsWriter::write(const std::map<std::string, double>& nums,
     first = false;
   }
   for (auto& [k, v] : strs) {
-    fprintf(f_, "%s\"%s\":\"%s\"", first ? "" : ",", k.c_str(), v.c_str());
+    // values that are themselves JSON objects embed as-is; everything else
+    // is quoted with escaping (an unescaped embed once produced malformed
+    // JSONL when the CLI passed its config dump as a plain string)
+    if (!v.empty() && v.front() == '{' && v.back() == '}') {
+      fprintf(f_, "%s\"%s\":%s", first ? "" : ",", k.c_str(), v.c_str());
+    } else {
+      fprintf(f_, "%s\"%s\":\"", first ? "" : ",", k.c_str());
+      for (char c : v) {
+        if (c == '"' || c == '\\') fputc('\\', f_);
+        if ((unsigned char)c >= 0x20) fputc(c, f_);
+      }
+      fprintf(f_, "\"");
+    }
     first = false;
   }
   fprintf(f_, "}\n");

@@ -188,3 +188,30 @@ print(json.dumps(losses))
         outs.append(json.loads(r.stdout.decode().strip().splitlines()[-1]))
     assert outs[0] == outs[1]  # deterministic
     assert len(set(round(x, 6) for x in outs[0])) > 2  # data actually varies
+
+
+def test_metrics_jsonl_valid(tmp_path):
+    """NEZHA_METRICS per-step records must be valid JSONL (an unescaped
+    embedded config string once produced malformed lines)."""
+    import subprocess, sys
+    mpath = tmp_path / "m.jsonl"
+    script = """
+import os, sys
+sys.path.insert(0, os.getcwd())
+import nezha_amd as nz
+ctx = nz.Ctx(nz.CPU)
+tr = nz.Trainer.mlp(ctx, layers=2, dim=64, batch=32, lr=1e-3, seed=5)
+for _ in range(3):
+    tr.step()
+    tr.loss()
+"""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, NEZHA_METRICS=str(mpath))
+    r = subprocess.run([sys.executable, "-c", script], env=env, cwd=repo,
+                       capture_output=True, timeout=120)
+    assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
+    lines = mpath.read_text().strip().splitlines()
+    assert len(lines) >= 3
+    for ln in lines:
+        rec = json.loads(ln)
+        assert "loss" in rec and "step" in rec
