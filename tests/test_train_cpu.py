@@ -212,6 +212,7 @@ for _ in range(3):
     assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
     lines = mpath.read_text().strip().splitlines()
     assert len(lines) >= 3
+    import json as _json
     for ln in lines:
-        rec = json.loads(ln)
+        rec = _json.loads(ln)
         assert "loss" in rec and "step" in rec
