@@ -208,14 +208,18 @@ __launch_bounds__(256) __global__ void attn_fwd_fused_k(
 //   Qt  [64][136]  17 KB (Q^T scatter)
 //   psum[128][32] f32 16 KB + rs[128] f32  (rowsum partials + fold)
 // ---------------------------------------------------------------------------
-__launch_bounds__(256) __global__ void attn_bwd_fused_k(
+__launch_bounds__(512) __global__ void attn_bwd_fused_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const uint16_t* __restrict__ probs,
     const uint16_t* __restrict__ dout, uint16_t* __restrict__ dq,
     uint16_t* __restrict__ dk, uint16_t* __restrict__ dv, float scale,
     int heads, int64_t qb, int64_t qh, int64_t qr, int64_t db, int64_t dhh,
     int64_t dr, int64_t gb, int64_t gh, int64_t gr) {
-  __shared__ __attribute__((aligned(16))) uint16_t smem[68352];
+  // 512 threads (8 waves as 2x4): the 256-thread version ran 4 waves on a
+  // 137 KB, 1-WG/CU kernel = 1 wave/SIMD — every barrier and LDS/HBM
+  // latency fully exposed. 8 waves give 2/SIMD at the same occupancy.
+  // LDS 153 KB: psum widens to [128][64] (each lane covers 2 columns).
+  __shared__ __attribute__((aligned(16))) uint16_t smem[76544];
   uint16_t* dOs = smem;                    // [128][64] swz
   uint16_t* Vs = smem + AS * AD;           // [128][64] swz
   uint16_t* dSn = smem;                    // [128][128] swz (overlay dO+V)
@@ -224,8 +228,8 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
   uint16_t* dOt = Pt + AS * PSTRIDE;       // [d][136]
   uint16_t* Kt = dOt + AD * PSTRIDE;       // [d][136] (Kt[d][s2])
   uint16_t* Qt = Kt + AD * PSTRIDE;        // [d][136] (Qt[d][s1])
-  float* psum = (float*)(Qt + AD * PSTRIDE);  // [128][32]
-  float* rs = psum + AS * 32;                 // [128]
+  float* psum = (float*)(Qt + AD * PSTRIDE);  // [128][64]
+  float* rs = psum + AS * 64;                 // [128]
 
   const int64_t bh = blockIdx.x;
   const int64_t b_ = bh / heads, h_ = bh % heads;
@@ -240,30 +244,27 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int wr = wid >> 1, wc = wid & 1;
+  const int wr = wid >> 2, wc = wid & 3;  // 2x4 wave grid
 
   // ---- P0: stage ----------------------------------------------------------
   const int e0 = (int)threadIdx.x * 8;
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
-    int e = e0 + c * 2048;
+  for (int c = 0; c < 2; ++c) {
+    int e = e0 + c * 4096;
     int row = e >> 6;
     int kk = (e & 63) ^ ((row & 7) << 3);
     att_glds16(dog + row * dr + kk, dOs + e);
     att_glds16(vg + row * qr + kk, Vs + e);
   }
-  // register loads + transposed scatters (issued while the DMA flies);
-  // pairs of loads per scatter group keep the live staging registers small
-  // (the all-loads-then-all-scatters version allocated 316 VGPRs -> 1
-  // wave/SIMD; this kernel wants 2)
+  // register loads + transposed scatters (issued while the DMA flies)
   {
 #pragma unroll
-    for (int c = 0; c < 8; c += 2) {
-      short8 p0 = *(const short8*)(pg + e0 + c * 2048);
-      short8 p1 = *(const short8*)(pg + e0 + (c + 1) * 2048);
+    for (int c = 0; c < 4; c += 2) {
+      short8 p0 = *(const short8*)(pg + e0 + c * 4096);
+      short8 p1 = *(const short8*)(pg + e0 + (c + 1) * 4096);
 #pragma unroll
       for (int u = 0; u < 2; ++u) {
-        int e = e0 + (c + u) * 2048;  // element in P [s1][s2]
+        int e = e0 + (c + u) * 4096;  // element in P [s1][s2]
         int s1 = e >> 7, s2 = e & 127;
         const short8& pv = u ? p1 : p0;
 #pragma unroll
@@ -272,8 +273,8 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
       }
     }
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      int e = e0 + c * 2048;          // element in [s][d]
+    for (int c = 0; c < 2; ++c) {
+      int e = e0 + c * 4096;          // element in [s][d]
       int s_ = e >> 6, d = e & 63;
       short8 dov = *(const short8*)(dog + s_ * dr + d);
       short8 kv = *(const short8*)(kg + s_ * qr + d);
@@ -288,11 +289,11 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
   }
   __syncthreads();  // glds drained + scatters visible
 
-  // ---- P1: dP = dO.V^T (f32 acc, quadrant per wave) -----------------------
-  f32x4 acc[4][4] = {};
+  // ---- P1: dP = dO.V^T (f32 acc; wave owns 64 s1 x 32 s2) -----------------
+  f32x4 acc[4][2] = {};
 #pragma unroll
   for (int kk = 0; kk < AD; kk += 32) {
-    short8 a_frag[4], b_frag[4];
+    short8 a_frag[4], b_frag[2];
     const int kbase = kk + (lane >> 4) * 8;
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm) {
@@ -301,21 +302,20 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
       a_frag[fm] = *(const short8*)(dOs + row * AD + ks);
     }
 #pragma unroll
-    for (int fn = 0; fn < 4; ++fn) {
-      int col = wc * 64 + fn * 16 + (lane & 15);
+    for (int fn = 0; fn < 2; ++fn) {
+      int col = wc * 32 + fn * 16 + (lane & 15);
       int ks = kbase ^ ((col & 7) << 3);
       b_frag[fn] = *(const short8*)(Vs + col * AD + ks);
     }
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn)
+      for (int fn = 0; fn < 2; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_frag[fm], b_frag[fn], acc[fm][fn], 0, 0, 0);
   }
 
-  // ---- P2: rowsum partials of dP o P (P read from LDS; each (row,col)
-  // element of Pt is read and later overwritten only by its owning lane) ----
+  // ---- P2: rowsum partials of dP o P --------------------------------------
 #pragma unroll
   for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
@@ -323,56 +323,57 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
       int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
       float part = 0.f;
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn) {
-        int col = wc * 64 + fn * 16 + (lane & 15);
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = wc * 32 + fn * 16 + (lane & 15);
         part += acc[fm][fn][rI] * b2f(Pt[col * PSTRIDE + row]);
       }
-      psum[row * 32 + wc * 16 + (lane & 15)] = part;
+      psum[row * 64 + wc * 16 + (lane & 15)] = part;
     }
 
-  // ---- P2b: dV = P^T.dO (from Pt / dOt; C rows = s2) ----------------------
+  // ---- P2b: dV = P^T.dO (wave owns 64 s2 x 16 d) --------------------------
   {
-    f32x4 acc2[4][2] = {};
+    f32x4 acc2[4] = {};
 #pragma unroll
     for (int kk = 0; kk < AS; kk += 32) {
-      short8 a_frag[4], b_frag[2];
+      short8 a_frag[4], b_frag;
       const int kbase = kk + (lane >> 4) * 8;
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm) {
         int row = wr * 64 + fm * 16 + (lane & 15);  // s2
         a_frag[fm] = *(const short8*)(Pt + row * PSTRIDE + kbase);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        int col = wc * 32 + fn * 16 + (lane & 15);  // d
-        b_frag[fn] = *(const short8*)(dOt + col * PSTRIDE + kbase);
+      {
+        int col = wc * 16 + (lane & 15);  // d
+        b_frag = *(const short8*)(dOt + col * PSTRIDE + kbase);
       }
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm)
-#pragma unroll
-        for (int fn = 0; fn < 2; ++fn)
-          acc2[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag[fm], b_frag[fn], acc2[fm][fn], 0, 0, 0);
+        acc2[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[fm], b_frag,
+                                                           acc2[fm], 0, 0, 0);
     }
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
-#pragma unroll
-        for (int rI = 0; rI < 4; ++rI) {
-          int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
-          int col = wc * 32 + fn * 16 + (lane & 15);
-          dvg[row * gr + col] = f2b(acc2[fm][fn][rI]);
-        }
+      for (int rI = 0; rI < 4; ++rI) {
+        int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+        int col = wc * 16 + (lane & 15);
+        dvg[row * gr + col] = f2b(acc2[fm][rI]);
+      }
   }
   __syncthreads();  // psum complete (and Pt reads done -> dSt may overlay)
 
-  // ---- P3: fold rowsums ---------------------------------------------------
+  // ---- P3: fold rowsums (4-way ILP) ---------------------------------------
   if (threadIdx.x < AS) {
-    float s_ = 0.f;
+    const float* pr = psum + threadIdx.x * 64;
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
 #pragma unroll
-    for (int i = 0; i < 32; ++i) s_ += psum[threadIdx.x * 32 + i];
-    rs[threadIdx.x] = s_;
+    for (int i = 0; i < 64; i += 4) {
+      a0 += pr[i];
+      a1 += pr[i + 1];
+      a2 += pr[i + 2];
+      a3 += pr[i + 3];
+    }
+    rs[threadIdx.x] = (a0 + a1) + (a2 + a3);
   }
   __syncthreads();
 
@@ -385,8 +386,8 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
       int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
       float r = rs[row];
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn) {
-        int col = wc * 64 + fn * 16 + (lane & 15);
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = wc * 32 + fn * 16 + (lane & 15);
         float pb = b2f(Pt[col * PSTRIDE + row]);
         uint16_t ds = f2b(scale * pb * (acc[fm][fn][rI] - r));
         dSn[row * AS + (col ^ ((row & 7) << 3))] = ds;
@@ -395,12 +396,12 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
     }
   __syncthreads();
 
-  // ---- P5: dQ = dS.K (A=dSn swz, B=Kt) ------------------------------------
+  // ---- P5: dQ = dS.K (A=dSn swz, B=Kt; wave owns 64 s1 x 16 d) ------------
   {
-    f32x4 acc2[4][2] = {};
+    f32x4 acc2[4] = {};
 #pragma unroll
     for (int kk = 0; kk < AS; kk += 32) {
-      short8 a_frag[4], b_frag[2];
+      short8 a_frag[4], b_frag;
       const int kbase = kk + (lane >> 4) * 8;
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm) {
@@ -408,64 +409,54 @@ __launch_bounds__(256) __global__ void attn_bwd_fused_k(
         int ks = kbase ^ ((row & 7) << 3);
         a_frag[fm] = *(const short8*)(dSn + row * AS + ks);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        int col = wc * 32 + fn * 16 + (lane & 15);  // d
-        b_frag[fn] = *(const short8*)(Kt + col * PSTRIDE + kbase);
+      {
+        int col = wc * 16 + (lane & 15);  // d
+        b_frag = *(const short8*)(Kt + col * PSTRIDE + kbase);
       }
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm)
-#pragma unroll
-        for (int fn = 0; fn < 2; ++fn)
-          acc2[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag[fm], b_frag[fn], acc2[fm][fn], 0, 0, 0);
+        acc2[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[fm], b_frag,
+                                                           acc2[fm], 0, 0, 0);
     }
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
-#pragma unroll
-        for (int rI = 0; rI < 4; ++rI) {
-          int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
-          int col = wc * 32 + fn * 16 + (lane & 15);
-          dqg[row * gr + col] = f2b(acc2[fm][fn][rI]);
-        }
+      for (int rI = 0; rI < 4; ++rI) {
+        int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+        int col = wc * 16 + (lane & 15);
+        dqg[row * gr + col] = f2b(acc2[fm][rI]);
+      }
   }
 
-  // ---- P5b: dK = dS^T.Q (A=dSt, B=Qt) -------------------------------------
+  // ---- P5b: dK = dS^T.Q (A=dSt, B=Qt; wave owns 64 s2 x 16 d) -------------
   {
-    f32x4 acc2[4][2] = {};
+    f32x4 acc2[4] = {};
 #pragma unroll
     for (int kk = 0; kk < AS; kk += 32) {
-      short8 a_frag[4], b_frag[2];
+      short8 a_frag[4], b_frag;
       const int kbase = kk + (lane >> 4) * 8;
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm) {
         int row = wr * 64 + fm * 16 + (lane & 15);  // s2
         a_frag[fm] = *(const short8*)(dSt + row * PSTRIDE + kbase);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        int col = wc * 32 + fn * 16 + (lane & 15);  // d
-        b_frag[fn] = *(const short8*)(Qt + col * PSTRIDE + kbase);
+      {
+        int col = wc * 16 + (lane & 15);  // d
+        b_frag = *(const short8*)(Qt + col * PSTRIDE + kbase);
       }
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm)
-#pragma unroll
-        for (int fn = 0; fn < 2; ++fn)
-          acc2[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag[fm], b_frag[fn], acc2[fm][fn], 0, 0, 0);
+        acc2[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[fm], b_frag,
+                                                           acc2[fm], 0, 0, 0);
     }
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
-#pragma unroll
-        for (int rI = 0; rI < 4; ++rI) {
-          int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
-          int col = wc * 32 + fn * 16 + (lane & 15);
-          dkg[row * gr + col] = f2b(acc2[fm][fn][rI]);
-        }
+      for (int rI = 0; rI < 4; ++rI) {
+        int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+        int col = wc * 16 + (lane & 15);
+        dkg[row * gr + col] = f2b(acc2[fm][rI]);
+      }
   }
 }
 
@@ -512,7 +503,7 @@ void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
   NZ_CHECK(q.dtype() == DType::BF16, "fused attention bwd: bf16 required");
   hipStream_t s = stream_of(q);
   const int64_t sh = (int64_t)AS * AD;
-  attn_bwd_fused_k<<<dim3((uint32_t)batch_heads), dim3(256), 0, s>>>(
+  attn_bwd_fused_k<<<dim3((uint32_t)batch_heads), dim3(512), 0, s>>>(
       (const uint16_t*)q.data(), (const uint16_t*)k.data(),
       (const uint16_t*)v.data(), (const uint16_t*)probs.data(),
       (const uint16_t*)dout.data(), (uint16_t*)dq.data(),
@@ -532,7 +523,7 @@ void attn_fused_bwd_packed(const Tensor& qkv, const Tensor& probs,
   int64_t H = heads * dh;
   const uint16_t* base = (const uint16_t*)qkv.data();
   uint16_t* gbase = (uint16_t*)dqkv.data();
-  attn_bwd_fused_k<<<dim3((uint32_t)(B * heads)), dim3(256), 0, s>>>(
+  attn_bwd_fused_k<<<dim3((uint32_t)(B * heads)), dim3(512), 0, s>>>(
       base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
       (const uint16_t*)dmerged.data(), gbase, gbase + H, gbase + 2 * H, scale,
       (int)heads, S * 3 * H, dh, 3 * H, S * H, dh, H, S * 3 * H, dh, 3 * H);
