@@ -38,6 +38,8 @@ struct GemmArgs {
   int64_t sA, sB, sC;     // batch strides (elements)
   float alpha, beta;
   int fuse_gelu;
+  int diag_no_store;  // NEZHA_PERS_DIAG=1: skip pers epilogue stores
+                      // (DIAGNOSTIC ONLY — output garbage; cost attribution)
 };
 
 // Stage op(A) tile [BM][BK] (rows m, cols k) into lds with row stride
@@ -452,6 +454,7 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_nt_pers_k(
     float* out32 = args.out_f32 ? args.out_f32 + b * args.sC : nullptr;
     uint16_t* save_pre = args.save_pre ? args.save_pre + b * args.sC : nullptr;
     const int64_t N = args.N;
+    if (args.diag_no_store) return;
     if (em0 + BM <= args.M && en0 + BN <= args.N) {
       const int64_t base =
           (em0 + wr * 64 + (lane >> 4) * 4) * N + en0 + wc * 64 + (lane & 15);
@@ -1943,6 +1946,11 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     return v ? atoi(v) : 224;
   }();
   static const bool kMinWgForced = getenv("NEZHA_GEMM_MIN_WG") != nullptr;
+  static const int kPersDiag = [] {
+    const char* v = getenv("NEZHA_PERS_DIAG");
+    return v ? atoi(v) : 0;
+  }();
+  const_cast<GemmArgs&>(args).diag_no_store = kPersDiag;
   // The 256^2 kernel runs ~1 workgroup per CU, so a grid that is not close
   // to a multiple of 256 leaves whole CUs idle for a full round (N=768 ->
   // 384 WGs = 25% idle; measured 679 vs 884 TF against the 128^2 kernel).
