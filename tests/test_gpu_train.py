@@ -41,6 +41,19 @@ def test_bert_training(hip_ctx):
     assert np.isfinite(l0) and tr.loss() < l0
 
 
+def test_bert_large_config(hip_ctx):
+    """BERT-large geometry (24x1024, 16 heads): exercises the N=1024
+    wave-LayerNorm path and the fused attention kernels at a different
+    head count than the flagship config."""
+    tr = nz.Trainer.bert(hip_ctx, layers=24, hidden=1024, ffn=4096, heads=16,
+                         seq=128, batch=32, lr=1e-4, seed=9)
+    tr.step()
+    l0 = tr.loss()
+    for _ in range(4):
+        tr.step()
+    assert np.isfinite(l0) and np.isfinite(tr.loss())
+
+
 def test_cpu_gpu_agreement():
     """Same model/seed/data on CPU f32 and GPU bf16: losses agree to bf16
     tolerance after the first step (validates the whole GPU path E2E)."""
