@@ -1,8 +1,11 @@
-// Fused attention forward (K5 upgraded from "composed" to a single-kernel
-// fusion for the BERT contract shape S=128, dh=64): one workgroup per
-// (batch, head) computes QK^T -> scaled softmax -> P.V entirely on-chip.
-// The S x S score tile never touches HBM; P is written once (backward
-// consumes it — the backward pass stays composed from K1/K4).
+// Fused attention (K5), forward AND backward, for the BERT contract shape
+// S=128, dh=64: one workgroup per (batch, head). Forward computes
+// QK^T -> scaled softmax -> P.V entirely on-chip (the S x S score tile
+// never touches HBM; P is written once for backward). Backward
+// (attn_bwd_fused_k below) computes dP, the softmax Jacobian, and
+// dQ/dK/dV in one kernel. Both kernels address q/k/v/outputs through
+// generalized strides so they consume the packed qkv [B,S,3H] projection
+// and produce the merged [B,S,H] layout directly.
 //
 // LDS plan (49.5 KB -> 3 blocks/CU):
 //   region Q  [128][64]  16 KB  (glds lane-linear, XOR-swizzled)  \ reused as
