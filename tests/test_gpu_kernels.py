@@ -39,6 +39,28 @@ def test_gemm_trans(hip_ctx, transA, transB):
     check(c.numpy(), ref)
 
 
+def test_gemm_ragged_pers(hip_ctx):
+    """Non-multiple-of-128 M at a size that routes to the persistent NT
+    kernel: exercises the staging clamp + edge epilogue."""
+    M, N, K = 20000, 768, 768
+    A, B = randb(M, K), randb(N, K)
+    a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
+    c = hip_ctx.tensor((M, N), nz.BF16)
+    nz.gemm(a, b, c, False, True)
+    check(c.numpy(), A @ B.T, rtol=0.03)
+
+
+def test_gemm_ragged_8phase(hip_ctx):
+    """Ragged M at a balanced >=1024-tile grid: the 256^2 kernel's clamped
+    staging + exterior epilogue path."""
+    M, N, K = 16512, 4096, 1536
+    A, B = randb(M, K), randb(N, K)
+    a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
+    c = hip_ctx.tensor((M, N), nz.BF16)
+    nz.gemm(a, b, c, False, True)
+    check(c.numpy(), A @ B.T, rtol=0.03)
+
+
 def test_gemm_large_shape(hip_ctx):
     M, N, K = 512, 384, 512
     A, B = randb(M, K), randb(K, N)
