@@ -44,13 +44,13 @@ __device__ __forceinline__ void att_glds16(const uint16_t* gsrc,
 // straight from the packed qkv projection [B, S, 3H] (qr=3H, qh=64,
 // qb=S*3H, with k/v base pointers offset by H / 2H) — the latter removes
 // the attn_split/attn_merge layout passes entirely. Same for `out`.
-__launch_bounds__(256) __global__ void attn_fwd_fused_k(
+__launch_bounds__(512) __global__ void attn_fwd_fused_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ probs,
     uint16_t* __restrict__ out, float scale, int heads, int64_t qb,
     int64_t qh, int64_t qr, int64_t ob, int64_t oh, int64_t orr) {
-  // LDS: [P 128*136 | VT 64*136]; Q and K overlay P's first 32 KB (the P
-  // region is 34 KB — VT must live BEYOND it, clobbering it was a bug).
+  // 512 threads (8 waves as 2x4) at the same 49.5 KB LDS / 3 WGs/CU:
+  // 6 waves/SIMD to hide the phase barriers (the 256-thread version ran 3).
   __shared__ __attribute__((aligned(16))) uint16_t smem[(AS + AD) * PSTRIDE];
   uint16_t* Qs = smem;
   uint16_t* Ks = smem + AS * AD;
@@ -69,25 +69,25 @@ __launch_bounds__(256) __global__ void attn_fwd_fused_k(
   const int wid = threadIdx.x >> 6;
 
   // ---- stage Q,K by glds (swizzled source), V to registers ---------------
-  const int e0 = (int)threadIdx.x * 8;      // 2 chunks x 8 elems per thread
+  const int e0 = (int)threadIdx.x * 8;
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
-    int e = e0 + c * 2048;
+  for (int c = 0; c < 2; ++c) {
+    int e = e0 + c * 4096;
     int row = e >> 6;
     int kk = (e & 63) ^ ((row & 7) << 3);
     att_glds16(qg + row * qr + kk, Qs + e);
     att_glds16(kg + row * qr + kk, Ks + e);
   }
-  short8 vreg[4];
+  short8 vreg[2];
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
-    int e = e0 + c * 2048;                  // element in [s][d]
+  for (int c = 0; c < 2; ++c) {
+    int e = e0 + c * 4096;                  // element in [s][d]
     vreg[c] = *(const short8*)(vg + (e >> 6) * qr + (e & 63));
   }
   // scatter V^T while the DMA flies: (s,d..d+7) -> VT[d+i][s]
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
-    int e = e0 + c * 2048;
+  for (int c = 0; c < 2; ++c) {
+    int e = e0 + c * 4096;
     int s = e >> 6, d = e & 63;
 #pragma unroll
     for (int i = 0; i < 8; ++i)
@@ -95,13 +95,13 @@ __launch_bounds__(256) __global__ void attn_fwd_fused_k(
   }
   __syncthreads();  // drains glds (vmcnt0) + V scatter visible
 
-  // ---- QK^T: wave (wr,wc) of 2x2 computes rows wr*64.., cols wc*64.. -----
+  // ---- QK^T: wave (wr,wc) of 2x4 computes 64 rows x 32 cols ---------------
   {
-    const int wr = wid >> 1, wc = wid & 1;
-    f32x4 acc[4][4] = {};
+    const int wr = wid >> 2, wc = wid & 3;
+    f32x4 acc[4][2] = {};
 #pragma unroll
     for (int kk = 0; kk < AD; kk += 32) {
-      short8 a_frag[4], b_frag[4];
+      short8 a_frag[4], b_frag[2];
       const int kbase = kk + (lane >> 4) * 8;
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm) {
@@ -110,15 +110,15 @@ __launch_bounds__(256) __global__ void attn_fwd_fused_k(
         a_frag[fm] = *(const short8*)(Qs + row * AD + ks);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn) {
-        int col = wc * 64 + fn * 16 + (lane & 15);
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = wc * 32 + fn * 16 + (lane & 15);
         int ks = kbase ^ ((col & 7) << 3);
         b_frag[fn] = *(const short8*)(Ks + col * AD + ks);
       }
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
-        for (int fn = 0; fn < 4; ++fn)
+        for (int fn = 0; fn < 2; ++fn)
           acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag[fm], b_frag[fn], acc[fm][fn], 0, 0, 0);
     }
@@ -127,18 +127,18 @@ __launch_bounds__(256) __global__ void attn_fwd_fused_k(
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn)
+      for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
         for (int rI = 0; rI < 4; ++rI) {
           int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
-          int col = wc * 64 + fn * 16 + (lane & 15);
+          int col = wc * 32 + fn * 16 + (lane & 15);
           Ps[row * PSTRIDE + col] = f2b(acc[fm][fn][rI]);
         }
   }
   __syncthreads();
 
-  // ---- softmax: wave w handles rows w, w+4, ... (2 elems per lane) -------
-  for (int row = wid; row < AS; row += 4) {
+  // ---- softmax: wave w handles rows w, w+8, ... (2 elems per lane) -------
+  for (int row = wid; row < AS; row += 8) {
     const uint16_t* pr = Ps + row * PSTRIDE;
     float v0 = scale * b2f(pr[lane * 2]);
     float v1 = scale * b2f(pr[lane * 2 + 1]);
@@ -155,41 +155,36 @@ __launch_bounds__(256) __global__ void attn_fwd_fused_k(
   }
   __syncthreads();
 
-  // ---- P.V: output 128x64; wave (wr,wc) of 2x2 computes 64x32 ------------
+  // ---- P.V: output 128x64; wave (wr,wc) of 2x4 computes 64x16 ------------
   {
-    const int wr = wid >> 1, wc = wid & 1;
-    f32x4 acc[4][2] = {};
+    const int wr = wid >> 2, wc = wid & 3;
+    f32x4 acc[4] = {};
 #pragma unroll
     for (int kk = 0; kk < AS; kk += 32) {
-      short8 a_frag[4], b_frag[2];
+      short8 a_frag[4], b_frag;
       const int kbase = kk + (lane >> 4) * 8;
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm) {
         int row = wr * 64 + fm * 16 + (lane & 15);
         a_frag[fm] = *(const short8*)(Ps + row * PSTRIDE + kbase);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        int col = wc * 32 + fn * 16 + (lane & 15);
-        b_frag[fn] = *(const short8*)(VTs + col * PSTRIDE + kbase);
+      {
+        int col = wc * 16 + (lane & 15);
+        b_frag = *(const short8*)(VTs + col * PSTRIDE + kbase);
       }
 #pragma unroll
       for (int fm = 0; fm < 4; ++fm)
-#pragma unroll
-        for (int fn = 0; fn < 2; ++fn)
-          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag[fm], b_frag[fn], acc[fm][fn], 0, 0, 0);
+        acc[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[fm], b_frag,
+                                                          acc[fm], 0, 0, 0);
     }
 #pragma unroll
     for (int fm = 0; fm < 4; ++fm)
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
-#pragma unroll
-        for (int rI = 0; rI < 4; ++rI) {
-          int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
-          int col = wc * 32 + fn * 16 + (lane & 15);
-          og[row * orr + col] = f2b(acc[fm][fn][rI]);
-        }
+      for (int rI = 0; rI < 4; ++rI) {
+        int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+        int col = wc * 16 + (lane & 15);
+        og[row * orr + col] = f2b(acc[fm][rI]);
+      }
   }
 }
 
@@ -475,7 +470,7 @@ void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
                     float scale) {
   NZ_CHECK(q.dtype() == DType::BF16, "fused attention: bf16 required");
   hipStream_t s = stream_of(q);
-  attn_fwd_fused_k<<<dim3((uint32_t)batch_heads), dim3(256), 0, s>>>(
+  attn_fwd_fused_k<<<dim3((uint32_t)batch_heads), dim3(512), 0, s>>>(
       (const uint16_t*)q.data(), (const uint16_t*)k.data(),
       (const uint16_t*)v.data(), (uint16_t*)probs.data(),
       (uint16_t*)ctx_out.data(), scale, 1, (int64_t)AS * AD, 0, AD,
@@ -492,7 +487,7 @@ void attn_fused_fwd_packed(const Tensor& qkv, Tensor& probs, Tensor& out,
   hipStream_t s = stream_of(qkv);
   int64_t H = heads * dh;
   const uint16_t* base = (const uint16_t*)qkv.data();
-  attn_fwd_fused_k<<<dim3((uint32_t)(B * heads)), dim3(256), 0, s>>>(
+  attn_fwd_fused_k<<<dim3((uint32_t)(B * heads)), dim3(512), 0, s>>>(
       base, base + H, base + 2 * H, (uint16_t*)probs.data(),
       (uint16_t*)out.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
       S * H, dh, H);
