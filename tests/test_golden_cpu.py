@@ -219,3 +219,39 @@ def test_dropout(cpu_ctx):
     # p=0 is identity with all-ones mask
     nz.dropout_fwd(x, y, mask, 0.0, seed=11)
     np.testing.assert_allclose(y.numpy(), X, rtol=1e-6)
+
+
+def test_attn_packed_cpu_fallback():
+    """ops::attn_fwd_packed / attn_bwd_packed CPU fallback vs a manual
+    torch composition of the same math (f32)."""
+    import torch
+    B, S, H, D = 2, 16, 2, 8
+    scale = 1.0 / np.sqrt(D)
+    rng = np.random.default_rng(3)
+    QKV = rng.normal(size=(B * S, 3 * H * D)).astype(np.float32)
+    DM = rng.normal(size=(B * S, H * D)).astype(np.float32)
+    ctx = nz.Ctx(nz.CPU)
+    qkv = ctx.from_numpy(QKV)
+    probs = ctx.tensor((B * H * S, S))
+    out = ctx.tensor((B * S, H * D))
+    nz.attn_fwd_packed(qkv, probs, out, B, S, H, D, scale)
+    # torch reference
+    qkv_t = torch.from_numpy(QKV).reshape(B, S, 3, H, D).permute(2, 0, 3, 1, 4)
+    q, k, v = qkv_t[0], qkv_t[1], qkv_t[2]  # [B,H,S,D]
+    q = q.clone().requires_grad_(True)
+    k = k.clone().requires_grad_(True)
+    v = v.clone().requires_grad_(True)
+    p_ref = torch.softmax(scale * (q @ k.transpose(-1, -2)), dim=-1)
+    o_ref = p_ref @ v
+    merged_ref = o_ref.permute(0, 2, 1, 3).reshape(B * S, H * D)
+    np.testing.assert_allclose(out.numpy(), merged_ref.detach().numpy(),
+                               rtol=1e-3, atol=1e-4)
+    dm_t = torch.from_numpy(DM)
+    merged_ref.backward(dm_t)
+    dqkv = ctx.tensor((B * S, 3 * H * D))
+    dm = ctx.from_numpy(DM)
+    nz.attn_bwd_packed(qkv, probs, dm, dqkv, B, S, H, D, scale)
+    dq_ref = torch.stack([q.grad, k.grad, v.grad])  # [3,B,H,S,D]
+    dqkv_ref = dq_ref.permute(1, 3, 0, 2, 4).reshape(B * S, 3 * H * D)
+    np.testing.assert_allclose(dqkv.numpy(), dqkv_ref.numpy(), rtol=1e-3,
+                               atol=1e-4)
