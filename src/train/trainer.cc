@@ -158,6 +158,12 @@ float Trainer::last_loss() {
   NZ_CHECK(loss_.defined(), "no step taken yet");
   float v = 0.f;
   loss_.download(&v, 4);
+  // under DP the reported loss is the mean over ranks (each rank trains a
+  // different shard — SURVEY.md §2.4 "loss/metric reduction" call site)
+  if (ddp_ && ddp_->comm() && ddp_->comm()->nranks() > 1) {
+    v = ddp_->comm()->all_reduce_scalar(v, ReduceOp::SUM) /
+        (float)ddp_->comm()->nranks();
+  }
   return v;
 }
 
