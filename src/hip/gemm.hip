@@ -570,6 +570,269 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_nt_pers_k(
 }
 
 // ---------------------------------------------------------------------------
+// RING persistent NT kernel: the latency-cover kernel at 2 WGs/CU.
+// Diagnosis (docs/perf): the 2-buffer persistent kernel's steady state is
+// ~1.2 us/iteration against 0.43 us of MFMA — each stage gets only ONE
+// compute-slot (~0.86 us across the 2 co-resident WGs) to cover ~1-2 us of
+// HBM latency. Deeper whole-tile buffering busts the 80 KB/WG budget that
+// 2 WGs/CU requires; HALF-tile rings fit exactly:
+//   A ring: 6 half-slots (64 rows x 64 k, 8 KB) = 3 k-tiles deep — A rows
+//           are compulsory HBM misses, they get ~2.5 iterations of cover;
+//   B ring: 4 half-slots = 2 k-tiles deep — B panels are L2-resident.
+// Total 48 + 32 = 80 KB. Iteration g: wait vmcnt(12) (the 6 younger
+// in-flight half-units, 2 glds each) -> compute k-tile g (4 waves, 64x64
+// quadrants, same XOR swizzle) -> lgkm barrier (all waves done reading g's
+// slots) -> stage A(g+3) both halves + B(g+2) both halves into the slots
+// g's data just vacated ((2g+h) mod ring); steady wait vmcnt(8). The
+// ring flows ACROSS tile
+// boundaries (persistent sweep, same XCD-chunked order as
+// gemm_bf16_nt_pers_k); at a boundary the wait is vmcnt(63) so the
+// epilogue's scalar stores (FIFO-younger than every needed glds) drain
+// under the next tile's compute. True tail: vmcnt(0).
+// ---------------------------------------------------------------------------
+__launch_bounds__(THREADS) __global__ void gemm_bf16_nt_ring_pers_k(
+    GemmArgs args, int nwg_n, int64_t per_b, int64_t total) {
+  constexpr int RBK = 64;
+  constexpr int HALF = 64 * RBK;              // elements per half-slot (8 KB)
+  __shared__ __attribute__((aligned(16))) uint16_t smem[10 * HALF];  // 80 KB
+  uint16_t* Aring = smem;                      // 6 slots
+  uint16_t* Bring = smem + 6 * HALF;           // 4 slots
+
+  const int orig = blockIdx.x;
+  const int xcd = orig & 7, slotq = orig >> 3;  // 64 slots per XCD
+  const int64_t q = total / 8, r8 = total % 8;
+  const int64_t lo = xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q;
+  const int64_t cnt = xcd < r8 ? q + 1 : q;
+  if (slotq >= cnt) return;
+  constexpr int S = 64;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  const int64_t ntiles = args.K / RBK;  // k-tiles per output tile
+  const int e0 = (int)threadIdx.x * 8;
+
+  f32x4 acc[4][4] = {};
+
+  // Tile states: staging reaches g+3 (A) so at most one boundary ahead.
+  struct TileSt {
+    const uint16_t* A;
+    const uint16_t* B;
+    int32_t m0, n0;
+  };
+  auto make_state = [&](int64_t tl) {
+    TileSt ts;
+    const int64_t b = tl / per_b, rtl = tl % per_b;
+    ts.A = args.A + b * args.sA;
+    ts.B = args.B + b * args.sB;
+    ts.m0 = (int32_t)((rtl / nwg_n) * BM);
+    ts.n0 = (int32_t)((rtl % nwg_n) * BN);
+    return ts;
+  };
+
+  // stage one half (h = 0/1) of operand A or B for LOCAL k-tile kt of
+  // tile-state ts, into ring slot rs. 2 glds per thread.
+  auto stage_half = [&](const TileSt& ts, int isB, int h, int rs, int64_t kt) {
+    const int64_t k0 = kt * RBK;
+    uint16_t* dst = (isB ? Bring : Aring) + rs * HALF;
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int e = e0 + c * 2048;
+      int row = e >> 6;  // 0..63 within the half
+      int k = (e & 63) ^ (((h * 64 + row) & 7) << 3);
+      if (!isB) {
+        int64_t ar = ts.m0 + h * 64 + row;
+        if (ar > Mc) ar = Mc;
+        glds16(ts.A + ar * lda + k0 + k, dst + e);
+      } else {
+        int64_t br = ts.n0 + h * 64 + row;
+        if (br > Nc) br = Nc;
+        glds16(ts.B + br * ldb + k0 + k, dst + e);
+      }
+    }
+  };
+
+  auto epilogue = [&](int64_t tl) {
+    const int64_t b = tl / per_b, rtl = tl % per_b;
+    const int64_t em0 = (rtl / nwg_n) * BM;
+    const int64_t en0 = (rtl % nwg_n) * BN;
+    uint16_t* C = args.C + b * args.sC;
+    float* out32 = args.out_f32 ? args.out_f32 + b * args.sC : nullptr;
+    uint16_t* save_pre = args.save_pre ? args.save_pre + b * args.sC : nullptr;
+    const int64_t N = args.N;
+    if (em0 + BM <= args.M && en0 + BN <= args.N) {
+      const int64_t base =
+          (em0 + wr * 64 + (lane >> 4) * 4) * N + en0 + wc * 64 + (lane & 15);
+      const uint16_t* bias_l =
+          args.bias ? args.bias + en0 + wc * 64 + (lane & 15) : nullptr;
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int rI = 0; rI < 4; ++rI) {
+          const int64_t rowbase = base + (int64_t)(fm * 16 + rI) * N;
+#pragma unroll
+          for (int fn = 0; fn < 4; ++fn) {
+            const int64_t idx = rowbase + fn * 16;
+            if (out32) {
+              out32[idx] = acc[fm][fn][rI];
+            } else {
+              float v = args.alpha * acc[fm][fn][rI];
+              if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+              if (bias_l) v += b2f(bias_l[fn * 16]);
+              if (save_pre) save_pre[idx] = f2b(v);
+              if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+              C[idx] = f2b(v);
+            }
+          }
+        }
+    } else {
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+          for (int rI = 0; rI < 4; ++rI) {
+            int64_t grow = em0 + wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+            int64_t gcol = en0 + wc * 64 + fn * 16 + (lane & 15);
+            if (grow < args.M && gcol < args.N) {
+              int64_t idx = grow * args.N + gcol;
+              if (out32) {
+                out32[idx] = acc[fm][fn][rI];
+              } else {
+                float v = args.alpha * acc[fm][fn][rI];
+                if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+                if (args.bias) v += b2f(args.bias[gcol]);
+                if (save_pre) save_pre[idx] = f2b(v);
+                if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+                C[idx] = f2b(v);
+              }
+            }
+          }
+    }
+#pragma unroll
+    for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) acc[fm][fn] = f32x4{};
+  };
+
+  // ---- persistent sweep with a GLOBAL k-iteration counter g (ring phase) --
+  int64_t tl = lo + slotq;
+  TileSt cur = make_state(tl);
+  const bool have2 = (tl + S) < (lo + cnt);
+  TileSt nxt = have2 ? make_state(tl + S) : cur;
+
+  // map global k-iter g (< 2 boundaries ahead) to (state, local kt);
+  // returns false when past the WG's work
+  const int64_t wg_iters = ((cnt - slotq + S - 1) / S) * ntiles;
+
+  // prologue: A(0) B(0) A(1) B(1) A(2) — 10 half-units, 20 glds; the first
+  // wait vmcnt(12) retires exactly A(0)+B(0)
+  stage_half(cur, 0, 0, 0, 0);
+  stage_half(cur, 0, 1, 1, 0);
+  stage_half(cur, 1, 0, 0, 0);
+  stage_half(cur, 1, 1, 1, 0);
+  stage_half(cur, 0, 0, 2, 1);
+  stage_half(cur, 0, 1, 3, 1);
+  stage_half(cur, 1, 0, 2, 1);
+  stage_half(cur, 1, 1, 3, 1);
+  stage_half(cur, 0, 0, 4, 2);
+  stage_half(cur, 0, 1, 5, 2);
+
+  int64_t g = 0;  // global k-iteration index for this WG
+  for (; tl < lo + cnt; tl += S) {
+    for (int64_t t = 0; t < ntiles; ++t, ++g) {
+      const bool boundary = (t == 0) && (g != 0);
+      const bool tail = (g + 2 >= wg_iters);  // B(g+2) and beyond skipped
+      if (tail) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      } else if (boundary) {
+        asm volatile("s_waitcnt vmcnt(63)" ::: "memory");
+      } else {
+        // younger-than-B(g) in the FIFO = B(g+1)+A(g+2) = 8 glds
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("" ::: "memory");
+
+      // compute k-tile g from ring slots (2g+h) mod 6 / mod 4
+      {
+        const int sa = (int)((2 * g) % 6), sb = (int)((2 * g) % 4);
+        const uint16_t* Ab0 = Aring + sa * HALF;
+        const uint16_t* Ab1 = Aring + ((sa + 1) % 6) * HALF;
+        const uint16_t* Bb0 = Bring + sb * HALF;
+        const uint16_t* Bb1 = Bring + ((sb + 1) % 4) * HALF;
+        const uint16_t* As = wr ? Ab1 : Ab0;   // wave's A half
+        const uint16_t* Bs = wc ? Bb1 : Bb0;   // wave's B half
+#pragma unroll
+        for (int kk = 0; kk < RBK; kk += 32) {
+          short8 a_frag[4], b_frag[4];
+          const int kbase = kk + (lane >> 4) * 8;
+#pragma unroll
+          for (int fm = 0; fm < 4; ++fm) {
+            int row = wr * 64 + fm * 16 + (lane & 15);  // global row for swz
+            int ks = kbase ^ ((row & 7) << 3);
+            a_frag[fm] = *(const short8*)(As + (row & 63) * RBK + ks);
+          }
+#pragma unroll
+          for (int fn = 0; fn < 4; ++fn) {
+            int col = wc * 64 + fn * 16 + (lane & 15);
+            int ks = kbase ^ ((col & 7) << 3);
+            b_frag[fn] = *(const short8*)(Bs + (col & 63) * RBK + ks);
+          }
+          __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+          for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+            for (int fn = 0; fn < 4; ++fn)
+              acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a_frag[fm], b_frag[fn], acc[fm][fn], 0, 0, 0);
+          __builtin_amdgcn_s_setprio(0);
+        }
+      }
+
+      // epilogue (tile's last k-iter) BEFORE the reuse barrier: its stores
+      // are FIFO-younger than every glds a later wait needs
+      if (t + 1 == ntiles) epilogue(tl);
+
+      // all waves done reading g's slots before they are restaged
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      asm volatile("" ::: "memory");
+
+      // stage A(g+3) and B(g+2) into the freed slots
+      {
+        const int64_t ga = g + 3, gb = g + 2;
+        if (ga < wg_iters) {
+          const int64_t ta = ga / ntiles - g / ntiles;  // 0=cur,1=nxt,2=skip
+          if (ta <= 1) {
+            const TileSt& ts = ta ? nxt : cur;
+            const int64_t kt = ga % ntiles;
+            const int rs = (int)((2 * ga) % 6);
+            stage_half(ts, 0, 0, rs, kt);
+            stage_half(ts, 0, 1, (rs + 1) % 6, kt);
+          }
+        }
+        if (gb < wg_iters) {
+          const int64_t tb = gb / ntiles - g / ntiles;
+          if (tb <= 1) {
+            const TileSt& ts = tb ? nxt : cur;
+            const int64_t kt = gb % ntiles;
+            const int rs = (int)((2 * gb) % 4);
+            stage_half(ts, 1, 0, rs, kt);
+            stage_half(ts, 1, 1, (rs + 1) % 4, kt);
+          }
+        }
+      }
+    }
+    cur = nxt;
+    const bool have_n2 = (tl + 2 * S) < (lo + cnt);
+    nxt = have_n2 ? make_state(tl + 2 * S) : cur;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Tier B+: deep-pipelined NT kernel. 256x128 tile, 512 threads (8 waves as
 // 4Mx2N, 64x64 per wave), BK=64, THREE LDS buffers (144 KB, 1 block/CU)
 // with counted s_waitcnt vmcnt(N) and raw s_barriers so two tiles of glds
@@ -2127,6 +2390,20 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
     // shapes and small batched tiles; neutral for long K.
     int64_t per_b = (int64_t)nwg_m * nwg_n;
     int64_t total = per_b * batch;
+    // Ring variant (half-tile rings, 3-tile A latency cover): measurement
+    // opt-in via NEZHA_GEMM_RING=1 — race-screen clean but measured ~1 us
+    // per k-iteration SLOWER than the 2-buffer persistent kernel at short
+    // K (367 vs 582 TF on qkv-fwd) and only equal at deep K; see
+    // docs/perf/README.md for the analysis.
+    static const bool kRingOn =
+        getenv("NEZHA_GEMM_RING") && atoi(getenv("NEZHA_GEMM_RING")) == 1;
+    if (kRingOn && !kPersOff && total >= 512 && !args.out_f32 &&
+        args.K / BK >= 3) {
+      gemm_bf16_nt_ring_pers_k<<<dim3(512), dim3(THREADS), 0, s>>>(
+          args, nwg_n, per_b, total);
+      NZ_KERNEL_CHECK();
+      return;
+    }
     if (!kPersOff && total >= 512 && !args.out_f32) {
       gemm_bf16_nt_pers_k<<<dim3(512), dim3(THREADS), 0, s>>>(args, nwg_n,
                                                               per_b, total);
