@@ -79,6 +79,7 @@ static int run_rank(const Flags& flags, int rank, int nranks, int uid_fd) {
   if (!ckpt_in.empty()) trainer.load_checkpoint(ckpt_in);
 
   MetricsWriter metrics(rank == 0 ? flags.get_str("metrics", "") : "");
+  if (rank == 0) flags.warn_unused();  // typo'd flags must not die silently
   for (int i = 0; i < warmup; ++i) trainer.step();
   ctx->sync();
   if (comm) comm->barrier();

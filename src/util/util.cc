@@ -25,6 +25,7 @@ void Flags::parse(int argc, const char** argv) {
 }
 
 std::string Flags::resolve(const std::string& k) const {
+  used_.insert(k);
   auto it = kv_.find(k);
   if (it != kv_.end()) return it->second;
   std::string env = "NEZHA_";
@@ -61,6 +62,14 @@ std::string Flags::dump_json() const {
   }
   os << "}";
   return os.str();
+}
+
+void Flags::warn_unused() const {
+  for (auto& [k, v] : kv_) {
+    if (!used_.count(k))
+      log_msg(LogLevel::WARN, "flag --%s=%s was never consulted (typo?)",
+              k.c_str(), v.c_str());
+  }
 }
 
 // ------------------------------------------------------------------- log ---

@@ -4,6 +4,7 @@
 
 #include <cstdarg>
 #include <map>
+#include <set>
 #include <sstream>
 #include <string>
 #include <vector>
@@ -23,9 +24,14 @@ class Flags {
   double get_float(const std::string& k, double dflt) const;
   bool get_bool(const std::string& k, bool dflt) const;
   std::string dump_json() const;  // resolved config, embedded in metrics
+  // Warn (rank-tagged) about parsed --flags no get_* ever consulted — a
+  // typo'd flag would otherwise be SILENTLY ignored (e.g. --ckpt-out vs
+  // --save). Call after all get_* lookups.
+  void warn_unused() const;
 
  private:
   std::map<std::string, std::string> kv_;
+  mutable std::set<std::string> used_;
   std::string resolve(const std::string& k) const;  // CLI > env > ""
 };
 
