@@ -475,3 +475,22 @@ def test_ce_loss_bf16_gpu(hip_ctx):
     nz.ce_loss(lo, lab, loss, dl)
     assert abs(loss.numpy()[0] - ref.item()) < 2e-2
     check(dl.numpy(), lt.grad.numpy(), rtol=0.03)
+
+
+def test_gemm_fuzz_shapes(hip_ctx):
+    """Seeded random-shape sweep across dispatch tiers vs torch."""
+    import torch
+    rng = np.random.default_rng(1234)
+    for _ in range(8):
+        M = int(rng.integers(1, 40)) * 64
+        N = int(rng.integers(1, 40)) * 64
+        K = int(rng.integers(1, 24)) * 64
+        ta, tb = bool(rng.integers(0, 2)), bool(rng.integers(0, 2))
+        A = randb(*(K, M) if ta else (M, K))
+        B = randb(*(N, K) if tb else (K, N))
+        a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
+        c = hip_ctx.tensor((M, N), nz.BF16)
+        nz.gemm(a, b, c, ta, tb)
+        At = torch.from_numpy(A.T if ta else A)
+        Bt = torch.from_numpy(B.T if tb else B)
+        check(c.numpy(), (At @ Bt).numpy(), rtol=0.035)

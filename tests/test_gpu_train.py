@@ -195,3 +195,15 @@ def test_ddp_8rank_one_device():
     binp = os.path.join(repo, "build", "test_ddp_fake_gpu")
     r = subprocess.run([binp], capture_output=True, timeout=600)
     assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
+
+
+def test_bert_noncontract_seq(hip_ctx):
+    """seq=64 routes attention through the composed fallback on GPU
+    (fused kernels are gated to the S=128 contract shape)."""
+    tr = nz.Trainer.bert(hip_ctx, layers=2, hidden=256, ffn=1024, heads=4,
+                         seq=64, batch=8, lr=1e-4, seed=11)
+    tr.step()
+    l0 = tr.loss()
+    for _ in range(9):
+        tr.step()
+    assert np.isfinite(l0) and tr.loss() < l0
