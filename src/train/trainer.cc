@@ -252,8 +252,18 @@ void Trainer::load_checkpoint(const std::string& path) {
         size_t off = size_t(b.offsets[i]) * es;
         auto& pa = fetch("param/" + p->name);
         NZ_CHECK(pa.numel() == p->numel(), "shape mismatch for " + p->name);
+        // dtype must match exactly: the memcpy below trusts the byte count,
+        // so an f32 checkpoint loaded into bf16 params (e.g. a CPU-saved
+        // file on GPU) would otherwise overflow the staging buffer
+        NZ_CHECK(pa.dtype == p->data.dtype(),
+                 "dtype mismatch for " + p->name +
+                     " (checkpoint saved on a different device class?)");
+        NZ_CHECK(pa.data.size() == size_t(p->numel()) * es,
+                 "byte-size mismatch for " + p->name);
         std::memcpy(data_host.data() + off, pa.data.data(), pa.data.size());
         auto& ma = fetch("master/" + p->name);
+        NZ_CHECK(ma.data.size() == size_t(p->numel()) * 4,
+                 "master byte-size mismatch for " + p->name);
         std::memcpy(master_host.data() + b.offsets[i] * 4, ma.data.data(),
                     ma.data.size());
         if (!m_host.empty()) {

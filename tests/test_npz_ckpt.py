@@ -4,6 +4,7 @@ import hashlib
 import zipfile
 
 import numpy as np
+import pytest
 
 import nezha_amd as nz
 
@@ -80,3 +81,16 @@ def test_ckpt_resume_determinism(cpu_ctx, tmp_path):
     ca, cb = np.load(final_a), np.load(final_b)
     for k in ca.files:
         np.testing.assert_array_equal(ca[k], cb[k], err_msg=k)
+
+
+def test_load_shape_mismatch_rejected(tmp_path):
+    """Loading a checkpoint into a differently-shaped model must raise a
+    clean error, not corrupt memory."""
+    ctx = nz.Ctx(nz.CPU)
+    a = nz.Trainer.mlp(ctx, layers=2, dim=64, batch=16, lr=1e-3, seed=1)
+    a.step()
+    p = str(tmp_path / "a.npz")
+    a.save(p)
+    b = nz.Trainer.mlp(ctx, layers=2, dim=128, batch=16, lr=1e-3, seed=1)
+    with pytest.raises(nz.NezhaError, match="mismatch|missing"):
+        b.load(p)
