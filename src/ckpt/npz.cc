@@ -173,8 +173,15 @@ std::map<std::string, NpyArray> npz_load(const std::string& path) {
   uint32_t cd_off = rd32(&buf[eocd + 16]);
 
   std::map<std::string, NpyArray> out;
+  // every offset below comes from the (untrusted) file: bound-check before
+  // dereferencing so a truncated/corrupt checkpoint errors instead of OOB
+  auto inb = [&](size_t off, size_t need) {
+    NZ_CHECK(off <= (size_t)fsize && need <= (size_t)fsize - off,
+             "corrupt zip: offset out of range in " + path);
+  };
   size_t p = cd_off;
   for (int e = 0; e < nent; ++e) {
+    inb(p, 46);
     NZ_CHECK(rd32(&buf[p]) == 0x02014b50, "bad central directory");
     uint16_t method = rd16(&buf[p + 10]);
     uint32_t csize = rd32(&buf[p + 20]);
@@ -183,14 +190,17 @@ std::map<std::string, NpyArray> npz_load(const std::string& path) {
     uint16_t xlen = rd16(&buf[p + 30]);
     uint16_t clen = rd16(&buf[p + 32]);
     uint32_t lho = rd32(&buf[p + 42]);
+    inb(p + 46, (size_t)nlen + xlen + clen);
     std::string name((const char*)&buf[p + 46], nlen);
     p += 46 + nlen + xlen + clen;
     NZ_CHECK(method == 0, "npz member " + name + " is compressed (unsupported)");
     // local header
+    inb(lho, 30);
     NZ_CHECK(rd32(&buf[lho]) == 0x04034b50, "bad local header");
     uint16_t lnlen = rd16(&buf[lho + 26]);
     uint16_t lxlen = rd16(&buf[lho + 28]);
-    size_t data_off = lho + 30 + lnlen + lxlen;
+    size_t data_off = lho + 30 + (size_t)lnlen + lxlen;
+    inb(data_off, csize);
     std::vector<char> member(buf.begin() + data_off, buf.begin() + data_off + csize);
     uint32_t crc = crc32(0, (const Bytef*)member.data(), (uInt)member.size());
     NZ_CHECK(crc == crc_expect, "crc mismatch in member " + name);

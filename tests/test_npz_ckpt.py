@@ -94,3 +94,26 @@ def test_load_shape_mismatch_rejected(tmp_path):
     b = nz.Trainer.mlp(ctx, layers=2, dim=128, batch=16, lr=1e-3, seed=1)
     with pytest.raises(nz.NezhaError, match="mismatch|missing"):
         b.load(p)
+
+
+def test_corrupt_npz_rejected(tmp_path):
+    """Truncated / bit-flipped checkpoints must raise clean errors."""
+    ctx = nz.Ctx(nz.CPU)
+    a = nz.Trainer.mlp(ctx, layers=2, dim=64, batch=16, lr=1e-3, seed=1)
+    a.step()
+    p = tmp_path / "a.npz"
+    a.save(str(p))
+    raw = p.read_bytes()
+    # truncation
+    (tmp_path / "t.npz").write_bytes(raw[: len(raw) // 2])
+    b = nz.Trainer.mlp(ctx, layers=2, dim=64, batch=16, lr=1e-3, seed=1)
+    with pytest.raises(nz.NezhaError):
+        b.load(str(tmp_path / "t.npz"))
+    # bit flip in the middle (crc or structure must catch it)
+    flipped = bytearray(raw)
+    flipped[len(raw) // 3] ^= 0xFF
+    (tmp_path / "f.npz").write_bytes(bytes(flipped))
+    with pytest.raises(nz.NezhaError):
+        b.load(str(tmp_path / "f.npz"))
+    # b must still be usable after failed loads
+    b.step()
