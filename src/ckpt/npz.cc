@@ -40,10 +40,12 @@ static bool parse_npy(const std::vector<char>& buf, NpyArray& out) {
     hlen = (uint8_t)buf[8] | ((uint8_t)buf[9] << 8);
     hoff = 10;
   } else {  // v2: 4-byte header len
+    if (buf.size() < 12) return false;
     hlen = (uint8_t)buf[8] | ((uint8_t)buf[9] << 8) | ((uint8_t)buf[10] << 16) |
            ((uint8_t)buf[11] << 24);
     hoff = 12;
   }
+  if (hlen > buf.size() || hoff > buf.size() - hlen) return false;  // corrupt
   std::string dict(buf.begin() + hoff, buf.begin() + hoff + hlen);
   auto find_val = [&](const std::string& key) -> std::string {
     auto p = dict.find("'" + key + "'");
@@ -64,7 +66,9 @@ static bool parse_npy(const std::vector<char>& buf, NpyArray& out) {
   while (pos < inner.size()) {
     while (pos < inner.size() && !isdigit((unsigned char)inner[pos])) ++pos;
     if (pos >= inner.size()) break;
-    out.shape.push_back(strtoll(inner.c_str() + pos, nullptr, 10));
+    long long d = strtoll(inner.c_str() + pos, nullptr, 10);
+    if (d < 0 || d > (1ll << 40)) return false;  // corrupt/absurd dim
+    out.shape.push_back(d);
     while (pos < inner.size() && isdigit((unsigned char)inner[pos])) ++pos;
   }
   size_t payload = hoff + hlen;

@@ -181,6 +181,10 @@ Tensor::Buf::~Buf() {
 
 Tensor::Tensor(Context* ctx, DType dt, Shape shape)
     : ctx_(ctx), dtype_(dt), shape_(std::move(shape)) {
+  for (int64_t d : shape_.dims)
+    NZ_CHECK(d >= 0, "negative tensor dimension " + std::to_string(d));
+  NZ_CHECK(shape_.numel() >= 0 && shape_.numel() < (int64_t(1) << 40),
+           "tensor numel out of range");
   buf_ = std::make_shared<Buf>();
   buf_->ctx = ctx;
   buf_->p = ctx->arena()->alloc(bytes());
