@@ -56,13 +56,23 @@ __device__ void stage_a(const uint16_t* __restrict__ A, uint16_t* __restrict__ l
       int m = e / BK, k = e % BK;
       int64_t gm = m0 + m, gk = k0 + k;
       short8 v = {};
-      if (gm < M) v = *(const short8*)(A + gm * lda + gk);
+      if (gm < M) {
+        if (gk + 7 < K) {
+          v = *(const short8*)(A + gm * lda + gk);
+        } else {
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            v[i] = (gk + i < K) ? (short)A[gm * lda + gk + i] : (short)0;
+        }
+      }
       *(short8*)(lds + m * LDS_STRIDE + k) = v;
     } else {
       int k = e / BM, m = e % BM;
       int64_t gk = k0 + k, gm = m0 + m;
       short8 v = {};
-      if (gm + 7 < M) {
+      if (gk >= K) {
+        // ragged-K zero row
+      } else if (gm + 7 < M) {
         v = *(const short8*)(A + gk * lda + gm);
       } else {
 #pragma unroll
@@ -89,7 +99,9 @@ __device__ void stage_b(const uint16_t* __restrict__ B, uint16_t* __restrict__ l
       int k = e / BN, n = e % BN;
       int64_t gk = k0 + k, gn = n0 + n;
       short8 v = {};
-      if (gn + 7 < N) {
+      if (gk >= K) {
+        // ragged-K zero row
+      } else if (gn + 7 < N) {
         v = *(const short8*)(B + gk * ldb + gn);
       } else {
 #pragma unroll
@@ -102,7 +114,15 @@ __device__ void stage_b(const uint16_t* __restrict__ B, uint16_t* __restrict__ l
       int n = e / BK, k = e % BK;
       int64_t gn = n0 + n, gk = k0 + k;
       short8 v = {};
-      if (gn < N) v = *(const short8*)(B + gn * ldb + gk);
+      if (gn < N) {
+        if (gk + 7 < K) {
+          v = *(const short8*)(B + gn * ldb + gk);
+        } else {
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            v[i] = (gk + i < K) ? (short)B[gn * ldb + gk + i] : (short)0;
+        }
+      }
       *(short8*)(lds + n * LDS_STRIDE + k) = v;
     }
   }
@@ -127,7 +147,7 @@ __launch_bounds__(THREADS) __global__ void gemm_bf16_k(GemmArgs args) {
 
   f32x4 acc[4][4] = {};
 
-  const int64_t ntiles = args.K / BK;
+  const int64_t ntiles = (args.K + BK - 1) / BK;  // ragged K: zero-padded
   stage_a<TA>(A, As[0], m0, 0, args.M, args.K, args.lda);
   stage_b<TB>(B, Bs[0], n0, 0, args.N, args.K, args.ldb);
   __syncthreads();
@@ -2198,8 +2218,22 @@ thread_local Context* g_splitk_ctx = nullptr;
 
 void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
                  hipStream_t s) {
-  NZ_CHECK(args.K % BK == 0, "hip gemm requires K % 64 == 0, got K=" +
-                                 std::to_string(args.K));
+  if (args.K % BK != 0) {
+    // ragged K (non-contract dims): the generic tiled kernel zero-pads the
+    // last k-block at the staging loads; all four transpose forms
+    dim3 grid((args.N + BN - 1) / BN, (args.M + BM - 1) / BM, batch);
+    dim3 block(THREADS);
+    if (!transA && !transB)
+      gemm_bf16_k<false, false><<<grid, block, 0, s>>>(args);
+    else if (!transA && transB)
+      gemm_bf16_k<false, true><<<grid, block, 0, s>>>(args);
+    else if (transA && !transB)
+      gemm_bf16_k<true, false><<<grid, block, 0, s>>>(args);
+    else
+      gemm_bf16_k<true, true><<<grid, block, 0, s>>>(args);
+    NZ_KERNEL_CHECK();
+    return;
+  }
   static const int kNtVariant = [] {
     const char* v = getenv("NEZHA_GEMM_VARIANT");
     return v ? atoi(v) : 2;  // 0=2-buf 128x128, 1=3-buf deep, 2=8-phase 256^2

@@ -494,3 +494,18 @@ def test_gemm_fuzz_shapes(hip_ctx):
         At = torch.from_numpy(A.T if ta else A)
         Bt = torch.from_numpy(B.T if tb else B)
         check(c.numpy(), (At @ Bt).numpy(), rtol=0.035)
+
+
+@pytest.mark.parametrize("transA,transB", [(False, False), (False, True),
+                                           (True, False)])
+def test_gemm_ragged_k(hip_ctx, transA, transB):
+    """K not divisible by 64 (non-contract dims) routes to the generic
+    kernel with zero-padded staging."""
+    M, N, K = 192, 256, 224
+    A = randb(*(K, M) if transA else (M, K))
+    B = randb(*(N, K) if transB else (K, N))
+    ref = (A.T if transA else A) @ (B.T if transB else B)
+    a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
+    c = hip_ctx.tensor((M, N), nz.BF16)
+    nz.gemm(a, b, c, transA, transB)
+    check(c.numpy(), ref, rtol=0.03)
