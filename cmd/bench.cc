@@ -91,6 +91,30 @@ int main(int argc, const char** argv) {
     bench_gemm(ctx, 4096, 4096, 4096, false, false, "square-4k (NN)", iters);
     bench_gemm(ctx, 8192, 8192, 8192, false, false, "square-8k (NN)", 5);
   }
+  if (which == "all" || which == "attn") {
+    // fused attention pair at the contract shape (B=256, H=12, S=128, dh=64)
+    int64_t Bb = 256, H = 12, S = 128, D = 64, HID = H * D;
+    Tensor qkv(&ctx, DType::BF16, Shape{Bb * S, 3 * HID});
+    Tensor probs(&ctx, DType::BF16, Shape{Bb * H * S, S});
+    Tensor out(&ctx, DType::BF16, Shape{Bb * S, HID});
+    Tensor dm(&ctx, DType::BF16, Shape{Bb * S, HID});
+    Tensor dqkv(&ctx, DType::BF16, Shape{Bb * S, 3 * HID});
+    ops::rand_normal(qkv, 11, 0.f, 0.5f);
+    ops::rand_normal(dm, 12, 0.f, 0.5f);
+    float sc = 0.125f;
+    double fw = time_op(ctx, iters, [&] {
+      ops::attn_fwd_packed(qkv, probs, out, Bb, S, H, D, sc);
+    });
+    double bw = time_op(ctx, iters, [&] {
+      ops::attn_bwd_packed(qkv, probs, dm, dqkv, Bb, S, H, D, sc);
+    });
+    double fwf = 2.0 * Bb * H * (2.0 * S * S * D);            // QK^T + PV
+    double bwf = 2.0 * Bb * H * (4.0 * S * S * D);            // dP,dV,dQ,dK
+    printf("attn fwd  (fused, packed)  B=256 H=12 S=128 dh=64 %8.3f ms  %7.1f TF/s\n",
+           fw * 1e3, fwf / fw / 1e12);
+    printf("attn bwd  (fused, packed)  B=256 H=12 S=128 dh=64 %8.3f ms  %7.1f TF/s\n",
+           bw * 1e3, bwf / bw / 1e12);
+  }
   if (which == "all" || which == "mem") {
     int64_t n = 256 << 20;  // 256M bf16 elements = 512 MB
     Tensor x(&ctx, DType::BF16, Shape{n});
