@@ -216,3 +216,15 @@ for _ in range(3):
     for ln in lines:
         rec = _json.loads(ln)
         assert "loss" in rec and "step" in rec
+
+
+def test_bert_exotic_configs_cpu():
+    """Non-contract geometries (dh!=64, odd seq, ragged ffn) train on the
+    composed CPU path — mirrors the GPU fuzz in profiles/fuzz_models2.log."""
+    ctx = nz.Ctx(nz.CPU)
+    for kw in [dict(hidden=128, ffn=256, heads=4, seq=24),     # dh=32
+               dict(hidden=96, ffn=100, heads=2, seq=10)]:     # dh=48, ragged
+        tr = nz.Trainer.bert(ctx, layers=1, batch=2, lr=1e-3, seed=5, **kw)
+        tr.step()
+        tr.step()
+        assert np.isfinite(tr.loss())
