@@ -255,3 +255,28 @@ def test_attn_packed_cpu_fallback():
     dqkv_ref = dq_ref.permute(1, 3, 0, 2, 4).reshape(B * S, 3 * H * D)
     np.testing.assert_allclose(dqkv.numpy(), dqkv_ref.numpy(), rtol=1e-3,
                                atol=1e-4)
+
+
+def test_adam_weight_decay(cpu_ctx):
+    """wd is classic Adam+L2 (g += wd*p), matching torch.optim.Adam's
+    weight_decay (NOT AdamW's decoupled form)."""
+    n = 512
+    P, Gr = randf(n), randf(n)
+    lr, b1, b2, eps, wd = 1e-3, 0.9, 0.999, 1e-8, 0.05
+    pt = torch.from_numpy(P.copy()).requires_grad_(True)
+    opt = torch.optim.Adam([pt], lr=lr, betas=(b1, b2), eps=eps,
+                           weight_decay=wd)
+    for i in range(3):
+        pt.grad = torch.from_numpy(Gr.copy())
+        opt.step()
+
+    p = cpu_ctx.from_numpy(P.copy())
+    g = cpu_ctx.from_numpy(Gr.copy())
+    master = cpu_ctx.from_numpy(P.copy())
+    m = cpu_ctx.tensor((n,)).fill(0)
+    v = cpu_ctx.tensor((n,)).fill(0)
+    for i in range(3):
+        nz.adam_step(p, g, master, m, v, lr, b1, b2, eps, step=i + 1,
+                     weight_decay=wd)
+    np.testing.assert_allclose(p.numpy(), pt.detach().numpy(), rtol=1e-5,
+                               atol=1e-6)
