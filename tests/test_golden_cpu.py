@@ -280,3 +280,23 @@ def test_adam_weight_decay(cpu_ctx):
                      weight_decay=wd)
     np.testing.assert_allclose(p.numpy(), pt.detach().numpy(), rtol=1e-5,
                                atol=1e-6)
+
+
+def test_sgd_momentum(cpu_ctx):
+    """SGD momentum buffer matches torch.optim.SGD (v = mu*v + g; p -= lr*v)."""
+    n = 256
+    P, Gr = randf(n), randf(n)
+    lr, mu = 0.1, 0.9
+    pt = torch.from_numpy(P.copy()).requires_grad_(True)
+    opt = torch.optim.SGD([pt], lr=lr, momentum=mu)
+    for _ in range(3):
+        pt.grad = torch.from_numpy(Gr.copy())
+        opt.step()
+    p = cpu_ctx.from_numpy(P.copy())
+    g = cpu_ctx.from_numpy(Gr.copy())
+    master = cpu_ctx.from_numpy(P.copy())
+    mom = cpu_ctx.tensor((n,)).fill(0)
+    for _ in range(3):
+        nz.sgd_step(p, g, master, mom, lr=lr, mu=mu)
+    np.testing.assert_allclose(p.numpy(), pt.detach().numpy(), rtol=1e-6,
+                               atol=1e-7)
