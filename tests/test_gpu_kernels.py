@@ -52,13 +52,17 @@ def test_gemm_ragged_pers(hip_ctx):
 
 def test_gemm_ragged_8phase(hip_ctx):
     """Ragged M at a balanced >=1024-tile grid: the 256^2 kernel's clamped
-    staging + exterior epilogue path."""
+    staging + exterior epilogue path. The reference is computed on a row
+    slice (including the ragged tail) — the full 208-GFLOP numpy matmul
+    took ~40 s of test time."""
     M, N, K = 16512, 4096, 1536
     A, B = randb(M, K), randb(N, K)
     a, b = hip_ctx.from_numpy_bf16(A), hip_ctx.from_numpy_bf16(B)
     c = hip_ctx.tensor((M, N), nz.BF16)
     nz.gemm(a, b, c, False, True)
-    check(c.numpy(), A @ B.T, rtol=0.03)
+    rows = np.concatenate([np.arange(0, 128), np.arange(M - 192, M),
+                           RNG.integers(0, M, 128)])
+    check(c.numpy()[rows], A[rows] @ B.T, rtol=0.03)
 
 
 def test_gemm_large_shape(hip_ctx):
