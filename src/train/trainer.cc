@@ -69,18 +69,20 @@ void Trainer::step_body(const Tensor* bc_dev) {
     if (ddp_) ddp_->finish_grads();
     opt_->step(ddp_ ? ddp_->grad_scale() : 1.0f, bc_dev);
   }
-  if (mx) {
+  const bool mwrite =
+      metrics_ && (!ddp_ || ddp_->comm()->rank() == 0);  // one writer per file
+  if (mx && mwrite) {
     ev_opt_.record(*ctx_->compute_stream());
     ev_opt_.sync();
     metrics_->write({{"step", (double)opt_->step_count()},
-                     {"loss", last_loss()},
+                     {"loss", last_loss_local()},
                      {"fwd_ms", ev_fwd_.elapsed_ms_since(ev_start_)},
                      {"bwd_ms", ev_bwd_.elapsed_ms_since(ev_fwd_)},
                      {"opt_ms", ev_opt_.elapsed_ms_since(ev_bwd_)},
                      {"mem_hw_gb", ctx_->arena()->high_water() / 1e9}});
-  } else if (metrics_) {
+  } else if (mwrite) {
     metrics_->write({{"step", (double)opt_->step_count()},
-                     {"loss", last_loss()}});
+                     {"loss", last_loss_local()}});
   }
 }
 
@@ -154,10 +156,15 @@ void Trainer::step() {
   opt_->set_step_count(next);
 }
 
-float Trainer::last_loss() {
+float Trainer::last_loss_local() {
   NZ_CHECK(loss_.defined(), "no step taken yet");
   float v = 0.f;
   loss_.download(&v, 4);
+  return v;
+}
+
+float Trainer::last_loss() {
+  float v = last_loss_local();
   // under DP the reported loss is the mean over ranks (each rank trains a
   // different shard — SURVEY.md §2.4 "loss/metric reduction" call site)
   if (ddp_ && ddp_->comm() && ddp_->comm()->nranks() > 1) {

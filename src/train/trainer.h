@@ -26,7 +26,9 @@ class Trainer {
   // One training step: fwd -> loss -> bwd (-> overlapped all-reduce) -> opt.
   // Asynchronous on HIP (no sync, no download).
   void step();
-  float last_loss();   // downloads the loss scalar (syncs on HIP)
+  float last_loss();   // downloads the loss scalar (syncs on HIP);
+                       // COLLECTIVE under DDP (mean over ranks)
+  float last_loss_local();  // this rank's loss only (no collective)
   int step_count() const { return opt_->step_count(); }
 
   Model* model() { return model_.get(); }
