@@ -44,7 +44,7 @@ __device__ __forceinline__ void att_glds16(const uint16_t* gsrc,
 // straight from the packed qkv projection [B, S, 3H] (qr=3H, qh=64,
 // qb=S*3H, with k/v base pointers offset by H / 2H) — the latter removes
 // the attn_split/attn_merge layout passes entirely. Same for `out`.
-__launch_bounds__(512) __global__ void attn_fwd_fused_k(
+__launch_bounds__(512, 2) __global__ void attn_fwd_fused_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ probs,
     uint16_t* __restrict__ out, float scale, int heads, int64_t qb,
