@@ -20,6 +20,30 @@ int main(int argc, const char** argv) {
            prop.maxSharedMemoryPerMultiProcessor / 1024,
            prop.totalGlobalMem / 1e9, prop.clockRate / 1000);
   }
+  if (n > 1) {
+    // xGMI topology: P2P access + link type/hops between device pairs
+    printf("P2P topology (link:hops, '-' = no direct access):\n     ");
+    for (int j = 0; j < n; ++j) printf("%5d", j);
+    printf("\n");
+    for (int i = 0; i < n; ++i) {
+      printf("  %2d ", i);
+      for (int j = 0; j < n; ++j) {
+        if (i == j) {
+          printf("    .");
+          continue;
+        }
+        int access = 0;
+        (void)hipDeviceCanAccessPeer(&access, i, j);
+        uint32_t link = 0, hops = 0;
+        if (access &&
+            hipExtGetLinkTypeAndHopCount(i, j, &link, &hops) == hipSuccess)
+          printf("  %u:%u", link, hops);
+        else
+          printf("    -");
+      }
+      printf("\n");
+    }
+  }
   if (flags.get_bool("comm-check", false)) {
     if (n == 0) {
       printf("comm-check: no GPU\n");
