@@ -86,3 +86,13 @@ tests-asan: $(SAN_BUILD)/test_core $(SAN_BUILD)/test_ddp_fake
 
 clean:
 	rm -rf $(BUILD) $(SAN_BUILD)
+
+.PHONY: test help
+test: all  ## run the CPU test tiers (C++ binaries + pytest)
+	./build/test_core && ./build/test_modules && ./build/test_ddp_fake
+	python3 -m pytest tests -q -m "not gpu"
+
+help:  ## list targets
+	@echo "targets: all (lib+cmds+tests), lib, cmds, tests, test (run CPU tiers),"
+	@echo "         tests-asan (ASan/UBSan tier), yardstick (hipBLASLt compare), clean"
+	@echo "GPU runs go through gpurun: see tools/gpurun/*.sh and docs/perf/README.md"
