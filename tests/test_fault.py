@@ -75,3 +75,17 @@ def test_restart_from_checkpoint(tmp_path):
     cr, cb = np.load(ref_path), np.load(out)
     for k in cr.files:
         np.testing.assert_array_equal(cr[k], cb[k], err_msg=k)
+
+
+def test_flags_precedence():
+    """CLI --steps beats NEZHA_STEPS env; env applies when flag absent."""
+    import json
+    binp = os.path.join(REPO, "build", "nezha-train")
+    base = [binp, "--model=mlp", "--gpus=1", "--warmup=0", "--dim=64",
+            "--layers=1", "--batch=16"]
+    env = dict(os.environ, NEZHA_STEPS="3")
+    r = subprocess.run(base + ["--steps=2"], env=env, capture_output=True,
+                       timeout=120)
+    assert json.loads(r.stdout.decode().strip().splitlines()[-1])["steps"] == 2
+    r = subprocess.run(base, env=env, capture_output=True, timeout=120)
+    assert json.loads(r.stdout.decode().strip().splitlines()[-1])["steps"] == 3
