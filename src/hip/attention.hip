@@ -150,8 +150,9 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused_k(
     uint16_t* pw = Ps + row * PSTRIDE;
     pw[lane * 2] = o0;
     pw[lane * 2 + 1] = o1;
-    // global P (backward input), coalesced ushort2
-    *(ushort2*)(pg + row * AS + lane * 2) = make_ushort2(o0, o1);
+    // global P write deferred to after PV: a store here sits in the vmcnt
+    // FIFO that the next __syncthreads drains (measured SQ_WAIT_ANY =
+    // 12.7x MFMA cycles on this kernel)
   }
   __syncthreads();
 
@@ -185,6 +186,17 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused_k(
         int col = wc * 16 + (lane & 15);
         og[row * orr + col] = f2b(acc[fm][rI]);
       }
+  }
+
+  // ---- stream P (backward input) out from LDS: 16-B row-contiguous -------
+  // 128x128 elems / 512 threads = 32 = 4 x short8 per thread; the stores
+  // drain at kernel end, overlapped by other workgroups.
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    int e = (c * 512 + (int)threadIdx.x) * 8;
+    int row = e >> 7, col = e & 127;
+    *(short8*)(pg + row * AS + col) =
+        *(const short8*)(Ps + row * PSTRIDE + col);
   }
 }
 
