@@ -10,19 +10,25 @@ import numpy as np
 import nezha_amd as nz
 
 ctx = nz.Ctx(nz.HIP, 0)
-for (M, N, K) in [(512, 512, 512), (2048, 1024, 768), (4096, 4096, 4096),
-                  (8192, 4096, 4096)]:
-    A = ctx.tensor((M, K), nz.BF16).rand_normal(1)
-    B = ctx.tensor((N, K), nz.BF16).rand_normal(2)
+# shapes chosen to hit every dispatch tier: tier-A (512^3), plain nt_k
+# (2048x1024), 8-phase (4096^3, 8192-), PERSISTENT short-K (32768x2304x768)
+# and the split-K 8-phase wgrad path (TN 768x2304xK32768)
+for (M, N, K, tA) in [(512, 512, 512, False), (2048, 1024, 768, False),
+                      (4096, 4096, 4096, False), (8192, 4096, 4096, False),
+                      (32768, 2304, 768, False), (768, 2304, 32768, True)]:
+    A = ctx.tensor((K, M) if tA else (M, K), nz.BF16).rand_normal(1)
+    B = ctx.tensor((K, N) if tA else (N, K), nz.BF16).rand_normal(2)
     outs = []
     for rep in range(3):
         C = ctx.tensor((M, N), nz.BF16)
-        nz.gemm(A, B, C, False, True)
+        nz.gemm(A, B, C, tA, not tA)
         outs.append(C.raw())
     for rep in range(1, 3):
         assert np.array_equal(outs[0], outs[rep]), f"NONDET {M}x{N}x{K} rep{rep}"
-    if M * N <= 2048 * 1024:  # oracle check on the smaller shapes
-        ref = A.numpy().astype(np.float32) @ B.numpy().astype(np.float32).T
+    if M * N <= 2048 * 1024 and K <= 1024:  # oracle on small+cheap shapes
+        a_np = A.numpy().astype(np.float32)
+        b_np = B.numpy().astype(np.float32)
+        ref = (a_np.T @ b_np) if tA else (a_np @ b_np.T)
         got = nz.bf16_to_f32(outs[0])
         scale = np.abs(ref).max()
         err = np.abs(got - ref).max() / scale
