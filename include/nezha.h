@@ -81,6 +81,7 @@ nz_status nz_op_mse_loss(nz_tensor* y, nz_tensor* target, nz_tensor* loss,
                          nz_tensor* dy, float grad_scale);
 nz_status nz_op_ce_loss(nz_tensor* logits, nz_tensor* labels, nz_tensor* loss,
                         nz_tensor* dlogits, float grad_scale);
+nz_status nz_op_add_out(nz_tensor* out, nz_tensor* x, nz_tensor* y);
 nz_status nz_op_add_inplace(nz_tensor* dst, nz_tensor* src, float alpha);
 nz_status nz_op_scale_inplace(nz_tensor* dst, float alpha);
 nz_status nz_op_cast_copy(nz_tensor* dst, nz_tensor* src);

@@ -79,6 +79,7 @@ _lib.nz_op_dropout_bwd.argtypes = [ctypes.c_void_p] * 3 + [ctypes.c_float]
 _lib.nz_op_mse_loss.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_float]
 _lib.nz_op_ce_loss.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_float]
 _lib.nz_op_add_inplace.argtypes = [ctypes.c_void_p] * 2 + [ctypes.c_float]
+_lib.nz_op_add_out.argtypes = [ctypes.c_void_p] * 3
 _lib.nz_op_scale_inplace.argtypes = [ctypes.c_void_p, ctypes.c_float]
 _lib.nz_op_cast_copy.argtypes = [ctypes.c_void_p] * 2
 _lib.nz_op_transpose2d.argtypes = [ctypes.c_void_p] * 2
@@ -330,6 +331,10 @@ def mse_loss(y, target, loss, dy=None, grad_scale=1.0):
 def ce_loss(logits, labels, loss, dlogits=None, grad_scale=1.0):
     _ck(_lib.nz_op_ce_loss(_h(logits), _h(labels), _h(loss), _h(dlogits),
                            grad_scale))
+
+
+def add_out(out, x, y):
+    _ck(_lib.nz_op_add_out(_h(out), _h(x), _h(y)))
 
 
 def add_inplace(dst, src, alpha=1.0):

@@ -219,6 +219,11 @@ extern "C" nz_status nz_op_ce_loss(nz_tensor* logits, nz_tensor* labels,
   ops::ce_loss(logits->t, labels->t, loss->t, dlogits ? &dlogits->t : nullptr, gs);
   NZ_API_END
 }
+extern "C" nz_status nz_op_add_out(nz_tensor* out, nz_tensor* x, nz_tensor* y) {
+  NZ_API_BEGIN
+  ops::add_out(out->t, x->t, y->t);
+  NZ_API_END
+}
 extern "C" nz_status nz_op_add_inplace(nz_tensor* dst, nz_tensor* src, float a) {
   NZ_API_BEGIN
   ops::add_inplace(dst->t, src->t, a);
