@@ -32,11 +32,23 @@ struct Param {
   }
 };
 
+// Guard for residual_add's grad aliasing (modules.cc): when one grad
+// buffer is aliased into both inputs of a residual, any accumulation into
+// the trunk input mutates the shared buffer — the branch input's grad must
+// be fully consumed BEFORE that happens. The guard turns a violation of
+// that replay-order invariant into a hard error instead of silent grad
+// corruption (a third model wired differently would otherwise train wrong).
+struct GradAliasGuard {
+  bool mutated = false;  // some alias holder accumulated into the buffer
+};
+
 struct Var {
   Tensor value;
   Tensor grad;            // lazily allocated, same dtype as value
   bool grad_init = false;
   bool needs_grad = true; // false for the input leaf
+  std::shared_ptr<GradAliasGuard> alias_guard;  // set by residual_add
+  bool alias_trunk = false;  // this var is the residual trunk input
 
   Tensor& ensure_grad() {
     if (!grad.defined())
@@ -47,7 +59,22 @@ struct Var {
     ensure_grad();
     float b = grad_init ? 1.0f : 0.0f;
     grad_init = true;
+    if (b == 1.0f) note_grad_mutation();
     return b;
+  }
+  // Every in-place accumulation into an existing grad must pass through
+  // grad_beta() or call this explicitly.
+  void note_grad_mutation() {
+    if (alias_guard) alias_guard->mutated = true;
+  }
+  // Upstream-grad read used by backward closures: fails if this var's grad
+  // aliases a residual buffer that has already been mutated by the trunk.
+  Tensor& grad_for_read() {
+    NZ_CHECK(grad.defined(), "backward without upstream grad");
+    NZ_CHECK(!alias_guard || alias_trunk || !alias_guard->mutated,
+             "residual grad alias read after trunk mutation (replay-order "
+             "invariant violated — model wiring incompatible with aliasing)");
+    return grad;
   }
 };
 using VarPtr = std::shared_ptr<Var>;

@@ -82,12 +82,16 @@ class FakeComm : public Comm {
       std::vector<void*> hptrs(ptrs.size());
       for (size_t i = 0; i < ptrs.size(); ++i) {
         host[i].resize(bytes);
-        (void)hipMemcpy(host[i].data(), ptrs[i], bytes, hipMemcpyDeviceToHost);
+        NZ_CHECK(hipMemcpy(host[i].data(), ptrs[i], bytes,
+                           hipMemcpyDeviceToHost) == hipSuccess,
+                 "FakeComm D2H staging copy failed");
         hptrs[i] = host[i].data();
       }
       reduce_buffers<float>(hptrs, count, op, dt == DType::BF16);
       for (size_t i = 0; i < ptrs.size(); ++i)
-        (void)hipMemcpy(ptrs[i], host[i].data(), bytes, hipMemcpyHostToDevice);
+        NZ_CHECK(hipMemcpy(ptrs[i], host[i].data(), bytes,
+                           hipMemcpyHostToDevice) == hipSuccess,
+                 "FakeComm H2D staging copy failed");
     });
   }
 
@@ -102,10 +106,14 @@ class FakeComm : public Comm {
         return;
       }
       std::vector<char> host(bytes);
-      (void)hipMemcpy(host.data(), ptrs[root], bytes, hipMemcpyDeviceToHost);
+      NZ_CHECK(hipMemcpy(host.data(), ptrs[root], bytes,
+                         hipMemcpyDeviceToHost) == hipSuccess,
+               "FakeComm D2H broadcast copy failed");
       for (int i = 0; i < (int)ptrs.size(); ++i)
         if (i != root)
-          (void)hipMemcpy(ptrs[i], host.data(), bytes, hipMemcpyHostToDevice);
+          NZ_CHECK(hipMemcpy(ptrs[i], host.data(), bytes,
+                             hipMemcpyHostToDevice) == hipSuccess,
+                   "FakeComm H2D broadcast copy failed");
     });
   }
 
@@ -114,8 +122,13 @@ class FakeComm : public Comm {
   }
 
   float all_reduce_scalar(float v, ReduceOp op) override {
+    // The scalar buffer is HOST memory even when this comm carries device
+    // buffers — force the host reduce path (the device staging path would
+    // hipMemcpy D2H from a host stack pointer and return garbage).
     float buf = v;
-    all_reduce(&buf, 1, DType::F32, op, nullptr);
+    st_->rendezvous(rank_, &buf, [&](std::vector<void*>& ptrs) {
+      reduce_buffers<float>(ptrs, 1, op, /*bf16=*/false);
+    });
     return buf;
   }
 

@@ -1126,7 +1126,8 @@ __global__ void adam_k(TP* __restrict__ param, const TG* __restrict__ grad,
                        float* __restrict__ master, float* __restrict__ m,
                        float* __restrict__ v, int64_t n, float lr, float b1,
                        float b2, float eps, float bc1, float bc2, float gs,
-                       float wd) {
+                       float wd, const float* __restrict__ bc_dev) {
+  if (bc_dev) { bc1 = bc_dev[0]; bc2 = bc_dev[1]; }  // graph-replay path
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
@@ -1156,15 +1157,16 @@ void adam_step(Tensor& param, const Tensor& grad, Tensor& master, Tensor& m,
         b1, b2, eps, bc1, bc2, gs, wd,
         bc_dev ? (const float*)bc_dev->data() : nullptr);
   else if (param.dtype() == DType::BF16)
-    // scalar fallback ignores bc_dev (not used on the graph path)
     adam_k<uint16_t, uint16_t><<<g, b, 0, s>>>(
         (uint16_t*)param.data(), (const uint16_t*)grad.data(),
         (float*)master.data(), (float*)m.data(), (float*)v.data(), n, lr, b1, b2,
-        eps, bc1, bc2, gs, wd);
+        eps, bc1, bc2, gs, wd,
+        bc_dev ? (const float*)bc_dev->data() : nullptr);
   else
     adam_k<float, float><<<g, b, 0, s>>>(
         (float*)param.data(), (const float*)grad.data(), (float*)master.data(),
-        (float*)m.data(), (float*)v.data(), n, lr, b1, b2, eps, bc1, bc2, gs, wd);
+        (float*)m.data(), (float*)v.data(), n, lr, b1, b2, eps, bc1, bc2, gs, wd,
+        bc_dev ? (const float*)bc_dev->data() : nullptr);
   NZ_KERNEL_CHECK();
 }
 

@@ -68,14 +68,16 @@ void BertEncoderModel::collect_params(std::vector<Param*>& out) {
 }
 
 double BertEncoderModel::flops_per_step() const {
-  // GEMM flops only, fwd+bwd ~ 3x fwd.
+  // GEMM flops only, fwd+bwd ~ 3x fwd — minus layer 0's qkv-proj dgrad,
+  // which Linear backward skips (the data input needs no grad).
   double T = double(cfg_.batch) * cfg_.seq;
+  double qkv = 2.0 * T * cfg_.hidden * (3.0 * cfg_.hidden);
   double per_layer =
-      2.0 * T * cfg_.hidden * (3.0 * cfg_.hidden)        // qkv proj
+      qkv                                                // qkv proj
       + 2.0 * T * cfg_.seq * cfg_.hidden * 2.0           // QK^T and P.V
       + 2.0 * T * cfg_.hidden * cfg_.hidden              // out proj
       + 2.0 * T * cfg_.hidden * cfg_.ffn * 2.0;          // ffn pair
-  return 3.0 * per_layer * double(cfg_.layers);
+  return 3.0 * per_layer * double(cfg_.layers) - qkv;
 }
 
 }  // namespace nz

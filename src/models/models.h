@@ -42,7 +42,10 @@ class MlpModel : public Model {
   Shape output_shape() const override { return Shape{batch_, dim_}; }
   int64_t samples_per_batch() const override { return batch_; }
   double flops_per_step() const override {
-    return 3.0 * 2.0 * double(batch_) * dim_ * dim_ * double(layers_.size());
+    // 3 GEMMs/layer (fwd, wgrad, dgrad) EXCEPT layer 0, whose dgrad is
+    // skipped because the input needs no grad (modules.cc Linear backward).
+    return (3.0 * double(layers_.size()) - 1.0) * 2.0 * double(batch_) *
+           dim_ * dim_;
   }
 
  private:

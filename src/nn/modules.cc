@@ -48,8 +48,7 @@ VarPtr Linear::forward(Tape& tape, VarPtr x) {
   tape.push([this, x, y, z]() {
     Context* ctx = ctx_;
     DType pd = param_dtype(ctx);
-    Tensor& dy = y->grad;
-    NZ_CHECK(dy.defined(), "linear backward without upstream grad");
+    Tensor& dy = y->grad_for_read();
     Tensor dz;
     if (fuse_gelu) {
       dz = Tensor(ctx, pd, dy.shape());
@@ -102,8 +101,7 @@ VarPtr LayerNorm::forward(Tape& tape, VarPtr x) {
   ops::layernorm_fwd(x->value, gamma.data, beta.data, y->value, mean, rstd, 1e-5f);
 
   tape.push([this, x, y, mean, rstd]() {
-    Tensor& dy = y->grad;
-    NZ_CHECK(dy.defined(), "layernorm backward without upstream grad");
+    Tensor& dy = y->grad_for_read();
     NZ_CHECK(!gamma.grad_init && !beta.grad_init, "LN params used twice");
     if (x->needs_grad) {
       x->ensure_grad();
@@ -149,8 +147,7 @@ VarPtr MultiHeadAttention::forward(Tape& tape, VarPtr x) {
   ops::attn_fwd_packed(qkv->value, probs, merged->value, B, S, Hh, D, scale);
 
   tape.push([qkv, merged, probs, B, S, Hh, D, scale]() {
-    Tensor& dmerged = merged->grad;
-    NZ_CHECK(dmerged.defined(), "attention backward without upstream grad");
+    Tensor& dmerged = merged->grad_for_read();
     qkv->ensure_grad();
     NZ_CHECK(!qkv->grad_init, "qkv grad written twice");
     ops::attn_bwd_packed(qkv->value, probs, dmerged, qkv->grad, B, S, Hh, D,
@@ -166,15 +163,16 @@ VarPtr Gelu::forward(Tape& tape, VarPtr x) {
   Tensor z = x->value;  // pre-activation IS the input (no bias)
   ops::bias_gelu_fwd(x->value, nullptr, y->value, nullptr);
   tape.push([this, x, y, z]() {
-    NZ_CHECK(y->grad.defined(), "gelu backward without upstream grad");
+    Tensor& dy = y->grad_for_read();
     if (!x->needs_grad) return;
     x->ensure_grad();
     if (!x->grad_init) {
-      ops::bias_gelu_bwd(y->grad, z, x->grad, nullptr);
+      ops::bias_gelu_bwd(dy, z, x->grad, nullptr);
       x->grad_init = true;
     } else {
       Tensor dz(ctx_, z.dtype(), z.shape());
-      ops::bias_gelu_bwd(y->grad, z, dz, nullptr);
+      ops::bias_gelu_bwd(dy, z, dz, nullptr);
+      x->note_grad_mutation();
       ops::add_inplace(x->grad, dz, 1.0f);
     }
   });
@@ -187,15 +185,16 @@ VarPtr Softmax::forward(Tape& tape, VarPtr x) {
   float sc = scale_;
   Context* ctx = ctx_;
   tape.push([ctx, sc, x, y]() {
-    NZ_CHECK(y->grad.defined(), "softmax backward without upstream grad");
+    Tensor& dy = y->grad_for_read();
     if (!x->needs_grad) return;
     x->ensure_grad();
     if (!x->grad_init) {
-      ops::softmax_bwd(y->grad, y->value, x->grad, sc);
+      ops::softmax_bwd(dy, y->value, x->grad, sc);
       x->grad_init = true;
     } else {
       Tensor dx(ctx, x->value.dtype(), x->value.shape());
-      ops::softmax_bwd(y->grad, y->value, dx, sc);
+      ops::softmax_bwd(dy, y->value, dx, sc);
+      x->note_grad_mutation();
       ops::add_inplace(x->grad, dx, 1.0f);
     }
   });
@@ -208,21 +207,28 @@ VarPtr residual_add(Tape& tape, VarPtr a, VarPtr b) {
   auto out = make_var(Tensor(ctx, a->value.dtype(), a->value.shape()));
   ops::add_out(out->value, a->value, b->value);
   tape.push([a, b, out]() {
-    NZ_CHECK(out->grad.defined(), "residual backward without upstream grad");
+    Tensor& dout = out->grad_for_read();
     // ALIAS out's grad into uninitialized inputs instead of copying: out is
     // dead after this entry, and the replay order guarantees b's grad is
     // fully consumed (start of the sublayer's reverse chain) before any
-    // accumulation into a's grad mutates the shared buffer.
-    auto attach = [&](Var& v) {
+    // accumulation into a's grad mutates the shared buffer. The invariant
+    // is ENFORCED: a GradAliasGuard is planted on both holders, every
+    // accumulation marks it (Var::grad_beta / note_grad_mutation), and any
+    // later read of the branch's aliased grad fails loudly (tape.h).
+    auto guard = std::make_shared<GradAliasGuard>();
+    auto attach = [&](Var& v, bool trunk) {
       if (!v.grad_init) {
-        v.grad = out->grad;
+        v.grad = dout;
         v.grad_init = true;
+        v.alias_guard = guard;
+        v.alias_trunk = trunk;
       } else {
-        ops::add_inplace(v.grad, out->grad, 1.0f);
+        v.note_grad_mutation();
+        ops::add_inplace(v.grad, dout, 1.0f);
       }
     };
-    attach(*b);
-    attach(*a);
+    attach(*b, /*trunk=*/false);
+    attach(*a, /*trunk=*/true);
   });
   return out;
 }

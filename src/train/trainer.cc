@@ -69,8 +69,10 @@ void Trainer::step_body(const Tensor* bc_dev) {
     if (ddp_) ddp_->finish_grads();
     opt_->step(ddp_ ? ddp_->grad_scale() : 1.0f, bc_dev);
   }
+  // one writer per file; never inside graph capture (last_loss_local syncs
+  // + D2H-copies, which would abort hipStreamBeginCapture)
   const bool mwrite =
-      metrics_ && (!ddp_ || ddp_->comm()->rank() == 0);  // one writer per file
+      metrics_ && !graph_mode_ && (!ddp_ || ddp_->comm()->rank() == 0);
   if (mx && mwrite) {
     ev_opt_.record(*ctx_->compute_stream());
     ev_opt_.sync();
@@ -227,14 +229,17 @@ void Trainer::save_checkpoint(const std::string& path) {
   std::memcpy(stepa.data.data(), &st, 4);
   arrays["meta/step"] = stepa;
   // RNG state (SURVEY.md §5.4 meta/rng): the generator is counter-based,
-  // so the data seed IS the full state.
+  // so the data seed IS the full state. Four 16-bit chunks, each exactly
+  // representable in f32, cover all 64 seed bits.
   NpyArray rnga;
   rnga.dtype = DType::F32;
-  rnga.shape = {2};
-  float sd[2] = {(float)(data_seed_ & 0xffffffu),
-                 (float)((data_seed_ >> 24) & 0xffffffu)};
-  rnga.data.resize(8);
-  std::memcpy(rnga.data.data(), sd, 8);
+  rnga.shape = {4};
+  float sd[4] = {(float)(data_seed_ & 0xffffu),
+                 (float)((data_seed_ >> 16) & 0xffffu),
+                 (float)((data_seed_ >> 32) & 0xffffu),
+                 (float)((data_seed_ >> 48) & 0xffffu)};
+  rnga.data.resize(16);
+  std::memcpy(rnga.data.data(), sd, 16);
   arrays["meta/data_seed"] = rnga;
   npz_save(path, arrays);
   if (ddp_) ddp_->comm()->barrier();
@@ -259,20 +264,38 @@ void Trainer::load_checkpoint(const std::string& path) {
         size_t off = size_t(b.offsets[i]) * es;
         auto& pa = fetch("param/" + p->name);
         NZ_CHECK(pa.numel() == p->numel(), "shape mismatch for " + p->name);
-        // dtype must match exactly: the memcpy below trusts the byte count,
-        // so an f32 checkpoint loaded into bf16 params (e.g. a CPU-saved
-        // file on GPU) would otherwise overflow the staging buffer
-        NZ_CHECK(pa.dtype == p->data.dtype(),
-                 "dtype mismatch for " + p->name +
-                     " (checkpoint saved on a different device class?)");
-        NZ_CHECK(pa.data.size() == size_t(p->numel()) * es,
-                 "byte-size mismatch for " + p->name);
-        std::memcpy(data_host.data() + off, pa.data.data(), pa.data.size());
+        if (pa.dtype == p->data.dtype()) {
+          NZ_CHECK(pa.data.size() == size_t(p->numel()) * es,
+                   "byte-size mismatch for " + p->name);
+          std::memcpy(data_host.data() + off, pa.data.data(), pa.data.size());
+        } else if (pa.dtype == DType::F32 && p->data.dtype() == DType::BF16) {
+          // CPU (f32) checkpoint loaded on GPU (bf16): round-convert; exact
+          // values are restored from the f32 master below anyway.
+          NZ_CHECK(pa.data.size() == size_t(p->numel()) * 4,
+                   "byte-size mismatch for " + p->name);
+          const float* srcf = (const float*)pa.data.data();
+          uint16_t* dst = (uint16_t*)(data_host.data() + off);
+          for (int64_t k = 0; k < p->numel(); ++k) dst[k] = f32_to_bf16(srcf[k]);
+        } else if (pa.dtype == DType::BF16 && p->data.dtype() == DType::F32) {
+          // GPU (bf16) checkpoint loaded on CPU (f32): widen.
+          NZ_CHECK(pa.data.size() == size_t(p->numel()) * 2,
+                   "byte-size mismatch for " + p->name);
+          const uint16_t* srcb = (const uint16_t*)pa.data.data();
+          float* dst = (float*)(data_host.data() + off);
+          for (int64_t k = 0; k < p->numel(); ++k) dst[k] = bf16_to_f32(srcb[k]);
+        } else {
+          NZ_CHECK(false, "unsupported dtype conversion for " + p->name);
+        }
         auto& ma = fetch("master/" + p->name);
         NZ_CHECK(ma.data.size() == size_t(p->numel()) * 4,
                  "master byte-size mismatch for " + p->name);
         std::memcpy(master_host.data() + b.offsets[i] * 4, ma.data.data(),
                     ma.data.size());
+        if (pa.dtype != p->data.dtype() && p->data.dtype() == DType::F32) {
+          // cross-device resume onto f32 params: the f32 master is exact
+          // where the bf16 param member is rounded — prefer the master
+          std::memcpy(data_host.data() + off, ma.data.data(), ma.data.size());
+        }
         if (!m_host.empty()) {
           std::memcpy(m_host.data() + b.offsets[i] * 4,
                       fetch("adam_m/" + p->name).data.data(), p->numel() * 4);
@@ -294,11 +317,17 @@ void Trainer::load_checkpoint(const std::string& path) {
       opt_->set_step_count((int)st);
     }
     auto itr = arrays.find("meta/data_seed");
-    if (itr != arrays.end() && itr->second.data.size() == 8) {
+    if (itr != arrays.end() && itr->second.data.size() == 16) {
+      float sd[4];
+      std::memcpy(sd, itr->second.data.data(), 16);
+      uint64_t seed = (uint64_t)sd[0] | ((uint64_t)sd[1] << 16) |
+                      ((uint64_t)sd[2] << 32) | ((uint64_t)sd[3] << 48);
+      reseed_data(seed);
+    } else if (itr != arrays.end() && itr->second.data.size() == 8) {
+      // legacy 2x24-bit format (covers only 48 seed bits)
       float sd[2];
       std::memcpy(sd, itr->second.data.data(), 8);
-      uint64_t seed = (uint64_t)sd[0] | ((uint64_t)sd[1] << 24);
-      reseed_data(seed);
+      reseed_data((uint64_t)sd[0] | ((uint64_t)sd[1] << 24));
     }
   }
   if (ddp_) {
@@ -314,6 +343,18 @@ void Trainer::load_checkpoint(const std::string& path) {
     float st = ddp_->comm()->all_reduce_scalar(
         is_root ? (float)opt_->step_count() : 0.f, ReduceOp::MAX);
     opt_->set_step_count((int)st);
+    // Broadcast the restored base data seed (rank 0's seed = base + 0) and
+    // re-derive each rank's shard seed with the launcher's stride convention
+    // (bench.py / cmd/train.cc use base + kDdpSeedStride * rank).
+    uint64_t base = 0;
+    for (int c = 0; c < 4; ++c) {
+      float chunk = ddp_->comm()->all_reduce_scalar(
+          is_root ? (float)((data_seed_ >> (16 * c)) & 0xffffu) : 0.f,
+          ReduceOp::MAX);
+      base |= (uint64_t)chunk << (16 * c);
+    }
+    if (!is_root)
+      reseed_data(base + kDdpSeedStride * (uint64_t)ddp_->comm()->rank());
     if (ctx_->device() == Device::HIP) ctx_->compute_stream()->sync();
   }
 }

@@ -12,6 +12,11 @@
 
 namespace nz {
 
+// Per-rank data-shard seed stride: launchers (bench.py, cmd/train.cc) seed
+// rank r with base + kDdpSeedStride*r; checkpoint restore re-derives the
+// same shards from rank 0's saved base seed.
+constexpr uint64_t kDdpSeedStride = 1000003ull;
+
 class Trainer {
  public:
   Trainer(Context* ctx, std::unique_ptr<Model> model, Optimizer::Kind optim,
