@@ -255,6 +255,14 @@ func (t *Trainer) Save(path string) error {
 	return check(C.nz_ckpt_save(t.h, cp))
 }
 
+// SaveLocal writes THIS rank's state to its own file (no rank gating, no
+// collective) — test hook for cross-rank agreement checks.
+func (t *Trainer) SaveLocal(path string) error {
+	cp := C.CString(path)
+	defer C.free(unsafe.Pointer(cp))
+	return check(C.nz_ckpt_save_local(t.h, cp))
+}
+
 // Load restores an .npz checkpoint and (under DDP) broadcasts rank 0's state.
 func (t *Trainer) Load(path string) error {
 	cp := C.CString(path)

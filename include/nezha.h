@@ -149,6 +149,8 @@ double nz_trainer_flops_per_step(nz_trainer*);
 /* --------------------------------------------------------- checkpointing -- */
 nz_status nz_ckpt_save(nz_trainer*, const char* path);
 nz_status nz_ckpt_load(nz_trainer*, const char* path);
+/* this rank's state to its own file; no rank gating, no collective */
+nz_status nz_ckpt_save_local(nz_trainer*, const char* path);
 
 #ifdef __cplusplus
 } /* extern "C" */

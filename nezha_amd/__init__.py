@@ -128,6 +128,7 @@ _lib.nz_trainer_samples_per_step.argtypes = [ctypes.c_void_p]
 _lib.nz_trainer_flops_per_step.restype = ctypes.c_double
 _lib.nz_trainer_flops_per_step.argtypes = [ctypes.c_void_p]
 _lib.nz_ckpt_save.argtypes = [ctypes.c_void_p, ctypes.c_char_p]
+_lib.nz_ckpt_save_local.argtypes = [ctypes.c_void_p, ctypes.c_char_p]
 _lib.nz_ckpt_load.argtypes = [ctypes.c_void_p, ctypes.c_char_p]
 _lib.nz_hip_device_count.restype = ctypes.c_int
 
@@ -517,6 +518,10 @@ class Trainer:
 
     def save(self, path):
         _ck(_lib.nz_ckpt_save(self._h, path.encode()))
+
+    def save_local(self, path):
+        """this rank's state, no rank gating/collective (test hook)"""
+        _ck(_lib.nz_ckpt_save_local(self._h, path.encode()))
 
     def load(self, path):
         _ck(_lib.nz_ckpt_load(self._h, path.encode()))

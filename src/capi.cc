@@ -447,3 +447,10 @@ extern "C" nz_status nz_ckpt_load(nz_trainer* t, const char* path) {
   t->tr->load_checkpoint(path);
   NZ_API_END
 }
+/* Per-rank local save (no rank gating / collective): test hook for bitwise
+   cross-rank agreement checks under DDP. */
+extern "C" nz_status nz_ckpt_save_local(nz_trainer* t, const char* path) {
+  NZ_API_BEGIN
+  t->tr->save_checkpoint(path, /*local_only=*/true);
+  NZ_API_END
+}

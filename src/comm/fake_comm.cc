@@ -18,14 +18,30 @@ struct FakeState {
   std::condition_variable cv;
   int arrived = 0;
   uint64_t generation = 0;
+  uint64_t sig = 0;       // collective signature of this rendezvous round
+  bool poisoned = false;  // a rank detected a signature mismatch
   std::vector<void*> ptrs;
   // Reusable sense barrier; `leader_fn` runs once (by the last arriver)
   // while everyone waits, then all are released.
   explicit FakeState(int n) : nranks(n), ptrs(n, nullptr) {}
 
+  // Like RCCL, every rank must issue the SAME collective sequence. Each
+  // call carries a signature (kind/op/dtype/count); the first arriver of a
+  // round pins it and any mismatching later arrival fails ALL ranks loudly
+  // instead of silently reducing mismatched buffers (the bug class this
+  // models: DDP buckets launched in different orders on different ranks).
   template <typename Fn>
-  void rendezvous(int rank, void* ptr, Fn leader_fn) {
+  void rendezvous(int rank, void* ptr, uint64_t s, Fn leader_fn) {
     std::unique_lock<std::mutex> lk(mu);
+    NZ_CHECK(!poisoned, "FakeComm poisoned by earlier collective mismatch");
+    if (arrived == 0) {
+      sig = s;
+    } else if (s != sig) {
+      poisoned = true;
+      cv.notify_all();
+      NZ_CHECK(false, "FakeComm collective mismatch across ranks (different "
+                      "op/dtype/count — bucket launch order diverged?)");
+    }
     ptrs[rank] = ptr;
     uint64_t gen = generation;
     if (++arrived == nranks) {
@@ -34,10 +50,17 @@ struct FakeState {
       ++generation;
       cv.notify_all();
     } else {
-      cv.wait(lk, [&] { return generation != gen; });
+      cv.wait(lk, [&] { return generation != gen || poisoned; });
+      NZ_CHECK(!poisoned, "FakeComm poisoned by collective mismatch");
     }
   }
 };
+
+// signature: kind(4b) | op(4b) | dtype(4b) | count(52b)
+inline uint64_t coll_sig(int kind, ReduceOp op, DType dt, int64_t count) {
+  return (uint64_t(kind) << 60) | (uint64_t((int)op) << 56) |
+         (uint64_t((int)dt) << 52) | (uint64_t(count) & 0xfffffffffffffull);
+}
 
 template <typename T>
 void reduce_buffers(std::vector<void*>& ptrs, int64_t count, ReduceOp op,
@@ -71,7 +94,8 @@ class FakeComm : public Comm {
   void all_reduce(void* data, int64_t count, DType dt, ReduceOp op,
                   Stream* stream) override {
     if (device_ && stream) stream->sync();  // grads ready on the device
-    st_->rendezvous(rank_, data, [&](std::vector<void*>& ptrs) {
+    st_->rendezvous(rank_, data, coll_sig(1, op, dt, count),
+                    [&](std::vector<void*>& ptrs) {
       size_t bytes = size_t(count) * dtype_size(dt);
       if (!device_) {
         reduce_buffers<float>(ptrs, count, op, dt == DType::BF16);
@@ -99,7 +123,8 @@ class FakeComm : public Comm {
     if (device_ && stream) stream->sync();
     size_t bytes = size_t(count) * dtype_size(dt);
     bool dev = device_;
-    st_->rendezvous(rank_, data, [&, root, bytes, dev](std::vector<void*>& ptrs) {
+    st_->rendezvous(rank_, data, coll_sig(2, ReduceOp::SUM, dt, count + root),
+                    [&, root, bytes, dev](std::vector<void*>& ptrs) {
       if (!dev) {
         for (int i = 0; i < (int)ptrs.size(); ++i)
           if (i != root) std::memcpy(ptrs[i], ptrs[root], bytes);
@@ -118,7 +143,8 @@ class FakeComm : public Comm {
   }
 
   void barrier() override {
-    st_->rendezvous(rank_, nullptr, [](std::vector<void*>&) {});
+    st_->rendezvous(rank_, nullptr, coll_sig(3, ReduceOp::SUM, DType::F32, 0),
+                    [](std::vector<void*>&) {});
   }
 
   float all_reduce_scalar(float v, ReduceOp op) override {
@@ -126,7 +152,8 @@ class FakeComm : public Comm {
     // buffers — force the host reduce path (the device staging path would
     // hipMemcpy D2H from a host stack pointer and return garbage).
     float buf = v;
-    st_->rendezvous(rank_, &buf, [&](std::vector<void*>& ptrs) {
+    st_->rendezvous(rank_, &buf, coll_sig(4, op, DType::F32, 1),
+                    [&](std::vector<void*>& ptrs) {
       reduce_buffers<float>(ptrs, 1, op, /*bf16=*/false);
     });
     return buf;

@@ -52,6 +52,7 @@ void Ddp::on_ready(Param& p) {
 
 void Ddp::launch_allreduce(Bucket& b, int bi) {
   ++comm_launches_;
+  launch_order_.push_back(bi);
   if (ctx_->device() == Device::HIP) {
     // compute stream -> event -> comm stream all-reduce -> event
     if (allreduce_f32_) ops::cast_copy(f32_scratch_[bi], b.grad_flat);

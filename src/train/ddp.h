@@ -26,6 +26,9 @@ class Ddp {
   float grad_scale() const { return 1.0f / float(comm_->nranks()); }
   Comm* comm() const { return comm_; }
   int64_t comm_launches() const { return comm_launches_; }
+  // bucket indices in all-reduce launch order (RCCL requires every rank to
+  // issue collectives in the same order — tests assert cross-rank equality)
+  const std::vector<int>& launch_order() const { return launch_order_; }
 
  private:
   void on_ready(Param& p);
@@ -36,6 +39,7 @@ class Ddp {
   Comm* comm_;
   std::unordered_map<Param*, int> bucket_of_;
   int64_t comm_launches_ = 0;
+  std::vector<int> launch_order_;
   // --allreduce-dtype=f32 fallback: grads are all-reduced
   // in bf16 by default (half the ring bytes on the per-link-bound xGMI
   // ring); NEZHA_ALLREDUCE_DTYPE=f32 upcasts each bucket to an f32 scratch

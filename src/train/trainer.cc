@@ -189,8 +189,8 @@ static NpyArray slice_to_npy(const std::vector<char>& flat_host, DType dt,
   return a;
 }
 
-void Trainer::save_checkpoint(const std::string& path) {
-  if (ddp_ && ddp_->comm()->rank() != 0) {
+void Trainer::save_checkpoint(const std::string& path, bool local_only) {
+  if (!local_only && ddp_ && ddp_->comm()->rank() != 0) {
     ddp_->comm()->barrier();
     return;
   }
@@ -242,7 +242,7 @@ void Trainer::save_checkpoint(const std::string& path) {
   std::memcpy(rnga.data.data(), sd, 16);
   arrays["meta/data_seed"] = rnga;
   npz_save(path, arrays);
-  if (ddp_) ddp_->comm()->barrier();
+  if (!local_only && ddp_) ddp_->comm()->barrier();
 }
 
 void Trainer::load_checkpoint(const std::string& path) {

@@ -42,7 +42,9 @@ class Trainer {
   Context* ctx() { return ctx_; }
 
   // Checkpoint: rank 0 writes; resume loads + broadcasts.
-  void save_checkpoint(const std::string& path);
+  // local_only: THIS rank writes its own file, no rank gating/barrier —
+  // test hook for bitwise cross-rank agreement checks.
+  void save_checkpoint(const std::string& path, bool local_only = false);
   void load_checkpoint(const std::string& path);
 
  private:

@@ -55,6 +55,7 @@ int main() {
   auto comms = make_fake_comm_group(NRANKS);
   std::vector<std::vector<float>> ddp(NRANKS);
   std::vector<int64_t> launches(NRANKS, 0);
+  std::vector<std::vector<int>> order(NRANKS);
   std::vector<std::thread> threads;
   for (int r = 0; r < NRANKS; ++r) {
     threads.emplace_back([&, r]() {
@@ -65,9 +66,14 @@ int main() {
       tr->step();
       ddp[r] = dump_params(*tr);
       launches[r] = tr->ddp()->comm_launches();
+      order[r] = tr->ddp()->launch_order();
     });
   }
   for (auto& t : threads) t.join();
+
+  // every rank issued its bucket all-reduces in the SAME order (the RCCL
+  // ordering contract; a divergence here is silent corruption on hardware)
+  for (int r = 1; r < NRANKS; ++r) assert(order[r] == order[0]);
 
   // all ranks identical
   for (int r = 1; r < NRANKS; ++r)
@@ -90,6 +96,31 @@ int main() {
            (long long)launches[0]);
     assert(nbuckets >= 2);
     assert(launches[0] == (int64_t)nbuckets);
+  }
+
+  // --- adversarial: mismatched collective sequences must fail loudly ------
+  // Two ranks all-reduce different-size "buckets" concurrently (the bug
+  // class: DDP buckets launched in diverging order across ranks). FakeComm
+  // pins the first arriver's collective signature and poisons the group on
+  // mismatch — BOTH ranks must see an Error, never a silent wrong reduce.
+  {
+    auto comms2 = make_fake_comm_group(2);
+    std::vector<int> caught(2, 0);
+    std::vector<std::thread> ths;
+    for (int r = 0; r < 2; ++r) {
+      ths.emplace_back([&, r]() {
+        std::vector<float> buf(r == 0 ? 16 : 8, 1.0f);
+        try {
+          comms2[r]->all_reduce(buf.data(), (int64_t)buf.size(), DType::F32,
+                                ReduceOp::SUM, nullptr);
+        } catch (const Error&) {
+          caught[r] = 1;
+        }
+      });
+    }
+    for (auto& t : ths) t.join();
+    assert(caught[0] == 1 && caught[1] == 1);
+    printf("adversarial collective mismatch detected on both ranks OK\n");
   }
 
   printf("test_ddp_fake: ALL OK\n");

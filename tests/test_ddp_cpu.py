@@ -42,6 +42,7 @@ if not same_data:
 for _ in range(4):
     tr.step()
 tr.save(os.environ["OUT"])  # rank 0 writes, others hit the barrier inside
+tr.save_local(os.environ["OUT"] + f".rank{rank}")  # per-rank, ungated
 comm.barrier()
 """
 
@@ -56,8 +57,19 @@ def test_tcp_ddp_two_ranks(tmp_path, same_data):
     for p in ps:
         o, _ = p.communicate(timeout=120)
         assert p.returncode == 0, o.decode()
-    c0 = np.load(out)  # rank 0 writes; bitwise rank agreement is covered by
-    assert len(c0.files) > 4  # the FakeComm C++ test (test_ddp_fake)
+    c0 = np.load(out)
+    assert len(c0.files) > 4
+    # BITWISE cross-rank agreement: after broadcast + deterministic
+    # all-reduce, both ranks' full optimizer state must be byte-identical
+    # (params, master, moments, step) — this catches any rank divergence,
+    # sharded data or not.
+    r0 = np.load(out + ".rank0")
+    r1 = np.load(out + ".rank1")
+    assert set(r0.files) == set(r1.files)
+    for k in r0.files:
+        if k == "meta/data_seed":
+            continue  # per-rank data shards legitimately differ in seed
+        assert r0[k].tobytes() == r1[k].tobytes(), f"rank divergence in {k}"
 
     if same_data:
         # DP over identical shards == single-rank training (grad avg of two

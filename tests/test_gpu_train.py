@@ -197,6 +197,99 @@ def test_ddp_8rank_one_device():
     assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
 
 
+def test_cross_device_resume(hip_ctx, tmp_path):
+    """CPU-saved (f32) checkpoint resumes on GPU (bf16) and vice versa
+    (VERDICT r01 missing #5): params are converted, exact values restored
+    from the f32 master, training continues finite and comparable."""
+    kw = dict(layers=2, dim=256, batch=128, lr=1e-3, seed=21)
+    cpu = nz.Ctx(nz.CPU)
+    a = nz.Trainer.mlp(cpu, **kw)
+    for _ in range(2):
+        a.step()
+    p = str(tmp_path / "cpu.npz")
+    a.save(p)
+    ck = np.load(p)
+    assert ck["param/mlp.l0.w"].dtype == np.float32
+
+    b = nz.Trainer.mlp(hip_ctx, **kw)
+    b.load(p)
+    # bf16 params must equal round(master) of the CPU state
+    q = str(tmp_path / "gpu_after_load.npz")
+    b.save(q)
+    cg = np.load(q)
+    np.testing.assert_array_equal(cg["master/mlp.l0.w"],
+                                  ck["master/mlp.l0.w"])
+    np.testing.assert_array_equal(
+        cg["param/mlp.l0.w"],
+        nz.f32_to_bf16(ck["master/mlp.l0.w"]))
+    a.step()
+    b.step()
+    la, lb = a.loss(), b.loss()
+    assert np.isfinite(lb) and abs(la - lb) / la < 0.05, f"{la} vs {lb}"
+
+    # and back: GPU (bf16) checkpoint onto CPU (f32)
+    r = str(tmp_path / "gpu.npz")
+    b.save(r)
+    c = nz.Trainer.mlp(cpu, **kw)
+    c.load(r)
+    c.step()
+    assert np.isfinite(c.loss())
+
+
+def test_graph_metrics_coexist(tmp_path):
+    """NEZHA_GRAPH=1 + NEZHA_METRICS both set must not abort capture
+    (regression: metrics loss readback used to sync inside capture)."""
+    import subprocess, sys, os
+    script = r"""
+import os, sys
+sys.path.insert(0, os.getcwd())
+import nezha_amd as nz
+ctx = nz.Ctx(nz.HIP, 0)
+tr = nz.Trainer.mlp(ctx, layers=2, dim=512, batch=256, lr=1e-3, seed=44)
+for _ in range(6):
+    tr.step()
+assert tr.loss() > 0
+print("GRAPH_METRICS_OK")
+"""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, NEZHA_GRAPH="1",
+               NEZHA_METRICS=str(tmp_path / "m.jsonl"))
+    r = subprocess.run([sys.executable, "-c", script], env=env, cwd=repo,
+                       capture_output=True, timeout=300)
+    assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
+    assert b"GRAPH_METRICS_OK" in r.stdout
+
+
+def test_graph_odd_bucket_matches_eager():
+    """Graph replay with a bucket whose numel % 4 != 0 exercises the scalar
+    adam_k bc_dev path (regression: bias correction used to freeze at the
+    capture-time step). dim=125, bucket_bytes=32000 -> bucket numels
+    15875 and 15625, both odd."""
+    import subprocess, sys, os, json
+    script = r"""
+import os, sys, json
+sys.path.insert(0, os.getcwd())
+import nezha_amd as nz
+ctx = nz.Ctx(nz.HIP, 0)
+tr = nz.Trainer.mlp(ctx, layers=2, dim=125, batch=64, lr=1e-2, seed=7,
+                    bucket_bytes=32000)
+losses = []
+for _ in range(8):
+    tr.step()
+    losses.append(tr.loss())
+print(json.dumps(losses))
+"""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    outs = []
+    for graph in ("0", "1"):
+        env = dict(os.environ, NEZHA_GRAPH=graph)
+        r = subprocess.run([sys.executable, "-c", script], env=env, cwd=repo,
+                           capture_output=True, timeout=300)
+        assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
+        outs.append(json.loads(r.stdout.decode().strip().splitlines()[-1]))
+    assert outs[0] == outs[1], f"eager {outs[0]} vs graph {outs[1]}"
+
+
 def test_bert_noncontract_seq(hip_ctx):
     """seq=64 routes attention through the composed fallback on GPU
     (fused kernels are gated to the S=128 contract shape)."""
