@@ -28,6 +28,10 @@ class Comm {
   virtual void broadcast(void* data, int64_t count, DType dt, int root,
                          Stream* stream) = 0;
   virtual void barrier() = 0;
+  // True if this backend's collectives may be recorded into a hipGraph
+  // (stream-ordered device work only — RCCL qualifies; the host-staged
+  // test comms do not).
+  virtual bool capturable() const { return false; }
   // Host-scalar convenience (stages through the device for RCCL).
   virtual float all_reduce_scalar(float v, ReduceOp op) = 0;
 };

@@ -70,6 +70,7 @@ class RcclComm : public Comm {
   }
   int rank() const override { return rank_; }
   int nranks() const override { return nranks_; }
+  bool capturable() const override { return true; }
 
   void all_reduce(void* data, int64_t count, DType dt, ReduceOp op,
                   Stream* stream) override {

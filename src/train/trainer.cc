@@ -103,17 +103,24 @@ void Trainer::step() {
     }
   }
 
-  if (!graph_mode_ || ddp_) {  // DDP path stays eager (RCCL + capture N/A)
+  // Graph capture works under DDP too when the comm backend is pure
+  // stream-ordered device work (RCCL): the comm-stream collectives are
+  // pulled into the capture through the ready/comm event edges, so a
+  // replayed step carries the same overlapped all-reduce schedule
+  // (VERDICT r01 next-step #7). Host-staged test comms are not capturable.
+  if (!graph_mode_ || (ddp_ && !ddp_->comm()->capturable())) {
     step_body(nullptr);
     return;
   }
 
   auto upload_bc = [&](int stepno) {
-    float bc[2] = {1.0f - powf(opt_->beta1, (float)stepno),
-                   1.0f - powf(opt_->beta2, (float)stepno)};
+    // member buffer, not stack: hipMemcpyAsync from pageable host memory
+    // must outlive the call
+    bc_host_[0] = 1.0f - powf(opt_->beta1, (float)stepno);
+    bc_host_[1] = 1.0f - powf(opt_->beta2, (float)stepno);
     hipStream_t st = ctx_->compute_stream()->handle();
-    NZ_CHECK(hipMemcpyAsync(bc_dev_.data(), bc, 8, hipMemcpyHostToDevice, st) ==
-                 hipSuccess,
+    NZ_CHECK(hipMemcpyAsync(bc_dev_.data(), bc_host_, 8,
+                            hipMemcpyHostToDevice, st) == hipSuccess,
              "bc upload failed");
   };
 
@@ -131,23 +138,46 @@ void Trainer::step() {
     upload_bc(next);
     ctx_->compute_stream()->sync();
     hipStream_t st = ctx_->compute_stream()->handle();
-    NZ_CHECK(hipStreamBeginCapture(st, hipStreamCaptureModeThreadLocal) ==
-                 hipSuccess,
-             "graph capture begin failed");
-    step_body(&bc_dev_);  // increments step_count to `next`
-    hipGraph_t graph = nullptr;
-    NZ_CHECK(hipStreamEndCapture(st, &graph) == hipSuccess,
-             "graph capture end failed");
-    hipGraphExec_t exec = nullptr;
-    NZ_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0) ==
-                 hipSuccess,
-             "graph instantiate failed");
-    (void)hipGraphDestroy(graph);
-    graph_exec_ = (void*)exec;
-    graph_captured_ = true;
+    // Capture failure must not kill the run (and under DDP must not
+    // desynchronize ranks): fall back to eager. The eager step issues the
+    // exact same collective sequence a replayed graph would, so a mixed
+    // graph/eager world stays collectively consistent.
+    int count_before = opt_->step_count();
+    try {
+      NZ_CHECK(hipStreamBeginCapture(st, hipStreamCaptureModeThreadLocal) ==
+                   hipSuccess,
+               "graph capture begin failed");
+      step_body(&bc_dev_);  // increments step_count to `next`
+      hipGraph_t graph = nullptr;
+      NZ_CHECK(hipStreamEndCapture(st, &graph) == hipSuccess,
+               "graph capture end failed");
+      hipGraphExec_t exec = nullptr;
+      hipError_t inst = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
+      (void)hipGraphDestroy(graph);
+      NZ_CHECK(inst == hipSuccess, "graph instantiate failed");
+      graph_exec_ = (void*)exec;
+      graph_captured_ = true;
+    } catch (const Error& e) {
+      NZ_LOG_ERROR("hipGraph capture failed (%s) — falling back to eager",
+                   e.what());
+      hipGraph_t dead = nullptr;
+      (void)hipStreamEndCapture(st, &dead);  // unwind a half-open capture
+      if (dead) (void)hipGraphDestroy(dead);
+      (void)hipGetLastError();  // clear sticky error
+      graph_mode_ = false;
+      opt_->set_step_count(count_before);
+      // a mid-capture throw can leave per-param/bucket host state dirty
+      for (auto& b : opt_->buckets())
+        for (Param* p : b.params) p->grad_init = false;
+      if (ddp_) ddp_->pre_step();
+      step_body(nullptr);
+      return;
+    }
     // the captured work did NOT run; launch it for this step
-    NZ_CHECK(hipGraphLaunch(exec, st) == hipSuccess, "graph launch failed");
-    NZ_LOG_INFO("hipGraph captured: training step replays as one graph");
+    NZ_CHECK(hipGraphLaunch((hipGraphExec_t)graph_exec_, st) == hipSuccess,
+             "graph launch failed");
+    NZ_LOG_INFO("hipGraph captured: training step replays as one graph%s",
+                ddp_ ? " (incl. bucketed RCCL all-reduce)" : "");
     return;
   }
   int next = opt_->step_count() + 1;

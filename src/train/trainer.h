@@ -69,6 +69,7 @@ class Trainer {
   bool graph_captured_ = false;
   void* graph_exec_ = nullptr;  // hipGraphExec_t (opaque here)
   Tensor bc_dev_;
+  float bc_host_[2] = {1.f, 1.f};  // staging for async bc upload
 };
 
 }  // namespace nz

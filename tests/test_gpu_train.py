@@ -290,6 +290,39 @@ print(json.dumps(losses))
     assert outs[0] == outs[1], f"eager {outs[0]} vs graph {outs[1]}"
 
 
+def test_graph_ddp_world1_matches_eager():
+    """NEZHA_GRAPH=1 with DDP attached (RCCL world=1): the captured step
+    includes the bucketed all-reduce; losses must match the eager DDP path
+    exactly. On a 1-GPU box world=1 is the
+    deepest RCCL-in-graph check available."""
+    import subprocess, sys, os, json
+    script = r"""
+import os, sys, json
+sys.path.insert(0, os.getcwd())
+import nezha_amd as nz
+ctx = nz.Ctx(nz.HIP, 0)
+tr = nz.Trainer.mlp(ctx, layers=2, dim=512, batch=256, lr=1e-3, seed=44,
+                    bucket_bytes=200000)
+uid = nz.Comm.get_uid()
+comm = nz.Comm.rccl(ctx, 1, 0, uid)
+tr.attach_ddp(comm)
+losses = []
+for _ in range(6):
+    tr.step()
+    losses.append(tr.loss())
+print(json.dumps(losses))
+"""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    outs = []
+    for graph in ("0", "1"):
+        env = dict(os.environ, NEZHA_GRAPH=graph)
+        r = subprocess.run([sys.executable, "-c", script], env=env, cwd=repo,
+                           capture_output=True, timeout=300)
+        assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
+        outs.append(json.loads(r.stdout.decode().strip().splitlines()[-1]))
+    assert outs[0] == outs[1], f"eager {outs[0]} vs graph {outs[1]}"
+
+
 def test_bert_noncontract_seq(hip_ctx):
     """seq=64 routes attention through the composed fallback on GPU
     (fused kernels are gated to the S=128 contract shape)."""
