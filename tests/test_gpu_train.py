@@ -319,7 +319,10 @@ print(json.dumps(losses))
         r = subprocess.run([sys.executable, "-c", script], env=env, cwd=repo,
                            capture_output=True, timeout=300)
         assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
-        outs.append(json.loads(r.stdout.decode().strip().splitlines()[-1]))
+        # RCCL prints a version banner to stdout — take the JSON array line
+        lines = [l for l in r.stdout.decode().splitlines()
+                 if l.startswith("[")]
+        outs.append(json.loads(lines[-1]))
     assert outs[0] == outs[1], f"eager {outs[0]} vs graph {outs[1]}"
 
 
