@@ -1676,6 +1676,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase6_k(GemmArgs args,
 // group (B0(0), A1(0), A0(0), B1(0)) for the pre-read and p0/p1.
 // Tails drain with vmcnt(0) as in variants 2/6.
 // ---------------------------------------------------------------------------
+template <int PPMODE>  // bit0: static G1 prio; bit1: stage before reads
 __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
     GemmArgs args, int nwg_n) {
   using namespace ph8;
@@ -1701,6 +1702,11 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
   const int wid = threadIdx.x >> 6;
   const int g = wid >> 2;        // ping-pong group (0: waves 0-3, 1: 4-7)
   const int ws = wid & 3;        // wave-in-group: 2x2 over the quadrant
+  // microarch "two waves per SIMD" item 4: the second-dispatched half is
+  // the arbitration loser; one static s_setprio(1) for it, no flips
+  if constexpr (PPMODE & 1) {
+    if (g == 1) __builtin_amdgcn_s_setprio(1);
+  }
   const int tig = (int)threadIdx.x & 255;  // thread-in-group
 
   const int64_t lda = args.lda, ldb = args.ldb;
@@ -1762,7 +1768,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
 
   // pure 32-MFMA cluster for quadrant column b
   auto mfma32 = [&](int b) {
-    __builtin_amdgcn_s_setprio(1);
+    if constexpr (!(PPMODE & 1)) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int kk2 = 0; kk2 < 2; ++kk2)
 #pragma unroll
@@ -1771,7 +1777,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
         for (int fn = 0; fn < 4; ++fn)
           acc[b][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[kk2][fm], bf[kk2][fn], acc[b][fm][fn], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
+    if constexpr (!(PPMODE & 1)) __builtin_amdgcn_s_setprio(0);
   };
 
   // prologue: 6 units, 3 per group, in each group's own FIFO order
@@ -1819,84 +1825,94 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
   else                                                        \
     asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
 
+  const bool dg_mfma = args.diag_no_store != 2;   // 2: stage-only ablation
+  const bool dg_stage = args.diag_no_store != 3;  // 3: compute-only ablation
   for (int64_t t = 0; t < ntiles; t += 2) {
-    const bool tail = (t + 4 >= ntiles);
-    const bool s2 = (t + 2 < ntiles), s3 = (t + 3 < ntiles);
+    const bool tail = (t + 4 >= ntiles) || !dg_stage;
+    const bool s2 = (t + 2 < ntiles) && dg_stage, s3 = (t + 3 < ntiles) && dg_stage;
     // p0: G0 computes Q(0,0)t | G1 reads af(A1,t)+bf(B0,t), stages A0(t+1)
     if (g == 0) {
-      mfma32(0);
+      if (dg_mfma) mfma32(0);
     } else {
+      if constexpr (PPMODE & 2) { if (dg_stage) stage(0, 0, t + 1); }
       read_af(t);
       read_bf(0, t);
-      stage(0, 0, t + 1);
+      if constexpr (!(PPMODE & 2)) { if (dg_stage) stage(0, 0, t + 1); }
     }
     PP_END(PP_W0(4));
     // p1: G1 computes Q(1,0)t | G0 reads bf(B1,t) [af reuse], stages B1(t+1)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma32(0);
+      if (dg_mfma) mfma32(0);
     } else {
+      if constexpr (PPMODE & 2) { if (dg_stage) stage(1, 1, t + 1); }
       read_bf(1, t);
-      stage(1, 1, t + 1);
+      if constexpr (!(PPMODE & 2)) { if (dg_stage) stage(1, 1, t + 1); }
     }
     PP_END();
     // p2: G0 computes Q(0,1)t | G1 reads bf(B1,t), stages B0(t+2)
     if (g == 0) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma32(1);
+      if (dg_mfma) mfma32(1);
     } else {
+      if constexpr (PPMODE & 2) { if (s2) stage(1, 0, t + 2); }
       read_bf(1, t);
-      if (s2) stage(1, 0, t + 2);
+      if constexpr (!(PPMODE & 2)) { if (s2) stage(1, 0, t + 2); }
     }
     PP_END(PP_W1(4));
     // p3: G1 computes Q(1,1)t | G0 reads af(A0,t+1)+bf(B0,t+1), stages A1(t+2)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma32(1);
+      if (dg_mfma) mfma32(1);
     } else {
+      if constexpr (PPMODE & 2) { if (s2) stage(0, 1, t + 2); }
       read_af(t + 1);
       read_bf(0, t + 1);
-      if (s2) stage(0, 1, t + 2);
+      if constexpr (!(PPMODE & 2)) { if (s2) stage(0, 1, t + 2); }
     }
     PP_END(PP_W0(8));
     // p4: G0 computes Q(0,0)t+1 | G1 reads af(A1,t+1)+bf(B0,t+1), stages A0(t+2)
     if (g == 0) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma32(0);
+      if (dg_mfma) mfma32(0);
     } else {
+      if constexpr (PPMODE & 2) { if (s2) stage(0, 0, t + 2); }
       read_af(t + 1);
       read_bf(0, t + 1);
-      if (s2) stage(0, 0, t + 2);
+      if constexpr (!(PPMODE & 2)) { if (s2) stage(0, 0, t + 2); }
     }
     PP_END(PP_W0(4));
     // p5: G1 computes Q(1,0)t+1 | G0 reads bf(B1,t+1), stages B1(t+2)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma32(0);
+      if (dg_mfma) mfma32(0);
     } else {
+      if constexpr (PPMODE & 2) { if (s2) stage(1, 1, t + 2); }
       read_bf(1, t + 1);
-      if (s2) stage(1, 1, t + 2);
+      if constexpr (!(PPMODE & 2)) { if (s2) stage(1, 1, t + 2); }
     }
     PP_END();
     // p6: G0 computes Q(0,1)t+1 | G1 reads bf(B1,t+1), stages B0(t+3)
     if (g == 0) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma32(1);
+      if (dg_mfma) mfma32(1);
     } else {
+      if constexpr (PPMODE & 2) { if (s3) stage(1, 0, t + 3); }
       read_bf(1, t + 1);
-      if (s3) stage(1, 0, t + 3);
+      if constexpr (!(PPMODE & 2)) { if (s3) stage(1, 0, t + 3); }
     }
     PP_END(PP_W1(4));
     // p7: G1 computes Q(1,1)t+1 | G0 reads af(A0,t+2)+bf(B0,t+2), stages A1(t+3)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma32(1);
+      if (dg_mfma) mfma32(1);
     } else {
+      if constexpr (PPMODE & 2) { if (s3) stage(0, 1, t + 3); }
       if (s2) {
         read_af(t + 2);
         read_bf(0, t + 2);
       }
-      if (s3) stage(0, 1, t + 3);
+      if constexpr (!(PPMODE & 2)) { if (s3) stage(0, 1, t + 3); }
     }
     PP_END(PP_WB(8));
   }
@@ -2942,7 +2958,7 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
   }
   static const int kNtVariant = [] {
     const char* v = getenv("NEZHA_GEMM_VARIANT");
-    return v ? atoi(v) : 2;  // 0=2-buf 128x128, 1=3-buf deep, 2=8-phase 256^2
+    return v ? atoi(v) : 7;  // 7=ping-pong 256^2 (r2 default); 2=8-phase
   }();
   static const int kNtMinWG = [] {
     const char* v = getenv("NEZHA_GEMM_MIN_WG");  // test hook: force 8-phase
@@ -2976,14 +2992,26 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
       return;
     }
   }
+  // ping-pong wants enough K-iterations to amortize its prologue; short-K
+  // shapes (BERT K=768) measure faster on the persistent kernel below
+  // (gpurun_out/r2_defaults.log: 536-526 vs 539-569 TF)
   if (!transA && transB && batch == 1 && kNtVariant == 7 &&
-      args.K % 128 == 0 && args.K / 64 >= 6) {
+      args.K % 128 == 0 && args.K >= 1536) {
     int nwg_m = (int)((args.M + 255) / 256);
     int nwg_n = (int)((args.N + 255) / 256);
     if (nwg_m * (int64_t)nwg_n >= kNtMinWG &&
         balanced_256(nwg_m * nwg_n)) {
       dim3 grid(nwg_m * nwg_n);
-      gemm_bf16_nt_pingpong_k<<<grid, dim3(512), 0, s>>>(args, nwg_n);
+      static const int kPP = [] {
+        const char* v = getenv("NEZHA_GEMM_PP");
+        return v ? atoi(v) : 1;  // static prio for the younger half
+      }();
+      switch (kPP & 3) {
+        case 1: gemm_bf16_nt_pingpong_k<1><<<grid, dim3(512), 0, s>>>(args, nwg_n); break;
+        case 2: gemm_bf16_nt_pingpong_k<2><<<grid, dim3(512), 0, s>>>(args, nwg_n); break;
+        case 3: gemm_bf16_nt_pingpong_k<3><<<grid, dim3(512), 0, s>>>(args, nwg_n); break;
+        default: gemm_bf16_nt_pingpong_k<0><<<grid, dim3(512), 0, s>>>(args, nwg_n); break;
+      }
       NZ_KERNEL_CHECK();
       return;
     }

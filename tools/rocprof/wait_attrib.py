@@ -14,7 +14,8 @@ with open(sys.argv[1]) as f:
         name = row.get("Kernel_Name", "")
         if "gemm" not in name:
             continue
-        if "8phase" in name or "nt_k" in name or "pers" in name:
+        if ("8phase" in name or "nt_k" in name or "pers" in name
+                or "pingpong" in name):
             kname = name.split("(")[0]
             c = row["Counter_Name"]
             v = float(row["Counter_Value"])
