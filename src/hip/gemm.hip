@@ -1828,7 +1828,10 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
   const bool dg_mfma = args.diag_no_store != 2;   // 2: stage-only ablation
   const bool dg_stage = args.diag_no_store != 3;  // 3: compute-only ablation
   for (int64_t t = 0; t < ntiles; t += 2) {
-    const bool tail = (t + 4 >= ntiles) || !dg_stage;
+    // drain only in the LAST iteration (its skipped stages break the
+    // steady vmcnt counts); the second-to-last still stages every unit,
+    // so its counts — and all its certificates — are exactly steady
+    const bool tail = (t + 2 >= ntiles) || !dg_stage;
     const bool s2 = (t + 2 < ntiles) && dg_stage, s3 = (t + 3 < ntiles) && dg_stage;
     // p0: G0 computes Q(0,0)t | G1 reads af(A1,t)+bf(B0,t), stages A0(t+1)
     if (g == 0) {
@@ -3133,7 +3136,10 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
           pa.sC = args.M * args.N;
           pa.out_f32 = (float*)parts.data();
           dim3 g8((uint32_t)nwg8, sl);
-          gemm_bf16_nt_8phase_k<<<g8, dim3(512), 0, s>>>(pa, nwg_n8);
+          if (kNtVariant >= 7)  // ping-pong slices (+15% on deep-K tiles)
+            gemm_bf16_nt_pingpong_k<1><<<g8, dim3(512), 0, s>>>(pa, nwg_n8);
+          else
+            gemm_bf16_nt_8phase_k<<<g8, dim3(512), 0, s>>>(pa, nwg_n8);
           NZ_KERNEL_CHECK();
           splitk_reduce_k<<<dim3(grid_1d(args.M * args.N, 256)), dim3(256), 0,
                             s>>>((const float*)parts.data(), args, sl);
