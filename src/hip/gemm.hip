@@ -1780,18 +1780,27 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
     if constexpr (!(PPMODE & 1)) __builtin_amdgcn_s_setprio(0);
   };
 
-  // prologue: 6 units, 3 per group, in each group's own FIFO order
+  // prologue: tiles 0 and 1 fully staged, 4 units per group in each
+  // group's in-loop FIFO order (deep schedule, r2: iteration t stages ALL
+  // of tiles t+2/t+3, giving every unit a 5-7 phase flight; the old
+  // schedule's 3-phase flights made DMA latency the pace-setter on
+  // short-K shapes)
   if (g == 1) {
-    stage(1, 0, 0);  // B0(0)
     stage(0, 0, 0);  // A0(0)
+    stage(1, 0, 0);  // B0(0)
+    stage(0, 0, 1);  // A0(1)
     stage(1, 0, 1);  // B0(1)
   } else {
     stage(0, 1, 0);  // A1(0)
     stage(1, 1, 0);  // B1(0)
     stage(0, 1, 1);  // A1(1)
+    stage(1, 1, 1);  // B1(1)
   }
-  // first two units of each group certified (B0(0) A0(0) | A1(0) B1(0))
-  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  // certificates for the pre-read + p0 reads: oldest units of each group
+  if (g == 0)
+    asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+  else
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
   // pre-read: G0 its p0 fragments (A0,B0 of tile 0); G1 reads at p0 proper
@@ -1806,6 +1815,10 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
 #define PP_END(wait_stmt)                                  \
   do {                                                     \
     wait_stmt;                                             \
+    /* drain THIS wave's LDS reads before crossing: the deep  */ \
+    /* schedule re-stages a slot ONE phase after its last read */ \
+    /* issue; an undrained ds_read would race the incoming DMA */ \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");          \
     __builtin_amdgcn_s_barrier();                          \
     asm volatile("" ::: "memory");                         \
   } while (0)
@@ -1837,20 +1850,20 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
     if (g == 0) {
       if (dg_mfma) mfma32(0);
     } else {
-      if constexpr (PPMODE & 2) { if (dg_stage) stage(0, 0, t + 1); }
+      if constexpr (PPMODE & 2) { if (s2) stage(0, 0, t + 2); }
       read_af(t);
       read_bf(0, t);
-      if constexpr (!(PPMODE & 2)) { if (dg_stage) stage(0, 0, t + 1); }
+      if constexpr (!(PPMODE & 2)) { if (s2) stage(0, 0, t + 2); }
     }
-    PP_END(PP_W0(4));
+    PP_END(PP_W0(8));
     // p1: G1 computes Q(1,0)t | G0 reads bf(B1,t) [af reuse], stages B1(t+1)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       if (dg_mfma) mfma32(0);
     } else {
-      if constexpr (PPMODE & 2) { if (dg_stage) stage(1, 1, t + 1); }
+      if constexpr (PPMODE & 2) { if (s2) stage(0, 1, t + 2); }
       read_bf(1, t);
-      if constexpr (!(PPMODE & 2)) { if (dg_stage) stage(1, 1, t + 1); }
+      if constexpr (!(PPMODE & 2)) { if (s2) stage(0, 1, t + 2); }
     }
     PP_END();
     // p2: G0 computes Q(0,1)t | G1 reads bf(B1,t), stages B0(t+2)
@@ -1862,37 +1875,37 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
       read_bf(1, t);
       if constexpr (!(PPMODE & 2)) { if (s2) stage(1, 0, t + 2); }
     }
-    PP_END(PP_W1(4));
+    PP_END(PP_W1(8));
     // p3: G1 computes Q(1,1)t | G0 reads af(A0,t+1)+bf(B0,t+1), stages A1(t+2)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       if (dg_mfma) mfma32(1);
     } else {
-      if constexpr (PPMODE & 2) { if (s2) stage(0, 1, t + 2); }
+      if constexpr (PPMODE & 2) { if (s2) stage(1, 1, t + 2); }
       read_af(t + 1);
       read_bf(0, t + 1);
-      if constexpr (!(PPMODE & 2)) { if (s2) stage(0, 1, t + 2); }
+      if constexpr (!(PPMODE & 2)) { if (s2) stage(1, 1, t + 2); }
     }
-    PP_END(PP_W0(8));
+    PP_END(PP_W0(12));
     // p4: G0 computes Q(0,0)t+1 | G1 reads af(A1,t+1)+bf(B0,t+1), stages A0(t+2)
     if (g == 0) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       if (dg_mfma) mfma32(0);
     } else {
-      if constexpr (PPMODE & 2) { if (s2) stage(0, 0, t + 2); }
+      if constexpr (PPMODE & 2) { if (s3) stage(0, 0, t + 3); }
       read_af(t + 1);
       read_bf(0, t + 1);
-      if constexpr (!(PPMODE & 2)) { if (s2) stage(0, 0, t + 2); }
+      if constexpr (!(PPMODE & 2)) { if (s3) stage(0, 0, t + 3); }
     }
-    PP_END(PP_W0(4));
+    PP_END(PP_W0(8));
     // p5: G1 computes Q(1,0)t+1 | G0 reads bf(B1,t+1), stages B1(t+2)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       if (dg_mfma) mfma32(0);
     } else {
-      if constexpr (PPMODE & 2) { if (s2) stage(1, 1, t + 2); }
+      if constexpr (PPMODE & 2) { if (s3) stage(0, 1, t + 3); }
       read_bf(1, t + 1);
-      if constexpr (!(PPMODE & 2)) { if (s2) stage(1, 1, t + 2); }
+      if constexpr (!(PPMODE & 2)) { if (s3) stage(0, 1, t + 3); }
     }
     PP_END();
     // p6: G0 computes Q(0,1)t+1 | G1 reads bf(B1,t+1), stages B0(t+3)
@@ -1904,20 +1917,20 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
       read_bf(1, t + 1);
       if constexpr (!(PPMODE & 2)) { if (s3) stage(1, 0, t + 3); }
     }
-    PP_END(PP_W1(4));
+    PP_END(PP_W1(8));
     // p7: G1 computes Q(1,1)t+1 | G0 reads af(A0,t+2)+bf(B0,t+2), stages A1(t+3)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       if (dg_mfma) mfma32(1);
     } else {
-      if constexpr (PPMODE & 2) { if (s3) stage(0, 1, t + 3); }
+      if constexpr (PPMODE & 2) { if (s3) stage(1, 1, t + 3); }
       if (s2) {
         read_af(t + 2);
         read_bf(0, t + 2);
       }
-      if constexpr (!(PPMODE & 2)) { if (s3) stage(0, 1, t + 3); }
+      if constexpr (!(PPMODE & 2)) { if (s3) stage(1, 1, t + 3); }
     }
-    PP_END(PP_WB(8));
+    PP_END(PP_W0(12); PP_W1(8));
   }
 #undef PP_END
 #undef PP_W0
@@ -1998,6 +2011,414 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
       }
     }
   }
+}
+
+// ---------------------------------------------------------------------------
+// Variant 7-PERSISTENT: ping-pong kernel with the pipeline flowing ACROSS
+// tile boundaries — the short-K answer (BERT K=768 panels were 32.6% of the
+// step at ~540 TF on the 2-buffer persistent kernel).
+//
+// Tile-boundary correspondence (same discovery as gemm_bf16_nt_8phase_pers_k,
+// restated for the ping-pong stage order): the s2/s3-guarded stages of a
+// tile's LAST iteration (B0/A1/A0/B1 of t+2 at p2..p5, B0/A1 of t+3 at
+// p6/p7) are — in ORDER, GROUP and LDS SLOT — exactly the 6-unit prologue
+// of the NEXT tile (G1: B0(0) A0(0) B0(1); G0: A1(0) B1(0) A1(1); parity
+// matches because ntiles is even). Substituting next-tile addresses there
+// keeps every per-group glds count, hence the whole per-group vmcnt
+// schedule, steady across the sweep: NO drain and NO prologue between
+// tiles. The p7 fragment reads flow too: read_af/bf of "t+2" are next
+// tile's A0(0)/B0(0) from the very slots the substitution staged.
+//
+// Epilogue between tiles: stores are FIFO-YOUNGER than every glds a later
+// wait needs, so the steady vmcnt counts remain CORRECT certificates (a
+// wait only gets stricter — it may force store drain, never miss a load).
+// Interior tiles stage each 128x128 C-quadrant through a 32 KiB spare LDS
+// region (the full 160 KiB budget: 128 KiB pipeline + 32 KiB spare) and
+// stream it out as b128 stores — 16 store instructions per wave instead
+// of the 128-scalar scatter whose issue tail stalled the 8phase_pers
+// experiment. Edge tiles fall back to the masked scalar scatter.
+// ---------------------------------------------------------------------------
+__launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_pers_k(
+    GemmArgs args, int nwg_n, int64_t per_b, int64_t total) {
+  using namespace ph8;
+  __shared__ __attribute__((aligned(16))) uint16_t smem[2 * BUF_E + 16384];
+  uint16_t* spare = smem + 2 * BUF_E;  // 32 KiB epilogue staging
+
+  const int orig = blockIdx.x;  // 256 WGs
+  const int xcd = orig & 7, slotq = orig >> 3;  // 32 slots per XCD
+  const int64_t q = total / 8, r8 = total % 8;
+  const int64_t lo = xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q;
+  const int64_t cnt = xcd < r8 ? q + 1 : q;
+  if (slotq >= cnt) return;
+  constexpr int SLOTS = 32;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int g = wid >> 2;   // ping-pong group
+  const int ws = wid & 3;   // wave-in-group (2x2)
+  const int tig = (int)threadIdx.x & 255;
+  if (g == 1) __builtin_amdgcn_s_setprio(1);  // static younger-half prio
+
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  const int64_t ntiles = args.K / PBK;  // even, >= 2 (launch guard)
+
+  // compact per-tile state (3 x i32; the pointers are recomputed at use so
+  // the sweep's live set stays tiny — the fat 2-pointer TileSt spilled)
+  struct TileSt {
+    int32_t bidx, m0, n0;
+  };
+  auto make_state = [&](int64_t tl) {
+    TileSt ts;
+    // readfirstlane: these are wave-uniform; pinning them scalar keeps the
+    // sweep state out of the VGPR budget (the main loop sits at the cap)
+    ts.bidx = __builtin_amdgcn_readfirstlane((int32_t)(tl / per_b));
+    const int64_t rtl = tl % per_b;
+    ts.m0 = __builtin_amdgcn_readfirstlane((int32_t)((rtl / nwg_n) * PBM));
+    ts.n0 = __builtin_amdgcn_readfirstlane((int32_t)((rtl % nwg_n) * PBN));
+    return ts;
+  };
+
+  // one half-tile staged by ONE group (4 glds per thread); kt indexes the
+  // tile-local K-step, buf = kt & 1 (every tile starts at buf 0)
+  auto stage = [&](const TileSt& ts, int isB, int h, int64_t kt) {
+    int64_t k0 = kt * PBK;
+    uint16_t* dst = slot(smem, (int)(kt & 1), isB, h);
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int e = tig * 8 + c * 2048;
+      int row = e >> 6;
+      int k = (e & 63) ^ ((row & 7) << 3);
+      if (!isB) {
+        int64_t ar = ts.m0 + h * 128 + row;
+        if (ar > Mc) ar = Mc;
+        glds16(args.A + ts.bidx * args.sA + ar * lda + k0 + k, dst + e);
+      } else {
+        int64_t br = ts.n0 + h * 128 + row;
+        if (br > Nc) br = Nc;
+        glds16(args.B + ts.bidx * args.sB + br * ldb + k0 + k, dst + e);
+      }
+    }
+  };
+
+  f32x4 acc[2][4][4] = {};
+  short8 af[2][4], bf[2][4];
+
+  auto read_af = [&](int64_t kt) {
+    const uint16_t* as = slot(smem, (int)(kt & 1), 0, g);
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2) {
+      const int kbase = kk2 * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        int row = (ws >> 1) * 64 + fm * 16 + (lane & 15);
+        int ks = kbase ^ ((row & 7) << 3);
+        af[kk2][fm] = *(const short8*)(as + row * PBK + ks);
+      }
+    }
+  };
+  auto read_bf = [&](int b, int64_t kt) {
+    const uint16_t* bs = slot(smem, (int)(kt & 1), 1, b);
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2) {
+      const int kbase = kk2 * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = (ws & 1) * 64 + fn * 16 + (lane & 15);
+        int ks = kbase ^ ((col & 7) << 3);
+        bf[kk2][fn] = *(const short8*)(bs + col * PBK + ks);
+      }
+    }
+  };
+
+  auto mfma32 = [&](int b) {
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2)
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn)
+          acc[b][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[kk2][fm], bf[kk2][fn], acc[b][fm][fn], 0, 0, 0);
+  };
+
+  // Epilogue. Interior tiles stage each 128x128 C-quadrant through the
+  // 32 KiB spare LDS and stream it out as 4 b128 stores per thread
+  // (16 store instructions per wave for the whole tile). The store COUNT
+  // matters, not just the issue rate: epilogue stores sit in the vmcnt
+  // FIFO between the boundary prefetch and the next tile's counted waits,
+  // so the first post-boundary waits add +16 to their counts instead of
+  // force-draining the pipeline (a scalar scatter's 128 stores/wave blew
+  // past vmcnt's 63 cap and serialized every tile boundary at ~290 TF).
+  // Raw barriers + lgkm-only waits: __syncthreads would drain the
+  // in-flight next-tile glds.
+  auto epilogue = [&](const TileSt& ts) {
+    uint16_t* C = args.C + (int64_t)ts.bidx * args.sC;
+    float* out32 =
+        args.out_f32 ? args.out_f32 + (int64_t)ts.bidx * args.sC : nullptr;
+    uint16_t* save_pre =
+        args.save_pre ? args.save_pre + (int64_t)ts.bidx * args.sC : nullptr;
+    const int64_t N = args.N;
+    const bool interior =
+        !out32 && (ts.m0 + PBM <= args.M) && (ts.n0 + PBN <= args.N);
+    if (interior) {
+      for (int pass = 0; pass < 4; ++pass) {
+        const int ha = pass >> 1, hb = pass & 1;
+        if (g == ha) {
+#pragma unroll
+          for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+            for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+              for (int rI = 0; rI < 4; ++rI) {
+                int lrow = (ws >> 1) * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+                int lcol = (ws & 1) * 64 + fn * 16 + (lane & 15);
+                spare[lrow * 128 + lcol] =
+                    f2b(args.alpha * acc[hb][fm][fn][rI]);
+              }
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        asm volatile("" ::: "memory");
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          int e = (c * 512 + (int)threadIdx.x) * 8;
+          int row = e >> 7, col = e & 127;
+          short8 vv = *(const short8*)(spare + e);
+          int64_t gcol = ts.n0 + hb * 128 + col;
+          int64_t idx = (ts.m0 + ha * 128 + row) * N + gcol;
+          short8 zout, yout;
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            float v = b2f((uint16_t)vv[i]);
+            if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
+            if (args.bias) v += b2f(args.bias[gcol + i]);
+            zout[i] = (short)f2b(v);
+            if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+            yout[i] = (short)f2b(v);
+          }
+          if (save_pre) *(short8*)(save_pre + idx) = zout;
+          *(short8*)(C + idx) = yout;
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        asm volatile("" ::: "memory");
+      }
+    } else {
+#pragma unroll
+      for (int b2 = 0; b2 < 2; ++b2)
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+            for (int rI = 0; rI < 4; ++rI) {
+              int64_t grow = ts.m0 + g * 128 + (ws >> 1) * 64 + fm * 16 +
+                             (lane >> 4) * 4 + rI;
+              int64_t gcol =
+                  ts.n0 + b2 * 128 + (ws & 1) * 64 + fn * 16 + (lane & 15);
+              if (grow < args.M && gcol < args.N) {
+                int64_t idx = grow * N + gcol;
+                if (out32) {
+                  out32[idx] = acc[b2][fm][fn][rI];
+                } else {
+                  float v = args.alpha * acc[b2][fm][fn][rI];
+                  if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+                  if (args.bias) v += b2f(args.bias[gcol]);
+                  if (save_pre) save_pre[idx] = f2b(v);
+                  if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+                  C[idx] = f2b(v);
+                }
+              }
+            }
+      // a non-interior epilogue's scalar stores overflow the countable
+      // window — drain them (rare: edge tiles / split-K slabs only)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      asm volatile("" ::: "memory");
+    }
+#pragma unroll
+    for (int b2 = 0; b2 < 2; ++b2)
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn) acc[b2][fm][fn] = f32x4{};
+  };
+
+#define PPP_END(wait_stmt)                                 \
+  do {                                                     \
+    wait_stmt;                                             \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");     \
+    __builtin_amdgcn_s_barrier();                          \
+    asm volatile("" ::: "memory");                         \
+  } while (0)
+// fresh = first iteration after a tile epilogue: its 16 b128 stores per
+// wave are still in the FIFO between the boundary prefetch and these
+// waits, so the counts widen by 16. FIFO completion order makes the wide
+// form safe even once the stores drain (stores drained => every older
+// load drained too).
+#define PPP_W0(n, nf)                                         \
+  if (tail)                                                   \
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");          \
+  else if (g == 0) {                                          \
+    if (fresh)                                                \
+      asm volatile("s_waitcnt vmcnt(" #nf ")" ::: "memory");  \
+    else                                                      \
+      asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory");   \
+  }
+#define PPP_W1(n, nf)                                         \
+  if (tail)                                                   \
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");          \
+  else if (g == 1) {                                          \
+    if (fresh)                                                \
+      asm volatile("s_waitcnt vmcnt(" #nf ")" ::: "memory");  \
+    else                                                      \
+      asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory");   \
+  }
+
+  int64_t tl = lo + slotq;
+  TileSt cur = make_state(tl);
+  bool have_next = (tl + SLOTS) < (lo + cnt);
+  TileSt nxt = have_next ? make_state(tl + SLOTS) : cur;
+
+  // first-tile prologue: tiles 0 and 1 fully staged, 4 units per group
+  // in the in-loop FIFO order (deep schedule: 5-7 phase flights)
+  if (g == 1) {
+    stage(cur, 0, 0, 0);  // A0(0)
+    stage(cur, 1, 0, 0);  // B0(0)
+    stage(cur, 0, 0, 1);  // A0(1)
+    stage(cur, 1, 0, 1);  // B0(1)
+  } else {
+    stage(cur, 0, 1, 0);  // A1(0)
+    stage(cur, 1, 1, 0);  // B1(0)
+    stage(cur, 0, 1, 1);  // A1(1)
+    stage(cur, 1, 1, 1);  // B1(1)
+  }
+  if (g == 0)
+    asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+  else
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  if (g == 0) {
+    read_af(0);
+    read_bf(0, 0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+  for (; tl < lo + cnt; tl += SLOTS) {
+    const bool not_first_tile = (tl != lo + slotq);
+    for (int64_t t = 0; t < ntiles; t += 2) {
+      const bool last_it = (t + 2 >= ntiles);
+      const bool fresh = (t == 0) && not_first_tile;
+      // stage targets: tile-local K-step kt < ntiles stays on cur; the
+      // overflow (kt - ntiles on nxt) is the next tile's prologue
+      const bool tail = last_it && !have_next;  // true end: drain
+      const bool s2 = !last_it || have_next;
+      const bool s3 = (t + 3 < ntiles) || have_next;
+      auto st = [&](int isB, int h, int64_t kt) {
+        // value-select the 3-int state (one inlined stage body, not two)
+        const bool ov = kt >= ntiles;
+        TileSt ts = ov ? nxt : cur;
+        stage(ts, isB, h, ov ? kt - ntiles : kt);
+      };
+      // p0: G0 Q(0,0)t | G1 reads af(A1,t)+bf(B0,t), stages A0(t+1)
+      if (g == 0) {
+        mfma32(0);
+      } else {
+        read_af(t);
+        read_bf(0, t);
+        if (s2) st(0, 0, t + 2);
+      }
+      PPP_END(PPP_W0(8, 24));
+      // p1: G1 Q(1,0)t | G0 reads bf(B1,t), stages B1(t+1)
+      if (g == 1) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma32(0);
+      } else {
+        read_bf(1, t);
+        if (s2) st(0, 1, t + 2);
+      }
+      PPP_END();
+      // p2: G0 Q(0,1)t | G1 reads bf(B1,t), stages B0(t+2)
+      if (g == 0) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma32(1);
+      } else {
+        read_bf(1, t);
+        if (s2) st(1, 0, t + 2);
+      }
+      PPP_END(PPP_W1(8, 24));
+      // p3: G1 Q(1,1)t | G0 reads af(A0,t+1)+bf(B0,t+1), stages A1(t+2)
+      if (g == 1) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma32(1);
+      } else {
+        read_af(t + 1);
+        read_bf(0, t + 1);
+        if (s2) st(1, 1, t + 2);
+      }
+      PPP_END(PPP_W0(12, 28));
+      // p4: G0 Q(0,0)t+1 | G1 reads af(A1,t+1)+bf(B0,t+1), stages A0(t+2)
+      if (g == 0) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma32(0);
+      } else {
+        read_af(t + 1);
+        read_bf(0, t + 1);
+        if (s3) st(0, 0, t + 3);
+      }
+      PPP_END(PPP_W0(8, 24));
+      // p5: G1 Q(1,0)t+1 | G0 reads bf(B1,t+1), stages B1(t+2)
+      if (g == 1) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma32(0);
+      } else {
+        read_bf(1, t + 1);
+        if (s3) st(0, 1, t + 3);
+      }
+      PPP_END();
+      // p6: G0 Q(0,1)t+1 | G1 reads bf(B1,t+1), stages B0(t+3)
+      if (g == 0) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma32(1);
+      } else {
+        read_bf(1, t + 1);
+        if (s3) st(1, 0, t + 3);
+      }
+      PPP_END(PPP_W1(8, 24));
+      // p7: G1 Q(1,1)t+1 | G0 reads af+bf of t+2 (within-tile only: the
+      // cross-boundary read happens AFTER the epilogue so af/bf are not
+      // live across it — keeping them live spilled 1.2 KB of scratch),
+      // stages A1(t+3)
+      if (g == 1) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma32(1);
+      } else {
+        if (!last_it) {
+          read_af(t + 2);
+          read_bf(0, t + 2);
+        }
+        if (s3) st(1, 1, t + 3);
+      }
+      PPP_END(PPP_W0(12, 28); PPP_W1(8, 24));
+    }
+    epilogue(cur);
+    cur = nxt;
+    have_next = (tl + 2 * SLOTS) < (lo + cnt);
+    nxt = have_next ? make_state(tl + 2 * SLOTS) : cur;
+    if (tl + SLOTS < lo + cnt && g == 0) {
+      // next tile's p0 fragments (slots staged by the boundary overflow;
+      // certified by the last steady end-p6 wait + the epilogue barriers)
+      read_af(0);
+      read_bf(0, 0);
+    }
+  }
+#undef PPP_END
+#undef PPP_W0
+#undef PPP_W1
 }
 
 // ---------------------------------------------------------------------------
@@ -2995,9 +3416,27 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
       return;
     }
   }
-  // ping-pong wants enough K-iterations to amortize its prologue; short-K
-  // shapes (BERT K=768) measure faster on the persistent kernel below
-  // (gpurun_out/r2_defaults.log: 536-526 vs 539-569 TF)
+  // short-K persistent ping-pong: EXPERIMENTAL opt-in (NEZHA_GEMM_PERS_PP=1).
+  // The cross-tile flow + deep schedule are simulator-verified, but the
+  // epilogue's register pressure fragments the MFMA clusters (303 scratch
+  // ops in the hot span) and it measures at/below the 2-WG/CU persistent
+  // kernel; the fix is an acc-halving BN=128 regeometry — future work.
+  static const bool kPersPP =
+      getenv("NEZHA_GEMM_PERS_PP") && atoi(getenv("NEZHA_GEMM_PERS_PP")) == 1;
+  if (!transA && transB && kPersPP && kNtVariant >= 7 && args.K % 128 == 0 &&
+      args.K >= 256 && args.K < 1536) {
+    int nwg_m8 = (int)((args.M + 255) / 256);
+    int nwg_n8 = (int)((args.N + 255) / 256);
+    int64_t per_b8 = (int64_t)nwg_m8 * nwg_n8;
+    int64_t total8 = per_b8 * batch;
+    if (total8 >= 512) {
+      gemm_bf16_nt_pingpong_pers_k<<<dim3(256), dim3(512), 0, s>>>(
+          args, nwg_n8, per_b8, total8);
+      NZ_KERNEL_CHECK();
+      return;
+    }
+  }
+  // ping-pong single-tile for deep K
   if (!transA && transB && batch == 1 && kNtVariant == 7 &&
       args.K % 128 == 0 && args.K >= 1536) {
     int nwg_m = (int)((args.M + 255) / 256);

@@ -36,5 +36,45 @@ for (M, N, K, tA) in [(512, 512, 512, False), (2048, 1024, 768, False),
         print(f"{M}x{N}x{K}: deterministic + oracle ok (rel err {err:.4f})", flush=True)
     else:
         print(f"{M}x{N}x{K}: deterministic over 3 runs", flush=True)
+
+# Cross-kernel oracle for shapes too big for numpy: the dispatch-selected
+# kernel vs the proven tier-A/128^2 route (NEZHA_GEMM_VARIANT=0). Both
+# accumulate f32 over the same K (different order), so rel err ~ bf16 eps.
+# Covers the persistent ping-pong (short-K) and ping-pong (deep-K) routes.
+for (M, N, K, tA) in [(32768, 1024, 768, False), (16384, 2304, 768, False),
+                      (8192, 4096, 4096, False)]:
+    A = ctx.tensor((K, M) if tA else (M, K), nz.BF16).rand_normal(11)
+    B = ctx.tensor((K, N) if tA else (N, K), nz.BF16).rand_normal(12)
+    C = ctx.tensor((M, N), nz.BF16)
+    nz.gemm(A, B, C, tA, not tA)
+    got = C.numpy()
+    import subprocess, sys as _sys
+    # reference kernel in a subprocess (variant env is latched per process)
+    import tempfile, os as _os
+    ref_env = dict(os.environ, NEZHA_GEMM_VARIANT="0")
+    code = (
+        "import os,sys; sys.path.insert(0, %r)\n"
+        "import numpy as np, nezha_amd as nz\n"
+        "ctx = nz.Ctx(nz.HIP, 0)\n"
+        "A = ctx.tensor((%d, %d), nz.BF16).rand_normal(11)\n"
+        "B = ctx.tensor((%d, %d), nz.BF16).rand_normal(12)\n"
+        "C = ctx.tensor((%d, %d), nz.BF16)\n"
+        "nz.gemm(A, B, C, %r, %r)\n"
+        "np.save(sys.argv[1], C.numpy())\n"
+    ) % (os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+         K if tA else M, M if tA else K,
+         K if tA else N, N if tA else K, M, N, tA, not tA)
+    with tempfile.NamedTemporaryFile(suffix=".npy", delete=False) as f:
+        refp = f.name
+    subprocess.run([_sys.executable, "-c", code, refp], env=ref_env,
+                   check=True)
+    ref = np.load(refp)
+    _os.unlink(refp)
+    scale = np.abs(ref).max() + 1e-9
+    err = np.abs(got - ref).max() / scale
+    assert err < 0.02, f"CROSS-KERNEL MISMATCH {M}x{N}x{K}: rel err {err}"
+    print(f"{M}x{N}x{K}: cross-kernel oracle ok (rel err {err:.4f})",
+          flush=True)
+
 print("RACE-SCREEN-OK variant=%s min_wg=%s" % (os.environ.get("NEZHA_GEMM_VARIANT", "2"),
                                                os.environ.get("NEZHA_GEMM_MIN_WG", "224")))
