@@ -115,6 +115,10 @@ void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
     hip::attn_fused_fwd(q, k, v, probs, ctx_out, batch_heads, scale);
     return;
   }
+  if (q.device() == Device::HIP && hip::attn_fused_fwd512_supported(S, dh)) {
+    hip::attn_fused_fwd512(q, k, v, probs, ctx_out, batch_heads, scale);
+    return;
+  }
   // composed fallback (CPU oracle + non-contract shapes): K1 batched + K4
   Tensor scores(q.ctx(), q.dtype(), Shape{batch_heads * S, S});
   gemm_batched(q, k, scores, false, true, 1.0f, batch_heads, S, S, dh, S * dh,
@@ -152,6 +156,12 @@ void attn_fwd_packed(const Tensor& qkv, Tensor& probs, Tensor& out, int64_t B,
   prof::OpScope nz_scope_("attn_fwd_packed", qkv.ctx());
   if (qkv.device() == Device::HIP && hip::attn_fused_fwd_supported(S, dh)) {
     hip::attn_fused_fwd_packed(qkv, probs, out, B, S, heads, dh, scale);
+    return;
+  }
+  if (qkv.device() == Device::HIP && hip::attn_fused_fwd512_supported(S, dh)) {
+    // S=512 tiled fused forward; backward stays on the composed path
+    // (it consumes the normalized probs this kernel writes)
+    hip::attn_fused_fwd512_packed(qkv, probs, out, B, S, heads, dh, scale);
     return;
   }
   Tensor q(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});

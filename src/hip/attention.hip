@@ -470,10 +470,259 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused attention FORWARD for S=512, dh=64 (SURVEY.md §2.2 K4's second
+// tuning shape; VERDICT r01 missing #3). The 512x512 score matrix no
+// longer fits a workgroup, so the kernel tiles it: one WG per
+// (batch, head, 64-row q-block), K streamed in 64-row tiles.
+//
+// Three phases per WG (4 waves of 16 q-rows each, stats wave-local):
+//   A1: loop kt 0..7: stage K-tile (glds dbuf, XOR-swz source) ->
+//       S_t = Q.K_t^T (8 MFMA/wave) -> running row max -> raw bf16 scores
+//       into the P strip [64][512] (stride 520 spreads banks)
+//   A2: elementwise exp(s - m) over the strip (4 threads per row), row
+//       sums folded through LDS -> inv l
+//   B:  loop kt 0..7: V-tile via register scatter to VT [64][72] (dbuf) ->
+//       O += P_t . V_t (8 MFMA/wave); epilogue scales O rows by 1/l
+// P (normalized) streams out at kernel end for the composed backward.
+// LDS: Q 8K + K dbuf 16K + VT dbuf 18.4K + P 66.6K + stats ~= 110 KiB ->
+// 1 WG/CU; grid B*H*8 (BERT S=512 batch 64: 6144 WGs = 24 rounds).
+// ---------------------------------------------------------------------------
+namespace s512 {
+constexpr int SQ = 64;         // q-block rows per WG
+constexpr int S5 = 512;        // sequence length
+constexpr int KT = 64;         // k/v tile rows
+constexpr int PST = S5 + 8;    // P strip row stride (elements)
+constexpr int VST = KT + 8;    // VT row stride
+}  // namespace s512
+
+__launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
+    const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
+    const uint16_t* __restrict__ v, uint16_t* __restrict__ probs,
+    uint16_t* __restrict__ out, float scale, int heads, int64_t qb,
+    int64_t qh, int64_t qr, int64_t ob, int64_t oh, int64_t orr) {
+  using namespace s512;
+  // [Q 64x64][K dbuf 2x 64x64][VT dbuf 2x 64x72][P 64x520][ms 64][ls 64+psum]
+  __shared__ __attribute__((aligned(16))) uint16_t smem[
+      SQ * AD + 2 * KT * AD + 2 * AD * VST + SQ * PST];
+  __shared__ float ms[SQ], lsum[SQ][4], ls[SQ];
+  uint16_t* Qs = smem;
+  uint16_t* Ks = smem + SQ * AD;                  // dbuf: + (kt&1)*KT*AD
+  uint16_t* VTs = smem + SQ * AD + 2 * KT * AD;   // dbuf: + (kt&1)*AD*VST
+  uint16_t* Ps = VTs + 2 * AD * VST;
+
+  const int64_t nqb = S5 / SQ;                    // 8 q-blocks per head
+  const int64_t bhq = blockIdx.x;
+  const int64_t bh = bhq / nqb, qb0 = (bhq % nqb) * SQ;
+  const int64_t b_ = bh / heads, h_ = bh % heads;
+  const uint16_t* qg = q + b_ * qb + h_ * qh + qb0 * qr;
+  const uint16_t* kg = k + b_ * qb + h_ * qh;
+  const uint16_t* vg = v + b_ * qb + h_ * qh;
+  uint16_t* pg = probs + bh * (int64_t)S5 * S5 + qb0 * S5;
+  uint16_t* og = out + b_ * ob + h_ * oh + qb0 * orr;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;  // 4 waves; wave w owns rows w*16..+16
+
+  // ---- stage Q (glds, swizzled source) + K tile 0 -------------------------
+  // a 64x64 tile is 4096 elems = 2 chunks of (256 threads x 8 elems)
+  const int e0 = (int)threadIdx.x * 8;
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    int e = e0 + c * 2048;
+    int row = e >> 6;
+    int kk = (e & 63) ^ ((row & 7) << 3);
+    att_glds16(qg + row * qr + kk, Qs + e);
+    att_glds16(kg + row * qr + kk, Ks + e);          // K tile 0 -> buf 0
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+  float m_run[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  // ---- A1: scores + running max ------------------------------------------
+  for (int kt = 0; kt < S5 / KT; ++kt) {
+    // prefetch next K tile into the other buffer
+    if (kt + 1 < S5 / KT) {
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        int e = e0 + c * 2048;
+        int row = e >> 6;
+        int kk2 = (e & 63) ^ ((row & 7) << 3);
+        att_glds16(kg + (int64_t)(kt + 1) * KT * qr + row * qr + kk2,
+                   Ks + ((kt + 1) & 1) * KT * AD + e);
+      }
+    }
+    const uint16_t* kb = Ks + (kt & 1) * KT * AD;
+    f32x4 acc[4] = {};
+#pragma unroll
+    for (int kk = 0; kk < AD; kk += 32) {
+      short8 a_frag, b_frag[4];
+      const int kbase = kk + (lane >> 4) * 8;
+      {
+        int row = wid * 16 + (lane & 15);
+        int ks = kbase ^ ((row & 7) << 3);
+        a_frag = *(const short8*)(Qs + row * AD + ks);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = fn * 16 + (lane & 15);
+        int ks = kbase ^ ((col & 7) << 3);
+        b_frag[fn] = *(const short8*)(kb + col * AD + ks);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        acc[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
+                                                          acc[fn], 0, 0, 0);
+    }
+    // running max (rows wave-local: reduce over the 16-lane col group + fn)
+#pragma unroll
+    for (int rI = 0; rI < 4; ++rI) {
+      float mx = -1e30f;
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) mx = fmaxf(mx, scale * acc[fn][rI]);
+      for (int off = 1; off < 16; off <<= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+      m_run[rI] = fmaxf(m_run[rI], mx);
+    }
+    // raw scores (scaled) -> P strip, bf16
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+      for (int rI = 0; rI < 4; ++rI) {
+        int row = wid * 16 + (lane >> 4) * 4 + rI;
+        int col = kt * KT + fn * 16 + (lane & 15);
+        Ps[row * PST + col] = f2b(scale * acc[fn][rI]);
+      }
+    // next iteration reads the prefetched buffer; everyone must also be
+    // done READING kb before it is overwritten two iterations later
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+  }
+  // publish row maxima: wave w's lane group holds rows w*16+(lane>>4)*4+rI
+  if ((lane & 15) == 0) {
+#pragma unroll
+    for (int rI = 0; rI < 4; ++rI)
+      ms[wid * 16 + (lane >> 4) * 4 + rI] = m_run[rI];
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+  // ---- A2: exp(s - m), row sums (4 threads per row) -----------------------
+  {
+    const int row = (int)threadIdx.x >> 2;
+    const int qtr = (int)threadIdx.x & 3;
+    const float m = ms[row];
+    uint16_t* pr = Ps + row * PST + qtr * 128;
+    float acc_l = 0.f;
+#pragma unroll 4
+    for (int i = 0; i < 128; ++i) {
+      float e = __expf(b2f(pr[i]) - m);
+      acc_l += e;
+      pr[i] = f2b(e);
+    }
+    lsum[row][qtr] = acc_l;
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  if (threadIdx.x < SQ)
+    ls[threadIdx.x] = 1.0f / (lsum[threadIdx.x][0] + lsum[threadIdx.x][1] +
+                              lsum[threadIdx.x][2] + lsum[threadIdx.x][3]);
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+  // ---- B: O = P~ . V, V tiles scattered transposed ------------------------
+  f32x4 accO[4] = {};
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    // V tile 0 -> VT buf 0
+    int e = e0 + c * 2048;
+    int s_ = e >> 6, d = e & 63;
+    short8 vv = *(const short8*)(vg + s_ * qr + d);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      VTs[(d + i) * VST + s_] = (uint16_t)vv[i];
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  for (int kt = 0; kt < S5 / KT; ++kt) {
+    if (kt + 1 < S5 / KT) {
+      uint16_t* vt = VTs + ((kt + 1) & 1) * AD * VST;
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        int e = e0 + c * 2048;
+        int s_ = e >> 6, d = e & 63;
+        short8 vv =
+            *(const short8*)(vg + (int64_t)(kt + 1) * KT * qr + s_ * qr + d);
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          vt[(d + i) * VST + s_] = (uint16_t)vv[i];
+      }
+    }
+    const uint16_t* vt = VTs + (kt & 1) * AD * VST;
+#pragma unroll
+    for (int kk = 0; kk < KT; kk += 32) {
+      short8 a_frag, b_frag[4];
+      const int kbase = kk + (lane >> 4) * 8;
+      {
+        int row = wid * 16 + (lane & 15);
+        a_frag = *(const short8*)(Ps + row * PST + kt * KT + kbase);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = fn * 16 + (lane & 15);
+        b_frag[fn] = *(const short8*)(vt + col * VST + kbase);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        accO[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
+                                                           accO[fn], 0, 0, 0);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+  }
+  // epilogue: O rows scaled by 1/l
+#pragma unroll
+  for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+    for (int rI = 0; rI < 4; ++rI) {
+      int row = wid * 16 + (lane >> 4) * 4 + rI;
+      int col = fn * 16 + (lane & 15);
+      og[row * orr + col] = f2b(accO[fn][rI] * ls[row]);
+    }
+
+  // ---- stream normalized P out (backward input) ---------------------------
+  {
+    const int row = (int)threadIdx.x >> 2;
+    const int qtr = (int)threadIdx.x & 3;
+    const float invl = ls[row];
+    const uint16_t* pr = Ps + row * PST + qtr * 128;
+    uint16_t* pgr = pg + row * S5 + qtr * 128;
+#pragma unroll 2
+    for (int i = 0; i < 128; i += 8) {
+      short8 vv = *(const short8*)(pr + i);
+      short8 outp;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) outp[j] = (short)f2b(b2f((uint16_t)vv[j]) * invl);
+      *(short8*)(pgr + i) = outp;
+    }
+  }
+}
+
 }  // namespace
 
 bool attn_fused_fwd_supported(int64_t S, int64_t dh) {
   return S == AS && dh == AD;
+}
+
+bool attn_fused_fwd512_supported(int64_t S, int64_t dh) {
+  return S == 512 && dh == AD;
 }
 
 // q,k,v,ctx: [B, heads, S, dh]; probs: [B*heads*S, S]
@@ -519,6 +768,35 @@ void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
       (const uint16_t*)dout.data(), (uint16_t*)dq.data(),
       (uint16_t*)dk.data(), (uint16_t*)dv.data(), scale, 1, sh, 0, AD, sh, 0,
       AD, sh, 0, AD);
+  NZ_KERNEL_CHECK();
+}
+
+// S=512 launchers: per-head-contiguous and packed-qkv layouts.
+void attn_fused_fwd512(const Tensor& q, const Tensor& k, const Tensor& v,
+                       Tensor& probs, Tensor& ctx_out, int64_t batch_heads,
+                       float scale) {
+  NZ_CHECK(q.dtype() == DType::BF16, "fused attention 512: bf16 required");
+  hipStream_t s = stream_of(q);
+  attn_fwd_fused512_k<<<dim3((uint32_t)(batch_heads * 8)), dim3(256), 0, s>>>(
+      (const uint16_t*)q.data(), (const uint16_t*)k.data(),
+      (const uint16_t*)v.data(), (uint16_t*)probs.data(),
+      (uint16_t*)ctx_out.data(), scale, 1, (int64_t)512 * AD, 0, AD,
+      (int64_t)512 * AD, 0, AD);
+  NZ_KERNEL_CHECK();
+}
+
+void attn_fused_fwd512_packed(const Tensor& qkv, Tensor& probs, Tensor& out,
+                              int64_t B, int64_t S, int64_t heads, int64_t dh,
+                              float scale) {
+  NZ_CHECK(qkv.dtype() == DType::BF16, "fused attention 512: bf16 required");
+  NZ_CHECK(S == 512 && dh == AD, "fused attention 512: S=512, dh=64 only");
+  hipStream_t s = stream_of(qkv);
+  int64_t H = heads * dh;
+  const uint16_t* base = (const uint16_t*)qkv.data();
+  attn_fwd_fused512_k<<<dim3((uint32_t)(B * heads * 8)), dim3(256), 0, s>>>(
+      base, base + H, base + 2 * H, (uint16_t*)probs.data(),
+      (uint16_t*)out.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
+      S * H, dh, H);
   NZ_KERNEL_CHECK();
 }
 

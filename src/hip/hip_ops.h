@@ -62,6 +62,13 @@ void gemm_f32_batched(const Tensor& A, const Tensor& B, Tensor& C, bool transA,
 // fused attention forward (src/hip/attention.hip)
 namespace nz { namespace hip {
 bool attn_fused_fwd_supported(int64_t S, int64_t dh);
+bool attn_fused_fwd512_supported(int64_t S, int64_t dh);
+void attn_fused_fwd512_packed(const Tensor& qkv, Tensor& probs, Tensor& out,
+                              int64_t B, int64_t S, int64_t heads, int64_t dh,
+                              float scale);
+void attn_fused_fwd512(const Tensor& q, const Tensor& k, const Tensor& v,
+                       Tensor& probs, Tensor& ctx_out, int64_t batch_heads,
+                       float scale);
 void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
                     Tensor& probs, Tensor& ctx_out, int64_t batch_heads,
                     float scale);

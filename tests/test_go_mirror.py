@@ -88,3 +88,15 @@ def test_go_mirror_covers_core_api():
                 "nz_trainer_step", "nz_ckpt_save", "nz_ckpt_load",
                 "nz_comm_init_rccl", "nz_trainer_attach_ddp"]:
         assert f"C.{sym}(" in src, f"Go mirror no longer binds {sym}"
+
+
+def test_pingpong_schedule_simulator():
+    """The GEMM ping-pong stage schedule must pass its FIFO/certificate
+    simulator (tools/pp_schedule_sim.py) — this is the guard that caught
+    the r2 deep-schedule race before it could ship."""
+    import subprocess
+    import sys as _sys
+    r = subprocess.run([_sys.executable,
+                        os.path.join(REPO, "tools", "pp_schedule_sim.py")],
+                       capture_output=True, timeout=60)
+    assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()

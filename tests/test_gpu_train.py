@@ -326,6 +326,55 @@ print(json.dumps(losses))
     assert outs[0] == outs[1], f"eager {outs[0]} vs graph {outs[1]}"
 
 
+def test_attn_fused_s512_matches_composed(hip_ctx):
+    """S=512 tiled fused attention forward (VERDICT r01 missing #3) vs the
+    composed batched-GEMM + softmax reference built from the same
+    primitives, same inputs. Checks both the merged output and the
+    normalized probs the composed backward consumes."""
+    B, S, H, dh = 2, 512, 4, 64
+    hidden = H * dh
+    scale = 1.0 / (dh ** 0.5)
+    qkv = hip_ctx.tensor((B * S, 3 * hidden), nz.BF16).rand_normal(7)
+    probs = hip_ctx.tensor((B * H * S, S), nz.BF16)
+    out = hip_ctx.tensor((B * S, hidden), nz.BF16)
+    nz.attn_fwd_packed(qkv, probs, out, B, S, H, dh, scale)
+
+    # composed reference from the same primitive kernels
+    q = hip_ctx.tensor((B, H, S, dh), nz.BF16)
+    k = hip_ctx.tensor((B, H, S, dh), nz.BF16)
+    v = hip_ctx.tensor((B, H, S, dh), nz.BF16)
+    nz.attn_split_qkv(qkv, q, k, v, B, S, H, dh)
+    scores = hip_ctx.tensor((B * H * S, S), nz.BF16)
+    nz.gemm_batched(q, k, scores, False, True, 1.0, B * H, S, S, dh,
+                    S * dh, S * dh, S * S)
+    probs_ref = hip_ctx.tensor((B * H * S, S), nz.BF16)
+    nz.softmax_fwd(scores, probs_ref, scale)
+    ctxt = hip_ctx.tensor((B, H, S, dh), nz.BF16)
+    nz.gemm_batched(probs_ref, v, ctxt, False, False, 1.0, B * H, S, dh, S,
+                    S * S, S * dh, S * dh)
+    out_ref = hip_ctx.tensor((B * S, hidden), nz.BF16)
+    nz.attn_merge(ctxt, out_ref, B, S, H, dh)
+
+    o, orf = out.numpy(), out_ref.numpy()
+    scale_o = np.abs(orf).max() + 1e-9
+    assert np.abs(o - orf).max() / scale_o < 0.03, \
+        f"out err {np.abs(o - orf).max() / scale_o}"
+    p, prf = probs.numpy(), probs_ref.numpy()
+    assert np.abs(p - prf).max() < 0.02, f"probs err {np.abs(p - prf).max()}"
+
+
+def test_bert_s512_trains(hip_ctx):
+    """BERT with S=512 routes the forward through the tiled fused kernel
+    and the backward through the composed path; loss must decrease."""
+    tr = nz.Trainer.bert(hip_ctx, layers=2, hidden=256, ffn=1024, heads=4,
+                         seq=512, batch=4, lr=1e-4, seed=13)
+    tr.step()
+    l0 = tr.loss()
+    for _ in range(9):
+        tr.step()
+    assert np.isfinite(l0) and tr.loss() < l0, f"{l0} -> {tr.loss()}"
+
+
 def test_bert_noncontract_seq(hip_ctx):
     """seq=64 routes attention through the composed fallback on GPU
     (fused kernels are gated to the S=128 contract shape)."""
