@@ -1056,6 +1056,27 @@ __device__ __forceinline__ uint16_t* slot(uint16_t* smem, int buf, int isB,
                                           int half) {
   return smem + buf * BUF_E + (isB * 2 + half) * HALF_E;
 }
+// compile-time vmcnt wait (asm needs a literal; N<0 emits nothing)
+template <int N>
+__device__ __forceinline__ void vm_wait() {
+  if constexpr (N == 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else if constexpr (N == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else if constexpr (N == 12) asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+}
+// one wait site of the ping-pong schedule: group GRP waits the per-mode
+// count (NT_N for TRMODE 0, NN_N for TRMODE 1, nothing for TN); tails
+// drain everything
+template <int TRMODE, int GRP, int NT_N, int NN_N>
+__device__ __forceinline__ void pp_site(int g, bool tail) {
+  if (tail) {
+    vm_wait<0>();
+    return;
+  }
+  if (g != GRP) return;
+  if constexpr (TRMODE == 0) vm_wait<NT_N>();
+  else if constexpr (TRMODE == 1) vm_wait<NN_N>();
+}
 }  // namespace ph8
 
 __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase_k(GemmArgs args,
@@ -1676,7 +1697,17 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase6_k(GemmArgs args,
 // group (B0(0), A1(0), A0(0), B1(0)) for the pre-read and p0/p1.
 // Tails drain with vmcnt(0) as in variants 2/6.
 // ---------------------------------------------------------------------------
-template <int PPMODE>  // bit0: static G1 prio; bit1: stage before reads
+// TRMODE: native-layout staging (r2: kills the NN/TN lowering transposes).
+//   bit0: B stored [K,N] (k-major) — staged via short8 row loads + ds_write
+//         transpose scatter into the standard [n][k] XOR image;
+//   bit1: A stored [K,M] — same for A.
+// Transposed stages are SYNCHRONOUS (their loads drain before the scatter),
+// so they need no vmcnt certificates. Per-mode wait schedules are
+// simulator-verified (tools/pp_schedule_sim.py style derivation):
+//   TRMODE=0 (NT): the deep schedule's 8/12 waits;
+//   TRMODE=1 (NN): only A-units in flight -> all four waits vmcnt(4);
+//   TRMODE=3 (TN): every unit synchronous -> no steady waits at all.
+template <int PPMODE, int TRMODE = 0>
 __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
     GemmArgs args, int nwg_n) {
   using namespace ph8;
@@ -1714,9 +1745,52 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
   const int64_t ntiles = args.K / PBK;  // even, >= 6 (launch guard)
 
   // one half-tile (16 KB) staged by ONE group: 4 glds per thread
+  // k-major transposed stage: operand stored [K, R]; each thread of the
+  // staging group covers (k-row, 8 consecutive r) and scatters into a
+  // ROTATED [r][k] XOR image: col = (k ^ ((r&7)<<3)) + (r>>3 & 7)*8 mod 64.
+  // Without the (r>>3) rotation every lane of a scatter instruction lands
+  // on the SAME bank (row stride 128 B, swizzle constant across lanes) —
+  // a 64-way conflict that measured 6-10x slower. Loads are hoisted ahead
+  // of the scatters so their latency overlaps.
+  auto stage_tr = [&](const uint16_t* X, int64_t ldx, int64_t r0,
+                      int64_t Rlast8, uint16_t* dst, int64_t k0) {
+    short8 vv[4];
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int e = tig * 8 + c * 2048;  // [k][r] order over the 64x128 half
+      int kk = e >> 7;
+      int rl = e & 127;
+      int64_t gr = r0 + rl;
+      if (gr > Rlast8) gr = Rlast8;  // 8-aligned clamp (R % 8 == 0 gated)
+      vv[c] = *(const short8*)(X + (k0 + kk) * ldx + gr);
+    }
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int e = tig * 8 + c * 2048;
+      int kk = e >> 7;
+      int rl = e & 127;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        int r = rl + i;
+        int col = ((kk ^ ((r & 7) << 3)) + ((r >> 3) << 3)) & 63;
+        dst[r * PBK + col] = (uint16_t)vv[c][i];
+      }
+    }
+  };
   auto stage = [&](int isB, int h, int64_t t) {
     int64_t k0 = t * PBK;
     uint16_t* dst = slot(smem, (int)(t & 1), isB, h);
+    if (!isB) {
+      if constexpr ((TRMODE & 2) != 0) {
+        stage_tr(A, lda, m0 + h * 128, args.M - 8, dst, k0);
+        return;
+      }
+    } else {
+      if constexpr ((TRMODE & 1) != 0) {
+        stage_tr(B, ldb, n0 + h * 128, args.N - 8, dst, k0);
+        return;
+      }
+    }
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       int e = tig * 8 + c * 2048;
@@ -1747,6 +1821,8 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
       for (int fm = 0; fm < 4; ++fm) {
         int row = (ws >> 1) * 64 + fm * 16 + (lane & 15);
         int ks = kbase ^ ((row & 7) << 3);
+        if constexpr ((TRMODE & 2) != 0)
+          ks = (ks + ((row >> 3) << 3)) & 63;  // stage_tr's rotated image
         af[kk2][fm] = *(const short8*)(as + row * PBK + ks);
       }
     }
@@ -1761,6 +1837,8 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
       for (int fn = 0; fn < 4; ++fn) {
         int col = (ws & 1) * 64 + fn * 16 + (lane & 15);
         int ks = kbase ^ ((col & 7) << 3);
+        if constexpr ((TRMODE & 1) != 0)
+          ks = (ks + ((col >> 3) << 3)) & 63;  // stage_tr's rotated image
         bf[kk2][fn] = *(const short8*)(bs + col * PBK + ks);
       }
     }
@@ -1797,10 +1875,16 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
     stage(1, 1, 1);  // B1(1)
   }
   // certificates for the pre-read + p0 reads: oldest units of each group
-  if (g == 0)
-    asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
-  else
-    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  if constexpr (TRMODE == 0) {
+    if (g == 0)
+      asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  } else if constexpr (TRMODE == 1) {
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  }
+  // transposed-stage scatters (and the pre-read below) are lgkm traffic
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
   // pre-read: G0 its p0 fragments (A0,B0 of tile 0); G1 reads at p0 proper
@@ -1855,7 +1939,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
       read_bf(0, t);
       if constexpr (!(PPMODE & 2)) { if (s2) stage(0, 0, t + 2); }
     }
-    PP_END(PP_W0(8));
+    PP_END((pp_site<TRMODE, 0, 8, -1>(g, tail)));
     // p1: G1 computes Q(1,0)t | G0 reads bf(B1,t) [af reuse], stages B1(t+1)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -1875,7 +1959,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
       read_bf(1, t);
       if constexpr (!(PPMODE & 2)) { if (s2) stage(1, 0, t + 2); }
     }
-    PP_END(PP_W1(8));
+    PP_END((pp_site<TRMODE, 1, 8, 4>(g, tail)));
     // p3: G1 computes Q(1,1)t | G0 reads af(A0,t+1)+bf(B0,t+1), stages A1(t+2)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -1886,7 +1970,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
       read_bf(0, t + 1);
       if constexpr (!(PPMODE & 2)) { if (s2) stage(1, 1, t + 2); }
     }
-    PP_END(PP_W0(12));
+    PP_END((pp_site<TRMODE, 0, 12, 4>(g, tail)));
     // p4: G0 computes Q(0,0)t+1 | G1 reads af(A1,t+1)+bf(B0,t+1), stages A0(t+2)
     if (g == 0) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -1897,7 +1981,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
       read_bf(0, t + 1);
       if constexpr (!(PPMODE & 2)) { if (s3) stage(0, 0, t + 3); }
     }
-    PP_END(PP_W0(8));
+    PP_END((pp_site<TRMODE, 0, 8, -1>(g, tail)));
     // p5: G1 computes Q(1,0)t+1 | G0 reads bf(B1,t+1), stages B1(t+2)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -1917,7 +2001,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
       read_bf(1, t + 1);
       if constexpr (!(PPMODE & 2)) { if (s3) stage(1, 0, t + 3); }
     }
-    PP_END(PP_W1(8));
+    PP_END((pp_site<TRMODE, 1, 8, 4>(g, tail)));
     // p7: G1 computes Q(1,1)t+1 | G0 reads af(A0,t+2)+bf(B0,t+2), stages A1(t+3)
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -1930,7 +2014,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
       }
       if constexpr (!(PPMODE & 2)) { if (s3) stage(1, 1, t + 3); }
     }
-    PP_END(PP_W0(12); PP_W1(8));
+    PP_END((pp_site<TRMODE, 0, 12, 4>(g, tail), pp_site<TRMODE, 1, 8, -1>(g, tail)));
   }
 #undef PP_END
 #undef PP_W0
@@ -3709,6 +3793,77 @@ void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
     args.bias = epi->bias ? (const uint16_t*)epi->bias->data() : nullptr;
     args.save_pre = epi->save_pre ? (uint16_t*)epi->save_pre->data() : nullptr;
     args.fuse_gelu = epi->gelu ? 1 : 0;
+  }
+  // r2: NATIVE NN/TN ping-pong — consume the stored layouts directly via
+  // transpose-in-stage (TRMODE), skipping the lowering transposes below
+  // entirely (they were 6-9% of both model steps: r2 kernel stats).
+  {
+    static const int kNtVariantN = [] {
+      const char* v = getenv("NEZHA_GEMM_VARIANT");
+      return v ? atoi(v) : 7;
+    }();
+    // EXPERIMENTAL opt-in (NEZHA_GEMM_NATIVE_TR=1): correct on every
+    // oracle (race screen NN/TN rows) but the transpose-in-stage register
+    // pressure (4 staged short8 + scatter address chains on a kernel
+    // already at 255 VGPR) spills ~500 B and fragments the MFMA clusters
+    // — 72-90 TF vs 970-1025 via the lowering (profiles/r2/r2_native2).
+    // Unblocking needs the BN=128 regeometry (halves acc VGPRs) that the
+    // persistent ping-pong also waits on — round-3 work.
+    static const bool kNativeOff = !(getenv("NEZHA_GEMM_NATIVE_TR") &&
+                                     atoi(getenv("NEZHA_GEMM_NATIVE_TR")) == 1);
+    int trmode = 0;
+    if (!transA && !transB) trmode = 1;       // NN: B is [K,N]
+    else if (transA && !transB) trmode = 3;   // TN: A is [K,M], B is [K,N]
+    if (trmode && kNtVariantN >= 7 && !kNativeOff && args.K % 128 == 0 &&
+        args.K >= 1536 && args.N % 8 == 0 &&
+        (trmode != 3 || args.M % 8 == 0)) {
+      int nwg_m = (int)((args.M + 255) / 256);
+      int nwg_n = (int)((args.N + 255) / 256);
+      int64_t nwg = (int64_t)nwg_m * nwg_n;
+      hipStream_t st = stream_of(C);
+      auto launch_native = [&](const GemmArgs& a, int64_t gy, int nn) {
+        dim3 grid((uint32_t)((a.M + 255) / 256 * ((a.N + 255) / 256)),
+                  (uint32_t)gy);
+        if (trmode == 1)
+          gemm_bf16_nt_pingpong_k<1, 1><<<grid, dim3(512), 0, st>>>(a, nn);
+        else
+          gemm_bf16_nt_pingpong_k<1, 3><<<grid, dim3(512), 0, st>>>(a, nn);
+        NZ_KERNEL_CHECK();
+      };
+      const_cast<GemmArgs&>(args).group_m = [] {
+        const char* v = getenv("NEZHA_GEMM_GROUP_M");
+        return v ? atoi(v) : 8;
+      }();
+      if (nwg >= 224 && (nwg % 256 == 0 || nwg >= 1024)) {
+        launch_native(args, 1, nwg_n);
+        return;
+      }
+      // small-grid deep-K (BERT wgrads): split-K slices on the native
+      // kernel, f32 partial slabs + fused reduce (same scheme as the NT
+      // split-K path, without the two giant operand transposes)
+      if (args.K >= 4096 && nwg >= 9) {
+        int sl = 1;
+        while (sl < 16 && nwg * sl * 2 <= 288 &&
+               args.K % ((int64_t)sl * 256) == 0 && args.K / (sl * 2) >= 768)
+          sl *= 2;
+        if (sl > 1 && nwg * sl >= 96) {
+          int64_t ksl = args.K / sl;
+          Tensor parts(C.ctx(), DType::F32, Shape{(int64_t)sl, args.M, args.N});
+          GemmArgs pa = args;
+          pa.K = ksl;
+          // per-slice k-offsets in ELEMENTS of each stored layout
+          pa.sA = (transA ? ksl * args.lda : ksl);
+          pa.sB = ksl * args.ldb;  // B is [K,N] in both native modes
+          pa.sC = args.M * args.N;
+          pa.out_f32 = (float*)parts.data();
+          launch_native(pa, sl, nwg_n);
+          splitk_reduce_k<<<dim3(grid_1d(args.M * args.N, 256)), dim3(256),
+                            0, st>>>((const float*)parts.data(), args, sl);
+          NZ_KERNEL_CHECK();
+          return;
+        }
+      }
+    }
   }
   Tensor at_scratch, bt_scratch;  // keep lowered operands alive to launch
   if (args.K % 64 == 0) {

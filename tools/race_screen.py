@@ -40,13 +40,22 @@ for (M, N, K, tA) in [(512, 512, 512, False), (2048, 1024, 768, False),
 # Cross-kernel oracle for shapes too big for numpy: the dispatch-selected
 # kernel vs the proven tier-A/128^2 route (NEZHA_GEMM_VARIANT=0). Both
 # accumulate f32 over the same K (different order), so rel err ~ bf16 eps.
-# Covers the persistent ping-pong (short-K) and ping-pong (deep-K) routes.
-for (M, N, K, tA) in [(32768, 1024, 768, False), (16384, 2304, 768, False),
-                      (8192, 4096, 4096, False)]:
+# Covers the persistent ping-pong (short-K), ping-pong (deep-K) and the
+# NATIVE NN/TN transpose-in-stage routes (r2) incl. native split-K TN.
+for (M, N, K, tA, tB) in [(32768, 1024, 768, False, True),
+                          (16384, 2304, 768, False, True),
+                          (8192, 4096, 4096, False, True),
+                          (8192, 4096, 4096, False, False),   # native NN
+                          (4096, 4096, 32768, True, False),   # native TN
+                          (768, 2304, 32768, True, False)]:   # native TN splitk
     A = ctx.tensor((K, M) if tA else (M, K), nz.BF16).rand_normal(11)
-    B = ctx.tensor((K, N) if tA else (N, K), nz.BF16).rand_normal(12)
+    B = ctx.tensor((N, K) if tB else (K, N), nz.BF16).rand_normal(12)
     C = ctx.tensor((M, N), nz.BF16)
-    nz.gemm(A, B, C, tA, not tA)
+    nz.gemm(A, B, C, tA, tB)
+    # determinism of the dispatched route
+    C2 = ctx.tensor((M, N), nz.BF16)
+    nz.gemm(A, B, C2, tA, tB)
+    assert np.array_equal(C.raw(), C2.raw()), f"NONDET {M}x{N}x{K} {tA}{tB}"
     got = C.numpy()
     import subprocess, sys as _sys
     # reference kernel in a subprocess (variant env is latched per process)
@@ -63,7 +72,7 @@ for (M, N, K, tA) in [(32768, 1024, 768, False), (16384, 2304, 768, False),
         "np.save(sys.argv[1], C.numpy())\n"
     ) % (os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
          K if tA else M, M if tA else K,
-         K if tA else N, N if tA else K, M, N, tA, not tA)
+         N if tB else K, K if tB else N, M, N, tA, tB)
     with tempfile.NamedTemporaryFile(suffix=".npy", delete=False) as f:
         refp = f.name
     subprocess.run([_sys.executable, "-c", code, refp], env=ref_env,
