@@ -1063,6 +1063,9 @@ __device__ __forceinline__ void vm_wait() {
   else if constexpr (N == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
   else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
   else if constexpr (N == 12) asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+  else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else if constexpr (N == 14) asm volatile("s_waitcnt vmcnt(14)" ::: "memory");
+  else if constexpr (N == 16) asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
 }
 // one wait site of the ping-pong schedule: group GRP waits the per-mode
 // count (NT_N for TRMODE 0, NN_N for TRMODE 1, nothing for TN); tails
@@ -2493,16 +2496,466 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_pers_k(
     cur = nxt;
     have_next = (tl + 2 * SLOTS) < (lo + cnt);
     nxt = have_next ? make_state(tl + 2 * SLOTS) : cur;
-    if (tl + SLOTS < lo + cnt && g == 0) {
-      // next tile's p0 fragments (slots staged by the boundary overflow;
-      // certified by the last steady end-p6 wait + the epilogue barriers)
-      read_af(0);
-      read_bf(0, 0);
+    if (tl + SLOTS < lo + cnt) {
+      if (g == 0) {
+        // next tile's p0 fragments (slots staged by the boundary overflow;
+        // certified by the last steady end-p6 wait + the epilogue barriers)
+        read_af(0);
+        read_bf(0, 0);
+      }
+      // drain + barrier: the new tile's p0 stage overwrites these slots
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      asm volatile("" ::: "memory");
     }
   }
 #undef PPP_END
 #undef PPP_W0
 #undef PPP_W1
+}
+
+// ---------------------------------------------------------------------------
+// pp128: BM=256 x BN=128 PERSISTENT ping-pong, template TRMODE (r2).
+//
+// The 256^2 ping-pong held 128 acc VGPRs per lane and sat exactly at the
+// 255-register cap: both the persistent sweep AND the transpose-in-stage
+// experiments spilled and fragmented the MFMA clusters. Halving BN halves
+// the accumulator (64 VGPR) and leaves room for everything:
+//   - persistent sweep state (short-K BERT panels — no per-tile ramp),
+//   - native NN/TN staging (no lowering transposes),
+//   - a 64 KiB spare for a fully countable LDS-staged epilogue
+//     (8 b128 stores per wave; boundary waits widen by +8).
+// Intensity drops to 87 FLOP/B (demand ~12 TB/s at 1.05 PF), so DEEP-K NT
+// stays on the 256^2 single-tile kernel; this one serves short-K NT and
+// all native-TR shapes (incl. split-K TN via the batch dim).
+//
+// Stage schedule: the deep pattern (iteration t stages all of t+2/t+3;
+// the last iteration's stages are the next tile's prologue). Units are
+// MIXED SIZE (A half 16 KiB = 4 glds/thread, B half 8 KiB = 2), so the
+// per-group vmcnt certificates were re-derived and simulator-verified:
+//   NT: end-p0 W0(6), p2 W1(6), p3 W0(8), p4 W0(6), p6 W1(6),
+//       p7 W0(8)+W1(6); prologue vmcnt(6) both groups
+//   NN (B transposed, sync): all four waits vmcnt(4); prologue vmcnt(4)
+//   TN (both transposed): no steady waits at all
+// fresh (first iteration after an epilogue): counts widen by the 8
+// epilogue stores per wave (6->14, 8->16, 4->12); FIFO completion order
+// keeps the wide form safe once the stores drain.
+// ---------------------------------------------------------------------------
+namespace pp128 {
+constexpr int QBM = 256, QBN = 128, QBK = 64;
+constexpr int AHALF_E = 128 * QBK;  // 8192 elems (16 KiB)
+constexpr int BHALF_E = 64 * QBK;   // 4096 elems (8 KiB)
+constexpr int QBUF_E = 2 * AHALF_E + 2 * BHALF_E;  // 24576 elems (48 KiB)
+__device__ __forceinline__ uint16_t* qslot(uint16_t* smem, int buf, int isB,
+                                           int half) {
+  return smem + buf * QBUF_E +
+         (isB ? 2 * AHALF_E + half * BHALF_E : half * AHALF_E);
+}
+// one wait site; mode-selected count, tails drain
+template <int TRMODE, int GRP, int NT_N, int NN_N>
+__device__ __forceinline__ void site(int g, bool tail, bool fresh) {
+  if (tail) {
+    ph8::vm_wait<0>();
+    return;
+  }
+  if (g != GRP) return;
+  if constexpr (TRMODE == 0) {
+    if (fresh) ph8::vm_wait<NT_N + 8>();
+    else ph8::vm_wait<NT_N>();
+  } else if constexpr (TRMODE == 1) {
+    if constexpr (NN_N >= 0) {
+      if (fresh) ph8::vm_wait<NN_N + 8>();
+      else ph8::vm_wait<NN_N>();
+    }
+  }
+}
+}  // namespace pp128
+
+template <int TRMODE>
+__launch_bounds__(512, 2) __global__ void gemm_bf16_pp128_pers_k(
+    GemmArgs args, int nwg_n, int64_t per_b, int64_t total) {
+  using namespace pp128;
+  using ph8::vm_wait;
+  // [2 pipeline buffers 96 KiB][spare 64 KiB] = the full 160 KiB
+  __shared__ __attribute__((aligned(16))) uint16_t smem[2 * QBUF_E + 32768];
+  uint16_t* spare = smem + 2 * QBUF_E;
+
+  const int orig = blockIdx.x;  // 256 WGs
+  const int xcd = orig & 7, slotq = orig >> 3;
+  const int64_t q = total / 8, r8 = total % 8;
+  const int64_t lo = xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q;
+  const int64_t cnt = xcd < r8 ? q + 1 : q;
+  if (slotq >= cnt) return;
+  constexpr int SLOTS = 32;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int g = wid >> 2;   // ping-pong group: G0 rows [0,128), G1 [128,256)
+  const int ws = wid & 3;   // 2x2 wave grid over a 128x64 quadrant
+  const int tig = (int)threadIdx.x & 255;
+  if (g == 1) __builtin_amdgcn_s_setprio(1);
+
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  const int64_t ntiles = args.K / QBK;  // even, >= 2 (launch guard)
+
+  struct TileSt {
+    int32_t bidx, m0, n0;
+  };
+  auto make_state = [&](int64_t tl) {
+    TileSt ts;
+    ts.bidx = __builtin_amdgcn_readfirstlane((int32_t)(tl / per_b));
+    const int64_t rtl = tl % per_b;
+    ts.m0 = __builtin_amdgcn_readfirstlane((int32_t)((rtl / nwg_n) * QBM));
+    ts.n0 = __builtin_amdgcn_readfirstlane((int32_t)((rtl % nwg_n) * QBN));
+    return ts;
+  };
+
+  // k-major transposed stage into the bank-rotated XOR image (see the
+  // 256^2 kernel's stage_tr comment); NELEM = half-tile elems, RW = r width
+  auto stage_tr = [&](const uint16_t* X, int64_t ldx, int64_t r0,
+                      int64_t Rlast8, uint16_t* dst, int64_t k0, int rw) {
+    const int chunks = rw == 128 ? 4 : 2;
+    for (int c = 0; c < chunks; ++c) {
+      int e = tig * 8 + c * 2048;
+      int kk = rw == 128 ? (e >> 7) : (e >> 6);
+      int rl = e & (rw - 1);
+      int64_t gr = r0 + rl;
+      if (gr > Rlast8) gr = Rlast8;
+      short8 vv = *(const short8*)(X + (k0 + kk) * ldx + gr);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        int r = rl + i;
+        int col = ((kk ^ ((r & 7) << 3)) + ((r >> 3) << 3)) & 63;
+        dst[r * QBK + col] = (uint16_t)vv[i];
+      }
+    }
+  };
+  auto stage = [&](const TileSt& ts, int isB, int h, int64_t kt) {
+    int64_t k0 = kt * QBK;
+    uint16_t* dst = qslot(smem, (int)(kt & 1), isB, h);
+    if (!isB) {
+      if constexpr ((TRMODE & 2) != 0) {
+        stage_tr(args.A + (int64_t)ts.bidx * args.sA, lda, ts.m0 + h * 128,
+                 args.M - 8, dst, k0, 128);
+        return;
+      }
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        int e = tig * 8 + c * 2048;
+        int row = e >> 6;
+        int k = (e & 63) ^ ((row & 7) << 3);
+        int64_t ar = ts.m0 + h * 128 + row;
+        if (ar > Mc) ar = Mc;
+        glds16(args.A + (int64_t)ts.bidx * args.sA + ar * lda + k0 + k,
+               dst + e);
+      }
+    } else {
+      if constexpr ((TRMODE & 1) != 0) {
+        stage_tr(args.B + (int64_t)ts.bidx * args.sB, ldb, ts.n0 + h * 64,
+                 args.N - 8, dst, k0, 64);
+        return;
+      }
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        int e = tig * 8 + c * 2048;
+        int row = e >> 6;
+        int k = (e & 63) ^ ((row & 7) << 3);
+        int64_t br = ts.n0 + h * 64 + row;
+        if (br > Nc) br = Nc;
+        glds16(args.B + (int64_t)ts.bidx * args.sB + br * ldb + k0 + k,
+               dst + e);
+      }
+    }
+  };
+
+  f32x4 acc[2][4][2] = {};    // [B-half][fm][fn]
+  short8 af[2][4], bf[2][2];  // fragment set: [kk][frag]
+
+  auto read_af = [&](int64_t kt) {
+    const uint16_t* as = qslot(smem, (int)(kt & 1), 0, g);
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2) {
+      const int kbase = kk2 * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        int row = (ws >> 1) * 64 + fm * 16 + (lane & 15);
+        int ks = kbase ^ ((row & 7) << 3);
+        if constexpr ((TRMODE & 2) != 0)
+          ks = (ks + ((row >> 3) << 3)) & 63;
+        af[kk2][fm] = *(const short8*)(as + row * QBK + ks);
+      }
+    }
+  };
+  auto read_bf = [&](int b, int64_t kt) {
+    const uint16_t* bs = qslot(smem, (int)(kt & 1), 1, b);
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2) {
+      const int kbase = kk2 * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = (ws & 1) * 32 + fn * 16 + (lane & 15);
+        int ks = kbase ^ ((col & 7) << 3);
+        if constexpr ((TRMODE & 1) != 0)
+          ks = (ks + ((col >> 3) << 3)) & 63;
+        bf[kk2][fn] = *(const short8*)(bs + col * QBK + ks);
+      }
+    }
+  };
+
+  auto mfma16 = [&](int b) {
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2)
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn)
+          acc[b][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[kk2][fm], bf[kk2][fn], acc[b][fm][fn], 0, 0, 0);
+  };
+
+  // countable epilogue: whole 256x128 C tile staged through the 64 KiB
+  // spare, streamed as 8 b128 stores per thread. Raw barriers + lgkm-only
+  // waits (never drain the in-flight next-tile pipeline).
+  auto epilogue = [&](const TileSt& ts) {
+    uint16_t* C = args.C + (int64_t)ts.bidx * args.sC;
+    float* out32 =
+        args.out_f32 ? args.out_f32 + (int64_t)ts.bidx * args.sC : nullptr;
+    uint16_t* save_pre =
+        args.save_pre ? args.save_pre + (int64_t)ts.bidx * args.sC : nullptr;
+    const int64_t N = args.N;
+    const bool interior =
+        !out32 && (ts.m0 + QBM <= args.M) && (ts.n0 + QBN <= args.N);
+    if (interior) {
+#pragma unroll
+      for (int b2 = 0; b2 < 2; ++b2)
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+            for (int rI = 0; rI < 4; ++rI) {
+              int lrow = g * 128 + (ws >> 1) * 64 + fm * 16 +
+                         (lane >> 4) * 4 + rI;
+              int lcol = b2 * 64 + (ws & 1) * 32 + fn * 16 + (lane & 15);
+              spare[lrow * 128 + lcol] = f2b(args.alpha * acc[b2][fm][fn][rI]);
+            }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      asm volatile("" ::: "memory");
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        int e = (c * 512 + (int)threadIdx.x) * 8;
+        int row = e >> 7, col = e & 127;
+        short8 vv = *(const short8*)(spare + e);
+        int64_t gcol = ts.n0 + col;
+        int64_t idx = (ts.m0 + row) * N + gcol;
+        short8 zout, yout;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          float v = b2f((uint16_t)vv[i]);
+          if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
+          if (args.bias) v += b2f(args.bias[gcol + i]);
+          zout[i] = (short)f2b(v);
+          if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+          yout[i] = (short)f2b(v);
+        }
+        if (save_pre) *(short8*)(save_pre + idx) = zout;
+        *(short8*)(C + idx) = yout;
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      asm volatile("" ::: "memory");
+    } else {
+#pragma unroll
+      for (int b2 = 0; b2 < 2; ++b2)
+#pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+            for (int rI = 0; rI < 4; ++rI) {
+              int64_t grow = ts.m0 + g * 128 + (ws >> 1) * 64 + fm * 16 +
+                             (lane >> 4) * 4 + rI;
+              int64_t gcol =
+                  ts.n0 + b2 * 64 + (ws & 1) * 32 + fn * 16 + (lane & 15);
+              if (grow < args.M && gcol < args.N) {
+                int64_t idx = grow * N + gcol;
+                if (out32) {
+                  out32[idx] = acc[b2][fm][fn][rI];
+                } else {
+                  float v = args.alpha * acc[b2][fm][fn][rI];
+                  if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+                  if (args.bias) v += b2f(args.bias[gcol]);
+                  if (save_pre) save_pre[idx] = f2b(v);
+                  if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+                  C[idx] = f2b(v);
+                }
+              }
+            }
+      // uncountable scalar stores: drain before the next tile's waits
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      asm volatile("" ::: "memory");
+    }
+#pragma unroll
+    for (int b2 = 0; b2 < 2; ++b2)
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn) acc[b2][fm][fn] = f32x4{};
+  };
+
+#define Q_END(wait_stmt)                                   \
+  do {                                                     \
+    wait_stmt;                                             \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");     \
+    __builtin_amdgcn_s_barrier();                          \
+    asm volatile("" ::: "memory");                         \
+  } while (0)
+
+  int64_t tl = lo + slotq;
+  TileSt cur = make_state(tl);
+  bool have_next = (tl + SLOTS) < (lo + cnt);
+  TileSt nxt = have_next ? make_state(tl + SLOTS) : cur;
+
+  // prologue: tiles 0 and 1 fully staged, 4 units per group (FIFO order)
+  if (g == 1) {
+    stage(cur, 0, 0, 0);  // A0(0)
+    stage(cur, 1, 0, 0);  // B0(0)
+    stage(cur, 0, 0, 1);  // A0(1)
+    stage(cur, 1, 0, 1);  // B0(1)
+  } else {
+    stage(cur, 0, 1, 0);  // A1(0)
+    stage(cur, 1, 1, 0);  // B1(0)
+    stage(cur, 0, 1, 1);  // A1(1)
+    stage(cur, 1, 1, 1);  // B1(1)
+  }
+  if constexpr (TRMODE == 0) vm_wait<6>();
+  else if constexpr (TRMODE == 1) vm_wait<4>();
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  if (g == 0) {
+    read_af(0);
+    read_bf(0, 0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+  for (; tl < lo + cnt; tl += SLOTS) {
+    const bool not_first_tile = (tl != lo + slotq);
+    for (int64_t t = 0; t < ntiles; t += 2) {
+      const bool last_it = (t + 2 >= ntiles);
+      const bool fresh = (t == 0) && not_first_tile;
+      const bool tail = last_it && !have_next;
+      const bool s2 = !last_it || have_next;
+      const bool s3 = (t + 3 < ntiles) || have_next;
+      auto st = [&](int isB, int h, int64_t kt) {
+        const bool ov = kt >= ntiles;
+        TileSt ts = ov ? nxt : cur;
+        stage(ts, isB, h, ov ? kt - ntiles : kt);
+      };
+      // p0: G0 Q(0,0)t | G1 reads af(A1,t)+bf(B0,t), stages A0(t+2)
+      if (g == 0) {
+        mfma16(0);
+      } else {
+        read_af(t);
+        read_bf(0, t);
+        if (s2) st(0, 0, t + 2);
+      }
+      Q_END((site<TRMODE, 0, 6, -1>(g, tail, fresh)));
+      // p1: G1 Q(1,0)t | G0 reads bf(B1,t), stages A1(t+2)
+      if (g == 1) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma16(0);
+      } else {
+        read_bf(1, t);
+        if (s2) st(0, 1, t + 2);
+      }
+      Q_END();
+      // p2: G0 Q(0,1)t | G1 reads bf(B1,t), stages B0(t+2)
+      if (g == 0) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma16(1);
+      } else {
+        read_bf(1, t);
+        if (s2) st(1, 0, t + 2);
+      }
+      Q_END((site<TRMODE, 1, 6, 4>(g, tail, fresh)));
+      // p3: G1 Q(1,1)t | G0 reads af(A0,t+1)+bf(B0,t+1), stages B1(t+2)
+      if (g == 1) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma16(1);
+      } else {
+        read_af(t + 1);
+        read_bf(0, t + 1);
+        if (s2) st(1, 1, t + 2);
+      }
+      Q_END((site<TRMODE, 0, 8, 4>(g, tail, fresh)));
+      // p4: G0 Q(0,0)t+1 | G1 reads af(A1,t+1)+bf(B0,t+1), stages A0(t+3)
+      if (g == 0) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma16(0);
+      } else {
+        read_af(t + 1);
+        read_bf(0, t + 1);
+        if (s3) st(0, 0, t + 3);
+      }
+      Q_END((site<TRMODE, 0, 6, -1>(g, tail, fresh)));
+      // p5: G1 Q(1,0)t+1 | G0 reads bf(B1,t+1), stages A1(t+3)
+      if (g == 1) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma16(0);
+      } else {
+        read_bf(1, t + 1);
+        if (s3) st(0, 1, t + 3);
+      }
+      Q_END();
+      // p6: G0 Q(0,1)t+1 | G1 reads bf(B1,t+1), stages B0(t+3)
+      if (g == 0) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma16(1);
+      } else {
+        read_bf(1, t + 1);
+        if (s3) st(1, 0, t + 3);
+      }
+      Q_END((site<TRMODE, 1, 6, 4>(g, tail, fresh)));
+      // p7: G1 Q(1,1)t+1 | G0 reads af+bf of t+2 (within-tile only; the
+      // cross-boundary read happens after the epilogue), stages B1(t+3)
+      if (g == 1) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        mfma16(1);
+      } else {
+        if (!last_it) {
+          read_af(t + 2);
+          read_bf(0, t + 2);
+        }
+        if (s3) st(1, 1, t + 3);
+      }
+      Q_END((site<TRMODE, 0, 8, 4>(g, tail, fresh),
+             site<TRMODE, 1, 6, -1>(g, tail, fresh)));
+    }
+    epilogue(cur);
+    cur = nxt;
+    have_next = (tl + 2 * SLOTS) < (lo + cnt);
+    nxt = have_next ? make_state(tl + 2 * SLOTS) : cur;
+    if (tl + SLOTS < lo + cnt) {
+      if (g == 0) {
+        read_af(0);
+        read_bf(0, 0);
+      }
+      // the pre-read must complete before the new tile's p0 stage
+      // overwrites the A0/B0 slots it is reading (race found by the
+      // screen: flaky NONDET on 32768x2304x768)
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      asm volatile("" ::: "memory");
+    }
+  }
+#undef Q_END
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 }
 
 // ---------------------------------------------------------------------------
@@ -3500,6 +3953,27 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
       return;
     }
   }
+  // short-K shapes (BERT K=768 panels): pp128 persistent ping-pong —
+  // BN=128 frees the registers the 256^2 variant lacked; pipeline flows
+  // across tiles (no per-tile ramp), countable LDS-staged epilogue.
+  // opt-in (NEZHA_GEMM_PP128=1): at 1 WG/CU its 16-MFMA phases leave the
+  // SIMD pipe idle between barriers and it measures 397-416 TF vs the r1
+  // persistent kernel's 537-542 on the BERT panels (profiles/r2/r2_pp128)
+  static const bool kPP128On = getenv("NEZHA_GEMM_PP128") &&
+                               atoi(getenv("NEZHA_GEMM_PP128")) == 1;
+  if (!transA && transB && kPP128On && kNtVariant >= 7 &&
+      args.K % 128 == 0 && args.K >= 256 && args.K < 1536) {
+    int nwg_m8 = (int)((args.M + 255) / 256);
+    int nwg_n8 = (int)((args.N + 127) / 128);
+    int64_t per_b8 = (int64_t)nwg_m8 * nwg_n8;
+    int64_t total8 = per_b8 * batch;
+    if (total8 >= 512) {
+      gemm_bf16_pp128_pers_k<0><<<dim3(256), dim3(512), 0, s>>>(
+          args, nwg_n8, per_b8, total8);
+      NZ_KERNEL_CHECK();
+      return;
+    }
+  }
   // short-K persistent ping-pong: EXPERIMENTAL opt-in (NEZHA_GEMM_PERS_PP=1).
   // The cross-tile flow + deep schedule are simulator-verified, but the
   // epilogue's register pressure fragments the MFMA clusters (303 scratch
@@ -3821,32 +4295,31 @@ void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
       int nwg_n = (int)((args.N + 255) / 256);
       int64_t nwg = (int64_t)nwg_m * nwg_n;
       hipStream_t st = stream_of(C);
-      auto launch_native = [&](const GemmArgs& a, int64_t gy, int nn) {
-        dim3 grid((uint32_t)((a.M + 255) / 256 * ((a.N + 255) / 256)),
-                  (uint32_t)gy);
+      int nwg_n128 = (int)((args.N + 127) / 128);
+      int64_t per_b128 = (int64_t)nwg_m * nwg_n128;
+      auto launch_native = [&](const GemmArgs& a, int64_t nslices) {
+        // pp128 persistent sweep; split-K slices ride the batch dimension
+        int64_t tot = per_b128 * nslices;
         if (trmode == 1)
-          gemm_bf16_nt_pingpong_k<1, 1><<<grid, dim3(512), 0, st>>>(a, nn);
+          gemm_bf16_pp128_pers_k<1><<<dim3(256), dim3(512), 0, st>>>(
+              a, nwg_n128, per_b128, tot);
         else
-          gemm_bf16_nt_pingpong_k<1, 3><<<grid, dim3(512), 0, st>>>(a, nn);
+          gemm_bf16_pp128_pers_k<3><<<dim3(256), dim3(512), 0, st>>>(
+              a, nwg_n128, per_b128, tot);
         NZ_KERNEL_CHECK();
       };
-      const_cast<GemmArgs&>(args).group_m = [] {
-        const char* v = getenv("NEZHA_GEMM_GROUP_M");
-        return v ? atoi(v) : 8;
-      }();
-      if (nwg >= 224 && (nwg % 256 == 0 || nwg >= 1024)) {
-        launch_native(args, 1, nwg_n);
+      if (per_b128 >= 256) {  // big grid: no K split needed
+        launch_native(args, 1);
         return;
       }
-      // small-grid deep-K (BERT wgrads): split-K slices on the native
-      // kernel, f32 partial slabs + fused reduce (same scheme as the NT
-      // split-K path, without the two giant operand transposes)
-      if (args.K >= 4096 && nwg >= 9) {
+      // small-grid deep-K (BERT wgrads): split-K slices, f32 partial
+      // slabs + fused reduce — no giant operand transposes
+      if (args.K >= 4096 && per_b128 >= 9) {
         int sl = 1;
-        while (sl < 16 && nwg * sl * 2 <= 288 &&
+        while (sl < 16 && per_b128 * sl * 2 <= 576 &&
                args.K % ((int64_t)sl * 256) == 0 && args.K / (sl * 2) >= 768)
           sl *= 2;
-        if (sl > 1 && nwg * sl >= 96) {
+        if (sl > 1 && per_b128 * sl >= 192) {
           int64_t ksl = args.K / sl;
           Tensor parts(C.ctx(), DType::F32, Shape{(int64_t)sl, args.M, args.N});
           GemmArgs pa = args;
@@ -3856,7 +4329,7 @@ void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
           pa.sB = ksl * args.ldb;  // B is [K,N] in both native modes
           pa.sC = args.M * args.N;
           pa.out_f32 = (float*)parts.data();
-          launch_native(pa, sl, nwg_n);
+          launch_native(pa, sl);
           splitk_reduce_k<<<dim3(grid_1d(args.M * args.N, 256)), dim3(256),
                             0, st>>>((const float*)parts.data(), args, sl);
           NZ_KERNEL_CHECK();
