@@ -1909,22 +1909,6 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
     __builtin_amdgcn_s_barrier();                          \
     asm volatile("" ::: "memory");                         \
   } while (0)
-#define PP_W0(n)                                              \
-  if (tail)                                                   \
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");          \
-  else if (g == 0)                                            \
-    asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
-#define PP_W1(n)                                              \
-  if (tail)                                                   \
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");          \
-  else if (g == 1)                                            \
-    asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
-#define PP_WB(n)                                              \
-  if (tail)                                                   \
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");          \
-  else                                                        \
-    asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
-
   const bool dg_mfma = args.diag_no_store != 2;   // 2: stage-only ablation
   const bool dg_stage = args.diag_no_store != 3;  // 3: compute-only ablation
   for (int64_t t = 0; t < ntiles; t += 2) {
@@ -2020,9 +2004,6 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
     PP_END((pp_site<TRMODE, 0, 12, 4>(g, tail), pp_site<TRMODE, 1, 8, -1>(g, tail)));
   }
 #undef PP_END
-#undef PP_W0
-#undef PP_W1
-#undef PP_WB
 
   // drain any straggling DMA before the epilogue reuses smem
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");

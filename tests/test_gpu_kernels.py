@@ -11,6 +11,19 @@ import nezha_amd as nz
 
 pytestmark = pytest.mark.gpu
 
+
+def test_gemm_race_screen_and_oracles():
+    """The full GEMM race screen (multi-run bitwise determinism on every
+    dispatch tier + numpy oracle + cross-kernel oracles, incl. the native
+    NN/TN routes) as a pytest so the driver's round-end GPU tier runs it."""
+    import subprocess, sys, os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run([sys.executable,
+                        os.path.join(repo, "tools", "race_screen.py")],
+                       cwd=repo, capture_output=True, timeout=600)
+    assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
+    assert b"RACE-SCREEN-OK" in r.stdout
+
 RNG = np.random.default_rng(42)
 
 
