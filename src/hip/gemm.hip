@@ -2940,6 +2940,321 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_pp128_pers_k(
 }
 
 // ---------------------------------------------------------------------------
+// pp64q: 128x128 ping-pong at 2 WORKGROUPS PER CU — the short-K kernel.
+//
+// Short-K diagnosis (profiles/r2): every 1-WG/CU structure plateaus near
+// 540 TF on the BERT K=768 panels while hipBLASLt reaches ~1014-1043 —
+// with one workgroup the per-tile prologue fill (~2 us of DMA against
+// ~9 us of tile work at K=768) and every wait bubble are exposed. At
+// 64 KiB LDS and a <=128-VGPR budget (launch_bounds(512,4)) TWO
+// independent 8-wave ping-pong workgroups co-reside per CU: 4 waves per
+// SIMD, and the de-synchronized pair fills each other's prologue/wait/
+// epilogue bubbles — the cross-WG hiding the 128^2 persistent kernel
+// relied on, combined with the ping-pong's pure 8-MFMA clusters.
+// acc[2][2][2] (32 VGPR) + 4+4 fragment b128s (32) leave margin.
+// Deep schedule, uniform 2-glds units; waits simulator-verified:
+//   p0 W0(4), p2 W1(4), p3 W0(6), p4 W0(4), p6 W1(4), p7 W0(6)+W1(4);
+// prologue vmcnt(4) both groups; last iteration drains.
+// ---------------------------------------------------------------------------
+namespace pp64q {
+constexpr int QM = 128, QN = 128, QK = 64;
+constexpr int HALF_E = 64 * QK;  // 4096 elems (8 KiB)
+constexpr int BUF_E = 4 * HALF_E;
+__device__ __forceinline__ uint16_t* qslot(uint16_t* smem, int buf, int isB,
+                                           int half) {
+  return smem + buf * BUF_E + (isB * 2 + half) * HALF_E;
+}
+template <int GRP, int NT_N>
+__device__ __forceinline__ void site(int g, bool tail) {
+  if (tail) {
+    ph8::vm_wait<0>();
+    return;
+  }
+  if (g == GRP) ph8::vm_wait<NT_N>();
+}
+}  // namespace pp64q
+
+__launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
+                                                            int nwg_n) {
+  using namespace pp64q;
+  __shared__ __attribute__((aligned(16))) uint16_t smem[2 * BUF_E];
+
+  int nwg = gridDim.x;
+  int orig = blockIdx.x;
+  int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+  int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+  int64_t tm, tn;
+  tile_coords(wgid, nwg / nwg_n, nwg_n, args.group_m, tm, tn);
+  const int64_t m0 = tm * QM;
+  const int64_t n0 = tn * QN;
+  const int64_t batch = blockIdx.y;
+  const uint16_t* __restrict__ A = args.A + batch * args.sA;
+  const uint16_t* __restrict__ B = args.B + batch * args.sB;
+  uint16_t* C = args.C + batch * args.sC;
+  float* out32 = args.out_f32 ? args.out_f32 + batch * args.sC : nullptr;
+  uint16_t* save_pre = args.save_pre;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int g = wid >> 2;   // group: A-half (rows g*64..+64)
+  const int ws = wid & 3;   // 2x2 wave grid over a 64x64 quadrant
+  const int tig = (int)threadIdx.x & 255;
+  if (g == 1) __builtin_amdgcn_s_setprio(1);
+
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  const int64_t ntiles = args.K / QK;  // even, >= 4 (launch guard)
+
+  auto stage = [&](int isB, int h, int64_t t) {
+    int64_t k0 = t * QK;
+    uint16_t* dst = qslot(smem, (int)(t & 1), isB, h);
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int e = tig * 8 + c * 2048;
+      int row = e >> 6;
+      int k = (e & 63) ^ ((row & 7) << 3);
+      if (!isB) {
+        int64_t ar = m0 + h * 64 + row;
+        if (ar > Mc) ar = Mc;
+        glds16(A + ar * lda + k0 + k, dst + e);
+      } else {
+        int64_t br = n0 + h * 64 + row;
+        if (br > Nc) br = Nc;
+        glds16(B + br * ldb + k0 + k, dst + e);
+      }
+    }
+  };
+
+  f32x4 acc[2][2][2] = {};    // [B-half][fm][fn]
+  short8 af[2][2], bf[2][2];  // [kk][frag]
+
+  auto read_af = [&](int64_t t) {
+    const uint16_t* as = qslot(smem, (int)(t & 1), 0, g);
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2) {
+      const int kbase = kk2 * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        int row = (ws >> 1) * 32 + fm * 16 + (lane & 15);
+        int ks = kbase ^ ((row & 7) << 3);
+        af[kk2][fm] = *(const short8*)(as + row * QK + ks);
+      }
+    }
+  };
+  auto read_bf = [&](int b, int64_t t) {
+    const uint16_t* bs = qslot(smem, (int)(t & 1), 1, b);
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2) {
+      const int kbase = kk2 * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = (ws & 1) * 32 + fn * 16 + (lane & 15);
+        int ks = kbase ^ ((col & 7) << 3);
+        bf[kk2][fn] = *(const short8*)(bs + col * QK + ks);
+      }
+    }
+  };
+
+  auto mfma8 = [&](int b) {
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2)
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn)
+          acc[b][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[kk2][fm], bf[kk2][fn], acc[b][fm][fn], 0, 0, 0);
+  };
+
+  // prologue: tiles 0 and 1, 4 units per group in FIFO order
+  if (g == 1) {
+    stage(0, 0, 0);  // A0(0)
+    stage(1, 0, 0);  // B0(0)
+    stage(0, 0, 1);  // A0(1)
+    stage(1, 0, 1);  // B0(1)
+  } else {
+    stage(0, 1, 0);  // A1(0)
+    stage(1, 1, 0);  // B1(0)
+    stage(0, 1, 1);  // A1(1)
+    stage(1, 1, 1);  // B1(1)
+  }
+  ph8::vm_wait<4>();
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  if (g == 0) {
+    read_af(0);
+    read_bf(0, 0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+#define Q4_END(wait_stmt)                                  \
+  do {                                                     \
+    wait_stmt;                                             \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");     \
+    __builtin_amdgcn_s_barrier();                          \
+    asm volatile("" ::: "memory");                         \
+  } while (0)
+
+  for (int64_t t = 0; t < ntiles; t += 2) {
+    const bool tail = (t + 2 >= ntiles);
+    const bool s2 = (t + 2 < ntiles), s3 = (t + 3 < ntiles);
+    // p0
+    if (g == 0) {
+      mfma8(0);
+    } else {
+      read_af(t);
+      read_bf(0, t);
+      if (s2) stage(0, 0, t + 2);
+    }
+    Q4_END((site<0, 4>(g, tail)));
+    // p1
+    if (g == 1) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      mfma8(0);
+    } else {
+      read_bf(1, t);
+      if (s2) stage(0, 1, t + 2);
+    }
+    Q4_END();
+    // p2
+    if (g == 0) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      mfma8(1);
+    } else {
+      read_bf(1, t);
+      if (s2) stage(1, 0, t + 2);
+    }
+    Q4_END((site<1, 4>(g, tail)));
+    // p3
+    if (g == 1) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      mfma8(1);
+    } else {
+      read_af(t + 1);
+      read_bf(0, t + 1);
+      if (s2) stage(1, 1, t + 2);
+    }
+    Q4_END((site<0, 6>(g, tail)));
+    // p4
+    if (g == 0) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      mfma8(0);
+    } else {
+      read_af(t + 1);
+      read_bf(0, t + 1);
+      if (s3) stage(0, 0, t + 3);
+    }
+    Q4_END((site<0, 4>(g, tail)));
+    // p5
+    if (g == 1) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      mfma8(0);
+    } else {
+      read_bf(1, t + 1);
+      if (s3) stage(0, 1, t + 3);
+    }
+    Q4_END();
+    // p6
+    if (g == 0) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      mfma8(1);
+    } else {
+      read_bf(1, t + 1);
+      if (s3) stage(1, 0, t + 3);
+    }
+    Q4_END((site<1, 4>(g, tail)));
+    // p7
+    if (g == 1) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      mfma8(1);
+    } else {
+      if (s2) {
+        read_af(t + 2);
+        read_bf(0, t + 2);
+      }
+      if (s3) stage(1, 1, t + 3);
+    }
+    Q4_END((site<0, 6>(g, tail), site<1, 4>(g, tail)));
+  }
+#undef Q4_END
+
+  // drain, then the LDS-staged epilogue reuses smem (32 KiB C tile)
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  if (out32) {
+#pragma unroll
+    for (int b2 = 0; b2 < 2; ++b2)
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+          for (int rI = 0; rI < 4; ++rI) {
+            int64_t grow = m0 + g * 64 + (ws >> 1) * 32 + fm * 16 +
+                           (lane >> 4) * 4 + rI;
+            int64_t gcol =
+                n0 + b2 * 64 + (ws & 1) * 32 + fn * 16 + (lane & 15);
+            if (grow < args.M && gcol < args.N)
+              out32[grow * args.N + gcol] = acc[b2][fm][fn][rI];
+          }
+    return;
+  }
+#pragma unroll
+  for (int b2 = 0; b2 < 2; ++b2)
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+        for (int rI = 0; rI < 4; ++rI) {
+          int lrow = g * 64 + (ws >> 1) * 32 + fm * 16 + (lane >> 4) * 4 + rI;
+          int lcol = b2 * 64 + (ws & 1) * 32 + fn * 16 + (lane & 15);
+          smem[lrow * 128 + lcol] = f2b(args.alpha * acc[b2][fm][fn][rI]);
+        }
+  __syncthreads();
+  const bool interior = (m0 + QM <= args.M) && (n0 + QN <= args.N);
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    int e = (c * 512 + (int)threadIdx.x) * 8;
+    int row = e >> 7, col = e & 127;
+    int64_t grow = m0 + row;
+    int64_t gcol = n0 + col;
+    short8 vv = *(const short8*)(smem + e);
+    if (interior) {
+      int64_t idx = grow * args.N + gcol;
+      short8 zout, yout;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        zout[i] = (short)f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        yout[i] = (short)f2b(v);
+      }
+      if (save_pre) *(short8*)(save_pre + idx) = zout;
+      *(short8*)(C + idx) = yout;
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        if (grow >= args.M || gcol + i >= args.N) continue;
+        int64_t idx = grow * args.N + gcol + i;
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        if (save_pre) save_pre[idx] = f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        C[idx] = f2b(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // PERSISTENT 8-phase kernel: 256 WGs (1/CU), each sweeping a strided run of
 // (batch, m, n) 256^2 tiles with the SAME 8-phase wait schedule as
 // gemm_bf16_nt_8phase_k — and the pipeline flowing ACROSS tile boundaries.
@@ -3937,6 +4252,22 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
   // short-K shapes (BERT K=768 panels): pp128 persistent ping-pong —
   // BN=128 frees the registers the 256^2 variant lacked; pipeline flows
   // across tiles (no per-tile ramp), countable LDS-staged epilogue.
+  // short-K NT (BERT K=768 panels): 128^2 ping-pong at 2 WGs/CU —
+  // the de-synchronized pair hides prologue/wait/epilogue bubbles
+  static const bool k64qOff = getenv("NEZHA_GEMM_64Q") &&
+                              atoi(getenv("NEZHA_GEMM_64Q")) == 0;
+  if (!transA && transB && batch == 1 && !k64qOff && kNtVariant >= 7 &&
+      args.K % 128 == 0 && args.K >= 256 && args.K < 1536) {
+    int nwg_m4 = (int)((args.M + 127) / 128);
+    int nwg_n4 = (int)((args.N + 127) / 128);
+    int64_t nwg4 = (int64_t)nwg_m4 * nwg_n4;
+    if (nwg4 >= 1024) {  // >= 2 rounds of the 512-WG chip
+      dim3 grid((uint32_t)nwg4);
+      gemm_bf16_pp64q_k<<<grid, dim3(512), 0, s>>>(args, nwg_n4);
+      NZ_KERNEL_CHECK();
+      return;
+    }
+  }
   // opt-in (NEZHA_GEMM_PP128=1): at 1 WG/CU its 16-MFMA phases leave the
   // SIMD pipe idle between barriers and it measures 397-416 TF vs the r1
   // persistent kernel's 537-542 on the BERT panels (profiles/r2/r2_pp128)
