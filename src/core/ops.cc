@@ -151,6 +151,14 @@ void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
                S * S, S * dh, S * dh);
 }
 
+static bool attn512_fused_off() {
+  static const bool off = [] {
+    const char* v = getenv("NEZHA_ATTN512_FUSED");
+    return v && atoi(v) == 0;
+  }();
+  return off;
+}
+
 void attn_fwd_packed(const Tensor& qkv, Tensor& probs, Tensor& out, int64_t B,
                      int64_t S, int64_t heads, int64_t dh, float scale) {
   prof::OpScope nz_scope_("attn_fwd_packed", qkv.ctx());
@@ -158,9 +166,10 @@ void attn_fwd_packed(const Tensor& qkv, Tensor& probs, Tensor& out, int64_t B,
     hip::attn_fused_fwd_packed(qkv, probs, out, B, S, heads, dh, scale);
     return;
   }
-  if (qkv.device() == Device::HIP && hip::attn_fused_fwd512_supported(S, dh)) {
-    // S=512 tiled fused forward; backward stays on the composed path
-    // (it consumes the normalized probs this kernel writes)
+  if (qkv.device() == Device::HIP &&
+      hip::attn_fused_fwd512_supported(S, dh) && !attn512_fused_off()) {
+    // S=512 tiled fused forward (NEZHA_ATTN512_FUSED=0 forces the
+    // composed fallback — also the test oracle's reference path)
     hip::attn_fused_fwd512_packed(qkv, probs, out, B, S, heads, dh, scale);
     return;
   }
@@ -180,6 +189,12 @@ void attn_bwd_packed(const Tensor& qkv, const Tensor& probs,
   if (qkv.device() == Device::HIP && hip::attn_fused_fwd_supported(S, dh)) {
     hip::attn_fused_bwd_packed(qkv, probs, dmerged, dqkv, B, S, heads, dh,
                                scale);
+    return;
+  }
+  if (qkv.device() == Device::HIP &&
+      hip::attn_fused_fwd512_supported(S, dh) && !attn512_fused_off()) {
+    hip::attn_fused_bwd512_packed(qkv, probs, dmerged, dqkv, B, S, heads, dh,
+                                  scale);
     return;
   }
   Tensor q(qkv.ctx(), qkv.dtype(), Shape{B, heads, S, dh});

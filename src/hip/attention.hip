@@ -715,6 +715,384 @@ __launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused attention BACKWARD for S=512, dh=64 — two kernels (the S x S score
+// matrix cannot live in one WG, and dQ accumulates over k-tiles while
+// dK/dV accumulate over q-tiles, so each gets the decomposition that makes
+// its accumulator workgroup-local):
+//   attn_bwd512_q_k   one WG per (b,h,64-row q-block):
+//       dP strip = dO . V_t^T streamed over k-tiles into LDS;
+//       rs[s1] = rowsum(dP o P) (P staged once, 64 KiB);
+//       dS strip = scale * P o (dP - rs)  -> written to the rs scratch's
+//       companion use + dQ = sum_t dS_t . K_t; rs goes to a global
+//       scratch for the kv kernel.
+//   attn_bwd512_kv_k  one WG per (b,h,64-row k-block), 2 WGs/CU:
+//       loops q-tiles; dP^T = V . dO^T (pure NT, both operands staged
+//       k-contiguous over d); dS^T = scale * P^T o (dP^T - rs[s1]);
+//       dV += Pt . dOt-image, dK += dSt . Qt-image. dP/dS are recomputed
+//       here (~40% extra attention-bwd FLOPs) to keep both kernels free
+//       of cross-WG accumulation — attention is ~3% of the S=512 step,
+//       the win is fusing away the composed path's HBM round trips.
+// ---------------------------------------------------------------------------
+__launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
+    const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
+    const uint16_t* __restrict__ v, const uint16_t* __restrict__ probs,
+    const uint16_t* __restrict__ dout, uint16_t* __restrict__ dq,
+    float* __restrict__ rs_g, float scale, int heads, int64_t qb, int64_t qh,
+    int64_t qr, int64_t db, int64_t dhh, int64_t dr, int64_t gb, int64_t gh,
+    int64_t gr) {
+  using namespace s512;
+  // [dO 8K][KV dbuf 2x9.2K][dS strip 66.6K][P 64K] ~= 157 KiB, 1 WG/CU
+  __shared__ __attribute__((aligned(16))) uint16_t smem[
+      SQ * AD + 2 * AD * VST + SQ * PST + SQ * S5];
+  __shared__ float rsl[SQ], rpart[SQ][4];
+  uint16_t* dOs = smem;
+  uint16_t* KVs = smem + SQ * AD;            // dbuf: + (kt&1)*AD*VST
+  uint16_t* dSs = KVs + 2 * AD * VST;        // [64][520] strip (dP then dS)
+  uint16_t* Ps = dSs + SQ * PST;             // [64][512] linear (glds image)
+
+  const int64_t nqb = S5 / SQ;
+  const int64_t bhq = blockIdx.x;
+  const int64_t bh = bhq / nqb, q0 = (bhq % nqb) * SQ;
+  const int64_t b_ = bh / heads, h_ = bh % heads;
+  const uint16_t* qg = q + b_ * qb + h_ * qh + q0 * qr;
+  const uint16_t* kg = k + b_ * qb + h_ * qh;
+  const uint16_t* vg = v + b_ * qb + h_ * qh;
+  const uint16_t* pg = probs + bh * (int64_t)S5 * S5 + q0 * S5;
+  const uint16_t* dog = dout + b_ * db + h_ * dhh + q0 * dr;
+  uint16_t* dqg = dq + b_ * gb + h_ * gh + q0 * gr;
+  float* rsg = rs_g + bh * S5 + q0;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;  // 4 waves; wave w owns rows w*16..+16
+  const int e0 = (int)threadIdx.x * 8;
+
+  // stage dO (glds, swizzled) + the full P block (glds, linear [64][512])
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    int e = e0 + c * 2048;
+    int row = e >> 6;
+    int kk2 = (e & 63) ^ ((row & 7) << 3);
+    att_glds16(dog + row * dr + kk2, dOs + e);
+  }
+#pragma unroll
+  for (int c = 0; c < 16; ++c) {
+    int e = e0 + c * 2048;  // 32768 elems = [64][512] linear
+    att_glds16(pg + (e >> 9) * S5 + (e & 511), Ps + e);
+  }
+  // V tile 0 scattered transposed? NO — dP = dO . V^T is NT over d:
+  // V [s2][d] is already k-contiguous; stage swizzled like fwd K tiles.
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    int e = e0 + c * 2048;
+    int row = e >> 6;
+    int kk2 = (e & 63) ^ ((row & 7) << 3);
+    att_glds16(vg + row * qr + kk2, KVs + e);  // V tile 0 -> buf 0 (swz, 64x64)
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+  // ---- dP strip: loop k-tiles ---------------------------------------------
+  for (int kt = 0; kt < S5 / KT; ++kt) {
+    if (kt + 1 < S5 / KT) {
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        int e = e0 + c * 2048;
+        int row = e >> 6;
+        int kk2 = (e & 63) ^ ((row & 7) << 3);
+        att_glds16(vg + (int64_t)(kt + 1) * KT * qr + row * qr + kk2,
+                   KVs + ((kt + 1) & 1) * AD * VST + e);
+      }
+    }
+    const uint16_t* vb = KVs + (kt & 1) * AD * VST;
+    f32x4 acc[4] = {};
+#pragma unroll
+    for (int kk = 0; kk < AD; kk += 32) {
+      short8 a_frag, b_frag[4];
+      const int kbase = kk + (lane >> 4) * 8;
+      {
+        int row = wid * 16 + (lane & 15);
+        int ks = kbase ^ ((row & 7) << 3);
+        a_frag = *(const short8*)(dOs + row * AD + ks);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = fn * 16 + (lane & 15);
+        int ks = kbase ^ ((col & 7) << 3);
+        b_frag[fn] = *(const short8*)(vb + col * AD + ks);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        acc[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
+                                                          acc[fn], 0, 0, 0);
+    }
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+      for (int rI = 0; rI < 4; ++rI) {
+        int row = wid * 16 + (lane >> 4) * 4 + rI;
+        int col = kt * KT + fn * 16 + (lane & 15);
+        dSs[row * PST + col] = f2b(acc[fn][rI]);
+      }
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+  }
+
+  // ---- rs = rowsum(dP o P); then dS = scale * P o (dP - rs) ---------------
+  {
+    const int row = (int)threadIdx.x >> 2;
+    const int qtr = (int)threadIdx.x & 3;
+    const uint16_t* dpr = dSs + row * PST + qtr * 128;
+    const uint16_t* ppr = Ps + row * S5 + qtr * 128;
+    float accr = 0.f;
+#pragma unroll 4
+    for (int i = 0; i < 128; ++i) accr += b2f(dpr[i]) * b2f(ppr[i]);
+    rpart[row][qtr] = accr;
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  if (threadIdx.x < SQ) {
+    float r = rpart[threadIdx.x][0] + rpart[threadIdx.x][1] +
+              rpart[threadIdx.x][2] + rpart[threadIdx.x][3];
+    rsl[threadIdx.x] = r;
+    rsg[threadIdx.x] = r;  // scratch for the kv kernel
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  {
+    const int row = (int)threadIdx.x >> 2;
+    const int qtr = (int)threadIdx.x & 3;
+    uint16_t* dpr = dSs + row * PST + qtr * 128;
+    const uint16_t* ppr = Ps + row * S5 + qtr * 128;
+    const float r = rsl[row];
+#pragma unroll 4
+    for (int i = 0; i < 128; ++i)
+      dpr[i] = f2b(scale * b2f(ppr[i]) * (b2f(dpr[i]) - r));
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+  // ---- dQ = sum_t dS_t . K_t (K tiles scattered transposed [d][s2]) -------
+  f32x4 accQ[4] = {};
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    int e = e0 + c * 2048;
+    int s_ = e >> 6, d = e & 63;
+    short8 kv2 = *(const short8*)(kg + s_ * qr + d);
+    uint16_t* ktile = KVs;  // buf 0
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      ktile[(d + i) * VST + s_] = (uint16_t)kv2[i];
+  }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  for (int kt = 0; kt < S5 / KT; ++kt) {
+    if (kt + 1 < S5 / KT) {
+      uint16_t* kdst = KVs + ((kt + 1) & 1) * AD * VST;
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        int e = e0 + c * 2048;
+        int s_ = e >> 6, d = e & 63;
+        short8 kv2 =
+            *(const short8*)(kg + (int64_t)(kt + 1) * KT * qr + s_ * qr + d);
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          kdst[(d + i) * VST + s_] = (uint16_t)kv2[i];
+      }
+    }
+    const uint16_t* ktile = KVs + (kt & 1) * AD * VST;
+#pragma unroll
+    for (int kk = 0; kk < KT; kk += 32) {
+      short8 a_frag, b_frag[4];
+      const int kbase = kk + (lane >> 4) * 8;
+      {
+        int row = wid * 16 + (lane & 15);
+        a_frag = *(const short8*)(dSs + row * PST + kt * KT + kbase);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = fn * 16 + (lane & 15);
+        b_frag[fn] = *(const short8*)(ktile + col * VST + kbase);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        accQ[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
+                                                           accQ[fn], 0, 0, 0);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+  }
+#pragma unroll
+  for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+    for (int rI = 0; rI < 4; ++rI) {
+      int row = wid * 16 + (lane >> 4) * 4 + rI;
+      int col = fn * 16 + (lane & 15);
+      dqg[row * gr + col] = f2b(accQ[fn][rI]);
+    }
+}
+
+__launch_bounds__(256, 2) __global__ void attn_bwd512_kv_k(
+    const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
+    const uint16_t* __restrict__ v, const uint16_t* __restrict__ probs,
+    const uint16_t* __restrict__ dout, uint16_t* __restrict__ dk,
+    uint16_t* __restrict__ dv, const float* __restrict__ rs_g, float scale,
+    int heads, int64_t qb, int64_t qh, int64_t qr, int64_t db, int64_t dhh,
+    int64_t dr, int64_t gb, int64_t gh, int64_t gr) {
+  using namespace s512;
+  // [V 8K][dO 8K][Pt 9.2K][dOt 9.2K][Qt 9.2K][dSt 9.2K] ~= 53 KiB -> 2/CU
+  __shared__ __attribute__((aligned(16))) uint16_t smem[
+      2 * KT * AD + 4 * AD * VST];
+  __shared__ float rsl[SQ];
+  uint16_t* Vs = smem;                       // [s2][d] swz
+  uint16_t* dOsw = smem + KT * AD;           // [s1][d] swz (per q-tile)
+  uint16_t* Pt = dOsw + KT * AD;             // [s2][72] (Pt[s2][s1])
+  uint16_t* dOt = Pt + AD * VST;             // [d][72]
+  uint16_t* Qt = dOt + AD * VST;             // [d][72]
+  uint16_t* dSt = Qt + AD * VST;             // [s2][72]
+
+  const int64_t nkb = S5 / KT;
+  const int64_t bhk = blockIdx.x;
+  const int64_t bh = bhk / nkb, k0r = (bhk % nkb) * KT;
+  const int64_t b_ = bh / heads, h_ = bh % heads;
+  const uint16_t* qg = q + b_ * qb + h_ * qh;
+  const uint16_t* kg = k + b_ * qb + h_ * qh + k0r * qr;
+  const uint16_t* vg = v + b_ * qb + h_ * qh + k0r * qr;
+  const uint16_t* pg = probs + bh * (int64_t)S5 * S5 + k0r;  // column block
+  const uint16_t* dog = dout + b_ * db + h_ * dhh;
+  uint16_t* dkg = dk + b_ * gb + h_ * gh + k0r * gr;
+  uint16_t* dvg = dv + b_ * gb + h_ * gh + k0r * gr;
+  const float* rsg = rs_g + bh * S5;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;  // wave w owns s2 rows w*16..+16
+  const int e0 = (int)threadIdx.x * 8;
+
+  // V block (fixed for this WG), swizzled
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    int e = e0 + c * 2048;
+    int row = e >> 6;
+    int kk2 = (e & 63) ^ ((row & 7) << 3);
+    att_glds16(vg + row * qr + kk2, Vs + e);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+  f32x4 accV[4] = {}, accK[4] = {};
+  for (int qt = 0; qt < S5 / SQ; ++qt) {
+    const int64_t q0 = (int64_t)qt * SQ;
+    // stage dO_qt (glds swz); P_qt + Q_qt via registers -> transposed images
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int e = e0 + c * 2048;
+      int row = e >> 6;
+      int kk2 = (e & 63) ^ ((row & 7) << 3);
+      att_glds16(dog + (q0 + row) * dr + kk2, dOsw + e);
+    }
+    if (threadIdx.x < SQ) rsl[threadIdx.x] = rsg[q0 + threadIdx.x];
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int e = e0 + c * 2048;
+      int s1 = e >> 6, d = e & 63;  // [s1][64] block coords
+      short8 pv = *(const short8*)(pg + (q0 + s1) * S5 + d);  // P[s1][k0r+d]
+      short8 qv = *(const short8*)(qg + (q0 + s1) * qr + d);
+      short8 dov = *(const short8*)(dog + (q0 + s1) * dr + d);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        Pt[(d + i) * VST + s1] = (uint16_t)pv[i];   // Pt[s2][s1]
+        Qt[(d + i) * VST + s1] = (uint16_t)qv[i];   // Qt[d][s1]
+        dOt[(d + i) * VST + s1] = (uint16_t)dov[i]; // dOt[d][s1]
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+
+    // dP^T[s2][s1] = V . dO^T (NT over d)
+    f32x4 accP[4] = {};
+#pragma unroll
+    for (int kk = 0; kk < AD; kk += 32) {
+      short8 a_frag, b_frag[4];
+      const int kbase = kk + (lane >> 4) * 8;
+      {
+        int row = wid * 16 + (lane & 15);  // s2
+        int ks = kbase ^ ((row & 7) << 3);
+        a_frag = *(const short8*)(Vs + row * AD + ks);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = fn * 16 + (lane & 15);  // s1
+        int ks = kbase ^ ((col & 7) << 3);
+        b_frag[fn] = *(const short8*)(dOsw + col * AD + ks);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn)
+        accP[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
+                                                           accP[fn], 0, 0, 0);
+    }
+    // dS^T = scale * P^T o (dP^T - rs[s1]) -> dSt image
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+      for (int rI = 0; rI < 4; ++rI) {
+        int s2 = wid * 16 + (lane >> 4) * 4 + rI;
+        int s1 = fn * 16 + (lane & 15);
+        float pb = b2f(Pt[s2 * VST + s1]);
+        dSt[s2 * VST + s1] = f2b(scale * pb * (accP[fn][rI] - rsl[s1]));
+      }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+
+    // dV += Pt . dOt-image; dK += dSt . Qt-image (both: k = s1)
+#pragma unroll
+    for (int kk = 0; kk < SQ; kk += 32) {
+      short8 a_frag, b_frag, a2_frag, b2_frag[0 + 4];
+      const int kbase = kk + (lane >> 4) * 8;
+      {
+        int row = wid * 16 + (lane & 15);  // s2
+        a_frag = *(const short8*)(Pt + row * VST + kbase);
+        a2_frag = *(const short8*)(dSt + row * VST + kbase);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = fn * 16 + (lane & 15);  // d
+        b2_frag[fn] = *(const short8*)(Qt + col * VST + kbase);
+      }
+#pragma unroll
+      for (int fn = 0; fn < 4; ++fn) {
+        int col = fn * 16 + (lane & 15);
+        b_frag = *(const short8*)(dOt + col * VST + kbase);
+        accV[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
+                                                           accV[fn], 0, 0, 0);
+        accK[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a2_frag, b2_frag[fn], accK[fn], 0, 0, 0);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+  }
+#pragma unroll
+  for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+    for (int rI = 0; rI < 4; ++rI) {
+      int row = wid * 16 + (lane >> 4) * 4 + rI;  // s2
+      int col = fn * 16 + (lane & 15);            // d
+      dvg[row * gr + col] = f2b(accV[fn][rI]);
+      dkg[row * gr + col] = f2b(accK[fn][rI]);
+    }
+}
+
 }  // namespace
 
 bool attn_fused_fwd_supported(int64_t S, int64_t dh) {
@@ -768,6 +1146,32 @@ void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
       (const uint16_t*)dout.data(), (uint16_t*)dq.data(),
       (uint16_t*)dk.data(), (uint16_t*)dv.data(), scale, 1, sh, 0, AD, sh, 0,
       AD, sh, 0, AD);
+  NZ_KERNEL_CHECK();
+}
+
+// S=512 backward launcher (packed layout): q-block kernel (dQ + rowsum
+// scratch) then kv-block kernel (dK, dV) on the same stream.
+void attn_fused_bwd512_packed(const Tensor& qkv, const Tensor& probs,
+                              const Tensor& dmerged, Tensor& dqkv, int64_t B,
+                              int64_t S, int64_t heads, int64_t dh,
+                              float scale) {
+  NZ_CHECK(qkv.dtype() == DType::BF16, "fused attn bwd 512: bf16 required");
+  NZ_CHECK(S == 512 && dh == AD, "fused attn bwd 512: S=512, dh=64 only");
+  hipStream_t s = stream_of(qkv);
+  int64_t H = heads * dh;
+  const uint16_t* base = (const uint16_t*)qkv.data();
+  uint16_t* gbase = (uint16_t*)dqkv.data();
+  Tensor rs(qkv.ctx(), DType::F32, Shape{B * heads * S});
+  attn_bwd512_q_k<<<dim3((uint32_t)(B * heads * 8)), dim3(256), 0, s>>>(
+      base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
+      (const uint16_t*)dmerged.data(), gbase, (float*)rs.data(), scale,
+      (int)heads, S * 3 * H, dh, 3 * H, S * H, dh, H, S * 3 * H, dh, 3 * H);
+  NZ_KERNEL_CHECK();
+  attn_bwd512_kv_k<<<dim3((uint32_t)(B * heads * 8)), dim3(256), 0, s>>>(
+      base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
+      (const uint16_t*)dmerged.data(), gbase + H, gbase + 2 * H,
+      (const float*)rs.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
+      S * H, dh, H, S * 3 * H, dh, 3 * H);
   NZ_KERNEL_CHECK();
 }
 

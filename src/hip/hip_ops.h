@@ -69,6 +69,10 @@ void attn_fused_fwd512_packed(const Tensor& qkv, Tensor& probs, Tensor& out,
 void attn_fused_fwd512(const Tensor& q, const Tensor& k, const Tensor& v,
                        Tensor& probs, Tensor& ctx_out, int64_t batch_heads,
                        float scale);
+void attn_fused_bwd512_packed(const Tensor& qkv, const Tensor& probs,
+                              const Tensor& dmerged, Tensor& dqkv, int64_t B,
+                              int64_t S, int64_t heads, int64_t dh,
+                              float scale);
 void attn_fused_fwd(const Tensor& q, const Tensor& k, const Tensor& v,
                     Tensor& probs, Tensor& ctx_out, int64_t batch_heads,
                     float scale);
