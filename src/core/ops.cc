@@ -187,6 +187,17 @@ void attn_bwd_packed(const Tensor& qkv, const Tensor& probs,
                      int64_t heads, int64_t dh, float scale) {
   prof::OpScope nz_scope_("attn_bwd_packed", qkv.ctx());
   if (qkv.device() == Device::HIP && hip::attn_fused_fwd_supported(S, dh)) {
+    // S=128: monolithic single-kernel bwd by default; NEZHA_ATTN128_BWD=pair
+    // routes through the two-kernel 2-WG/CU pair (A/B hook)
+    static const bool kPair128 = [] {
+      const char* v = getenv("NEZHA_ATTN128_BWD");
+      return v && std::string(v) == "pair";
+    }();
+    if (kPair128) {
+      hip::attn_fused_bwd512_packed(qkv, probs, dmerged, dqkv, B, S, heads,
+                                    dh, scale);
+      return;
+    }
     hip::attn_fused_bwd_packed(qkv, probs, dmerged, dqkv, B, S, heads, dh,
                                scale);
     return;

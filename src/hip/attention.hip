@@ -734,6 +734,7 @@ __launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
 //       of cross-WG accumulation — attention is ~3% of the S=512 step,
 //       the win is fusing away the composed path's HBM round trips.
 // ---------------------------------------------------------------------------
+template <int SEQ>
 __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const uint16_t* __restrict__ probs,
@@ -742,26 +743,27 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
     int64_t qr, int64_t db, int64_t dhh, int64_t dr, int64_t gb, int64_t gh,
     int64_t gr) {
   using namespace s512;
+  constexpr int QPST = SEQ + 8;
   // [dO 8K][KV dbuf 2x9.2K][dS strip 66.6K][P 64K] ~= 157 KiB, 1 WG/CU
   __shared__ __attribute__((aligned(16))) uint16_t smem[
-      SQ * AD + 2 * AD * VST + SQ * PST + SQ * S5];
+      SQ * AD + 2 * AD * VST + SQ * QPST + SQ * SEQ];
   __shared__ float rsl[SQ], rpart[SQ][4];
   uint16_t* dOs = smem;
   uint16_t* KVs = smem + SQ * AD;            // dbuf: + (kt&1)*AD*VST
   uint16_t* dSs = KVs + 2 * AD * VST;        // [64][520] strip (dP then dS)
-  uint16_t* Ps = dSs + SQ * PST;             // [64][512] linear (glds image)
+  uint16_t* Ps = dSs + SQ * QPST;             // [64][512] linear (glds image)
 
-  const int64_t nqb = S5 / SQ;
+  const int64_t nqb = SEQ / SQ;
   const int64_t bhq = blockIdx.x;
   const int64_t bh = bhq / nqb, q0 = (bhq % nqb) * SQ;
   const int64_t b_ = bh / heads, h_ = bh % heads;
   const uint16_t* qg = q + b_ * qb + h_ * qh + q0 * qr;
   const uint16_t* kg = k + b_ * qb + h_ * qh;
   const uint16_t* vg = v + b_ * qb + h_ * qh;
-  const uint16_t* pg = probs + bh * (int64_t)S5 * S5 + q0 * S5;
+  const uint16_t* pg = probs + bh * (int64_t)SEQ * SEQ + q0 * SEQ;
   const uint16_t* dog = dout + b_ * db + h_ * dhh + q0 * dr;
   uint16_t* dqg = dq + b_ * gb + h_ * gh + q0 * gr;
-  float* rsg = rs_g + bh * S5 + q0;
+  float* rsg = rs_g + bh * SEQ + q0;
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;  // 4 waves; wave w owns rows w*16..+16
@@ -776,9 +778,9 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
     att_glds16(dog + row * dr + kk2, dOs + e);
   }
 #pragma unroll
-  for (int c = 0; c < 16; ++c) {
-    int e = e0 + c * 2048;  // 32768 elems = [64][512] linear
-    att_glds16(pg + (e >> 9) * S5 + (e & 511), Ps + e);
+  for (int c = 0; c < SEQ / 32; ++c) {
+    int e = e0 + c * 2048;  // 64*SEQ elems, [64][SEQ] linear
+    att_glds16(pg + (e / SEQ) * SEQ + (e % SEQ), Ps + e);
   }
   // V tile 0 scattered transposed? NO — dP = dO . V^T is NT over d:
   // V [s2][d] is already k-contiguous; stage swizzled like fwd K tiles.
@@ -794,8 +796,8 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
   asm volatile("" ::: "memory");
 
   // ---- dP strip: loop k-tiles ---------------------------------------------
-  for (int kt = 0; kt < S5 / KT; ++kt) {
-    if (kt + 1 < S5 / KT) {
+  for (int kt = 0; kt < SEQ / KT; ++kt) {
+    if (kt + 1 < SEQ / KT) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
         int e = e0 + c * 2048;
@@ -833,7 +835,7 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
       for (int rI = 0; rI < 4; ++rI) {
         int row = wid * 16 + (lane >> 4) * 4 + rI;
         int col = kt * KT + fn * 16 + (lane & 15);
-        dSs[row * PST + col] = f2b(acc[fn][rI]);
+        dSs[row * QPST + col] = f2b(acc[fn][rI]);
       }
     asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
@@ -844,11 +846,11 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
   {
     const int row = (int)threadIdx.x >> 2;
     const int qtr = (int)threadIdx.x & 3;
-    const uint16_t* dpr = dSs + row * PST + qtr * 128;
-    const uint16_t* ppr = Ps + row * S5 + qtr * 128;
+    const uint16_t* dpr = dSs + row * QPST + qtr * (SEQ / 4);
+    const uint16_t* ppr = Ps + row * SEQ + qtr * (SEQ / 4);
     float accr = 0.f;
 #pragma unroll 4
-    for (int i = 0; i < 128; ++i) accr += b2f(dpr[i]) * b2f(ppr[i]);
+    for (int i = 0; i < SEQ / 4; ++i) accr += b2f(dpr[i]) * b2f(ppr[i]);
     rpart[row][qtr] = accr;
   }
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -866,11 +868,11 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
   {
     const int row = (int)threadIdx.x >> 2;
     const int qtr = (int)threadIdx.x & 3;
-    uint16_t* dpr = dSs + row * PST + qtr * 128;
-    const uint16_t* ppr = Ps + row * S5 + qtr * 128;
+    uint16_t* dpr = dSs + row * QPST + qtr * (SEQ / 4);
+    const uint16_t* ppr = Ps + row * SEQ + qtr * (SEQ / 4);
     const float r = rsl[row];
 #pragma unroll 4
-    for (int i = 0; i < 128; ++i)
+    for (int i = 0; i < SEQ / 4; ++i)
       dpr[i] = f2b(scale * b2f(ppr[i]) * (b2f(dpr[i]) - r));
   }
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -892,8 +894,8 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
-  for (int kt = 0; kt < S5 / KT; ++kt) {
-    if (kt + 1 < S5 / KT) {
+  for (int kt = 0; kt < SEQ / KT; ++kt) {
+    if (kt + 1 < SEQ / KT) {
       uint16_t* kdst = KVs + ((kt + 1) & 1) * AD * VST;
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
@@ -913,7 +915,7 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
       const int kbase = kk + (lane >> 4) * 8;
       {
         int row = wid * 16 + (lane & 15);
-        a_frag = *(const short8*)(dSs + row * PST + kt * KT + kbase);
+        a_frag = *(const short8*)(dSs + row * QPST + kt * KT + kbase);
       }
 #pragma unroll
       for (int fn = 0; fn < 4; ++fn) {
@@ -939,6 +941,7 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
     }
 }
 
+template <int SEQ>
 __launch_bounds__(256, 2) __global__ void attn_bwd512_kv_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const uint16_t* __restrict__ probs,
@@ -958,18 +961,18 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_kv_k(
   uint16_t* Qt = dOt + AD * VST;             // [d][72]
   uint16_t* dSt = Qt + AD * VST;             // [s2][72]
 
-  const int64_t nkb = S5 / KT;
+  const int64_t nkb = SEQ / KT;
   const int64_t bhk = blockIdx.x;
   const int64_t bh = bhk / nkb, k0r = (bhk % nkb) * KT;
   const int64_t b_ = bh / heads, h_ = bh % heads;
   const uint16_t* qg = q + b_ * qb + h_ * qh;
   const uint16_t* kg = k + b_ * qb + h_ * qh + k0r * qr;
   const uint16_t* vg = v + b_ * qb + h_ * qh + k0r * qr;
-  const uint16_t* pg = probs + bh * (int64_t)S5 * S5 + k0r;  // column block
+  const uint16_t* pg = probs + bh * (int64_t)SEQ * SEQ + k0r;  // column block
   const uint16_t* dog = dout + b_ * db + h_ * dhh;
   uint16_t* dkg = dk + b_ * gb + h_ * gh + k0r * gr;
   uint16_t* dvg = dv + b_ * gb + h_ * gh + k0r * gr;
-  const float* rsg = rs_g + bh * S5;
+  const float* rsg = rs_g + bh * SEQ;
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;  // wave w owns s2 rows w*16..+16
@@ -988,7 +991,7 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_kv_k(
   asm volatile("" ::: "memory");
 
   f32x4 accV[4] = {}, accK[4] = {};
-  for (int qt = 0; qt < S5 / SQ; ++qt) {
+  for (int qt = 0; qt < SEQ / SQ; ++qt) {
     const int64_t q0 = (int64_t)qt * SQ;
     // stage dO_qt (glds swz); P_qt + Q_qt via registers -> transposed images
 #pragma unroll
@@ -1003,7 +1006,7 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_kv_k(
     for (int c = 0; c < 2; ++c) {
       int e = e0 + c * 2048;
       int s1 = e >> 6, d = e & 63;  // [s1][64] block coords
-      short8 pv = *(const short8*)(pg + (q0 + s1) * S5 + d);  // P[s1][k0r+d]
+      short8 pv = *(const short8*)(pg + (q0 + s1) * SEQ + d);  // P[s1][k0r+d]
       short8 qv = *(const short8*)(qg + (q0 + s1) * qr + d);
       short8 dov = *(const short8*)(dog + (q0 + s1) * dr + d);
 #pragma unroll
@@ -1156,22 +1159,37 @@ void attn_fused_bwd512_packed(const Tensor& qkv, const Tensor& probs,
                               int64_t S, int64_t heads, int64_t dh,
                               float scale) {
   NZ_CHECK(qkv.dtype() == DType::BF16, "fused attn bwd 512: bf16 required");
-  NZ_CHECK(S == 512 && dh == AD, "fused attn bwd 512: S=512, dh=64 only");
+  NZ_CHECK((S == 512 || S == 128) && dh == AD,
+           "fused attn bwd pair: S in {128,512}, dh=64 only");
   hipStream_t s = stream_of(qkv);
   int64_t H = heads * dh;
   const uint16_t* base = (const uint16_t*)qkv.data();
   uint16_t* gbase = (uint16_t*)dqkv.data();
   Tensor rs(qkv.ctx(), DType::F32, Shape{B * heads * S});
-  attn_bwd512_q_k<<<dim3((uint32_t)(B * heads * 8)), dim3(256), 0, s>>>(
-      base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
-      (const uint16_t*)dmerged.data(), gbase, (float*)rs.data(), scale,
-      (int)heads, S * 3 * H, dh, 3 * H, S * H, dh, H, S * 3 * H, dh, 3 * H);
-  NZ_KERNEL_CHECK();
-  attn_bwd512_kv_k<<<dim3((uint32_t)(B * heads * 8)), dim3(256), 0, s>>>(
-      base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
-      (const uint16_t*)dmerged.data(), gbase + H, gbase + 2 * H,
-      (const float*)rs.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
-      S * H, dh, H, S * 3 * H, dh, 3 * H);
+  uint32_t nblk = (uint32_t)(B * heads * (S / 64));
+  if (S == 512) {
+    attn_bwd512_q_k<512><<<dim3(nblk), dim3(256), 0, s>>>(
+        base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
+        (const uint16_t*)dmerged.data(), gbase, (float*)rs.data(), scale,
+        (int)heads, S * 3 * H, dh, 3 * H, S * H, dh, H, S * 3 * H, dh, 3 * H);
+    NZ_KERNEL_CHECK();
+    attn_bwd512_kv_k<512><<<dim3(nblk), dim3(256), 0, s>>>(
+        base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
+        (const uint16_t*)dmerged.data(), gbase + H, gbase + 2 * H,
+        (const float*)rs.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
+        S * H, dh, H, S * 3 * H, dh, 3 * H);
+  } else {
+    attn_bwd512_q_k<128><<<dim3(nblk), dim3(256), 0, s>>>(
+        base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
+        (const uint16_t*)dmerged.data(), gbase, (float*)rs.data(), scale,
+        (int)heads, S * 3 * H, dh, 3 * H, S * H, dh, H, S * 3 * H, dh, 3 * H);
+    NZ_KERNEL_CHECK();
+    attn_bwd512_kv_k<128><<<dim3(nblk), dim3(256), 0, s>>>(
+        base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
+        (const uint16_t*)dmerged.data(), gbase + H, gbase + 2 * H,
+        (const float*)rs.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
+        S * H, dh, H, S * 3 * H, dh, 3 * H);
+  }
   NZ_KERNEL_CHECK();
 }
 
