@@ -24,6 +24,20 @@ def test_gemm_race_screen_and_oracles():
     assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
     assert b"RACE-SCREEN-OK" in r.stdout
 
+def test_attn128_bwd_pair_matches_mono():
+    """The templated two-kernel attention backward at SEQ=128
+    (NEZHA_ATTN128_BWD=pair) vs the monolithic S=128 kernel — guards the
+    QPST/SEQ-4 strip addressing (a leftover s512 PST constant once made
+    the SEQ=128 instantiation read outside the dS strip)."""
+    import subprocess, sys, os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run([sys.executable,
+                        os.path.join(repo, "tools", "attn_pair_check.py")],
+                       cwd=repo, capture_output=True, timeout=600)
+    assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
+    assert b"ATTN128-PAIR-OK" in r.stdout
+
+
 RNG = np.random.default_rng(42)
 
 
