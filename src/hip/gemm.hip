@@ -2981,10 +2981,24 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
 
   int nwg = gridDim.x;
   int orig = blockIdx.x;
-  int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
-  int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
   int64_t tm, tn;
-  tile_coords(wgid, nwg / nwg_n, nwg_n, args.group_m, tm, tn);
+  if (args.group_m < 0) {
+    // XCD-slab raster (short-K): consecutive blockIdx dispatch round-robin
+    // over the 8 XCDs, so XCD x = orig%8 owns M-slabs {x, x+8, ...} and
+    // sweeps ALL N-tiles of a slab before the next (N-first order). The
+    // ~64 live WGs of an XCD then share ~4 A-slabs (~0.8 MB) in that
+    // XCD's L2 — the cold A stream becomes L2-hit and the glds stage
+    // rate is no longer HBM-latency-starved (requires nwg_m % 8 == 0;
+    // dispatch guards it).
+    int i = orig >> 3, x = orig & 7;
+    tm = (int64_t)(i / nwg_n) * 8 + x;
+    tn = i % nwg_n;
+  } else {
+    int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+    int wgid =
+        (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+    tile_coords(wgid, nwg / nwg_n, nwg_n, args.group_m, tm, tn);
+  }
   const int64_t m0 = tm * QM;
   const int64_t n0 = tn * QN;
   const int64_t batch = blockIdx.y;
@@ -3098,12 +3112,16 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
     asm volatile("" ::: "memory");                         \
   } while (0)
 
+  // diag ablation (NEZHA_PERS_DIAG, same contract as the v7 kernel;
+  // OUTPUT GARBAGE): 2 = stage-only, 3 = compute-only
+  const bool dg_mfma = args.diag_no_store != 2;
+  const bool dg_stage = args.diag_no_store != 3;
   for (int64_t t = 0; t < ntiles; t += 2) {
-    const bool tail = (t + 2 >= ntiles);
-    const bool s2 = (t + 2 < ntiles), s3 = (t + 3 < ntiles);
+    const bool tail = (t + 2 >= ntiles) || !dg_stage;
+    const bool s2 = (t + 2 < ntiles) && dg_stage, s3 = (t + 3 < ntiles) && dg_stage;
     // p0
     if (g == 0) {
-      mfma8(0);
+      if (dg_mfma) mfma8(0);
     } else {
       read_af(t);
       read_bf(0, t);
@@ -3113,7 +3131,7 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
     // p1
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma8(0);
+      if (dg_mfma) mfma8(0);
     } else {
       read_bf(1, t);
       if (s2) stage(0, 1, t + 2);
@@ -3122,7 +3140,7 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
     // p2
     if (g == 0) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma8(1);
+      if (dg_mfma) mfma8(1);
     } else {
       read_bf(1, t);
       if (s2) stage(1, 0, t + 2);
@@ -3131,7 +3149,7 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
     // p3
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma8(1);
+      if (dg_mfma) mfma8(1);
     } else {
       read_af(t + 1);
       read_bf(0, t + 1);
@@ -3141,7 +3159,7 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
     // p4
     if (g == 0) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma8(0);
+      if (dg_mfma) mfma8(0);
     } else {
       read_af(t + 1);
       read_bf(0, t + 1);
@@ -3151,7 +3169,7 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
     // p5
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma8(0);
+      if (dg_mfma) mfma8(0);
     } else {
       read_bf(1, t + 1);
       if (s3) stage(0, 1, t + 3);
@@ -3160,7 +3178,7 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
     // p6
     if (g == 0) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma8(1);
+      if (dg_mfma) mfma8(1);
     } else {
       read_bf(1, t + 1);
       if (s3) stage(1, 0, t + 3);
@@ -3169,7 +3187,7 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
     // p7
     if (g == 1) {
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      mfma8(1);
+      if (dg_mfma) mfma8(1);
     } else {
       if (s2) {
         read_af(t + 2);
@@ -3180,6 +3198,298 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
     Q4_END((site<0, 6>(g, tail), site<1, 4>(g, tail)));
   }
 #undef Q4_END
+
+  // drain, then the LDS-staged epilogue reuses smem (32 KiB C tile)
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  if (out32) {
+#pragma unroll
+    for (int b2 = 0; b2 < 2; ++b2)
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+          for (int rI = 0; rI < 4; ++rI) {
+            int64_t grow = m0 + g * 64 + (ws >> 1) * 32 + fm * 16 +
+                           (lane >> 4) * 4 + rI;
+            int64_t gcol =
+                n0 + b2 * 64 + (ws & 1) * 32 + fn * 16 + (lane & 15);
+            if (grow < args.M && gcol < args.N)
+              out32[grow * args.N + gcol] = acc[b2][fm][fn][rI];
+          }
+    return;
+  }
+#pragma unroll
+  for (int b2 = 0; b2 < 2; ++b2)
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+        for (int rI = 0; rI < 4; ++rI) {
+          int lrow = g * 64 + (ws >> 1) * 32 + fm * 16 + (lane >> 4) * 4 + rI;
+          int lcol = b2 * 64 + (ws & 1) * 32 + fn * 16 + (lane & 15);
+          smem[lrow * 128 + lcol] = f2b(args.alpha * acc[b2][fm][fn][rI]);
+        }
+  __syncthreads();
+  const bool interior = (m0 + QM <= args.M) && (n0 + QN <= args.N);
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    int e = (c * 512 + (int)threadIdx.x) * 8;
+    int row = e >> 7, col = e & 127;
+    int64_t grow = m0 + row;
+    int64_t gcol = n0 + col;
+    short8 vv = *(const short8*)(smem + e);
+    if (interior) {
+      int64_t idx = grow * args.N + gcol;
+      short8 zout, yout;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        zout[i] = (short)f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        yout[i] = (short)f2b(v);
+      }
+      if (save_pre) *(short8*)(save_pre + idx) = zout;
+      *(short8*)(C + idx) = yout;
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        if (grow >= args.M || gcol + i >= args.N) continue;
+        int64_t idx = grow * args.N + gcol + i;
+        float v = b2f((uint16_t)vv[i]);
+        if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
+        if (args.bias) v += b2f(args.bias[gcol + i]);
+        if (save_pre) save_pre[idx] = f2b(v);
+        if (args.fuse_gelu) v = gelu_fwd_scalar(v);
+        C[idx] = f2b(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 4-PHASE variant of the 2-WG/CU 128^2 ping-pong (NEZHA_GEMM_64Q=4): same
+// tile geometry and epilogue as gemm_bf16_pp64q_k, but 4 fat phases per 2
+// k-iterations instead of 8 thin ones — each compute phase is a 16-MFMA
+// cluster (both B-halves) and the barrier/lgkm machinery per FLOP halves.
+// The pp64q diag ablation showed compute-only pace 4x above the MFMA issue
+// floor (0.370 vs 0.092 ms on the BERT qkv panel) with the thin phases;
+// this variant attacks exactly that.
+//
+// Staging plan (verified by tools/pp64q4_sim.py — run it after ANY edit
+// to the schedule below):
+//   - A0/B0 (read by G0/both, staged by G1): DOUBLE buffered, slot t%2.
+//     G1 stages A0B0(t+2) at P0 and A0B0(t+3) at P2.
+//   - A1/B1 (read by G1/both, staged by G0): TRIPLE buffered, slot t%3 —
+//     G0 stages A1B1(t+3) at P1 and A1B1(t+4) at P3; the third buffer is
+//     what lets it run a full iteration deeper without overwriting a slot
+//     another phase still reads. 2x2+3x2 half-units = 80 KiB = the whole
+//     per-WG LDS budget at 2 WG/CU.
+//   - Certificates: BOTH groups wait vmcnt<=4 (keep 2 stage-calls in
+//     flight) at end-P0 and end-P2; end-P1/P3 are barrier-only. Tail
+//     iteration drains vmcnt(0). Every read-then-stage phase drains lgkm
+//     between its ds_reads and its glds issues (same-parity slot reuse).
+// ---------------------------------------------------------------------------
+__launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q4_k(GemmArgs args,
+                                                             int nwg_n) {
+  using namespace pp64q;
+  __shared__ __attribute__((aligned(16))) uint16_t smem[10 * HALF_E];
+
+  int nwg = gridDim.x;
+  int orig = blockIdx.x;
+  int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+  int wgid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+  int64_t tm, tn;
+  tile_coords(wgid, nwg / nwg_n, nwg_n, args.group_m, tm, tn);
+  const int64_t m0 = tm * QM;
+  const int64_t n0 = tn * QN;
+  const int64_t batch = blockIdx.y;
+  const uint16_t* __restrict__ A = args.A + batch * args.sA;
+  const uint16_t* __restrict__ B = args.B + batch * args.sB;
+  uint16_t* C = args.C + batch * args.sC;
+  float* out32 = args.out_f32 ? args.out_f32 + batch * args.sC : nullptr;
+  uint16_t* save_pre = args.save_pre;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int g = wid >> 2;   // group: A-half (rows g*64..+64)
+  const int ws = wid & 3;   // 2x2 wave grid over a 64x64 quadrant
+  const int tig = (int)threadIdx.x & 255;
+  if (g == 1) __builtin_amdgcn_s_setprio(1);
+
+  const int64_t lda = args.lda, ldb = args.ldb;
+  const int64_t Mc = args.M - 1, Nc = args.N - 1;
+  const int64_t ntiles = args.K / QK;  // even, >= 4 (launch guard)
+
+  // slot index: A0 -> 0 + t%2, B0 -> 2 + t%2, A1 -> 4 + t%3, B1 -> 7 + t%3
+  auto slot4 = [&](int isB, int h, int64_t t) -> uint16_t* {
+    int s = h ? (4 + 3 * isB + (int)(t % 3)) : (2 * isB + (int)(t & 1));
+    return smem + s * HALF_E;
+  };
+  auto stage4 = [&](int isB, int h, int64_t t) {
+    int64_t k0 = t * QK;
+    uint16_t* dst = slot4(isB, h, t);
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int e = tig * 8 + c * 2048;
+      int row = e >> 6;
+      int k = (e & 63) ^ ((row & 7) << 3);
+      if (!isB) {
+        int64_t ar = m0 + h * 64 + row;
+        if (ar > Mc) ar = Mc;
+        glds16(A + ar * lda + k0 + k, dst + e);
+      } else {
+        int64_t br = n0 + h * 64 + row;
+        if (br > Nc) br = Nc;
+        glds16(B + br * ldb + k0 + k, dst + e);
+      }
+    }
+  };
+
+  f32x4 acc[2][2][2] = {};       // [B-half][fm][fn]
+  short8 af[2][2], bf[2][2][2];  // af[kk2][fm]; bf[b][kk2][fn]
+
+  auto read_af4 = [&](int64_t t) {
+    const uint16_t* as = slot4(0, g, t);
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2) {
+      const int kbase = kk2 * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        int row = (ws >> 1) * 32 + fm * 16 + (lane & 15);
+        int ks = kbase ^ ((row & 7) << 3);
+        af[kk2][fm] = *(const short8*)(as + row * QK + ks);
+      }
+    }
+  };
+  auto read_bf4 = [&](int b, int64_t t) {
+    const uint16_t* bs = slot4(1, b, t);
+#pragma unroll
+    for (int kk2 = 0; kk2 < 2; ++kk2) {
+      const int kbase = kk2 * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = (ws & 1) * 32 + fn * 16 + (lane & 15);
+        int ks = kbase ^ ((col & 7) << 3);
+        bf[b][kk2][fn] = *(const short8*)(bs + col * QK + ks);
+      }
+    }
+  };
+
+  auto mfma16 = [&]() {
+#pragma unroll
+    for (int b = 0; b < 2; ++b)
+#pragma unroll
+      for (int kk2 = 0; kk2 < 2; ++kk2)
+#pragma unroll
+        for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn)
+            acc[b][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[kk2][fm], bf[b][kk2][fn], acc[b][fm][fn], 0, 0, 0);
+  };
+
+  // prologue: G1 stages A0B0 of tiles 0,1; G0 stages A1B1 of 0,1,2
+  // (the triple buffer starts one tile deeper); full drain, then G0
+  // pre-reads its tile-0 fragments.
+  if (g == 1) {
+    stage4(0, 0, 0); stage4(1, 0, 0);
+    stage4(0, 0, 1); stage4(1, 0, 1);
+  } else {
+    stage4(0, 1, 0); stage4(1, 1, 0);
+    stage4(0, 1, 1); stage4(1, 1, 1);
+    stage4(0, 1, 2); stage4(1, 1, 2);
+  }
+  ph8::vm_wait<0>();
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  if (g == 0) {
+    read_af4(0);
+    read_bf4(0, 0);
+    read_bf4(1, 0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+
+#define Q4P_END(vmstmt)                                    \
+  do {                                                     \
+    vmstmt;                                                \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");     \
+    __builtin_amdgcn_s_barrier();                          \
+    asm volatile("" ::: "memory");                         \
+  } while (0)
+
+  const bool dg_mfma = args.diag_no_store != 2;
+  const bool dg_stage = args.diag_no_store != 3;
+  const bool dg_drain = args.diag_no_store == 9;  // debug: vm(0) every phase
+  // debug bisect: 8 = add vm(4) at the barrier-only ends (P1/P3);
+  //               7 = tighten P0/P2 to vm(0), keep P1/P3 barrier-only
+  const bool dg_w13 = args.diag_no_store == 8;
+  const bool dg_w02 = args.diag_no_store == 7;
+  for (int64_t t = 0; t < ntiles; t += 2) {
+    const bool tail = (t + 2 >= ntiles) || !dg_stage || dg_drain;
+    const bool s2 = (t + 2 < ntiles) && dg_stage;
+    const bool s3 = (t + 3 < ntiles) && dg_stage;
+    const bool s4 = (t + 4 < ntiles) && dg_stage;
+    // P0: G0 computes t | G1 reads t-frags, stages A0B0(t+2)
+    if (g == 0) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if (dg_mfma) mfma16();
+    } else {
+      read_af4(t);
+      read_bf4(0, t);
+      read_bf4(1, t);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if (s2) { stage4(0, 0, t + 2); stage4(1, 0, t + 2); }
+    }
+    Q4P_END(((tail || dg_w02) ? ph8::vm_wait<0>() : ph8::vm_wait<4>()));
+    // P1: G1 computes t | G0 reads (t+1)-frags, stages A1B1(t+3)
+    if (g == 1) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if (dg_mfma) mfma16();
+    } else {
+      read_af4(t + 1);
+      read_bf4(0, t + 1);
+      read_bf4(1, t + 1);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if (s3) { stage4(0, 1, t + 3); stage4(1, 1, t + 3); }
+    }
+    Q4P_END((tail ? ph8::vm_wait<0>() : (dg_w13 ? ph8::vm_wait<4>() : (void)0)));
+    // P2: G0 computes t+1 | G1 reads (t+1)-frags, stages A0B0(t+3)
+    if (g == 0) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if (dg_mfma) mfma16();
+    } else {
+      read_af4(t + 1);
+      read_bf4(0, t + 1);
+      read_bf4(1, t + 1);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if (s3) { stage4(0, 0, t + 3); stage4(1, 0, t + 3); }
+    }
+    Q4P_END(((tail || dg_w02) ? ph8::vm_wait<0>() : ph8::vm_wait<4>()));
+    // P3: G1 computes t+1 | G0 reads (t+2)-frags, stages A1B1(t+4)
+    if (g == 1) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if (dg_mfma) mfma16();
+    } else {
+      if (s2) {
+        read_af4(t + 2);
+        read_bf4(0, t + 2);
+        read_bf4(1, t + 2);
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if (s4) { stage4(0, 1, t + 4); stage4(1, 1, t + 4); }
+    }
+    Q4P_END((tail ? ph8::vm_wait<0>() : (dg_w13 ? ph8::vm_wait<4>() : (void)0)));
+  }
+#undef Q4P_END
 
   // drain, then the LDS-staged epilogue reuses smem (32 KiB C tile)
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -4254,16 +4564,43 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
   // across tiles (no per-tile ramp), countable LDS-staged epilogue.
   // short-K NT (BERT K=768 panels): 128^2 ping-pong at 2 WGs/CU —
   // the de-synchronized pair hides prologue/wait/epilogue bubbles
-  static const bool k64qOff = getenv("NEZHA_GEMM_64Q") &&
-                              atoi(getenv("NEZHA_GEMM_64Q")) == 0;
+  // NEZHA_GEMM_64Q: 0 = off (fall through to the r1 persistent kernel),
+  // 1 = 8-thin-phase pp64q (DEFAULT), 4 = 4-fat-phase pp64q4.
+  // The 4-phase variant measured 501-523 TF vs the 8-phase's 519-533 on
+  // the BERT panels (profiles/r2/r2_pp64q4.log) — merging phases does not
+  // help because the pace-setter is DMA-service latency surfacing through
+  // the stage-wait certificates, not barrier count — AND it has a rare
+  // box-dependent nondeterminism (profiles/r2/r2_q4race*.log: a single
+  // 128 B glds beat of a B0 fragment goes stale; full-drain mode is clean
+  // but run-to-run variance defeated the wait-site bisection). Kept as an
+  // opt-in research artifact; NOT on any default path.
+  static const int k64qMode = [] {
+    const char* v = getenv("NEZHA_GEMM_64Q");
+    return v ? atoi(v) : 1;
+  }();
+  static const bool k64qOff = k64qMode == 0;
   if (!transA && transB && batch == 1 && !k64qOff && kNtVariant >= 7 &&
       args.K % 128 == 0 && args.K >= 256 && args.K < 1536) {
     int nwg_m4 = (int)((args.M + 127) / 128);
     int nwg_n4 = (int)((args.N + 127) / 128);
     int64_t nwg4 = (int64_t)nwg_m4 * nwg_n4;
     if (nwg4 >= 1024) {  // >= 2 rounds of the 512-WG chip
+      // XCD-slab raster (see kernel comment): measured a wash on the
+      // BERT panels (511 vs 532 TF, profiles/r2/r2_skraster.log) — the
+      // per-WG glds stream rate, not L2 locality, paces short-K staging.
+      // Kept as an opt-in (NEZHA_GEMM_SKRASTER=1) for future A/B.
+      static const bool kSkRaster = [] {
+        const char* v = getenv("NEZHA_GEMM_SKRASTER");
+        return v && atoi(v) == 1;
+      }();
+      GemmArgs a4 = args;
+      // slab raster only implemented in the 8-phase kernel
+      if (kSkRaster && nwg_m4 % 8 == 0 && k64qMode != 4) a4.group_m = -1;
       dim3 grid((uint32_t)nwg4);
-      gemm_bf16_pp64q_k<<<grid, dim3(512), 0, s>>>(args, nwg_n4);
+      if (k64qMode == 4)
+        gemm_bf16_pp64q4_k<<<grid, dim3(512), 0, s>>>(a4, nwg_n4);
+      else
+        gemm_bf16_pp64q_k<<<grid, dim3(512), 0, s>>>(a4, nwg_n4);
       NZ_KERNEL_CHECK();
       return;
     }
@@ -4306,9 +4643,15 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
       return;
     }
   }
-  // ping-pong single-tile for deep K
+  // ping-pong single-tile for deep K (NEZHA_GEMM_V7_MINK lowers the K
+  // threshold for A/B against the short-K kernels; default keeps 256^2
+  // on K >= 1536 where the ramp is amortized)
+  static const int kV7MinK = [] {
+    const char* v = getenv("NEZHA_GEMM_V7_MINK");
+    return v ? atoi(v) : 1536;
+  }();
   if (!transA && transB && batch == 1 && kNtVariant == 7 &&
-      args.K % 128 == 0 && args.K >= 1536) {
+      args.K % 128 == 0 && args.K >= kV7MinK && args.K / 64 >= 4) {
     int nwg_m = (int)((args.M + 255) / 256);
     int nwg_n = (int)((args.N + 255) / 256);
     if (nwg_m * (int64_t)nwg_n >= kNtMinWG &&
