@@ -218,16 +218,19 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused_k(
 //   Qt  [64][136]  17 KB (Q^T scatter)
 //   psum[128][32] f32 16 KB + rs[128] f32  (rowsum partials + fold)
 // ---------------------------------------------------------------------------
-__launch_bounds__(512) __global__ void attn_bwd_fused_k(
+__launch_bounds__(1024) __global__ void attn_bwd_fused_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const uint16_t* __restrict__ probs,
     const uint16_t* __restrict__ dout, uint16_t* __restrict__ dq,
     uint16_t* __restrict__ dk, uint16_t* __restrict__ dv, float scale,
     int heads, int64_t qb, int64_t qh, int64_t qr, int64_t db, int64_t dhh,
     int64_t dr, int64_t gb, int64_t gh, int64_t gr) {
-  // 512 threads (8 waves as 2x4): the 256-thread version ran 4 waves on a
-  // 137 KB, 1-WG/CU kernel = 1 wave/SIMD — every barrier and LDS/HBM
-  // latency fully exposed. 8 waves give 2/SIMD at the same occupancy.
+  // 1024 threads (16 waves as 4x4): this kernel is 153 KB of LDS and
+  // therefore pinned at 1 WG/CU, so waves-per-SIMD is the only occupancy
+  // axis. 256 threads ran 1 wave/SIMD (every barrier and LDS/HBM latency
+  // fully exposed), 512 ran 2 (r1), 16 waves give 4/SIMD — measured
+  // profile: the per-(b,h) chain is latency- not MFMA-bound (4.6% duty
+  // at 8 waves), so more resident waves per SIMD is pure overlap.
   // LDS 153 KB: psum widens to [128][64] (each lane covers 2 columns).
   __shared__ __attribute__((aligned(16))) uint16_t smem[76544];
   uint16_t* dOs = smem;                    // [128][64] swz
@@ -254,13 +257,12 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int wr = wid >> 2, wc = wid & 3;  // 2x4 wave grid
+  const int wr = wid >> 2, wc = wid & 3;  // 4x4 wave grid (32-row granules)
 
   // ---- P0: stage ----------------------------------------------------------
   const int e0 = (int)threadIdx.x * 8;
-#pragma unroll
-  for (int c = 0; c < 2; ++c) {
-    int e = e0 + c * 4096;
+  {
+    int e = e0;
     int row = e >> 6;
     int kk = (e & 63) ^ ((row & 7) << 3);
     att_glds16(dog + row * dr + kk, dOs + e);
@@ -269,12 +271,12 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
   // register loads + transposed scatters (issued while the DMA flies)
   {
 #pragma unroll
-    for (int c = 0; c < 4; c += 2) {
-      short8 p0 = *(const short8*)(pg + e0 + c * 4096);
-      short8 p1 = *(const short8*)(pg + e0 + (c + 1) * 4096);
+    for (int c = 0; c < 2; c += 2) {
+      short8 p0 = *(const short8*)(pg + e0 + c * 8192);
+      short8 p1 = *(const short8*)(pg + e0 + (c + 1) * 8192);
 #pragma unroll
       for (int u = 0; u < 2; ++u) {
-        int e = e0 + (c + u) * 4096;  // element in P [s1][s2]
+        int e = e0 + (c + u) * 8192;  // element in P [s1][s2]
         int s1 = e >> 7, s2 = e & 127;
         const short8& pv = u ? p1 : p0;
 #pragma unroll
@@ -282,9 +284,8 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
           Pt[(s2 + i) * PSTRIDE + s1] = (uint16_t)pv[i];
       }
     }
-#pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      int e = e0 + c * 4096;          // element in [s][d]
+    {
+      int e = e0;                     // element in [s][d]
       int s_ = e >> 6, d = e & 63;
       short8 dov = *(const short8*)(dog + s_ * dr + d);
       short8 kv = *(const short8*)(kg + s_ * qr + d);
@@ -299,15 +300,15 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
   }
   __syncthreads();  // glds drained + scatters visible
 
-  // ---- P1: dP = dO.V^T (f32 acc; wave owns 64 s1 x 32 s2) -----------------
-  f32x4 acc[4][2] = {};
+  // ---- P1: dP = dO.V^T (f32 acc; wave owns 32 s1 x 32 s2) -----------------
+  f32x4 acc[2][2] = {};
 #pragma unroll
   for (int kk = 0; kk < AD; kk += 32) {
-    short8 a_frag[4], b_frag[2];
+    short8 a_frag[2], b_frag[2];
     const int kbase = kk + (lane >> 4) * 8;
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm) {
-      int row = wr * 64 + fm * 16 + (lane & 15);
+    for (int fm = 0; fm < 2; ++fm) {
+      int row = wr * 32 + fm * 16 + (lane & 15);
       int ks = kbase ^ ((row & 7) << 3);
       a_frag[fm] = *(const short8*)(dOs + row * AD + ks);
     }
@@ -318,7 +319,7 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
       b_frag[fn] = *(const short8*)(Vs + col * AD + ks);
     }
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
+    for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -327,10 +328,10 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
 
   // ---- P2: rowsum partials of dP o P --------------------------------------
 #pragma unroll
-  for (int fm = 0; fm < 4; ++fm)
+  for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
     for (int rI = 0; rI < 4; ++rI) {
-      int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+      int row = wr * 32 + fm * 16 + (lane >> 4) * 4 + rI;
       float part = 0.f;
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn) {
@@ -340,16 +341,16 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
       psum[row * 64 + wc * 16 + (lane & 15)] = part;
     }
 
-  // ---- P2b: dV = P^T.dO (wave owns 64 s2 x 16 d) --------------------------
+  // ---- P2b: dV = P^T.dO (wave owns 32 s2 x 16 d) --------------------------
   {
-    f32x4 acc2[4] = {};
+    f32x4 acc2[2] = {};
 #pragma unroll
     for (int kk = 0; kk < AS; kk += 32) {
-      short8 a_frag[4], b_frag;
+      short8 a_frag[2], b_frag;
       const int kbase = kk + (lane >> 4) * 8;
 #pragma unroll
-      for (int fm = 0; fm < 4; ++fm) {
-        int row = wr * 64 + fm * 16 + (lane & 15);  // s2
+      for (int fm = 0; fm < 2; ++fm) {
+        int row = wr * 32 + fm * 16 + (lane & 15);  // s2
         a_frag[fm] = *(const short8*)(Pt + row * PSTRIDE + kbase);
       }
       {
@@ -357,15 +358,15 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
         b_frag = *(const short8*)(dOt + col * PSTRIDE + kbase);
       }
 #pragma unroll
-      for (int fm = 0; fm < 4; ++fm)
+      for (int fm = 0; fm < 2; ++fm)
         acc2[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[fm], b_frag,
                                                            acc2[fm], 0, 0, 0);
     }
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
+    for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
       for (int rI = 0; rI < 4; ++rI) {
-        int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+        int row = wr * 32 + fm * 16 + (lane >> 4) * 4 + rI;
         int col = wc * 16 + (lane & 15);
         dvg[row * gr + col] = f2b(acc2[fm][rI]);
       }
@@ -390,10 +391,10 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
   // ---- P4: dS = scale * P o (dP - rs); write dSn (swz) + dSt (transposed,
   // overwriting Pt in place: read P just before the store, same lane) ------
 #pragma unroll
-  for (int fm = 0; fm < 4; ++fm)
+  for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
     for (int rI = 0; rI < 4; ++rI) {
-      int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+      int row = wr * 32 + fm * 16 + (lane >> 4) * 4 + rI;
       float r = rs[row];
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn) {
@@ -408,14 +409,14 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
 
   // ---- P5: dQ = dS.K (A=dSn swz, B=Kt; wave owns 64 s1 x 16 d) ------------
   {
-    f32x4 acc2[4] = {};
+    f32x4 acc2[2] = {};
 #pragma unroll
     for (int kk = 0; kk < AS; kk += 32) {
-      short8 a_frag[4], b_frag;
+      short8 a_frag[2], b_frag;
       const int kbase = kk + (lane >> 4) * 8;
 #pragma unroll
-      for (int fm = 0; fm < 4; ++fm) {
-        int row = wr * 64 + fm * 16 + (lane & 15);  // s1
+      for (int fm = 0; fm < 2; ++fm) {
+        int row = wr * 32 + fm * 16 + (lane & 15);  // s1
         int ks = kbase ^ ((row & 7) << 3);
         a_frag[fm] = *(const short8*)(dSn + row * AS + ks);
       }
@@ -424,15 +425,15 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
         b_frag = *(const short8*)(Kt + col * PSTRIDE + kbase);
       }
 #pragma unroll
-      for (int fm = 0; fm < 4; ++fm)
+      for (int fm = 0; fm < 2; ++fm)
         acc2[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[fm], b_frag,
                                                            acc2[fm], 0, 0, 0);
     }
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
+    for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
       for (int rI = 0; rI < 4; ++rI) {
-        int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+        int row = wr * 32 + fm * 16 + (lane >> 4) * 4 + rI;
         int col = wc * 16 + (lane & 15);
         dqg[row * gr + col] = f2b(acc2[fm][rI]);
       }
@@ -440,14 +441,14 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
 
   // ---- P5b: dK = dS^T.Q (A=dSt, B=Qt; wave owns 64 s2 x 16 d) -------------
   {
-    f32x4 acc2[4] = {};
+    f32x4 acc2[2] = {};
 #pragma unroll
     for (int kk = 0; kk < AS; kk += 32) {
-      short8 a_frag[4], b_frag;
+      short8 a_frag[2], b_frag;
       const int kbase = kk + (lane >> 4) * 8;
 #pragma unroll
-      for (int fm = 0; fm < 4; ++fm) {
-        int row = wr * 64 + fm * 16 + (lane & 15);  // s2
+      for (int fm = 0; fm < 2; ++fm) {
+        int row = wr * 32 + fm * 16 + (lane & 15);  // s2
         a_frag[fm] = *(const short8*)(dSt + row * PSTRIDE + kbase);
       }
       {
@@ -455,15 +456,15 @@ __launch_bounds__(512) __global__ void attn_bwd_fused_k(
         b_frag = *(const short8*)(Qt + col * PSTRIDE + kbase);
       }
 #pragma unroll
-      for (int fm = 0; fm < 4; ++fm)
+      for (int fm = 0; fm < 2; ++fm)
         acc2[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[fm], b_frag,
                                                            acc2[fm], 0, 0, 0);
     }
 #pragma unroll
-    for (int fm = 0; fm < 4; ++fm)
+    for (int fm = 0; fm < 2; ++fm)
 #pragma unroll
       for (int rI = 0; rI < 4; ++rI) {
-        int row = wr * 64 + fm * 16 + (lane >> 4) * 4 + rI;
+        int row = wr * 32 + fm * 16 + (lane >> 4) * 4 + rI;
         int col = wc * 16 + (lane & 15);
         dkg[row * gr + col] = f2b(acc2[fm][rI]);
       }
@@ -1143,7 +1144,7 @@ void attn_fused_bwd(const Tensor& q, const Tensor& k, const Tensor& v,
   NZ_CHECK(q.dtype() == DType::BF16, "fused attention bwd: bf16 required");
   hipStream_t s = stream_of(q);
   const int64_t sh = (int64_t)AS * AD;
-  attn_bwd_fused_k<<<dim3((uint32_t)batch_heads), dim3(512), 0, s>>>(
+  attn_bwd_fused_k<<<dim3((uint32_t)batch_heads), dim3(1024), 0, s>>>(
       (const uint16_t*)q.data(), (const uint16_t*)k.data(),
       (const uint16_t*)v.data(), (const uint16_t*)probs.data(),
       (const uint16_t*)dout.data(), (uint16_t*)dq.data(),
@@ -1233,7 +1234,7 @@ void attn_fused_bwd_packed(const Tensor& qkv, const Tensor& probs,
   int64_t H = heads * dh;
   const uint16_t* base = (const uint16_t*)qkv.data();
   uint16_t* gbase = (uint16_t*)dqkv.data();
-  attn_bwd_fused_k<<<dim3((uint32_t)(B * heads)), dim3(512), 0, s>>>(
+  attn_bwd_fused_k<<<dim3((uint32_t)(B * heads)), dim3(1024), 0, s>>>(
       base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
       (const uint16_t*)dmerged.data(), gbase, gbase + H, gbase + 2 * H, scale,
       (int)heads, S * 3 * H, dh, 3 * H, S * H, dh, H, S * 3 * H, dh, 3 * H);
