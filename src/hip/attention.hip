@@ -497,7 +497,7 @@ constexpr int PST = S5 + 8;    // P strip row stride (elements)
 constexpr int VST = KT + 8;    // VT row stride
 }  // namespace s512
 
-__launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
+__launch_bounds__(512, 2) __global__ void attn_fwd_fused512_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ probs,
     uint16_t* __restrict__ out, float scale, int heads, int64_t qb,
@@ -506,7 +506,7 @@ __launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
   // [Q 64x64][K dbuf 2x 64x64][VT dbuf 2x 64x72][P 64x520][ms 64][ls 64+psum]
   __shared__ __attribute__((aligned(16))) uint16_t smem[
       SQ * AD + 2 * KT * AD + 2 * AD * VST + SQ * PST];
-  __shared__ float ms[SQ], lsum[SQ][4], ls[SQ];
+  __shared__ float ms2[2][SQ], lsum[SQ][8], ls[SQ];
   uint16_t* Qs = smem;
   uint16_t* Ks = smem + SQ * AD;                  // dbuf: + (kt&1)*KT*AD
   uint16_t* VTs = smem + SQ * AD + 2 * KT * AD;   // dbuf: + (kt&1)*AD*VST
@@ -523,14 +523,15 @@ __launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
   uint16_t* og = out + b_ * ob + h_ * oh + qb0 * orr;
 
   const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;  // 4 waves; wave w owns rows w*16..+16
+  const int wid = threadIdx.x >> 6;  // 8 waves (2/SIMD at 1 WG/CU)
+  const int wr = wid & 3;            // row group: 16 q-rows
+  const int wc = wid >> 2;           // column half: 2 of the 4 fn tiles
 
   // ---- stage Q (glds, swizzled source) + K tile 0 -------------------------
-  // a 64x64 tile is 4096 elems = 2 chunks of (256 threads x 8 elems)
+  // a 64x64 tile is 4096 elems = one pass of (512 threads x 8 elems)
   const int e0 = (int)threadIdx.x * 8;
-#pragma unroll
-  for (int c = 0; c < 2; ++c) {
-    int e = e0 + c * 2048;
+  {
+    int e = e0;
     int row = e >> 6;
     int kk = (e & 63) ^ ((row & 7) << 3);
     att_glds16(qg + row * qr + kk, Qs + e);
@@ -545,54 +546,52 @@ __launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
   for (int kt = 0; kt < S5 / KT; ++kt) {
     // prefetch next K tile into the other buffer
     if (kt + 1 < S5 / KT) {
-#pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        int e = e0 + c * 2048;
-        int row = e >> 6;
-        int kk2 = (e & 63) ^ ((row & 7) << 3);
-        att_glds16(kg + (int64_t)(kt + 1) * KT * qr + row * qr + kk2,
-                   Ks + ((kt + 1) & 1) * KT * AD + e);
-      }
+      int e = e0;
+      int row = e >> 6;
+      int kk2 = (e & 63) ^ ((row & 7) << 3);
+      att_glds16(kg + (int64_t)(kt + 1) * KT * qr + row * qr + kk2,
+                 Ks + ((kt + 1) & 1) * KT * AD + e);
     }
     const uint16_t* kb = Ks + (kt & 1) * KT * AD;
-    f32x4 acc[4] = {};
+    f32x4 acc[2] = {};
 #pragma unroll
     for (int kk = 0; kk < AD; kk += 32) {
-      short8 a_frag, b_frag[4];
+      short8 a_frag, b_frag[2];
       const int kbase = kk + (lane >> 4) * 8;
       {
-        int row = wid * 16 + (lane & 15);
+        int row = wr * 16 + (lane & 15);
         int ks = kbase ^ ((row & 7) << 3);
         a_frag = *(const short8*)(Qs + row * AD + ks);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn) {
-        int col = fn * 16 + (lane & 15);
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = (wc * 2 + fn) * 16 + (lane & 15);
         int ks = kbase ^ ((col & 7) << 3);
         b_frag[fn] = *(const short8*)(kb + col * AD + ks);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn)
+      for (int fn = 0; fn < 2; ++fn)
         acc[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
                                                           acc[fn], 0, 0, 0);
     }
-    // running max (rows wave-local: reduce over the 16-lane col group + fn)
+    // running max over this wave's half of the columns (combined with the
+    // other column-half's partial at publish time via ms2[2][SQ])
 #pragma unroll
     for (int rI = 0; rI < 4; ++rI) {
       float mx = -1e30f;
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn) mx = fmaxf(mx, scale * acc[fn][rI]);
+      for (int fn = 0; fn < 2; ++fn) mx = fmaxf(mx, scale * acc[fn][rI]);
       for (int off = 1; off < 16; off <<= 1)
         mx = fmaxf(mx, __shfl_xor(mx, off, 64));
       m_run[rI] = fmaxf(m_run[rI], mx);
     }
     // raw scores (scaled) -> P strip, bf16
 #pragma unroll
-    for (int fn = 0; fn < 4; ++fn)
+    for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
       for (int rI = 0; rI < 4; ++rI) {
-        int row = wid * 16 + (lane >> 4) * 4 + rI;
-        int col = kt * KT + fn * 16 + (lane & 15);
+        int row = wr * 16 + (lane >> 4) * 4 + rI;
+        int col = kt * KT + (wc * 2 + fn) * 16 + (lane & 15);
         Ps[row * PST + col] = f2b(scale * acc[fn][rI]);
       }
     // next iteration reads the prefetched buffer; everyone must also be
@@ -601,11 +600,11 @@ __launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
     __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
   }
-  // publish row maxima: wave w's lane group holds rows w*16+(lane>>4)*4+rI
+  // publish row maxima (one partial per column-half)
   if ((lane & 15) == 0) {
 #pragma unroll
     for (int rI = 0; rI < 4; ++rI)
-      ms[wid * 16 + (lane >> 4) * 4 + rI] = m_run[rI];
+      ms2[wc][wr * 16 + (lane >> 4) * 4 + rI] = m_run[rI];
   }
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
@@ -613,35 +612,36 @@ __launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
 
   // ---- A2: exp(s - m), row sums (4 threads per row) -----------------------
   {
-    const int row = (int)threadIdx.x >> 2;
-    const int qtr = (int)threadIdx.x & 3;
-    const float m = ms[row];
-    uint16_t* pr = Ps + row * PST + qtr * 128;
+    const int row = (int)threadIdx.x >> 3;
+    const int oct = (int)threadIdx.x & 7;
+    const float m = fmaxf(ms2[0][row], ms2[1][row]);
+    uint16_t* pr = Ps + row * PST + oct * 64;
     float acc_l = 0.f;
 #pragma unroll 4
-    for (int i = 0; i < 128; ++i) {
+    for (int i = 0; i < 64; ++i) {
       float e = __expf(b2f(pr[i]) - m);
       acc_l += e;
       pr[i] = f2b(e);
     }
-    lsum[row][qtr] = acc_l;
+    lsum[row][oct] = acc_l;
   }
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
-  if (threadIdx.x < SQ)
-    ls[threadIdx.x] = 1.0f / (lsum[threadIdx.x][0] + lsum[threadIdx.x][1] +
-                              lsum[threadIdx.x][2] + lsum[threadIdx.x][3]);
+  if (threadIdx.x < SQ) {
+    const float* lr = lsum[threadIdx.x];
+    ls[threadIdx.x] = 1.0f / (((lr[0] + lr[1]) + (lr[2] + lr[3])) +
+                              ((lr[4] + lr[5]) + (lr[6] + lr[7])));
+  }
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
 
   // ---- B: O = P~ . V, V tiles scattered transposed ------------------------
-  f32x4 accO[4] = {};
-#pragma unroll
-  for (int c = 0; c < 2; ++c) {
-    // V tile 0 -> VT buf 0
-    int e = e0 + c * 2048;
+  f32x4 accO[2] = {};
+  {
+    // V tile 0 -> VT buf 0 (single pass at 512 threads)
+    int e = e0;
     int s_ = e >> 6, d = e & 63;
     short8 vv = *(const short8*)(vg + s_ * qr + d);
 #pragma unroll
@@ -654,33 +654,30 @@ __launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
   for (int kt = 0; kt < S5 / KT; ++kt) {
     if (kt + 1 < S5 / KT) {
       uint16_t* vt = VTs + ((kt + 1) & 1) * AD * VST;
+      int e = e0;
+      int s_ = e >> 6, d = e & 63;
+      short8 vv =
+          *(const short8*)(vg + (int64_t)(kt + 1) * KT * qr + s_ * qr + d);
 #pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        int e = e0 + c * 2048;
-        int s_ = e >> 6, d = e & 63;
-        short8 vv =
-            *(const short8*)(vg + (int64_t)(kt + 1) * KT * qr + s_ * qr + d);
-#pragma unroll
-        for (int i = 0; i < 8; ++i)
-          vt[(d + i) * VST + s_] = (uint16_t)vv[i];
-      }
+      for (int i = 0; i < 8; ++i)
+        vt[(d + i) * VST + s_] = (uint16_t)vv[i];
     }
     const uint16_t* vt = VTs + (kt & 1) * AD * VST;
 #pragma unroll
     for (int kk = 0; kk < KT; kk += 32) {
-      short8 a_frag, b_frag[4];
+      short8 a_frag, b_frag[2];
       const int kbase = kk + (lane >> 4) * 8;
       {
-        int row = wid * 16 + (lane & 15);
+        int row = wr * 16 + (lane & 15);
         a_frag = *(const short8*)(Ps + row * PST + kt * KT + kbase);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn) {
-        int col = fn * 16 + (lane & 15);
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = (wc * 2 + fn) * 16 + (lane & 15);
         b_frag[fn] = *(const short8*)(vt + col * VST + kbase);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn)
+      for (int fn = 0; fn < 2; ++fn)
         accO[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
                                                            accO[fn], 0, 0, 0);
     }
@@ -690,23 +687,23 @@ __launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
   }
   // epilogue: O rows scaled by 1/l
 #pragma unroll
-  for (int fn = 0; fn < 4; ++fn)
+  for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
     for (int rI = 0; rI < 4; ++rI) {
-      int row = wid * 16 + (lane >> 4) * 4 + rI;
-      int col = fn * 16 + (lane & 15);
+      int row = wr * 16 + (lane >> 4) * 4 + rI;
+      int col = (wc * 2 + fn) * 16 + (lane & 15);
       og[row * orr + col] = f2b(accO[fn][rI] * ls[row]);
     }
 
   // ---- stream normalized P out (backward input) ---------------------------
   {
-    const int row = (int)threadIdx.x >> 2;
-    const int qtr = (int)threadIdx.x & 3;
+    const int row = (int)threadIdx.x >> 3;
+    const int oct = (int)threadIdx.x & 7;
     const float invl = ls[row];
-    const uint16_t* pr = Ps + row * PST + qtr * 128;
-    uint16_t* pgr = pg + row * S5 + qtr * 128;
+    const uint16_t* pr = Ps + row * PST + oct * 64;
+    uint16_t* pgr = pg + row * S5 + oct * 64;
 #pragma unroll 2
-    for (int i = 0; i < 128; i += 8) {
+    for (int i = 0; i < 64; i += 8) {
       short8 vv = *(const short8*)(pr + i);
       short8 outp;
 #pragma unroll
@@ -736,7 +733,7 @@ __launch_bounds__(256, 2) __global__ void attn_fwd_fused512_k(
 //       the win is fusing away the composed path's HBM round trips.
 // ---------------------------------------------------------------------------
 template <int SEQ>
-__launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
+__launch_bounds__(512, 2) __global__ void attn_bwd512_q_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const uint16_t* __restrict__ probs,
     const uint16_t* __restrict__ dout, uint16_t* __restrict__ dq,
@@ -748,7 +745,7 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
   // [dO 8K][KV dbuf 2x9.2K][dS strip 66.6K][P 64K] ~= 157 KiB, 1 WG/CU
   __shared__ __attribute__((aligned(16))) uint16_t smem[
       SQ * AD + 2 * AD * VST + SQ * QPST + SQ * SEQ];
-  __shared__ float rsl[SQ], rpart[SQ][4];
+  __shared__ float rsl[SQ], rpart[SQ][8];
   uint16_t* dOs = smem;
   uint16_t* KVs = smem + SQ * AD;            // dbuf: + (kt&1)*AD*VST
   uint16_t* dSs = KVs + 2 * AD * VST;        // [64][520] strip (dP then dS)
@@ -767,27 +764,27 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
   float* rsg = rs_g + bh * SEQ + q0;
 
   const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;  // 4 waves; wave w owns rows w*16..+16
+  const int wid = threadIdx.x >> 6;  // 8 waves (2/SIMD at 1 WG/CU)
+  const int wr = wid & 3;            // row group: 16 rows
+  const int wc = wid >> 2;           // column half: 2 of the 4 fn tiles
   const int e0 = (int)threadIdx.x * 8;
 
   // stage dO (glds, swizzled) + the full P block (glds, linear [64][512])
-#pragma unroll
-  for (int c = 0; c < 2; ++c) {
-    int e = e0 + c * 2048;
+  {
+    int e = e0;
     int row = e >> 6;
     int kk2 = (e & 63) ^ ((row & 7) << 3);
     att_glds16(dog + row * dr + kk2, dOs + e);
   }
 #pragma unroll
-  for (int c = 0; c < SEQ / 32; ++c) {
-    int e = e0 + c * 2048;  // 64*SEQ elems, [64][SEQ] linear
+  for (int c = 0; c < SEQ / 64; ++c) {
+    int e = e0 + c * 4096;  // 64*SEQ elems, [64][SEQ] linear
     att_glds16(pg + (e / SEQ) * SEQ + (e % SEQ), Ps + e);
   }
   // V tile 0 scattered transposed? NO — dP = dO . V^T is NT over d:
   // V [s2][d] is already k-contiguous; stage swizzled like fwd K tiles.
-#pragma unroll
-  for (int c = 0; c < 2; ++c) {
-    int e = e0 + c * 2048;
+  {
+    int e = e0;
     int row = e >> 6;
     int kk2 = (e & 63) ^ ((row & 7) << 3);
     att_glds16(vg + row * qr + kk2, KVs + e);  // V tile 0 -> buf 0 (swz, 64x64)
@@ -799,43 +796,40 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
   // ---- dP strip: loop k-tiles ---------------------------------------------
   for (int kt = 0; kt < SEQ / KT; ++kt) {
     if (kt + 1 < SEQ / KT) {
-#pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        int e = e0 + c * 2048;
-        int row = e >> 6;
-        int kk2 = (e & 63) ^ ((row & 7) << 3);
-        att_glds16(vg + (int64_t)(kt + 1) * KT * qr + row * qr + kk2,
-                   KVs + ((kt + 1) & 1) * AD * VST + e);
-      }
+      int e = e0;
+      int row = e >> 6;
+      int kk2 = (e & 63) ^ ((row & 7) << 3);
+      att_glds16(vg + (int64_t)(kt + 1) * KT * qr + row * qr + kk2,
+                 KVs + ((kt + 1) & 1) * AD * VST + e);
     }
     const uint16_t* vb = KVs + (kt & 1) * AD * VST;
-    f32x4 acc[4] = {};
+    f32x4 acc[2] = {};
 #pragma unroll
     for (int kk = 0; kk < AD; kk += 32) {
-      short8 a_frag, b_frag[4];
+      short8 a_frag, b_frag[2];
       const int kbase = kk + (lane >> 4) * 8;
       {
-        int row = wid * 16 + (lane & 15);
+        int row = wr * 16 + (lane & 15);
         int ks = kbase ^ ((row & 7) << 3);
         a_frag = *(const short8*)(dOs + row * AD + ks);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn) {
-        int col = fn * 16 + (lane & 15);
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = (wc * 2 + fn) * 16 + (lane & 15);
         int ks = kbase ^ ((col & 7) << 3);
         b_frag[fn] = *(const short8*)(vb + col * AD + ks);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn)
+      for (int fn = 0; fn < 2; ++fn)
         acc[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
                                                           acc[fn], 0, 0, 0);
     }
 #pragma unroll
-    for (int fn = 0; fn < 4; ++fn)
+    for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
       for (int rI = 0; rI < 4; ++rI) {
-        int row = wid * 16 + (lane >> 4) * 4 + rI;
-        int col = kt * KT + fn * 16 + (lane & 15);
+        int row = wr * 16 + (lane >> 4) * 4 + rI;
+        int col = kt * KT + (wc * 2 + fn) * 16 + (lane & 15);
         dSs[row * QPST + col] = f2b(acc[fn][rI]);
       }
     asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
@@ -845,21 +839,22 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
 
   // ---- rs = rowsum(dP o P); then dS = scale * P o (dP - rs) ---------------
   {
-    const int row = (int)threadIdx.x >> 2;
-    const int qtr = (int)threadIdx.x & 3;
-    const uint16_t* dpr = dSs + row * QPST + qtr * (SEQ / 4);
-    const uint16_t* ppr = Ps + row * SEQ + qtr * (SEQ / 4);
+    const int row = (int)threadIdx.x >> 3;
+    const int oct = (int)threadIdx.x & 7;
+    const uint16_t* dpr = dSs + row * QPST + oct * (SEQ / 8);
+    const uint16_t* ppr = Ps + row * SEQ + oct * (SEQ / 8);
     float accr = 0.f;
 #pragma unroll 4
-    for (int i = 0; i < SEQ / 4; ++i) accr += b2f(dpr[i]) * b2f(ppr[i]);
-    rpart[row][qtr] = accr;
+    for (int i = 0; i < SEQ / 8; ++i) accr += b2f(dpr[i]) * b2f(ppr[i]);
+    rpart[row][oct] = accr;
   }
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
   if (threadIdx.x < SQ) {
-    float r = rpart[threadIdx.x][0] + rpart[threadIdx.x][1] +
-              rpart[threadIdx.x][2] + rpart[threadIdx.x][3];
+    const float* rp = rpart[threadIdx.x];
+    float r = ((rp[0] + rp[1]) + (rp[2] + rp[3])) +
+              ((rp[4] + rp[5]) + (rp[6] + rp[7]));
     rsl[threadIdx.x] = r;
     rsg[threadIdx.x] = r;  // scratch for the kv kernel
   }
@@ -867,13 +862,13 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
   {
-    const int row = (int)threadIdx.x >> 2;
-    const int qtr = (int)threadIdx.x & 3;
-    uint16_t* dpr = dSs + row * QPST + qtr * (SEQ / 4);
-    const uint16_t* ppr = Ps + row * SEQ + qtr * (SEQ / 4);
+    const int row = (int)threadIdx.x >> 3;
+    const int oct = (int)threadIdx.x & 7;
+    uint16_t* dpr = dSs + row * QPST + oct * (SEQ / 8);
+    const uint16_t* ppr = Ps + row * SEQ + oct * (SEQ / 8);
     const float r = rsl[row];
 #pragma unroll 4
-    for (int i = 0; i < SEQ / 4; ++i)
+    for (int i = 0; i < SEQ / 8; ++i)
       dpr[i] = f2b(scale * b2f(ppr[i]) * (b2f(dpr[i]) - r));
   }
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -881,10 +876,9 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
   asm volatile("" ::: "memory");
 
   // ---- dQ = sum_t dS_t . K_t (K tiles scattered transposed [d][s2]) -------
-  f32x4 accQ[4] = {};
-#pragma unroll
-  for (int c = 0; c < 2; ++c) {
-    int e = e0 + c * 2048;
+  f32x4 accQ[2] = {};
+  {
+    int e = e0;
     int s_ = e >> 6, d = e & 63;
     short8 kv2 = *(const short8*)(kg + s_ * qr + d);
     uint16_t* ktile = KVs;  // buf 0
@@ -898,33 +892,30 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
   for (int kt = 0; kt < SEQ / KT; ++kt) {
     if (kt + 1 < SEQ / KT) {
       uint16_t* kdst = KVs + ((kt + 1) & 1) * AD * VST;
+      int e = e0;
+      int s_ = e >> 6, d = e & 63;
+      short8 kv2 =
+          *(const short8*)(kg + (int64_t)(kt + 1) * KT * qr + s_ * qr + d);
 #pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        int e = e0 + c * 2048;
-        int s_ = e >> 6, d = e & 63;
-        short8 kv2 =
-            *(const short8*)(kg + (int64_t)(kt + 1) * KT * qr + s_ * qr + d);
-#pragma unroll
-        for (int i = 0; i < 8; ++i)
-          kdst[(d + i) * VST + s_] = (uint16_t)kv2[i];
-      }
+      for (int i = 0; i < 8; ++i)
+        kdst[(d + i) * VST + s_] = (uint16_t)kv2[i];
     }
     const uint16_t* ktile = KVs + (kt & 1) * AD * VST;
 #pragma unroll
     for (int kk = 0; kk < KT; kk += 32) {
-      short8 a_frag, b_frag[4];
+      short8 a_frag, b_frag[2];
       const int kbase = kk + (lane >> 4) * 8;
       {
-        int row = wid * 16 + (lane & 15);
+        int row = wr * 16 + (lane & 15);
         a_frag = *(const short8*)(dSs + row * QPST + kt * KT + kbase);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn) {
-        int col = fn * 16 + (lane & 15);
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = (wc * 2 + fn) * 16 + (lane & 15);
         b_frag[fn] = *(const short8*)(ktile + col * VST + kbase);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn)
+      for (int fn = 0; fn < 2; ++fn)
         accQ[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
                                                            accQ[fn], 0, 0, 0);
     }
@@ -933,17 +924,17 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_q_k(
     asm volatile("" ::: "memory");
   }
 #pragma unroll
-  for (int fn = 0; fn < 4; ++fn)
+  for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
     for (int rI = 0; rI < 4; ++rI) {
-      int row = wid * 16 + (lane >> 4) * 4 + rI;
-      int col = fn * 16 + (lane & 15);
+      int row = wr * 16 + (lane >> 4) * 4 + rI;
+      int col = (wc * 2 + fn) * 16 + (lane & 15);
       dqg[row * gr + col] = f2b(accQ[fn][rI]);
     }
 }
 
 template <int SEQ>
-__launch_bounds__(256, 2) __global__ void attn_bwd512_kv_k(
+__launch_bounds__(512, 2) __global__ void attn_bwd512_kv_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const uint16_t* __restrict__ probs,
     const uint16_t* __restrict__ dout, uint16_t* __restrict__ dk,
@@ -976,13 +967,14 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_kv_k(
   const float* rsg = rs_g + bh * SEQ;
 
   const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;  // wave w owns s2 rows w*16..+16
+  const int wid = threadIdx.x >> 6;  // 8 waves (4/SIMD at 2 WG/CU)
+  const int wr = wid & 3;            // s2 row group: 16 rows
+  const int wc = wid >> 2;           // column half: 2 of the 4 fn tiles
   const int e0 = (int)threadIdx.x * 8;
 
   // V block (fixed for this WG), swizzled
-#pragma unroll
-  for (int c = 0; c < 2; ++c) {
-    int e = e0 + c * 2048;
+  {
+    int e = e0;
     int row = e >> 6;
     int kk2 = (e & 63) ^ ((row & 7) << 3);
     att_glds16(vg + row * qr + kk2, Vs + e);
@@ -991,21 +983,19 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_kv_k(
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
 
-  f32x4 accV[4] = {}, accK[4] = {};
+  f32x4 accV[2] = {}, accK[2] = {};
   for (int qt = 0; qt < SEQ / SQ; ++qt) {
     const int64_t q0 = (int64_t)qt * SQ;
     // stage dO_qt (glds swz); P_qt + Q_qt via registers -> transposed images
-#pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      int e = e0 + c * 2048;
+    {
+      int e = e0;
       int row = e >> 6;
       int kk2 = (e & 63) ^ ((row & 7) << 3);
       att_glds16(dog + (q0 + row) * dr + kk2, dOsw + e);
     }
     if (threadIdx.x < SQ) rsl[threadIdx.x] = rsg[q0 + threadIdx.x];
-#pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      int e = e0 + c * 2048;
+    {
+      int e = e0;
       int s1 = e >> 6, d = e & 63;  // [s1][64] block coords
       short8 pv = *(const short8*)(pg + (q0 + s1) * SEQ + d);  // P[s1][k0r+d]
       short8 qv = *(const short8*)(qg + (q0 + s1) * qr + d);
@@ -1022,34 +1012,34 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_kv_k(
     asm volatile("" ::: "memory");
 
     // dP^T[s2][s1] = V . dO^T (NT over d)
-    f32x4 accP[4] = {};
+    f32x4 accP[2] = {};
 #pragma unroll
     for (int kk = 0; kk < AD; kk += 32) {
-      short8 a_frag, b_frag[4];
+      short8 a_frag, b_frag[2];
       const int kbase = kk + (lane >> 4) * 8;
       {
-        int row = wid * 16 + (lane & 15);  // s2
+        int row = wr * 16 + (lane & 15);  // s2
         int ks = kbase ^ ((row & 7) << 3);
         a_frag = *(const short8*)(Vs + row * AD + ks);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn) {
-        int col = fn * 16 + (lane & 15);  // s1
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = (wc * 2 + fn) * 16 + (lane & 15);  // s1
         int ks = kbase ^ ((col & 7) << 3);
         b_frag[fn] = *(const short8*)(dOsw + col * AD + ks);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn)
+      for (int fn = 0; fn < 2; ++fn)
         accP[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
                                                            accP[fn], 0, 0, 0);
     }
     // dS^T = scale * P^T o (dP^T - rs[s1]) -> dSt image
 #pragma unroll
-    for (int fn = 0; fn < 4; ++fn)
+    for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
       for (int rI = 0; rI < 4; ++rI) {
-        int s2 = wid * 16 + (lane >> 4) * 4 + rI;
-        int s1 = fn * 16 + (lane & 15);
+        int s2 = wr * 16 + (lane >> 4) * 4 + rI;
+        int s1 = (wc * 2 + fn) * 16 + (lane & 15);
         float pb = b2f(Pt[s2 * VST + s1]);
         dSt[s2 * VST + s1] = f2b(scale * pb * (accP[fn][rI] - rsl[s1]));
       }
@@ -1060,21 +1050,21 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_kv_k(
     // dV += Pt . dOt-image; dK += dSt . Qt-image (both: k = s1)
 #pragma unroll
     for (int kk = 0; kk < SQ; kk += 32) {
-      short8 a_frag, b_frag, a2_frag, b2_frag[0 + 4];
+      short8 a_frag, b_frag, a2_frag, b2_frag[2];
       const int kbase = kk + (lane >> 4) * 8;
       {
-        int row = wid * 16 + (lane & 15);  // s2
+        int row = wr * 16 + (lane & 15);  // s2
         a_frag = *(const short8*)(Pt + row * VST + kbase);
         a2_frag = *(const short8*)(dSt + row * VST + kbase);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn) {
-        int col = fn * 16 + (lane & 15);  // d
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = (wc * 2 + fn) * 16 + (lane & 15);  // d
         b2_frag[fn] = *(const short8*)(Qt + col * VST + kbase);
       }
 #pragma unroll
-      for (int fn = 0; fn < 4; ++fn) {
-        int col = fn * 16 + (lane & 15);
+      for (int fn = 0; fn < 2; ++fn) {
+        int col = (wc * 2 + fn) * 16 + (lane & 15);
         b_frag = *(const short8*)(dOt + col * VST + kbase);
         accV[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
                                                            accV[fn], 0, 0, 0);
@@ -1087,11 +1077,11 @@ __launch_bounds__(256, 2) __global__ void attn_bwd512_kv_k(
     asm volatile("" ::: "memory");
   }
 #pragma unroll
-  for (int fn = 0; fn < 4; ++fn)
+  for (int fn = 0; fn < 2; ++fn)
 #pragma unroll
     for (int rI = 0; rI < 4; ++rI) {
-      int row = wid * 16 + (lane >> 4) * 4 + rI;  // s2
-      int col = fn * 16 + (lane & 15);            // d
+      int row = wr * 16 + (lane >> 4) * 4 + rI;   // s2
+      int col = (wc * 2 + fn) * 16 + (lane & 15); // d
       dvg[row * gr + col] = f2b(accV[fn][rI]);
       dkg[row * gr + col] = f2b(accK[fn][rI]);
     }
@@ -1169,23 +1159,23 @@ void attn_fused_bwd512_packed(const Tensor& qkv, const Tensor& probs,
   Tensor rs(qkv.ctx(), DType::F32, Shape{B * heads * S});
   uint32_t nblk = (uint32_t)(B * heads * (S / 64));
   if (S == 512) {
-    attn_bwd512_q_k<512><<<dim3(nblk), dim3(256), 0, s>>>(
+    attn_bwd512_q_k<512><<<dim3(nblk), dim3(512), 0, s>>>(
         base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
         (const uint16_t*)dmerged.data(), gbase, (float*)rs.data(), scale,
         (int)heads, S * 3 * H, dh, 3 * H, S * H, dh, H, S * 3 * H, dh, 3 * H);
     NZ_KERNEL_CHECK();
-    attn_bwd512_kv_k<512><<<dim3(nblk), dim3(256), 0, s>>>(
+    attn_bwd512_kv_k<512><<<dim3(nblk), dim3(512), 0, s>>>(
         base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
         (const uint16_t*)dmerged.data(), gbase + H, gbase + 2 * H,
         (const float*)rs.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
         S * H, dh, H, S * 3 * H, dh, 3 * H);
   } else {
-    attn_bwd512_q_k<128><<<dim3(nblk), dim3(256), 0, s>>>(
+    attn_bwd512_q_k<128><<<dim3(nblk), dim3(512), 0, s>>>(
         base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
         (const uint16_t*)dmerged.data(), gbase, (float*)rs.data(), scale,
         (int)heads, S * 3 * H, dh, 3 * H, S * H, dh, H, S * 3 * H, dh, 3 * H);
     NZ_KERNEL_CHECK();
-    attn_bwd512_kv_k<128><<<dim3(nblk), dim3(256), 0, s>>>(
+    attn_bwd512_kv_k<128><<<dim3(nblk), dim3(512), 0, s>>>(
         base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
         (const uint16_t*)dmerged.data(), gbase + H, gbase + 2 * H,
         (const float*)rs.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
@@ -1200,7 +1190,7 @@ void attn_fused_fwd512(const Tensor& q, const Tensor& k, const Tensor& v,
                        float scale) {
   NZ_CHECK(q.dtype() == DType::BF16, "fused attention 512: bf16 required");
   hipStream_t s = stream_of(q);
-  attn_fwd_fused512_k<<<dim3((uint32_t)(batch_heads * 8)), dim3(256), 0, s>>>(
+  attn_fwd_fused512_k<<<dim3((uint32_t)(batch_heads * 8)), dim3(512), 0, s>>>(
       (const uint16_t*)q.data(), (const uint16_t*)k.data(),
       (const uint16_t*)v.data(), (uint16_t*)probs.data(),
       (uint16_t*)ctx_out.data(), scale, 1, (int64_t)512 * AD, 0, AD,
@@ -1216,7 +1206,7 @@ void attn_fused_fwd512_packed(const Tensor& qkv, Tensor& probs, Tensor& out,
   hipStream_t s = stream_of(qkv);
   int64_t H = heads * dh;
   const uint16_t* base = (const uint16_t*)qkv.data();
-  attn_fwd_fused512_k<<<dim3((uint32_t)(B * heads * 8)), dim3(256), 0, s>>>(
+  attn_fwd_fused512_k<<<dim3((uint32_t)(B * heads * 8)), dim3(512), 0, s>>>(
       base, base + H, base + 2 * H, (uint16_t*)probs.data(),
       (uint16_t*)out.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
       S * H, dh, H);
