@@ -497,7 +497,7 @@ constexpr int PST = S5 + 8;    // P strip row stride (elements)
 constexpr int VST = KT + 8;    // VT row stride
 }  // namespace s512
 
-__launch_bounds__(512, 2) __global__ void attn_fwd_fused512_k(
+__launch_bounds__(1024, 4) __global__ void attn_fwd_fused512_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, uint16_t* __restrict__ probs,
     uint16_t* __restrict__ out, float scale, int heads, int64_t qb,
@@ -506,7 +506,7 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused512_k(
   // [Q 64x64][K dbuf 2x 64x64][VT dbuf 2x 64x72][P 64x520][ms 64][ls 64+psum]
   __shared__ __attribute__((aligned(16))) uint16_t smem[
       SQ * AD + 2 * KT * AD + 2 * AD * VST + SQ * PST];
-  __shared__ float ms2[2][SQ], lsum[SQ][8], ls[SQ];
+  __shared__ float ms2[4][SQ], lsum[SQ][16], ls[SQ];
   uint16_t* Qs = smem;
   uint16_t* Ks = smem + SQ * AD;                  // dbuf: + (kt&1)*KT*AD
   uint16_t* VTs = smem + SQ * AD + 2 * KT * AD;   // dbuf: + (kt&1)*AD*VST
@@ -523,14 +523,14 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused512_k(
   uint16_t* og = out + b_ * ob + h_ * oh + qb0 * orr;
 
   const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;  // 8 waves (2/SIMD at 1 WG/CU)
+  const int wid = threadIdx.x >> 6;  // 16 waves (4/SIMD at 1 WG/CU)
   const int wr = wid & 3;            // row group: 16 q-rows
-  const int wc = wid >> 2;           // column half: 2 of the 4 fn tiles
+  const int wc = wid >> 2;           // column quarter: one 16-col fn tile
 
   // ---- stage Q (glds, swizzled source) + K tile 0 -------------------------
-  // a 64x64 tile is 4096 elems = one pass of (512 threads x 8 elems)
+  // a 64x64 tile is 4096 elems = half of (1024 threads x 8 elems)
   const int e0 = (int)threadIdx.x * 8;
-  {
+  if (e0 < 4096) {
     int e = e0;
     int row = e >> 6;
     int kk = (e & 63) ^ ((row & 7) << 3);
@@ -545,7 +545,7 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused512_k(
   // ---- A1: scores + running max ------------------------------------------
   for (int kt = 0; kt < S5 / KT; ++kt) {
     // prefetch next K tile into the other buffer
-    if (kt + 1 < S5 / KT) {
+    if (kt + 1 < S5 / KT && e0 < 4096) {
       int e = e0;
       int row = e >> 6;
       int kk2 = (e & 63) ^ ((row & 7) << 3);
@@ -553,54 +553,47 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused512_k(
                  Ks + ((kt + 1) & 1) * KT * AD + e);
     }
     const uint16_t* kb = Ks + (kt & 1) * KT * AD;
-    f32x4 acc[2] = {};
+    f32x4 acc = {};
 #pragma unroll
     for (int kk = 0; kk < AD; kk += 32) {
-      short8 a_frag, b_frag[2];
+      short8 a_frag, b_frag;
       const int kbase = kk + (lane >> 4) * 8;
       {
         int row = wr * 16 + (lane & 15);
         int ks = kbase ^ ((row & 7) << 3);
         a_frag = *(const short8*)(Qs + row * AD + ks);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        int col = (wc * 2 + fn) * 16 + (lane & 15);
+      {
+        int col = wc * 16 + (lane & 15);
         int ks = kbase ^ ((col & 7) << 3);
-        b_frag[fn] = *(const short8*)(kb + col * AD + ks);
+        b_frag = *(const short8*)(kb + col * AD + ks);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
-        acc[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
-                                                          acc[fn], 0, 0, 0);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc, 0,
+                                                    0, 0);
     }
-    // running max over this wave's half of the columns (combined with the
-    // other column-half's partial at publish time via ms2[2][SQ])
+    // running max over this wave's column quarter (combined with the other
+    // quarters' partials at publish time via ms2[4][SQ])
 #pragma unroll
     for (int rI = 0; rI < 4; ++rI) {
-      float mx = -1e30f;
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn) mx = fmaxf(mx, scale * acc[fn][rI]);
+      float mx = scale * acc[rI];
       for (int off = 1; off < 16; off <<= 1)
         mx = fmaxf(mx, __shfl_xor(mx, off, 64));
       m_run[rI] = fmaxf(m_run[rI], mx);
     }
     // raw scores (scaled) -> P strip, bf16
 #pragma unroll
-    for (int fn = 0; fn < 2; ++fn)
-#pragma unroll
-      for (int rI = 0; rI < 4; ++rI) {
-        int row = wr * 16 + (lane >> 4) * 4 + rI;
-        int col = kt * KT + (wc * 2 + fn) * 16 + (lane & 15);
-        Ps[row * PST + col] = f2b(scale * acc[fn][rI]);
-      }
+    for (int rI = 0; rI < 4; ++rI) {
+      int row = wr * 16 + (lane >> 4) * 4 + rI;
+      int col = kt * KT + wc * 16 + (lane & 15);
+      Ps[row * PST + col] = f2b(scale * acc[rI]);
+    }
     // next iteration reads the prefetched buffer; everyone must also be
     // done READING kb before it is overwritten two iterations later
     asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
   }
-  // publish row maxima (one partial per column-half)
+  // publish row maxima (one partial per column quarter)
   if ((lane & 15) == 0) {
 #pragma unroll
     for (int rI = 0; rI < 4; ++rI)
@@ -612,35 +605,39 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused512_k(
 
   // ---- A2: exp(s - m), row sums (4 threads per row) -----------------------
   {
-    const int row = (int)threadIdx.x >> 3;
-    const int oct = (int)threadIdx.x & 7;
-    const float m = fmaxf(ms2[0][row], ms2[1][row]);
-    uint16_t* pr = Ps + row * PST + oct * 64;
+    const int row = (int)threadIdx.x >> 4;
+    const int hx = (int)threadIdx.x & 15;
+    const float m = fmaxf(fmaxf(ms2[0][row], ms2[1][row]),
+                          fmaxf(ms2[2][row], ms2[3][row]));
+    uint16_t* pr = Ps + row * PST + hx * 32;
     float acc_l = 0.f;
 #pragma unroll 4
-    for (int i = 0; i < 64; ++i) {
+    for (int i = 0; i < 32; ++i) {
       float e = __expf(b2f(pr[i]) - m);
       acc_l += e;
       pr[i] = f2b(e);
     }
-    lsum[row][oct] = acc_l;
+    lsum[row][hx] = acc_l;
   }
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
   if (threadIdx.x < SQ) {
     const float* lr = lsum[threadIdx.x];
-    ls[threadIdx.x] = 1.0f / (((lr[0] + lr[1]) + (lr[2] + lr[3])) +
-                              ((lr[4] + lr[5]) + (lr[6] + lr[7])));
+    float s0 = ((lr[0] + lr[1]) + (lr[2] + lr[3])) +
+               ((lr[4] + lr[5]) + (lr[6] + lr[7]));
+    float s1 = ((lr[8] + lr[9]) + (lr[10] + lr[11])) +
+               ((lr[12] + lr[13]) + (lr[14] + lr[15]));
+    ls[threadIdx.x] = 1.0f / (s0 + s1);
   }
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
 
   // ---- B: O = P~ . V, V tiles scattered transposed ------------------------
-  f32x4 accO[2] = {};
-  {
-    // V tile 0 -> VT buf 0 (single pass at 512 threads)
+  f32x4 accO = {};
+  if (e0 < 4096) {
+    // V tile 0 -> VT buf 0
     int e = e0;
     int s_ = e >> 6, d = e & 63;
     short8 vv = *(const short8*)(vg + s_ * qr + d);
@@ -652,7 +649,7 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused512_k(
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
   for (int kt = 0; kt < S5 / KT; ++kt) {
-    if (kt + 1 < S5 / KT) {
+    if (kt + 1 < S5 / KT && e0 < 4096) {
       uint16_t* vt = VTs + ((kt + 1) & 1) * AD * VST;
       int e = e0;
       int s_ = e >> 6, d = e & 63;
@@ -665,21 +662,18 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused512_k(
     const uint16_t* vt = VTs + (kt & 1) * AD * VST;
 #pragma unroll
     for (int kk = 0; kk < KT; kk += 32) {
-      short8 a_frag, b_frag[2];
+      short8 a_frag, b_frag;
       const int kbase = kk + (lane >> 4) * 8;
       {
         int row = wr * 16 + (lane & 15);
         a_frag = *(const short8*)(Ps + row * PST + kt * KT + kbase);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        int col = (wc * 2 + fn) * 16 + (lane & 15);
-        b_frag[fn] = *(const short8*)(vt + col * VST + kbase);
+      {
+        int col = wc * 16 + (lane & 15);
+        b_frag = *(const short8*)(vt + col * VST + kbase);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
-        accO[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
-                                                           accO[fn], 0, 0, 0);
+      accO = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, accO, 0,
+                                                     0, 0);
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
@@ -687,23 +681,21 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused512_k(
   }
   // epilogue: O rows scaled by 1/l
 #pragma unroll
-  for (int fn = 0; fn < 2; ++fn)
-#pragma unroll
-    for (int rI = 0; rI < 4; ++rI) {
-      int row = wr * 16 + (lane >> 4) * 4 + rI;
-      int col = (wc * 2 + fn) * 16 + (lane & 15);
-      og[row * orr + col] = f2b(accO[fn][rI] * ls[row]);
-    }
+  for (int rI = 0; rI < 4; ++rI) {
+    int row = wr * 16 + (lane >> 4) * 4 + rI;
+    int col = wc * 16 + (lane & 15);
+    og[row * orr + col] = f2b(accO[rI] * ls[row]);
+  }
 
   // ---- stream normalized P out (backward input) ---------------------------
   {
-    const int row = (int)threadIdx.x >> 3;
-    const int oct = (int)threadIdx.x & 7;
+    const int row = (int)threadIdx.x >> 4;
+    const int hx = (int)threadIdx.x & 15;
     const float invl = ls[row];
-    const uint16_t* pr = Ps + row * PST + oct * 64;
-    uint16_t* pgr = pg + row * S5 + oct * 64;
+    const uint16_t* pr = Ps + row * PST + hx * 32;
+    uint16_t* pgr = pg + row * S5 + hx * 32;
 #pragma unroll 2
-    for (int i = 0; i < 64; i += 8) {
+    for (int i = 0; i < 32; i += 8) {
       short8 vv = *(const short8*)(pr + i);
       short8 outp;
 #pragma unroll
@@ -733,7 +725,7 @@ __launch_bounds__(512, 2) __global__ void attn_fwd_fused512_k(
 //       the win is fusing away the composed path's HBM round trips.
 // ---------------------------------------------------------------------------
 template <int SEQ>
-__launch_bounds__(512, 2) __global__ void attn_bwd512_q_k(
+__launch_bounds__(1024, 4) __global__ void attn_bwd512_q_k(
     const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
     const uint16_t* __restrict__ v, const uint16_t* __restrict__ probs,
     const uint16_t* __restrict__ dout, uint16_t* __restrict__ dq,
@@ -745,7 +737,7 @@ __launch_bounds__(512, 2) __global__ void attn_bwd512_q_k(
   // [dO 8K][KV dbuf 2x9.2K][dS strip 66.6K][P 64K] ~= 157 KiB, 1 WG/CU
   __shared__ __attribute__((aligned(16))) uint16_t smem[
       SQ * AD + 2 * AD * VST + SQ * QPST + SQ * SEQ];
-  __shared__ float rsl[SQ], rpart[SQ][8];
+  __shared__ float rsl[SQ], rpart[SQ][16];
   uint16_t* dOs = smem;
   uint16_t* KVs = smem + SQ * AD;            // dbuf: + (kt&1)*AD*VST
   uint16_t* dSs = KVs + 2 * AD * VST;        // [64][520] strip (dP then dS)
@@ -764,26 +756,26 @@ __launch_bounds__(512, 2) __global__ void attn_bwd512_q_k(
   float* rsg = rs_g + bh * SEQ + q0;
 
   const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;  // 8 waves (2/SIMD at 1 WG/CU)
+  const int wid = threadIdx.x >> 6;  // 16 waves (4/SIMD at 1 WG/CU)
   const int wr = wid & 3;            // row group: 16 rows
-  const int wc = wid >> 2;           // column half: 2 of the 4 fn tiles
+  const int wc = wid >> 2;           // column quarter: one 16-col fn tile
   const int e0 = (int)threadIdx.x * 8;
 
   // stage dO (glds, swizzled) + the full P block (glds, linear [64][512])
-  {
+  if (e0 < 4096) {
     int e = e0;
     int row = e >> 6;
     int kk2 = (e & 63) ^ ((row & 7) << 3);
     att_glds16(dog + row * dr + kk2, dOs + e);
   }
 #pragma unroll
-  for (int c = 0; c < SEQ / 64; ++c) {
-    int e = e0 + c * 4096;  // 64*SEQ elems, [64][SEQ] linear
+  for (int c = 0; c < SEQ / 128; ++c) {
+    int e = e0 + c * 8192;  // 64*SEQ elems, [64][SEQ] linear
     att_glds16(pg + (e / SEQ) * SEQ + (e % SEQ), Ps + e);
   }
   // V tile 0 scattered transposed? NO — dP = dO . V^T is NT over d:
   // V [s2][d] is already k-contiguous; stage swizzled like fwd K tiles.
-  {
+  if (e0 < 4096) {
     int e = e0;
     int row = e >> 6;
     int kk2 = (e & 63) ^ ((row & 7) << 3);
@@ -795,7 +787,7 @@ __launch_bounds__(512, 2) __global__ void attn_bwd512_q_k(
 
   // ---- dP strip: loop k-tiles ---------------------------------------------
   for (int kt = 0; kt < SEQ / KT; ++kt) {
-    if (kt + 1 < SEQ / KT) {
+    if (kt + 1 < SEQ / KT && e0 < 4096) {
       int e = e0;
       int row = e >> 6;
       int kk2 = (e & 63) ^ ((row & 7) << 3);
@@ -803,35 +795,30 @@ __launch_bounds__(512, 2) __global__ void attn_bwd512_q_k(
                  KVs + ((kt + 1) & 1) * AD * VST + e);
     }
     const uint16_t* vb = KVs + (kt & 1) * AD * VST;
-    f32x4 acc[2] = {};
+    f32x4 acc = {};
 #pragma unroll
     for (int kk = 0; kk < AD; kk += 32) {
-      short8 a_frag, b_frag[2];
+      short8 a_frag, b_frag;
       const int kbase = kk + (lane >> 4) * 8;
       {
         int row = wr * 16 + (lane & 15);
         int ks = kbase ^ ((row & 7) << 3);
         a_frag = *(const short8*)(dOs + row * AD + ks);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        int col = (wc * 2 + fn) * 16 + (lane & 15);
+      {
+        int col = wc * 16 + (lane & 15);
         int ks = kbase ^ ((col & 7) << 3);
-        b_frag[fn] = *(const short8*)(vb + col * AD + ks);
+        b_frag = *(const short8*)(vb + col * AD + ks);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
-        acc[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
-                                                          acc[fn], 0, 0, 0);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc, 0,
+                                                    0, 0);
     }
 #pragma unroll
-    for (int fn = 0; fn < 2; ++fn)
-#pragma unroll
-      for (int rI = 0; rI < 4; ++rI) {
-        int row = wr * 16 + (lane >> 4) * 4 + rI;
-        int col = kt * KT + (wc * 2 + fn) * 16 + (lane & 15);
-        dSs[row * QPST + col] = f2b(acc[fn][rI]);
-      }
+    for (int rI = 0; rI < 4; ++rI) {
+      int row = wr * 16 + (lane >> 4) * 4 + rI;
+      int col = kt * KT + wc * 16 + (lane & 15);
+      dSs[row * QPST + col] = f2b(acc[rI]);
+    }
     asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
@@ -839,22 +826,24 @@ __launch_bounds__(512, 2) __global__ void attn_bwd512_q_k(
 
   // ---- rs = rowsum(dP o P); then dS = scale * P o (dP - rs) ---------------
   {
-    const int row = (int)threadIdx.x >> 3;
-    const int oct = (int)threadIdx.x & 7;
-    const uint16_t* dpr = dSs + row * QPST + oct * (SEQ / 8);
-    const uint16_t* ppr = Ps + row * SEQ + oct * (SEQ / 8);
+    const int row = (int)threadIdx.x >> 4;
+    const int hx = (int)threadIdx.x & 15;
+    const uint16_t* dpr = dSs + row * QPST + hx * (SEQ / 16);
+    const uint16_t* ppr = Ps + row * SEQ + hx * (SEQ / 16);
     float accr = 0.f;
 #pragma unroll 4
-    for (int i = 0; i < SEQ / 8; ++i) accr += b2f(dpr[i]) * b2f(ppr[i]);
-    rpart[row][oct] = accr;
+    for (int i = 0; i < SEQ / 16; ++i) accr += b2f(dpr[i]) * b2f(ppr[i]);
+    rpart[row][hx] = accr;
   }
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
   if (threadIdx.x < SQ) {
     const float* rp = rpart[threadIdx.x];
-    float r = ((rp[0] + rp[1]) + (rp[2] + rp[3])) +
-              ((rp[4] + rp[5]) + (rp[6] + rp[7]));
+    float r = (((rp[0] + rp[1]) + (rp[2] + rp[3])) +
+               ((rp[4] + rp[5]) + (rp[6] + rp[7]))) +
+              (((rp[8] + rp[9]) + (rp[10] + rp[11])) +
+               ((rp[12] + rp[13]) + (rp[14] + rp[15])));
     rsl[threadIdx.x] = r;
     rsg[threadIdx.x] = r;  // scratch for the kv kernel
   }
@@ -862,13 +851,13 @@ __launch_bounds__(512, 2) __global__ void attn_bwd512_q_k(
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
   {
-    const int row = (int)threadIdx.x >> 3;
-    const int oct = (int)threadIdx.x & 7;
-    uint16_t* dpr = dSs + row * QPST + oct * (SEQ / 8);
-    const uint16_t* ppr = Ps + row * SEQ + oct * (SEQ / 8);
+    const int row = (int)threadIdx.x >> 4;
+    const int hx = (int)threadIdx.x & 15;
+    uint16_t* dpr = dSs + row * QPST + hx * (SEQ / 16);
+    const uint16_t* ppr = Ps + row * SEQ + hx * (SEQ / 16);
     const float r = rsl[row];
 #pragma unroll 4
-    for (int i = 0; i < SEQ / 8; ++i)
+    for (int i = 0; i < SEQ / 16; ++i)
       dpr[i] = f2b(scale * b2f(ppr[i]) * (b2f(dpr[i]) - r));
   }
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -876,8 +865,8 @@ __launch_bounds__(512, 2) __global__ void attn_bwd512_q_k(
   asm volatile("" ::: "memory");
 
   // ---- dQ = sum_t dS_t . K_t (K tiles scattered transposed [d][s2]) -------
-  f32x4 accQ[2] = {};
-  {
+  f32x4 accQ = {};
+  if (e0 < 4096) {
     int e = e0;
     int s_ = e >> 6, d = e & 63;
     short8 kv2 = *(const short8*)(kg + s_ * qr + d);
@@ -890,7 +879,7 @@ __launch_bounds__(512, 2) __global__ void attn_bwd512_q_k(
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
   for (int kt = 0; kt < SEQ / KT; ++kt) {
-    if (kt + 1 < SEQ / KT) {
+    if (kt + 1 < SEQ / KT && e0 < 4096) {
       uint16_t* kdst = KVs + ((kt + 1) & 1) * AD * VST;
       int e = e0;
       int s_ = e >> 6, d = e & 63;
@@ -903,34 +892,29 @@ __launch_bounds__(512, 2) __global__ void attn_bwd512_q_k(
     const uint16_t* ktile = KVs + (kt & 1) * AD * VST;
 #pragma unroll
     for (int kk = 0; kk < KT; kk += 32) {
-      short8 a_frag, b_frag[2];
+      short8 a_frag, b_frag;
       const int kbase = kk + (lane >> 4) * 8;
       {
         int row = wr * 16 + (lane & 15);
         a_frag = *(const short8*)(dSs + row * QPST + kt * KT + kbase);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
-        int col = (wc * 2 + fn) * 16 + (lane & 15);
-        b_frag[fn] = *(const short8*)(ktile + col * VST + kbase);
+      {
+        int col = wc * 16 + (lane & 15);
+        b_frag = *(const short8*)(ktile + col * VST + kbase);
       }
-#pragma unroll
-      for (int fn = 0; fn < 2; ++fn)
-        accQ[fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag[fn],
-                                                           accQ[fn], 0, 0, 0);
+      accQ = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, accQ, 0,
+                                                     0, 0);
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
   }
 #pragma unroll
-  for (int fn = 0; fn < 2; ++fn)
-#pragma unroll
-    for (int rI = 0; rI < 4; ++rI) {
-      int row = wr * 16 + (lane >> 4) * 4 + rI;
-      int col = (wc * 2 + fn) * 16 + (lane & 15);
-      dqg[row * gr + col] = f2b(accQ[fn][rI]);
-    }
+  for (int rI = 0; rI < 4; ++rI) {
+    int row = wr * 16 + (lane >> 4) * 4 + rI;
+    int col = wc * 16 + (lane & 15);
+    dqg[row * gr + col] = f2b(accQ[rI]);
+  }
 }
 
 template <int SEQ>
@@ -1159,7 +1143,7 @@ void attn_fused_bwd512_packed(const Tensor& qkv, const Tensor& probs,
   Tensor rs(qkv.ctx(), DType::F32, Shape{B * heads * S});
   uint32_t nblk = (uint32_t)(B * heads * (S / 64));
   if (S == 512) {
-    attn_bwd512_q_k<512><<<dim3(nblk), dim3(512), 0, s>>>(
+    attn_bwd512_q_k<512><<<dim3(nblk), dim3(1024), 0, s>>>(
         base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
         (const uint16_t*)dmerged.data(), gbase, (float*)rs.data(), scale,
         (int)heads, S * 3 * H, dh, 3 * H, S * H, dh, H, S * 3 * H, dh, 3 * H);
@@ -1170,7 +1154,7 @@ void attn_fused_bwd512_packed(const Tensor& qkv, const Tensor& probs,
         (const float*)rs.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
         S * H, dh, H, S * 3 * H, dh, 3 * H);
   } else {
-    attn_bwd512_q_k<128><<<dim3(nblk), dim3(512), 0, s>>>(
+    attn_bwd512_q_k<128><<<dim3(nblk), dim3(1024), 0, s>>>(
         base, base + H, base + 2 * H, (const uint16_t*)probs.data(),
         (const uint16_t*)dmerged.data(), gbase, (float*)rs.data(), scale,
         (int)heads, S * 3 * H, dh, 3 * H, S * H, dh, H, S * 3 * H, dh, 3 * H);
@@ -1190,7 +1174,7 @@ void attn_fused_fwd512(const Tensor& q, const Tensor& k, const Tensor& v,
                        float scale) {
   NZ_CHECK(q.dtype() == DType::BF16, "fused attention 512: bf16 required");
   hipStream_t s = stream_of(q);
-  attn_fwd_fused512_k<<<dim3((uint32_t)(batch_heads * 8)), dim3(512), 0, s>>>(
+  attn_fwd_fused512_k<<<dim3((uint32_t)(batch_heads * 8)), dim3(1024), 0, s>>>(
       (const uint16_t*)q.data(), (const uint16_t*)k.data(),
       (const uint16_t*)v.data(), (uint16_t*)probs.data(),
       (uint16_t*)ctx_out.data(), scale, 1, (int64_t)512 * AD, 0, AD,
@@ -1206,7 +1190,7 @@ void attn_fused_fwd512_packed(const Tensor& qkv, Tensor& probs, Tensor& out,
   hipStream_t s = stream_of(qkv);
   int64_t H = heads * dh;
   const uint16_t* base = (const uint16_t*)qkv.data();
-  attn_fwd_fused512_k<<<dim3((uint32_t)(B * heads * 8)), dim3(512), 0, s>>>(
+  attn_fwd_fused512_k<<<dim3((uint32_t)(B * heads * 8)), dim3(1024), 0, s>>>(
       base, base + H, base + 2 * H, (uint16_t*)probs.data(),
       (uint16_t*)out.data(), scale, (int)heads, S * 3 * H, dh, 3 * H,
       S * H, dh, H);
