@@ -100,3 +100,16 @@ def test_pingpong_schedule_simulator():
                         os.path.join(REPO, "tools", "pp_schedule_sim.py")],
                        capture_output=True, timeout=60)
     assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
+
+
+def test_pp64q4_schedule_simulator():
+    """The 4-phase pp64q variant's stage schedule (triple-buffered A1/B1,
+    vmcnt<=4 at end-P0/P2) must pass its own verifier
+    (tools/pp64q4_sim.py). The kernel is an opt-in (NEZHA_GEMM_64Q=4),
+    but the simulator keeps its documented schedule honest."""
+    import subprocess
+    import sys as _sys
+    r = subprocess.run([_sys.executable,
+                        os.path.join(REPO, "tools", "pp64q4_sim.py")],
+                       capture_output=True, timeout=60)
+    assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
