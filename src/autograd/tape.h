@@ -47,6 +47,15 @@ struct Var {
   Tensor grad;            // lazily allocated, same dtype as value
   bool grad_init = false;
   bool needs_grad = true; // false for the input leaf
+  // dgrad-side GELU fusion (modules.cc Linear): the producer of this var's
+  // value is a fused-GELU Linear; dgelu_pre_t holds its saved pre-activation.
+  // A downstream consumer that can fuse gelu' into its dgrad GEMM epilogue
+  // sets dgelu_applied; the producer's backward then skips the separate
+  // multiply. grad_writes counts grad_beta() calls so the producer can
+  // verify the fused grad was the ONLY contribution.
+  Tensor dgelu_pre_t;
+  bool dgelu_applied = false;
+  int grad_writes = 0;
   std::shared_ptr<GradAliasGuard> alias_guard;  // set by residual_add
   bool alias_trunk = false;  // this var is the residual trunk input
 
@@ -59,6 +68,7 @@ struct Var {
     ensure_grad();
     float b = grad_init ? 1.0f : 0.0f;
     grad_init = true;
+    ++grad_writes;
     if (b == 1.0f) note_grad_mutation();
     return b;
   }

@@ -24,6 +24,12 @@ void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
           float alpha, float beta, const GemmEpilogue* epi) {
   NZ_DISPATCH(gemm, A, A, B, C, transA, transB, alpha, beta, epi);
 }
+bool gemm_can_fuse_dgelu(const Tensor& like, int64_t M, int64_t N, int64_t K,
+                         bool transA, bool transB) {
+  if (like.device() != Device::HIP) return false;
+  return hip::gemm_can_fuse_dgelu(M, N, K, transA, transB);
+}
+
 void gemm_batched(const Tensor& A, const Tensor& B, Tensor& C, bool transA,
                   bool transB, float alpha, int64_t batch, int64_t M, int64_t N,
                   int64_t K, int64_t sA, int64_t sB, int64_t sC) {

@@ -17,9 +17,17 @@ struct GemmEpilogue {
   const Tensor* bias = nullptr;   // [N] or null
   bool gelu = false;
   Tensor* save_pre = nullptr;     // [M,N] or null
+  // backward-path fusion: C = (alpha*acc) * gelu'(dgelu_pre[idx]) [+ beta*C
+  // + bias]. HIP-only, and only on routes gemm_can_fuse_dgelu approves —
+  // callers must gate on that predicate (hip::gemm fails loudly otherwise).
+  const Tensor* dgelu_pre = nullptr;  // [M,N] pre-activation, or null
 };
 void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
           float alpha, float beta, const GemmEpilogue* epi = nullptr);
+// True iff the HIP dispatch for this shape lands on a kernel implementing
+// the fused GELU-derivative epilogue (false on CPU).
+bool gemm_can_fuse_dgelu(const Tensor& like, int64_t M, int64_t N, int64_t K,
+                         bool transA, bool transB);
 
 // Strided-batched GEMM (attention: QK^T and P·V). Strides in elements.
 void gemm_batched(const Tensor& A, const Tensor& B, Tensor& C, bool transA,

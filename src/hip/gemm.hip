@@ -38,6 +38,10 @@ struct GemmArgs {
   int64_t sA, sB, sC;     // batch strides (elements)
   float alpha, beta;
   int fuse_gelu;
+  // dgrad-side fused GELU derivative: C = (alpha*acc) * gelu'(dgelu_pre)
+  // [+ beta*C + bias] — set only for routes whose kernels implement it
+  // (gemm_can_fuse_dgelu mirrors the dispatch below).
+  const uint16_t* dgelu_pre;
   int diag_no_store;  // NEZHA_PERS_DIAG=1: skip pers epilogue stores
                       // (DIAGNOSTIC ONLY — output garbage; cost attribution)
   int group_m = 0;    // >0: GROUP_M-blocked tile raster (see tile_coords)
@@ -1311,6 +1315,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase_k(GemmArgs args,
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx + i]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         zout[i] = (short)f2b(v);
@@ -1325,6 +1330,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase_k(GemmArgs args,
         if (grow >= args.M || gcol + i >= args.N) continue;
         int64_t idx = grow * args.N + gcol + i;
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         if (save_pre) save_pre[idx] = f2b(v);
@@ -1630,6 +1636,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase6_k(GemmArgs args,
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx + i]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         zout[i] = (short)f2b(v);
@@ -1644,6 +1651,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase6_k(GemmArgs args,
         if (grow >= args.M || gcol + i >= args.N) continue;
         int64_t idx = grow * args.N + gcol + i;
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         if (save_pre) save_pre[idx] = f2b(v);
@@ -2057,6 +2065,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx + i]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         zout[i] = (short)f2b(v);
@@ -2071,6 +2080,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_pingpong_k(
         if (grow >= args.M || gcol + i >= args.N) continue;
         int64_t idx = grow * args.N + gcol + i;
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         if (save_pre) save_pre[idx] = f2b(v);
@@ -3248,6 +3258,7 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx + i]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         zout[i] = (short)f2b(v);
@@ -3262,6 +3273,7 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q_k(GemmArgs args,
         if (grow >= args.M || gcol + i >= args.N) continue;
         int64_t idx = grow * args.N + gcol + i;
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         if (save_pre) save_pre[idx] = f2b(v);
@@ -3540,6 +3552,7 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q4_k(GemmArgs args,
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx + i]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         zout[i] = (short)f2b(v);
@@ -3554,6 +3567,7 @@ __launch_bounds__(512, 4) __global__ void gemm_bf16_pp64q4_k(GemmArgs args,
         if (grow >= args.M || gcol + i >= args.N) continue;
         int64_t idx = grow * args.N + gcol + i;
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         if (save_pre) save_pre[idx] = f2b(v);
@@ -4055,6 +4069,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase2_k(GemmArgs args,
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx + i]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         zout[i] = (short)f2b(v);
@@ -4069,6 +4084,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase2_k(GemmArgs args,
         if (grow >= args.M || gcol + i >= args.N) continue;
         int64_t idx = grow * args.N + gcol + i;
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         if (save_pre) save_pre[idx] = f2b(v);
@@ -4239,6 +4255,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase4_k(GemmArgs args,
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx + i]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         zout[i] = (short)f2b(v);
@@ -4253,6 +4270,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase4_k(GemmArgs args,
         if (grow >= args.M || gcol + i >= args.N) continue;
         int64_t idx = grow * args.N + gcol + i;
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         if (save_pre) save_pre[idx] = f2b(v);
@@ -4430,6 +4448,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase32_k(
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx + i]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx + i]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         zout[i] = (short)f2b(v);
@@ -4444,6 +4463,7 @@ __launch_bounds__(512, 2) __global__ void gemm_bf16_nt_8phase32_k(
         if (grow >= args.M || gcol + i >= args.N) continue;
         int64_t idx = grow * args.N + gcol + i;
         float v = b2f((uint16_t)vv[i]);
+        if (args.dgelu_pre) v *= gelu_bwd_scalar(b2f(args.dgelu_pre[idx]));
         if (args.beta != 0.f) v += args.beta * b2f(C[idx]);
         if (args.bias) v += b2f(args.bias[gcol + i]);
         if (save_pre) save_pre[idx] = f2b(v);
@@ -4593,7 +4613,7 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
         const char* v = getenv("NEZHA_GEMM_SKRASTER");
         return v && atoi(v) == 1;
       }();
-      GemmArgs a4 = args;
+      GemmArgs a4 = args;  // inherits dgelu_pre (route supports it)
       // slab raster only implemented in the 8-phase kernel
       if (kSkRaster && nwg_m4 % 8 == 0 && k64qMode != 4) a4.group_m = -1;
       dim3 grid((uint32_t)nwg4);
@@ -4661,11 +4681,12 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
         const char* v = getenv("NEZHA_GEMM_PP");
         return v ? atoi(v) : 1;  // static prio for the younger half
       }();
+      GemmArgs a7 = args;  // inherits dgelu_pre (route supports it)
       switch (kPP & 3) {
-        case 1: gemm_bf16_nt_pingpong_k<1><<<grid, dim3(512), 0, s>>>(args, nwg_n); break;
-        case 2: gemm_bf16_nt_pingpong_k<2><<<grid, dim3(512), 0, s>>>(args, nwg_n); break;
-        case 3: gemm_bf16_nt_pingpong_k<3><<<grid, dim3(512), 0, s>>>(args, nwg_n); break;
-        default: gemm_bf16_nt_pingpong_k<0><<<grid, dim3(512), 0, s>>>(args, nwg_n); break;
+        case 1: gemm_bf16_nt_pingpong_k<1><<<grid, dim3(512), 0, s>>>(a7, nwg_n); break;
+        case 2: gemm_bf16_nt_pingpong_k<2><<<grid, dim3(512), 0, s>>>(a7, nwg_n); break;
+        case 3: gemm_bf16_nt_pingpong_k<3><<<grid, dim3(512), 0, s>>>(a7, nwg_n); break;
+        default: gemm_bf16_nt_pingpong_k<0><<<grid, dim3(512), 0, s>>>(a7, nwg_n); break;
       }
       NZ_KERNEL_CHECK();
       return;
@@ -4889,6 +4910,37 @@ void launch_gemm(const GemmArgs& args, bool transA, bool transB, int64_t batch,
 // ~2 passes over the operand (~tens of us) vs the GEMM's hundreds, and NT
 // then runs ~4x the scatter-staged generic kernel. TT stays on the generic
 // tiled kernel (test-only form).
+// Mirrors the NT dispatch in hip::gemm: true iff the route chosen for this
+// shape is one of the kernels implementing the fused GELU-derivative
+// epilogue (pp64q / pp64q4 for short K, the v7 ping-pong for deep K).
+// KEEP IN SYNC with the branch conditions in hip::gemm — callers gate the
+// dgelu request on this, and hip::gemm NZ_CHECKs it.
+bool gemm_can_fuse_dgelu(int64_t M, int64_t N, int64_t K, bool transA,
+                         bool transB) {
+  if (transA || !transB) return false;
+  if (K % 128 != 0) return false;
+  static const int variant = [] {
+    const char* v = getenv("NEZHA_GEMM_VARIANT");
+    return v ? atoi(v) : 7;
+  }();
+  if (variant < 7) return false;
+  if (K >= 256 && K < 1536) {  // pp64q / pp64q4 route
+    static const int mode64q = [] {
+      const char* v = getenv("NEZHA_GEMM_64Q");
+      return v ? atoi(v) : 1;
+    }();
+    if (mode64q == 0) return false;
+    int64_t nwg4 = ((M + 127) / 128) * ((N + 127) / 128);
+    return nwg4 >= 1024;
+  }
+  // deep-K (v7 256^2) route: the kernel implements the epilogue, but the
+  // inline pre-activation read stalls its staged epilogue and measured a
+  // NET LOSS on the MLP dgrads (2487 vs 2547 k samples/s,
+  // profiles/r2/r2_dgelu.log) — the standalone gelu_bwd pass streams at
+  // full bandwidth instead. Fusion is therefore short-K (pp64q) only.
+  return false;
+}
+
 static bool can_fast_transpose(int64_t R, int64_t C) {
   return R % 64 == 0 && C % 64 == 0;
 }
@@ -4918,11 +4970,21 @@ void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
   args.ldb = B.cols();
   args.alpha = alpha;
   args.beta = beta;
+  const uint16_t* dgelu_src = nullptr;
   if (epi) {
     args.bias = epi->bias ? (const uint16_t*)epi->bias->data() : nullptr;
     args.save_pre = epi->save_pre ? (uint16_t*)epi->save_pre->data() : nullptr;
     args.fuse_gelu = epi->gelu ? 1 : 0;
+    dgelu_src = epi->dgelu_pre ? (const uint16_t*)epi->dgelu_pre->data()
+                               : nullptr;
   }
+  // dgelu_pre is handed only to routes whose kernels implement the fused
+  // derivative; callers must consult gemm_can_fuse_dgelu first (it mirrors
+  // the dispatch conditions below — keep the two in sync).
+  NZ_CHECK(!dgelu_src || gemm_can_fuse_dgelu(args.M, args.N, args.K, transA,
+                                             transB),
+           "dgelu epilogue requested on an unsupported GEMM route");
+  args.dgelu_pre = dgelu_src;
   // r2: NATIVE NN/TN ping-pong — consume the stored layouts directly via
   // transpose-in-stage (TRMODE), skipping the lowering transposes below
   // entirely (they were 6-9% of both model steps: r2 kernel stats).

@@ -13,6 +13,8 @@ void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
 void gemm_batched(const Tensor& A, const Tensor& B, Tensor& C, bool transA,
                   bool transB, float alpha, int64_t batch, int64_t M, int64_t N,
                   int64_t K, int64_t sA, int64_t sB, int64_t sC);
+bool gemm_can_fuse_dgelu(int64_t M, int64_t N, int64_t K, bool transA,
+                         bool transB);
 void bias_gelu_fwd(const Tensor& x, const Tensor* bias, Tensor& y, Tensor* save_pre);
 void bias_gelu_bwd(const Tensor& dy, const Tensor& z, Tensor& dz, Tensor* dbias);
 void layernorm_fwd(const Tensor& x, const Tensor& gamma, const Tensor& beta,

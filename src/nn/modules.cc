@@ -31,6 +31,16 @@ void Linear::collect_params(std::vector<Param*>& out) {
   if (has_bias) out.push_back(&b);
 }
 
+// NEZHA_FUSE_DGELU=0 disables folding gelu' into downstream dgrad GEMMs
+// (useful for kernel-variant A/B runs where the route would not support it).
+static bool fuse_dgelu_enabled() {
+  static const bool on = [] {
+    const char* v = getenv("NEZHA_FUSE_DGELU");
+    return !v || atoi(v) != 0;
+  }();
+  return on;
+}
+
 VarPtr Linear::forward(Tape& tape, VarPtr x) {
   int64_t M = x->value.rows();
   DType pd = param_dtype(ctx_);
@@ -44,13 +54,26 @@ VarPtr Linear::forward(Tape& tape, VarPtr x) {
     epi.save_pre = &z;
   }
   ops::gemm(x->value, w.data, y->value, false, false, 1.0f, 0.0f, &epi);
+  if (fuse_gelu) y->dgelu_pre_t = z;  // downstream dgrad may fuse gelu' in
 
   tape.push([this, x, y, z]() {
     Context* ctx = ctx_;
     DType pd = param_dtype(ctx);
     Tensor& dy = y->grad_for_read();
     Tensor dz;
-    if (fuse_gelu) {
+    if (fuse_gelu && y->dgelu_applied) {
+      // the downstream consumer's dgrad GEMM already multiplied gelu'(z)
+      // into dy in its epilogue; it must have been the only contributor.
+      NZ_CHECK(y->grad_writes == 1,
+               "dgelu fusion with multiple grad writers on " + w.name);
+      dz = dy;
+      if (has_bias) {
+        NZ_CHECK(!b.grad_init, "param " + b.name + " used twice");
+        ops::colsum(dy, b.grad, b.grad_beta());
+        b.grad_init = true;
+        b.mark_ready();
+      }
+    } else if (fuse_gelu) {
       dz = Tensor(ctx, pd, dy.shape());
       NZ_CHECK(!has_bias || !b.grad_init, "param " + b.name + " used twice");
       ops::bias_gelu_bwd(dy, z, dz, has_bias ? &b.grad : nullptr);
@@ -65,10 +88,21 @@ VarPtr Linear::forward(Tape& tape, VarPtr x) {
     // wgrad: dW[in,out] = x^T . dz
     ops::gemm(x->value, dz, w.grad, true, false, 1.0f, w.grad_beta(), nullptr);
     w.mark_ready();
-    // dgrad: dx[M,in] = dz . W^T
+    // dgrad: dx[M,in] = dz . W^T — if x is a fused-GELU output and the
+    // route supports it, fold gelu'(pre) into this GEMM's epilogue
+    // (saves the separate bias_gelu_bwd read+write pass over dx).
     if (x->needs_grad) {
       x->ensure_grad();
-      ops::gemm(dz, w.data, x->grad, false, true, 1.0f, x->grad_beta(), nullptr);
+      float bx = x->grad_beta();
+      ops::GemmEpilogue epi2;
+      bool fuse_dg = fuse_dgelu_enabled() && bx == 0.0f &&
+                     x->dgelu_pre_t.defined() && !x->dgelu_applied &&
+                     ops::gemm_can_fuse_dgelu(dz, dz.rows(), in_dim, out_dim,
+                                              false, true);
+      if (fuse_dg) epi2.dgelu_pre = &x->dgelu_pre_t;
+      ops::gemm(dz, w.data, x->grad, false, true, 1.0f, bx,
+                fuse_dg ? &epi2 : nullptr);
+      if (fuse_dg) x->dgelu_applied = true;
     }
   });
   return y;

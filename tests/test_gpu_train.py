@@ -423,3 +423,27 @@ np.save(sys.argv[1], dqkv.numpy())
     scale_g = np.abs(ref).max() + 1e-9
     err = np.abs(got - ref).max() / scale_g
     assert err < 0.04, f"dqkv rel err {err}"
+
+
+def test_dgelu_fusion_matches_unfused(tmp_path):
+    """Folding gelu' into the downstream dgrad GEMM epilogue
+    (NEZHA_FUSE_DGELU, default on) must train to the same losses as the
+    separate bias_gelu_bwd pass — both models, few steps, loose bf16
+    rounding tolerance (the fused path multiplies before the bf16
+    round of dx, the composed path after)."""
+    import subprocess, sys, os, json
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for model in ("mlp", "bert"):
+        losses = {}
+        for fuse in ("1", "0"):
+            env = dict(os.environ, NEZHA_FUSE_DGELU=fuse)
+            r = subprocess.run(
+                [sys.executable, os.path.join(repo, "bench.py"),
+                 "--model", model, "--steps", "4", "--warmup", "1",
+                 "--batch", "1024" if model == "mlp" else "64"],
+                env=env, cwd=repo, capture_output=True, timeout=600)
+            assert r.returncode == 0, r.stdout.decode() + r.stderr.decode()
+            line = [l for l in r.stdout.decode().splitlines()
+                    if l.startswith("{")][-1]
+            losses[fuse] = json.loads(line)["loss"]
+        assert abs(losses["1"] - losses["0"]) < 5e-3, losses
