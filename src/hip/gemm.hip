@@ -5009,8 +5009,6 @@ void gemm(const Tensor& A, const Tensor& B, Tensor& C, bool transA, bool transB,
         args.K >= 1536 && args.N % 8 == 0 &&
         (trmode != 3 || args.M % 8 == 0)) {
       int nwg_m = (int)((args.M + 255) / 256);
-      int nwg_n = (int)((args.N + 255) / 256);
-      int64_t nwg = (int64_t)nwg_m * nwg_n;
       hipStream_t st = stream_of(C);
       int nwg_n128 = (int)((args.N + 127) / 128);
       int64_t per_b128 = (int64_t)nwg_m * nwg_n128;
