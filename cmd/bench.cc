@@ -127,6 +127,10 @@ int main(int argc, const char** argv) {
     Tensor zt(&ctx, DType::BF16, Shape{64, 1 << 22});
     bench_mem(ctx, "transpose2d bf16", z.numel(), 2.0 * z.numel() * 2,
               [&] { ops::transpose2d(z, zt); }, iters);
+    Tensor z2(&ctx, DType::BF16, Shape{65536, 3072});
+    Tensor z2t(&ctx, DType::BF16, Shape{3072, 65536});
+    bench_mem(ctx, "transpose2d bf16 (wgrad shape)", z2.numel(),
+              2.0 * z2.numel() * 2, [&] { ops::transpose2d(z2, z2t); }, iters);
     // adam on a 64 MB bucket
     int64_t bn = 32 << 20;
     Tensor p(&ctx, DType::BF16, Shape{bn}), g(&ctx, DType::BF16, Shape{bn});

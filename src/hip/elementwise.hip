@@ -222,11 +222,16 @@ __global__ void transpose_vec_k(const T* __restrict__ src, T* __restrict__ dst,
   T* dp = dst + blockIdx.z * sDst;
   int64_t c0 = (int64_t)blockIdx.x * 64, r0 = (int64_t)blockIdx.y * 64;
   constexpr int CHUNKS_PER_ROW = 64 / VEC;
+  // column-group XOR swizzle: without it the gather below reads 8 rows a
+  // VEC-group apart whose addresses land on ONE bank (row stride x 8 is
+  // 0 mod 32 dwords for any 16B-aligned stride) — an 8-way conflict on
+  // every gathered vector. Permuting the stored group per row spreads the
+  // gather to ~2-way while keeping the stores 16 B vectors.
 #pragma unroll
   for (int rep = 0; rep < 64 * CHUNKS_PER_ROW / 256; ++rep) {
     int cidx = rep * 256 + threadIdx.x;
     int row = cidx / CHUNKS_PER_ROW, cg = cidx % CHUNKS_PER_ROW;
-    *(vec_t*)&tile[row][cg * VEC] =
+    *(vec_t*)&tile[row][(cg ^ (row & (CHUNKS_PER_ROW - 1))) * VEC] =
         *(const vec_t*)(sp + (r0 + row) * C + c0 + cg * VEC);
   }
   __syncthreads();
@@ -234,9 +239,13 @@ __global__ void transpose_vec_k(const T* __restrict__ src, T* __restrict__ dst,
   for (int rep = 0; rep < 64 * CHUNKS_PER_ROW / 256; ++rep) {
     int cidx = rep * 256 + threadIdx.x;
     int orow = cidx / CHUNKS_PER_ROW, og = cidx % CHUNKS_PER_ROW;
+    const int lg = orow / VEC, off = orow % VEC;
     T vals[VEC];
 #pragma unroll
-    for (int i = 0; i < VEC; ++i) vals[i] = tile[og * VEC + i][orow];
+    for (int i = 0; i < VEC; ++i) {
+      int r = og * VEC + i;
+      vals[i] = tile[r][(lg ^ (r & (CHUNKS_PER_ROW - 1))) * VEC + off];
+    }
     *(vec_t*)(dp + (c0 + orow) * R + r0 + og * VEC) = *(vec_t*)vals;
   }
 }
