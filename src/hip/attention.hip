@@ -13,7 +13,7 @@
 //   region VT [64][136]  17 KB  (V transposed via register scatter; +8 pad
 //                                gives row stride 272 B = 4 banks mod 64 ->
 //                                conflict-free column-of-rows reads)
-// Waves: 4 (256 threads). QK^T: 2x2 wave grid, 64x64 each (NT, 32 MFMA).
+// Waves: 8 (512 threads). QK^T: 2x4 wave grid, 64x32 each (NT, 16 MFMA).
 // Softmax: one wave per 4 rows, row in registers (2 elems/lane), wave
 // shuffles for max/sum; writes P to LDS + global. PV: 2x2 wave grid,
 // 64x32 each over K=S (NT against VT, 32 MFMA).
@@ -477,11 +477,12 @@ __launch_bounds__(1024) __global__ void attn_bwd_fused_k(
 // longer fits a workgroup, so the kernel tiles it: one WG per
 // (batch, head, 64-row q-block), K streamed in 64-row tiles.
 //
-// Three phases per WG (4 waves of 16 q-rows each, stats wave-local):
+// Three phases per WG (16 waves as 4 row-groups x 4 column quarters;
+// row stats combine one partial per column quarter via ms2[4][64]):
 //   A1: loop kt 0..7: stage K-tile (glds dbuf, XOR-swz source) ->
 //       S_t = Q.K_t^T (8 MFMA/wave) -> running row max -> raw bf16 scores
 //       into the P strip [64][512] (stride 520 spreads banks)
-//   A2: elementwise exp(s - m) over the strip (4 threads per row), row
+//   A2: elementwise exp(s - m) over the strip (16 threads per row), row
 //       sums folded through LDS -> inv l
 //   B:  loop kt 0..7: V-tile via register scatter to VT [64][72] (dbuf) ->
 //       O += P_t . V_t (8 MFMA/wave); epilogue scales O rows by 1/l
@@ -603,7 +604,7 @@ __launch_bounds__(1024, 4) __global__ void attn_fwd_fused512_k(
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
 
-  // ---- A2: exp(s - m), row sums (4 threads per row) -----------------------
+  // ---- A2: exp(s - m), row sums (16 threads per row) ----------------------
   {
     const int row = (int)threadIdx.x >> 4;
     const int hx = (int)threadIdx.x & 15;
