@@ -239,7 +239,7 @@ def main():
         if peak:
             out["mfu_est"] = tr.flops_per_step / (elapsed / args.steps) / peak
         print(json.dumps(out))
-    if os.environ.get("NEZHA_TRACE"):
+    if rank == 0 and os.environ.get("NEZHA_TRACE"):
         ctx.prof_flush()  # write the chrome://tracing JSON (op scopes)
 
 

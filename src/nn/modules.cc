@@ -95,8 +95,11 @@ VarPtr Linear::forward(Tape& tape, VarPtr x) {
       x->ensure_grad();
       float bx = x->grad_beta();
       ops::GemmEpilogue epi2;
+      // never fuse into an aliased (residual-trunk) grad buffer: its
+      // accumulation contract is outside this Linear's view
       bool fuse_dg = fuse_dgelu_enabled() && bx == 0.0f &&
                      x->dgelu_pre_t.defined() && !x->dgelu_applied &&
+                     !x->alias_guard &&
                      ops::gemm_can_fuse_dgelu(dz, dz.rows(), in_dim, out_dim,
                                               false, true);
       if (fuse_dg) epi2.dgelu_pre = &x->dgelu_pre_t;
